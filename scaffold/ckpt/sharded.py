@@ -1,0 +1,186 @@
+"""Sharded checkpoint save/load (contracts C3, C6, C12).
+
+Save path on GPU: device tensors stream to host through PINNED staging
+buffers with async D2H copies (hipMemcpyAsync under torch's
+``Tensor.copy_(non_blocking=True)``) on a DEDICATED side stream, so the
+compute stream is never blocked (C6).  Chunks are event-signalled; the
+host thread serializes a chunk to the shard file only after its event
+fires.  On CPU the same code degenerates to plain copies.
+
+Format: safetensors per shard + manifest.json (layout.py).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Tuple
+
+import torch
+from safetensors.torch import save_file, safe_open
+
+from . import layout
+
+
+def _named_state_tensors(model, optimizer) -> List[Tuple[str, torch.Tensor]]:
+    """Flat name->tensor list: model params+buffers, optimizer masters/moments."""
+    out: List[Tuple[str, torch.Tensor]] = []
+    for name, p in model.named_parameters():
+        out.append((f"model.{name}", p.data))
+    for name, b in model.named_buffers():
+        out.append((f"buffer.{name}", b.data))
+    if optimizer is not None:
+        pnames = [n for n, _ in model.named_parameters()]
+        for i, t in enumerate(optimizer.master):
+            out.append((f"opt.master.{pnames[i]}", t))
+        for i, t in enumerate(optimizer.exp_avg):
+            out.append((f"opt.exp_avg.{pnames[i]}", t))
+        for i, t in enumerate(optimizer.exp_avg_sq):
+            out.append((f"opt.exp_avg_sq.{pnames[i]}", t))
+    return out
+
+
+class _PinnedStager:
+    """Double-buffered pinned ring: D2H on a side stream, event-gated reads."""
+
+    def __init__(self, device: torch.device, nbuf: int = 4, buf_bytes: int = 64 << 20):
+        self.on_gpu = device.type == "cuda"
+        if self.on_gpu:
+            self.stream = torch.cuda.Stream(device)
+            self.bufs = [
+                torch.empty(buf_bytes, dtype=torch.uint8, pin_memory=True)
+                for _ in range(nbuf)
+            ]
+            self.events = [torch.cuda.Event() for _ in range(nbuf)]
+            self.buf_bytes = buf_bytes
+            self.slot = 0
+
+    def to_host(self, t: torch.Tensor) -> torch.Tensor:
+        """Return a CPU copy of t; async through the pinned ring on GPU."""
+        if not self.on_gpu or not t.is_cuda:
+            return t.detach().cpu()
+        flat = t.detach().contiguous().view(torch.uint8).view(-1)
+        n = flat.numel()
+        if n > self.buf_bytes:
+            # big tensor: one dedicated pinned buffer, still on side stream
+            host = torch.empty(n, dtype=torch.uint8, pin_memory=True)
+            with torch.cuda.stream(self.stream):
+                host.copy_(flat, non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record(self.stream)
+            ev.synchronize()
+            return host.view(t.dtype).view(t.shape).clone()
+        s = self.slot
+        self.slot = (s + 1) % len(self.bufs)
+        self.events[s].synchronize()  # previous use of this slot done
+        buf = self.bufs[s][:n]
+        with torch.cuda.stream(self.stream):
+            buf.copy_(flat, non_blocking=True)
+        self.events[s].record(self.stream)
+        self.events[s].synchronize()
+        return buf.view(t.dtype).view(t.shape).clone()
+
+
+def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
+                 cfg_hash: str, async_d2h: bool = True) -> None:
+    os.makedirs(ckpt_dir, exist_ok=True)
+    world = ctx.world_size
+    rank = ctx.rank
+
+    named = _named_state_tensors(model, optimizer)
+    owner = layout.partition_by_size(
+        [(n, t.numel() * t.element_size()) for n, t in named], world
+    )
+
+    stager = _PinnedStager(ctx.device) if async_d2h else None
+    shard: Dict[str, torch.Tensor] = {}
+    tensor_meta: Dict[str, dict] = {}
+    for name, t in named:
+        tensor_meta[name] = {
+            "shard": owner[name],
+            "shape": list(t.shape),
+            "dtype": str(t.dtype).replace("torch.", ""),
+        }
+        if owner[name] == rank:
+            host = stager.to_host(t) if stager else t.detach().cpu()
+            shard[name] = host.contiguous()
+
+    # rank-local RNG + data state (every shard)
+    shard[f"rng.torch.{rank}"] = torch.random.get_rng_state()
+    if ctx.on_gpu:
+        shard[f"rng.cuda.{rank}"] = torch.cuda.get_rng_state(ctx.device)
+    if data is not None:
+        shard[f"rng.data.{rank}"] = data.state_dict()["gen_state"].cpu()
+    sf = os.path.join(ckpt_dir, layout.shard_file(rank))
+    save_file(shard, sf, metadata={"rank": str(rank), "step": str(step)})
+    with open(sf, "rb") as fh:
+        os.fsync(fh.fileno())
+
+    # all shards durable before rank 0 commits the manifest
+    if ctx.initialized:
+        import torch.distributed as dist
+
+        dist.barrier()
+    if rank == 0:
+        meta = dict(tensor_meta)
+        meta["opt.step"] = {"shard": -1, "value": optimizer.step_count if optimizer else 0}
+        layout.write_manifest(
+            ckpt_dir, step=step, world_size=world, cfg_hash=cfg_hash, tensors=meta
+        )
+    if ctx.initialized:
+        import torch.distributed as dist
+
+        dist.barrier()
+
+
+def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = "",
+                 strict_hash: bool = False) -> int:
+    m = layout.read_manifest(ckpt_dir)
+    if strict_hash and cfg_hash and m["config_hash"] != cfg_hash:
+        raise ValueError(
+            f"config hash mismatch: ckpt {m['config_hash']} vs current {cfg_hash}"
+        )
+    saved_world = m["world_size"]
+
+    # open every shard once; load-by-name (reshard-capable)
+    handles = {}
+    for r in range(saved_world):
+        p = os.path.join(ckpt_dir, layout.shard_file(r))
+        handles[r] = safe_open(p, framework="pt", device="cpu")
+
+    def fetch(name: str) -> torch.Tensor:
+        info = m["tensors"][name]
+        return handles[info["shard"]].get_tensor(name)
+
+    with torch.no_grad():
+        pnames = []
+        for name, p in model.named_parameters():
+            pnames.append(name)
+            p.data.copy_(fetch(f"model.{name}").to(p.device, p.dtype))
+        for name, b in model.named_buffers():
+            key = f"buffer.{name}"
+            if key in m["tensors"]:
+                b.data.copy_(fetch(key).to(b.device, b.dtype))
+        if optimizer is not None:
+            for i, name in enumerate(pnames):
+                optimizer.master[i].copy_(fetch(f"opt.master.{name}").to(optimizer.master[i].device))
+                optimizer.exp_avg[i].copy_(fetch(f"opt.exp_avg.{name}").to(optimizer.exp_avg[i].device))
+                optimizer.exp_avg_sq[i].copy_(fetch(f"opt.exp_avg_sq.{name}").to(optimizer.exp_avg_sq[i].device))
+            optimizer.step_count = int(m["tensors"].get("opt.step", {}).get("value", m["step"]))
+            # working params refresh from masters
+            for p, mt in zip(optimizer.params, optimizer.master):
+                p.data.copy_(mt.to(p.dtype))
+
+    # rank-local RNG: same-topology fast path only
+    if saved_world == ctx.world_size:
+        h = handles[ctx.rank]
+        keys = set(h.keys())
+        if f"rng.torch.{ctx.rank}" in keys:
+            torch.random.set_rng_state(h.get_tensor(f"rng.torch.{ctx.rank}"))
+        if ctx.on_gpu and f"rng.cuda.{ctx.rank}" in keys:
+            torch.cuda.set_rng_state(h.get_tensor(f"rng.cuda.{ctx.rank}"), ctx.device)
+        if data is not None and f"rng.data.{ctx.rank}" in keys:
+            data.load_state_dict({"gen_state": h.get_tensor(f"rng.data.{ctx.rank}")})
+    elif data is not None:
+        data.reset(step=m["step"])
+
+    return int(m["step"])

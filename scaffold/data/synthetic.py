@@ -1,0 +1,110 @@
+"""Synthetic data generators (contract configs use synthetic data only).
+
+Device-side generation: batches are produced directly on the training
+device from a per-rank seeded generator, so the data path costs no host
+PCIe traffic and no dataloader workers — the right design for a bench
+whose data is synthetic by contract (BASELINE.json configs; no network
+for datasets in this environment).
+"""
+
+from __future__ import annotations
+
+from typing import Iterator, Tuple
+
+import torch
+
+from ..config import TrainConfig
+
+
+class SyntheticBase:
+    """Infinite per-rank deterministic batch stream."""
+
+    def __init__(self, cfg: TrainConfig, rank: int, device: torch.device, dtype: torch.dtype):
+        self.cfg = cfg
+        self.rank = rank
+        self.device = device
+        self.dtype = dtype
+        self.gen = torch.Generator(device=device)
+        self.reset()
+
+    def reset(self, step: int = 0) -> None:
+        # Seed folds in rank so ranks see distinct data, and step so a
+        # resume regenerates the same stream position deterministically.
+        self.gen.manual_seed(self.cfg.data.seed * 1000003 + self.rank * 997 + step)
+
+    def state_dict(self) -> dict:
+        return {"gen_state": self.gen.get_state()}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.gen.set_state(sd["gen_state"].to(self.gen.get_state().device))
+
+    def __iter__(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        while True:
+            yield self.next_batch()
+
+    def next_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        raise NotImplementedError
+
+
+class SyntheticImageData(SyntheticBase):
+    """ImageNet-shaped batches: (B,3,H,W) images + class labels."""
+
+    def next_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        d = self.cfg.data
+        x = torch.randn(
+            d.batch_size, d.channels, d.image_size, d.image_size,
+            generator=self.gen, device=self.device, dtype=torch.float32,
+        ).to(self.dtype)
+        y = torch.randint(
+            0, d.num_classes, (d.batch_size,), generator=self.gen, device=self.device
+        )
+        return x, y
+
+
+class SyntheticTokenData(SyntheticBase):
+    """Token streams for LM configs: (B,S) int64 inputs, (B,S) targets."""
+
+    def next_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        d = self.cfg.data
+        vocab = self.cfg.model.vocab_size
+        tokens = torch.randint(
+            0, vocab, (d.batch_size, d.seq_len + 1), generator=self.gen, device=self.device
+        )
+        return tokens[:, :-1].contiguous(), tokens[:, 1:].contiguous()
+
+
+class SyntheticVectorData(SyntheticBase):
+    """Flat feature vectors for the MLP smoke config (C10).
+
+    Targets follow a fixed random linear map of the inputs so that a
+    training run has signal and the loss measurably decreases.
+    """
+
+    def __init__(self, cfg: TrainConfig, rank: int, device: torch.device, dtype: torch.dtype):
+        super().__init__(cfg, rank, device, dtype)
+        teacher_gen = torch.Generator(device=device)
+        teacher_gen.manual_seed(cfg.data.seed)  # same teacher on all ranks
+        self.teacher = torch.randn(
+            cfg.data.vector_dim, cfg.model.out_dim, generator=teacher_gen,
+            device=device, dtype=torch.float32,
+        )
+
+    def next_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        d = self.cfg.data
+        x = torch.randn(
+            d.batch_size, d.vector_dim, generator=self.gen, device=self.device,
+            dtype=torch.float32,
+        )
+        y = x @ self.teacher
+        return x.to(self.dtype), y.to(self.dtype)
+
+
+def build_data(cfg: TrainConfig, rank: int, device: torch.device, dtype: torch.dtype) -> SyntheticBase:
+    kind = cfg.data.kind
+    if kind == "synthetic_image":
+        return SyntheticImageData(cfg, rank, device, dtype)
+    if kind == "synthetic_tokens":
+        return SyntheticTokenData(cfg, rank, device, dtype)
+    if kind == "synthetic_vector":
+        return SyntheticVectorData(cfg, rank, device, dtype)
+    raise ValueError(f"unknown data kind {kind!r}")

@@ -1,0 +1,201 @@
+"""BucketedDDP — the in-repo data-parallel engine (contract C4).
+
+Design, MI355X-first:
+
+* **Gradient-as-bucket-view**: each bucket owns ONE contiguous flat
+  buffer; every param's ``.grad`` is a view into it, set up front.
+  Autograd accumulates straight into the comm buffer, so there is no
+  flatten/unflatten copy on the hot path at all — zero extra HBM
+  traffic (288 GB HBM3E at ~8 TB/s is the usual bound; don't spend it
+  on copies).
+* **Bucket partition in reverse parameter order** (grads become ready
+  roughly in reverse-forward order during backward), with a small first
+  bucket so the first all-reduce fires early.
+* **Overlap**: when a bucket's last grad accumulates
+  (``register_post_accumulate_grad_hook``), its SUM all-reduce is
+  launched ``async_op=True``.  ProcessGroupNCCL (RCCL on ROCm) orders
+  the collective against the compute stream and runs it on its own
+  stream, so it proceeds concurrently with the remaining backward
+  kernels over xGMI.  ``finish()`` waits all works (stream-ordered, no
+  host sync).
+* **Averaging** is folded into the optimizer's fused AdamW
+  ``grad_scale=1/world`` (one less full-gradient pass).
+* Bucket sizes default to 25 MiB, first bucket 4 MiB — chosen so that
+  >=7 RCCL channels (one per xGMI link) stay busy; sweepable via
+  ``dist.bucket_cap_mb`` (bench harness mode).
+"""
+
+from __future__ import annotations
+
+import contextlib
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+import torch.nn as nn
+
+from .comm import Comm
+
+
+@dataclass
+class GradBucket:
+    index: int
+    params: List[torch.nn.Parameter]
+    flat: torch.Tensor = None  # type: ignore[assignment]
+    views: List[torch.Tensor] = field(default_factory=list)
+    pending: int = 0
+    work: object = None
+    launched: bool = False
+
+    @property
+    def numel(self) -> int:
+        return sum(p.numel() for p in self.params)
+
+    @property
+    def nbytes(self) -> int:
+        return self.flat.numel() * self.flat.element_size()
+
+
+def partition_buckets(
+    params: List[torch.nn.Parameter],
+    bucket_cap_mb: float,
+    first_bucket_mb: float,
+) -> List[List[torch.nn.Parameter]]:
+    """Split params (REVERSED construction order) into size-capped groups.
+
+    Pure function — unit-tested on CPU (SURVEY §4 unit tier).
+    """
+    buckets: List[List[torch.nn.Parameter]] = []
+    cur: List[torch.nn.Parameter] = []
+    cur_bytes = 0
+    cap = first_bucket_mb * (1 << 20)
+    for p in reversed(params):
+        p_bytes = p.numel() * p.element_size()
+        if cur and cur_bytes + p_bytes > cap:
+            buckets.append(cur)
+            cur = []
+            cur_bytes = 0
+            cap = bucket_cap_mb * (1 << 20)
+        cur.append(p)
+        cur_bytes += p_bytes
+    if cur:
+        buckets.append(cur)
+    return buckets
+
+
+class BucketedDDP(nn.Module):
+    def __init__(
+        self,
+        module: nn.Module,
+        comm: Optional[Comm] = None,
+        bucket_cap_mb: float = 25,
+        first_bucket_mb: float = 4,
+        overlap: bool = True,
+    ):
+        super().__init__()
+        self.module = module
+        self.comm = comm if comm is not None else Comm()
+        self.overlap = overlap
+        self._sync_enabled = True
+        self._require_finish = False
+
+        self.trainable = [p for p in module.parameters() if p.requires_grad]
+
+        # 1) rank-0 params/buffers win (SURVEY §3.1 broadcast at wrap)
+        if self.comm.is_distributed:
+            with torch.no_grad():
+                for t in list(module.parameters()) + list(module.buffers()):
+                    if t.is_floating_point() or t.dtype in (torch.int64, torch.int32):
+                        self.comm.broadcast(t.data, src=0)
+
+        # 2) buckets + flat buffers + grad views
+        groups = partition_buckets(self.trainable, bucket_cap_mb, first_bucket_mb)
+        self.buckets: List[GradBucket] = []
+        self._param_bucket: Dict[int, GradBucket] = {}
+        for i, group in enumerate(groups):
+            numel = sum(p.numel() for p in group)
+            dtype = group[0].dtype
+            device = group[0].device
+            flat = torch.zeros(numel, dtype=dtype, device=device)
+            b = GradBucket(index=i, params=group, flat=flat)
+            off = 0
+            for p in group:
+                view = flat[off : off + p.numel()].view_as(p)
+                p.grad = view  # autograd accumulates in place into the bucket
+                b.views.append(view)
+                self._param_bucket[id(p)] = b
+                off += p.numel()
+            b.pending = len(group)
+            self.buckets.append(b)
+
+        # 3) readiness hooks
+        self._hooks = [
+            p.register_post_accumulate_grad_hook(self._make_hook(p))
+            for p in self.trainable
+        ]
+
+    # ------------------------------------------------------------- hooks
+    def _make_hook(self, p: torch.nn.Parameter):
+        def hook(param: torch.Tensor) -> None:
+            if not self._sync_enabled:
+                return
+            b = self._param_bucket[id(p)]
+            b.pending -= 1
+            if b.pending == 0 and self.overlap:
+                self._launch(b)
+
+        return hook
+
+    def _launch(self, b: GradBucket) -> None:
+        if b.launched:
+            return
+        b.work = self.comm.all_reduce(b.flat, async_op=True)
+        b.launched = True
+
+    # -------------------------------------------------------------- API
+    def forward(self, *args, **kwargs):
+        if self._sync_enabled:
+            self._require_finish = True
+        return self.module(*args, **kwargs)
+
+    def finish(self) -> None:
+        """End-of-backward: launch stragglers, wait all reduce works."""
+        if not self._sync_enabled:
+            return
+        for b in self.buckets:
+            if not b.launched:
+                self._launch(b)
+        for b in self.buckets:
+            if b.work is not None:
+                b.work.wait()  # stream-ordered on RCCL; host wait on gloo
+        self._reset()
+        self._require_finish = False
+
+    def _reset(self) -> None:
+        for b in self.buckets:
+            b.pending = len(b.params)
+            b.work = None
+            b.launched = False
+
+    def zero_grad(self) -> None:
+        """Zero the flat buffers (cheap contiguous memset per bucket)."""
+        for b in self.buckets:
+            b.flat.zero_()
+
+    @contextlib.contextmanager
+    def no_sync(self):
+        """Skip gradient reduction (gradient-accumulation micro-steps)."""
+        prev = self._sync_enabled
+        self._sync_enabled = False
+        try:
+            yield
+        finally:
+            self._sync_enabled = prev
+
+    @property
+    def grad_scale(self) -> float:
+        """1/world — folded into the fused AdamW step (SUM + pre-scale)."""
+        return 1.0 / max(1, self.comm.world_size)
+
+    def bucket_sizes_mb(self) -> List[float]:
+        return [b.nbytes / (1 << 20) for b in self.buckets]

@@ -1,0 +1,99 @@
+"""GPT-2-small (124M): LayerNorm, causal attention, GELU MLP, tied embeddings.
+
+Contract config 4.  Every hot op goes through scaffold.ops: LayerNorm
+(K2), flash attention (K4), softmax-cross-entropy (K3); GEMMs through
+torch.matmul (hipBLASLt) with the hand MFMA path arriving per SURVEY
+§2.3 K5.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+
+
+class LayerNorm(nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.bias = nn.Parameter(torch.zeros(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.layer_norm(x, self.weight, self.bias, self.eps)
+
+
+class CausalSelfAttention(nn.Module):
+    def __init__(self, d_model: int, n_heads: int):
+        super().__init__()
+        assert d_model % n_heads == 0
+        self.n_heads = n_heads
+        self.head_dim = d_model // n_heads
+        self.qkv = nn.Linear(d_model, 3 * d_model)
+        self.proj = nn.Linear(d_model, d_model)
+
+    def forward(self, x):
+        B, S, C = x.shape
+        qkv = self.qkv(x).view(B, S, 3, self.n_heads, self.head_dim)
+        q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # (B,H,S,D)
+        o = ops.attention(q, k, v, causal=True)
+        return self.proj(o.transpose(1, 2).reshape(B, S, C))
+
+
+class Block(nn.Module):
+    def __init__(self, d_model: int, n_heads: int, d_ff: int, eps: float):
+        super().__init__()
+        self.ln1 = LayerNorm(d_model, eps)
+        self.attn = CausalSelfAttention(d_model, n_heads)
+        self.ln2 = LayerNorm(d_model, eps)
+        self.fc = nn.Linear(d_model, d_ff)
+        self.proj = nn.Linear(d_ff, d_model)
+
+    def forward(self, x):
+        x = x + self.attn(self.ln1(x))
+        x = x + self.proj(F.gelu(self.fc(self.ln2(x)), approximate="tanh"))
+        return x
+
+
+class GPT2(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        m = cfg.model
+        d_ff = m.d_ff or 4 * m.d_model
+        self.wte = nn.Embedding(m.vocab_size, m.d_model)
+        self.wpe = nn.Embedding(m.max_seq_len, m.d_model)
+        self.blocks = nn.ModuleList(
+            Block(m.d_model, m.n_heads, d_ff, m.norm_eps) for _ in range(m.n_layers)
+        )
+        self.ln_f = LayerNorm(m.d_model, m.norm_eps)
+        self.lm_head = nn.Linear(m.d_model, m.vocab_size, bias=False)
+        if m.tie_embeddings:
+            self.lm_head.weight = self.wte.weight
+        self.apply(self._init)
+        # scaled init for residual projections (GPT-2 recipe)
+        for name, p in self.named_parameters():
+            if name.endswith("proj.weight"):
+                nn.init.normal_(p, std=0.02 / math.sqrt(2 * m.n_layers))
+
+    @staticmethod
+    def _init(module):
+        if isinstance(module, nn.Linear):
+            nn.init.normal_(module.weight, std=0.02)
+            if module.bias is not None:
+                nn.init.zeros_(module.bias)
+        elif isinstance(module, nn.Embedding):
+            nn.init.normal_(module.weight, std=0.02)
+
+    def forward(self, tokens: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+        B, S = tokens.shape
+        pos = torch.arange(S, device=tokens.device)
+        x = self.wte(tokens) + self.wpe(pos)
+        for blk in self.blocks:
+            x = blk(x)
+        logits = self.lm_head(self.ln_f(x))
+        return ops.softmax_cross_entropy(logits, targets)

@@ -1,0 +1,97 @@
+"""Llama-3 family: RMSNorm, RoPE, GQA attention, SwiGLU (contract config 5).
+
+The llama3_8b preset (32 layers, d=4096, 32/8 heads, d_ff=14336,
+V=128256) sizes the 288 GB-per-GPU HBM check: 8.0B params -> bf16
+working 16 GB + fp32 master/m/v 96 GB + activations, comfortably
+single-GPU DDP on MI355X.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-5):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        return ops.rms_norm(x, self.weight, self.eps)
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, d_model: int, n_heads: int, n_kv_heads: int):
+        super().__init__()
+        self.n_heads = n_heads
+        self.n_kv = n_kv_heads
+        self.head_dim = d_model // n_heads
+        self.wq = nn.Linear(d_model, n_heads * self.head_dim, bias=False)
+        self.wk = nn.Linear(d_model, n_kv_heads * self.head_dim, bias=False)
+        self.wv = nn.Linear(d_model, n_kv_heads * self.head_dim, bias=False)
+        self.wo = nn.Linear(n_heads * self.head_dim, d_model, bias=False)
+
+    def forward(self, x, cos, sin):
+        B, S, _ = x.shape
+        q = self.wq(x).view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
+        k = self.wk(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        v = self.wv(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        q = ops.rope_apply(q, cos, sin)
+        k = ops.rope_apply(k, cos, sin)
+        o = ops.attention(q, k, v, causal=True)
+        return self.wo(o.transpose(1, 2).reshape(B, S, -1))
+
+
+class LlamaBlock(nn.Module):
+    def __init__(self, d_model: int, n_heads: int, n_kv_heads: int, d_ff: int, eps: float):
+        super().__init__()
+        self.attn_norm = RMSNorm(d_model, eps)
+        self.attn = LlamaAttention(d_model, n_heads, n_kv_heads)
+        self.mlp_norm = RMSNorm(d_model, eps)
+        self.w_gate = nn.Linear(d_model, d_ff, bias=False)
+        self.w_up = nn.Linear(d_model, d_ff, bias=False)
+        self.w_down = nn.Linear(d_ff, d_model, bias=False)
+
+    def forward(self, x, cos, sin):
+        x = x + self.attn(self.attn_norm(x), cos, sin)
+        h = self.mlp_norm(x)
+        x = x + self.w_down(F.silu(self.w_gate(h)) * self.w_up(h))
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(self, cfg):
+        super().__init__()
+        m = cfg.model
+        d_ff = m.d_ff or int(8 * m.d_model / 3 + 255) // 256 * 256
+        n_kv = m.n_kv_heads or m.n_heads
+        self.cfg = cfg
+        self.embed = nn.Embedding(m.vocab_size, m.d_model)
+        self.blocks = nn.ModuleList(
+            LlamaBlock(m.d_model, m.n_heads, n_kv, d_ff, m.norm_eps)
+            for _ in range(m.n_layers)
+        )
+        self.norm_f = RMSNorm(m.d_model, m.norm_eps)
+        self.lm_head = nn.Linear(m.d_model, m.vocab_size, bias=False)
+        head_dim = m.d_model // m.n_heads
+        cos, sin = ops.rope_tables(m.max_seq_len, head_dim, m.rope_theta, "cpu")
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+        for p in self.parameters():
+            if p.dim() > 1:
+                nn.init.normal_(p, std=0.02)
+
+    def forward(self, tokens: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+        B, S = tokens.shape
+        cos = self.rope_cos[:S]
+        sin = self.rope_sin[:S]
+        x = self.embed(tokens)
+        for blk in self.blocks:
+            x = blk(x, cos, sin)
+        logits = self.lm_head(self.norm_f(x))
+        return ops.softmax_cross_entropy(logits, targets)

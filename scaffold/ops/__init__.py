@@ -1,0 +1,254 @@
+"""Hot ops (contract C5): fused AdamW, LayerNorm, softmax-cross-entropy,
+attention, RMSNorm/RoPE.
+
+Dispatch rule (C8 — no multi-backend dispatch on the compute path):
+  * tensors on a CUDA (ROCm) device -> the hand-written CDNA4 HIP kernels
+    from the in-tree extension, REQUIRED (loud error if the .so is
+    missing) so GPU runs can never silently fall back to eager;
+  * tensors on CPU -> plain PyTorch eager, the no-GPU test path (C10)
+    and the fp32 numerics reference the kernel tests compare against.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import backend
+from .backend import require_ext, has_ext, ext  # re-export
+
+__all__ = [
+    "layer_norm",
+    "rms_norm",
+    "rope_apply",
+    "softmax_cross_entropy",
+    "attention",
+    "fused_adamw_step",
+    "require_ext",
+    "has_ext",
+]
+
+
+# ---------------------------------------------------------------- LayerNorm
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        m = require_ext()
+        x2 = x.contiguous().view(-1, x.shape[-1])
+        y, mean, rstd = m.layernorm_fwd(x2, weight, bias, eps)
+        ctx.save_for_backward(x2, weight, mean, rstd)
+        ctx.x_shape = x.shape
+        return y.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, weight, mean, rstd = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        m = require_ext()
+        dx, dw, db = m.layernorm_bwd(dy2, x2, weight, mean, rstd)
+        return dx.view(ctx.x_shape), dw, db, None
+
+
+def layer_norm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+               eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        return _LayerNorm.apply(x, weight, bias, eps)
+    return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
+# ----------------------------------------------------------------- RMSNorm
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        m = require_ext()
+        x2 = x.contiguous().view(-1, x.shape[-1])
+        y, rstd = m.rmsnorm_fwd(x2, weight, eps)
+        ctx.save_for_backward(x2, weight, rstd)
+        ctx.x_shape = x.shape
+        return y.view(x.shape)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, weight, rstd = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        m = require_ext()
+        dx, dw = m.rmsnorm_bwd(dy2, x2, weight, rstd)
+        return dx.view(ctx.x_shape), dw, None
+
+
+def rms_norm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        return _RMSNorm.apply(x, weight, eps)
+    xf = x.float()
+    y = xf * torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (y * weight.float()).to(x.dtype)
+
+
+# -------------------------------------------------------------------- RoPE
+class _Rope(torch.autograd.Function):
+    """Rotary embedding on (B, H, S, D); cos/sin are (S, D/2) fp32."""
+
+    @staticmethod
+    def forward(ctx, x, cos, sin):
+        m = require_ext()
+        y = m.rope_apply(x.contiguous(), cos, sin, False)
+        ctx.save_for_backward(cos, sin)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos, sin = ctx.saved_tensors
+        m = require_ext()
+        dx = m.rope_apply(dy.contiguous(), cos, sin, True)  # inverse rotation
+        return dx, None, None
+
+
+def _rope_eager(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                inverse: bool = False) -> torch.Tensor:
+    # x: (B,H,S,D); pairs (x[2i], x[2i+1]) rotated by angle theta_i at pos s.
+    xf = x.float()
+    x1 = xf[..., 0::2]
+    x2 = xf[..., 1::2]
+    c = cos.view(1, 1, *cos.shape)
+    s = sin.view(1, 1, *sin.shape)
+    if inverse:
+        s = -s
+    o1 = x1 * c - x2 * s
+    o2 = x1 * s + x2 * c
+    out = torch.empty_like(xf)
+    out[..., 0::2] = o1
+    out[..., 1::2] = o2
+    return out.to(x.dtype)
+
+
+def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda:
+        return _Rope.apply(x, cos, sin)
+    if x.requires_grad:
+        # eager autograd handles the backward symbolically
+        return _rope_eager(x, cos, sin)
+    return _rope_eager(x, cos, sin)
+
+
+def rope_tables(seq_len: int, dim: int, theta: float, device, dtype=torch.float32
+                ) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Host-precomputed cos/sin tables (S, D/2) — Appendix B guidance:
+    never compute trig per element on device."""
+    inv = 1.0 / (theta ** (torch.arange(0, dim, 2, dtype=torch.float32) / dim))
+    t = torch.arange(seq_len, dtype=torch.float32)
+    ang = torch.outer(t, inv)
+    return ang.cos().to(device=device, dtype=dtype), ang.sin().to(device=device, dtype=dtype)
+
+
+# ------------------------------------------------- softmax cross-entropy
+class _SoftmaxXent(torch.autograd.Function):
+    """Fused log-softmax + NLL; saves softmax stats, fuses dlogits in bwd."""
+
+    @staticmethod
+    def forward(ctx, logits, targets):
+        m = require_ext()
+        logits2 = logits.contiguous().view(-1, logits.shape[-1])
+        t = targets.contiguous().view(-1)
+        loss, lse = m.softmax_xent_fwd(logits2, t)
+        ctx.save_for_backward(logits2, t, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits2, t, lse = ctx.saved_tensors
+        m = require_ext()
+        dlogits = m.softmax_xent_bwd(logits2, t, lse, float(dloss))
+        return dlogits.view(*t.shape, logits2.shape[-1]), None
+
+
+def softmax_cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+    """Mean cross-entropy over all positions. logits (..., V), targets (...)."""
+    if logits.is_cuda:
+        return _SoftmaxXent.apply(logits, targets)
+    return F.cross_entropy(
+        logits.view(-1, logits.shape[-1]).float(), targets.view(-1)
+    )
+
+
+# --------------------------------------------------------------- attention
+class _Attention(torch.autograd.Function):
+    """Flash-style causal attention on (B, H, S, D), GQA via n_kv_heads."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        m = require_ext()
+        o, lse = m.attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), causal)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        m = require_ext()
+        dq, dk, dv = m.attn_bwd(do.contiguous(), q, k, v, o, lse, ctx.causal)
+        return dq, dk, dv, None
+
+
+def _attention_eager(q, k, v, causal):
+    # GQA: expand kv heads
+    Hq, Hkv = q.shape[1], k.shape[1]
+    if Hkv != Hq:
+        rep = Hq // Hkv
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    return F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              causal: bool = True) -> torch.Tensor:
+    if q.is_cuda:
+        return _Attention.apply(q, k, v, causal)
+    return _attention_eager(q, k, v, causal)
+
+
+# ------------------------------------------------------------- fused AdamW
+def fused_adamw_step(
+    params_fp32: list,
+    grads: list,
+    exp_avgs: list,
+    exp_avg_sqs: list,
+    params_bf16: Optional[list],
+    *,
+    lr: float,
+    beta1: float,
+    beta2: float,
+    eps: float,
+    weight_decay: float,
+    step: int,
+    grad_scale: float = 1.0,
+) -> None:
+    """One decoupled-AdamW step over many tensors in one kernel launch.
+
+    fp32 master params; grads may be bf16 or fp32; optionally also writes
+    the bf16 working copy (params_bf16) in the same pass (K8 folded into
+    K1 per SURVEY §2.3).
+    """
+    if params_fp32 and params_fp32[0].is_cuda:
+        m = require_ext()
+        m.fused_adamw(
+            params_fp32, grads, exp_avgs, exp_avg_sqs,
+            params_bf16 if params_bf16 is not None else [],
+            lr, beta1, beta2, eps, weight_decay, step, grad_scale,
+        )
+        return
+    # CPU eager reference (used by tests as ground truth)
+    bc1 = 1.0 - beta1 ** step
+    bc2 = 1.0 - beta2 ** step
+    for i, p in enumerate(params_fp32):
+        g = grads[i].float() * grad_scale
+        ea, eas = exp_avgs[i], exp_avg_sqs[i]
+        p.mul_(1.0 - lr * weight_decay)
+        ea.mul_(beta1).add_(g, alpha=1 - beta1)
+        eas.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+        denom = (eas / bc2).sqrt_().add_(eps)
+        p.addcdiv_(ea / bc1, denom, value=-lr)
+        if params_bf16 is not None and params_bf16:
+            params_bf16[i].copy_(p.to(params_bf16[i].dtype))

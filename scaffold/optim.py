@@ -1,0 +1,111 @@
+"""AdamW with fp32 master weights and a single fused multi-tensor step.
+
+The model holds bf16 (or fp32) working parameters; the optimizer owns
+fp32 masters + moments and updates everything in ONE multi-tensor HIP
+kernel launch per step on GPU (K1, pure-bandwidth — SURVEY §2.3), which
+also refreshes the bf16 working copy in the same pass (K8 folded in).
+"""
+
+from __future__ import annotations
+
+from typing import Iterable, List, Optional
+
+import torch
+
+from . import ops
+
+
+class MasterAdamW:
+    def __init__(
+        self,
+        params: Iterable[torch.nn.Parameter],
+        lr: float = 1e-3,
+        betas=(0.9, 0.95),
+        eps: float = 1e-8,
+        weight_decay: float = 0.01,
+        fused: bool = True,
+    ):
+        self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
+        if not self.params:
+            raise ValueError("no trainable parameters")
+        self.lr = lr
+        self.beta1, self.beta2 = betas
+        self.eps = eps
+        self.weight_decay = weight_decay
+        self.fused = fused
+        self.step_count = 0
+        self.mixed = any(p.dtype != torch.float32 for p in self.params)
+        # fp32 masters + moments
+        self.master: List[torch.Tensor] = [
+            p.detach().to(torch.float32).clone() for p in self.params
+        ]
+        self.exp_avg = [torch.zeros_like(m) for m in self.master]
+        self.exp_avg_sq = [torch.zeros_like(m) for m in self.master]
+
+    def zero_grad(self, set_to_none: bool = False) -> None:
+        for p in self.params:
+            if p.grad is not None:
+                if set_to_none:
+                    p.grad = None
+                else:
+                    p.grad.zero_()
+
+    @torch.no_grad()
+    def step(self, grad_scale: float = 1.0) -> None:
+        self.step_count += 1
+        grads = []
+        masters, eas, eass, working = [], [], [], []
+        for i, p in enumerate(self.params):
+            if p.grad is None:
+                continue
+            grads.append(p.grad)
+            masters.append(self.master[i])
+            eas.append(self.exp_avg[i])
+            eass.append(self.exp_avg_sq[i])
+            working.append(p.data)
+        bf16_out = working if self.mixed else None
+        ops.fused_adamw_step(
+            masters, grads, eas, eass, bf16_out,
+            lr=self.lr, beta1=self.beta1, beta2=self.beta2, eps=self.eps,
+            weight_decay=self.weight_decay, step=self.step_count,
+            grad_scale=grad_scale,
+        )
+        if not self.mixed:
+            # masters ARE authoritative; copy into working fp32 params
+            for w, m in zip(working, masters):
+                w.copy_(m)
+
+    def state_dict(self) -> dict:
+        return {
+            "step": self.step_count,
+            "lr": self.lr,
+            "master": self.master,
+            "exp_avg": self.exp_avg,
+            "exp_avg_sq": self.exp_avg_sq,
+        }
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.step_count = sd["step"]
+        self.lr = sd.get("lr", self.lr)
+        for dst, src in zip(self.master, sd["master"]):
+            dst.copy_(src.to(dst.device))
+        for dst, src in zip(self.exp_avg, sd["exp_avg"]):
+            dst.copy_(src.to(dst.device))
+        for dst, src in zip(self.exp_avg_sq, sd["exp_avg_sq"]):
+            dst.copy_(src.to(dst.device))
+        # refresh working params from masters
+        with torch.no_grad():
+            for p, m in zip(self.params, self.master):
+                p.data.copy_(m.to(p.dtype))
+
+
+def build_optimizer(cfg, model: torch.nn.Module) -> MasterAdamW:
+    o = cfg.optim
+    return MasterAdamW(
+        model.parameters(),
+        lr=o.lr,
+        betas=tuple(o.betas),
+        eps=o.eps,
+        weight_decay=o.weight_decay,
+        fused=o.fused,
+    )

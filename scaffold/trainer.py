@@ -1,0 +1,167 @@
+"""Trainer — the drop-in API surface (contract C1): Trainer(cfg).fit().
+
+Step loop: synthetic batch (device-side) -> forward ->
+backward (bucket all-reduces fire from grad hooks, overlapped) ->
+ddp.finish() -> fused AdamW step (grad averaged via grad_scale=1/world)
+-> metrics / checkpoint cadence.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Optional
+
+import torch
+
+from .config import TrainConfig, config_hash
+from .data.synthetic import build_data
+from .dist.comm import Comm
+from .dist.ddp import BucketedDDP
+from .metrics import MetricLogger
+from .models import build_model
+from .optim import build_optimizer, MasterAdamW
+from .runtime import DistContext, init_distributed, teardown
+
+
+DTYPES = {"bf16": torch.bfloat16, "fp32": torch.float32}
+
+
+class Trainer:
+    def __init__(
+        self,
+        cfg: TrainConfig,
+        ctx: Optional[DistContext] = None,
+        model: Optional[torch.nn.Module] = None,
+        data=None,
+        comm: Optional[Comm] = None,
+    ):
+        self.cfg = cfg
+        self.ctx = ctx if ctx is not None else init_distributed(cfg)
+        self.dtype = DTYPES[cfg.dtype]
+        self.device = self.ctx.device
+        # Deterministic init regardless of how ctx was constructed; rank 0's
+        # init wins via the wrap-time broadcast, so model init matches a
+        # single-process run with the same seed.
+        from .runtime import set_seed
+
+        set_seed(cfg.seed, self.ctx.rank)
+
+        raw_model = model if model is not None else build_model(cfg)
+        raw_model = raw_model.to(self.device)
+        if self.dtype is not torch.float32:
+            raw_model = self._to_mixed(raw_model)
+        self.raw_model = raw_model
+
+        self.ddp = BucketedDDP(
+            raw_model,
+            comm=comm,
+            bucket_cap_mb=cfg.dist.bucket_cap_mb,
+            first_bucket_mb=cfg.dist.first_bucket_mb,
+            overlap=cfg.dist.overlap,
+        )
+        self.model = self.ddp
+        self.optimizer: MasterAdamW = build_optimizer(cfg, raw_model)
+        self.data = data if data is not None else build_data(
+            cfg, self.ctx.rank, self.device, self.dtype
+        )
+        self.logger = MetricLogger(self.ctx, cfg)
+        self.step = 0
+        self.cfg_hash = config_hash(cfg)
+        if cfg.ckpt.resume:
+            self._resume(cfg.ckpt.resume)
+
+    def _to_mixed(self, model: torch.nn.Module) -> torch.nn.Module:
+        """bf16 working weights; BatchNorm stays fp32 (numerics)."""
+        model = model.to(self.dtype)
+        for mod in model.modules():
+            if isinstance(mod, (torch.nn.BatchNorm1d, torch.nn.BatchNorm2d)):
+                mod.float()
+        return model
+
+    # --------------------------------------------------------------- fit
+    def fit(self, steps: Optional[int] = None):
+        total = steps if steps is not None else self.cfg.steps
+        accum = max(1, self.cfg.grad_accum)
+        batch_iter = iter(self.data)
+        while self.step < total:
+            t0 = time.perf_counter()
+            loss_val = self._train_step(batch_iter, accum)
+            dt = time.perf_counter() - t0
+            self.step += 1
+            self.logger.log_step(
+                self.step, loss_val, self.cfg.data.batch_size * accum, dt
+            )
+            if (
+                self.cfg.ckpt.interval
+                and self.step % self.cfg.ckpt.interval == 0
+                and self.cfg.ckpt.dir
+            ):
+                self.save_checkpoint()
+        if self.cfg.ckpt.save_final and self.cfg.ckpt.dir:
+            self.save_checkpoint()
+        return self
+
+    def _train_step(self, batch_iter, accum: int) -> float:
+        self.ddp.zero_grad()
+        loss_val = 0.0
+        for micro in range(accum):
+            x, y = next(batch_iter)
+            if micro < accum - 1:
+                with self.ddp.no_sync():
+                    loss = self.model(x, y)
+                    loss.backward()
+            else:
+                loss = self.model(x, y)
+                loss.backward()
+                self.ddp.finish()
+            loss_val += loss.detach().float().item() / accum
+        self.optimizer.step(grad_scale=self.ddp.grad_scale / accum)
+        return loss_val
+
+    # -------------------------------------------------------------- eval
+    @torch.no_grad()
+    def evaluate(self, eval_steps: int = 10) -> float:
+        self.raw_model.eval()
+        total = 0.0
+        it = iter(self.data)
+        for _ in range(eval_steps):
+            x, y = next(it)
+            total += self.model.module(x, y).float().item()
+        self.raw_model.train()
+        t = torch.tensor([total / eval_steps])
+        self.ddp.comm.all_reduce(t)
+        return t.item() / max(1, self.ddp.comm.world_size)
+
+    # --------------------------------------------------------- checkpoint
+    def save_checkpoint(self, path: Optional[str] = None) -> str:
+        from .ckpt.sharded import save_sharded
+
+        out = path or os.path.join(self.cfg.ckpt.dir, f"step_{self.step:08d}")
+        save_sharded(
+            out,
+            step=self.step,
+            model=self.raw_model,
+            optimizer=self.optimizer,
+            data=self.data,
+            ctx=self.ctx,
+            cfg_hash=self.cfg_hash,
+            async_d2h=self.cfg.ckpt.async_d2h,
+        )
+        return out
+
+    def _resume(self, path: str) -> None:
+        from .ckpt.sharded import load_sharded
+
+        self.step = load_sharded(
+            path,
+            model=self.raw_model,
+            optimizer=self.optimizer,
+            data=self.data,
+            ctx=self.ctx,
+            cfg_hash=self.cfg_hash,
+        )
+
+    def close(self) -> None:
+        self.logger.close()
+        teardown(self.ctx)
