@@ -1,0 +1,105 @@
+// Common helpers for the scaffold's CDNA4 (gfx950) HIP kernels.
+// Target: MI355X only — wave64, 256 CUs / 8 XCDs, 160 KiB LDS/CU.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+// torch's current stream for this device (ROCm-native name; avoids the
+// hipify rewrite of at::cuda::getCurrentCUDAStream)
+inline hipStream_t scaffold_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define WAVE 64
+#define DEVINL __device__ __forceinline__
+
+#define HIP_CHECK(expr)                                                        \
+  do {                                                                         \
+    hipError_t _e = (expr);                                                    \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));       \
+  } while (0)
+
+#define CHECK_CUDA(x) TORCH_CHECK(x.is_cuda(), #x " must be a HIP tensor")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+using bf16 = __hip_bfloat16;
+
+// vector types for wide loads (G13: always vectorize bf16 as short4/short8)
+typedef short bf16x4 __attribute__((ext_vector_type(4)));   // 8 B
+typedef short bf16x8 __attribute__((ext_vector_type(8)));   // 16 B
+typedef float f32x4 __attribute__((ext_vector_type(4)));    // 16 B
+typedef float f32x16 __attribute__((ext_vector_type(16)));
+
+DEVINL float bf2f(short u) {
+  union { unsigned int i; float f; } c;
+  c.i = ((unsigned int)(unsigned short)u) << 16;
+  return c.f;
+}
+
+DEVINL short f2bf(float f) {
+  // round-to-nearest-even bf16 conversion
+  union { float f; unsigned int i; } c;
+  c.f = f;
+  unsigned int x = c.i;
+  unsigned int rounded = x + 0x7FFF + ((x >> 16) & 1);
+  if ((x & 0x7F800000u) == 0x7F800000u) rounded = x;  // inf/nan passthrough
+  return (short)(rounded >> 16);
+}
+
+// wave-level reductions (wave64)
+DEVINL float wave_reduce_sum(float v) {
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+DEVINL float wave_reduce_max(float v) {
+  #pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+// block reduction via LDS; `red` shared array must hold >= blockDim.x/64 floats
+DEVINL float block_reduce_sum(float v, float* red) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+  v = wave_reduce_sum(v);
+  if (lane == 0) red[wid] = v;
+  __syncthreads();
+  v = (lane < nwaves) ? red[lane] : 0.f;
+  v = wave_reduce_sum(v);
+  // broadcast through lane 0 of wave 0
+  if (threadIdx.x == 0) red[0] = v;
+  __syncthreads();
+  float out = red[0];
+  __syncthreads();
+  return out;
+}
+
+DEVINL float block_reduce_max(float v, float* red) {
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+  v = wave_reduce_max(v);
+  if (lane == 0) red[wid] = v;
+  __syncthreads();
+  v = (lane < nwaves) ? red[lane] : -INFINITY;
+  v = wave_reduce_max(v);
+  if (threadIdx.x == 0) red[0] = v;
+  __syncthreads();
+  float out = red[0];
+  __syncthreads();
+  return out;
+}
+
+// Grid sizing guideline 11: cap grid, grid-stride the rest.
+DEVINL long grid_stride() { return (long)gridDim.x * blockDim.x; }
+
+inline int capped_blocks(long total_threads, int block = 256, int cap = 2048) {
+  long b = (total_threads + block - 1) / block;
+  return (int)std::min<long>(b, cap);
+}

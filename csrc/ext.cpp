@@ -1,0 +1,48 @@
+// pybind11 glue for the scaffold's gfx950 HIP kernels.
+#include <torch/extension.h>
+
+#include <vector>
+
+void fused_adamw(std::vector<torch::Tensor> masters,
+                 std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> exp_avgs,
+                 std::vector<torch::Tensor> exp_avg_sqs,
+                 std::vector<torch::Tensor> outs, double lr, double beta1,
+                 double beta2, double eps, double weight_decay, long step,
+                 double grad_scale);
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+std::vector<torch::Tensor> softmax_xent_fwd(torch::Tensor logits,
+                                            torch::Tensor targets);
+torch::Tensor softmax_xent_bwd(torch::Tensor logits, torch::Tensor targets,
+                               torch::Tensor lse, double dloss);
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor w,
+                                       double eps);
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                       torch::Tensor w, torch::Tensor rstd);
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cos_t,
+                         torch::Tensor sin_t, bool inverse);
+std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
+                                    torch::Tensor v, bool causal);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
+                                    torch::Tensor k, torch::Tensor v,
+                                    torch::Tensor o, torch::Tensor lse,
+                                    bool causal);
+torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B, long which);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (K1)");
+  m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (K2)");
+  m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (K2)");
+  m.def("softmax_xent_fwd", &softmax_xent_fwd, "softmax cross-entropy fwd (K3)");
+  m.def("softmax_xent_bwd", &softmax_xent_bwd, "softmax cross-entropy bwd (K3)");
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (K7)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (K7)");
+  m.def("rope_apply", &rope_apply, "RoPE apply/inverse (K7)");
+  m.def("attn_fwd", &attn_fwd, "flash attention forward (K4)");
+  m.def("attn_bwd", &attn_bwd, "flash attention backward (K4)");
+  m.def("probe_mfma", &probe_mfma, "MFMA fragment-layout probe (tests)");
+}

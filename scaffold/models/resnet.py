@@ -17,6 +17,17 @@ import torch.nn as nn
 from .. import ops
 
 
+class Fp32BatchNorm(nn.BatchNorm2d):
+    """BatchNorm computed in fp32 regardless of activation dtype.
+
+    Trainer._to_mixed keeps BN params/running-stats fp32; this wrapper
+    casts the bf16 activation up and back so the module works under the
+    scaffold's explicit-bf16 regime (no autocast)."""
+
+    def forward(self, x):
+        return super().forward(x.float()).to(x.dtype)
+
+
 class Bottleneck(nn.Module):
     expansion = 4
 
@@ -24,11 +35,11 @@ class Bottleneck(nn.Module):
         super().__init__()
         out_ch = width * self.expansion
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = nn.BatchNorm2d(width)
+        self.bn1 = Fp32BatchNorm(width)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1, bias=False)
-        self.bn2 = nn.BatchNorm2d(width)
+        self.bn2 = Fp32BatchNorm(width)
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
-        self.bn3 = nn.BatchNorm2d(out_ch)
+        self.bn3 = Fp32BatchNorm(out_ch)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
 
@@ -50,7 +61,7 @@ class ResNet50(nn.Module):
         num_classes = cfg.model.num_classes
         self.in_ch = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = Fp32BatchNorm(64)
         self.relu = nn.ReLU(inplace=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, self.layers[0], stride=1)
@@ -77,7 +88,7 @@ class ResNet50(nn.Module):
         if stride != 1 or self.in_ch != out_ch:
             downsample = nn.Sequential(
                 nn.Conv2d(self.in_ch, out_ch, 1, stride=stride, bias=False),
-                nn.BatchNorm2d(out_ch),
+                Fp32BatchNorm(out_ch),
             )
         layers = [Bottleneck(self.in_ch, width, stride, downsample)]
         self.in_ch = out_ch

@@ -29,9 +29,7 @@ _SOURCES = [
     "kernels/rmsnorm_rope.hip",
     "kernels/attn_fwd.hip",
     "kernels/attn_bwd.hip",
-    "kernels/gemm_bf16.hip",
-    "kernels/bucket_copy.hip",
-    "kernels/ckpt_stage.cpp",
+    "kernels/probe.hip",
 ]
 
 

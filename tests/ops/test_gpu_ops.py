@@ -1,0 +1,213 @@
+"""Op-level GPU numerics tier: each HIP kernel vs a plain
+PyTorch fp32 reference of the same op.  All tests @pytest.mark.gpu."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from scaffold.ops.backend import require_ext
+
+    return require_ext()
+
+
+def _bf(x):
+    return x.to(torch.bfloat16)
+
+
+# ------------------------------------------------------------------ MFMA probe
+def test_mfma_layout_32x32x16(ext):
+    torch.manual_seed(0)
+    A = torch.randn(32, 16).cuda()
+    B = torch.randn(16, 32).cuda()  # asymmetric (guide G9)
+    C = ext.probe_mfma(_bf(A), _bf(B), 0)
+    ref = _bf(A).float() @ _bf(B).float()
+    assert torch.allclose(C, ref, rtol=1e-2, atol=1e-2), (C - ref).abs().max()
+
+
+def test_mfma_layout_16x16x32(ext):
+    torch.manual_seed(1)
+    A = torch.randn(16, 32).cuda()
+    B = torch.randn(32, 16).cuda()
+    C = ext.probe_mfma(_bf(A), _bf(B), 1)
+    ref = _bf(A).float() @ _bf(B).float()
+    assert torch.allclose(C, ref, rtol=1e-2, atol=1e-2), (C - ref).abs().max()
+
+
+# ------------------------------------------------------------------- AdamW
+@pytest.mark.parametrize("grad_dtype", [torch.bfloat16, torch.float32])
+def test_fused_adamw_vs_cpu(ext, grad_dtype):
+    from scaffold import ops
+
+    torch.manual_seed(2)
+    shapes = [(1000,), (128, 77), (3,), (513,)]
+    masters = [torch.randn(s) for s in shapes]
+    grads = [torch.randn(s).to(grad_dtype) for s in shapes]
+    ms = [torch.rand(s) * 0.1 for s in shapes]
+    vs = [torch.rand(s) * 0.01 for s in shapes]
+    outs = [m.to(torch.bfloat16) for m in masters]
+    kw = dict(lr=1e-2, beta1=0.9, beta2=0.95, eps=1e-8, weight_decay=0.01,
+              step=3, grad_scale=0.125)
+
+    # CPU reference (the eager path in scaffold.ops)
+    cm = [t.clone() for t in masters]
+    cg = [t.clone() for t in grads]
+    cms = [t.clone() for t in ms]
+    cvs = [t.clone() for t in vs]
+    cout = [t.clone() for t in outs]
+    ops.fused_adamw_step(cm, cg, cms, cvs, cout, **kw)
+
+    gm = [t.cuda() for t in masters]
+    gg = [t.cuda() for t in grads]
+    gms = [t.cuda() for t in ms]
+    gvs = [t.cuda() for t in vs]
+    gout = [t.cuda() for t in outs]
+    ext.fused_adamw(gm, gg, gms, gvs, gout, kw["lr"], kw["beta1"], kw["beta2"],
+                    kw["eps"], kw["weight_decay"], kw["step"], kw["grad_scale"])
+    torch.cuda.synchronize()
+    for i in range(len(shapes)):
+        assert torch.allclose(gm[i].cpu(), cm[i], rtol=1e-5, atol=1e-6), f"master {i}"
+        assert torch.allclose(gms[i].cpu(), cms[i], rtol=1e-5, atol=1e-6), f"m {i}"
+        assert torch.allclose(gvs[i].cpu(), cvs[i], rtol=1e-5, atol=1e-6), f"v {i}"
+        assert torch.allclose(gout[i].cpu().float(), cout[i].float(),
+                              rtol=1e-2, atol=1e-2), f"out {i}"
+
+
+# --------------------------------------------------------------- LayerNorm
+@pytest.mark.parametrize("R,D", [(128, 768), (64, 1024), (7, 333)])
+def test_layernorm_fwd_bwd(ext, R, D):
+    torch.manual_seed(3)
+    x32 = torch.randn(R, D, requires_grad=True)
+    w32 = torch.randn(D, requires_grad=True)
+    b32 = torch.randn(D, requires_grad=True)
+    y32 = torch.nn.functional.layer_norm(x32, (D,), w32, b32, 1e-5)
+    dy = torch.randn(R, D)
+    y32.backward(dy)
+
+    from scaffold import ops
+
+    xg = _bf(x32.detach()).cuda().requires_grad_(True)
+    wg = _bf(w32.detach()).cuda().requires_grad_(True)
+    bg = _bf(b32.detach()).cuda().requires_grad_(True)
+    yg = ops.layer_norm(xg, wg, bg, 1e-5)
+    yg.backward(_bf(dy).cuda())
+    torch.cuda.synchronize()
+
+    assert torch.allclose(yg.detach().cpu().float(), y32.detach().float(),
+                          rtol=5e-2, atol=5e-2)
+    assert torch.allclose(xg.grad.cpu().float(), x32.grad.float(),
+                          rtol=1e-1, atol=1e-1)
+    # column reductions accumulate over R rows -> looser bands
+    assert torch.allclose(wg.grad.cpu().float(), w32.grad.float(),
+                          rtol=1e-1, atol=R * 2e-2)
+    assert torch.allclose(bg.grad.cpu().float(), b32.grad.float(),
+                          rtol=1e-1, atol=R * 2e-2)
+
+
+# ----------------------------------------------------------------- RMSNorm
+def test_rmsnorm_fwd_bwd(ext):
+    torch.manual_seed(4)
+    R, D = 256, 512
+    x32 = torch.randn(R, D, requires_grad=True)
+    w32 = torch.randn(D, requires_grad=True)
+    rms = x32 * torch.rsqrt(x32.pow(2).mean(-1, keepdim=True) + 1e-5) * w32
+    dy = torch.randn(R, D)
+    rms.backward(dy)
+
+    from scaffold import ops
+
+    xg = _bf(x32.detach()).cuda().requires_grad_(True)
+    wg = _bf(w32.detach()).cuda().requires_grad_(True)
+    yg = ops.rms_norm(xg, wg, 1e-5)
+    yg.backward(_bf(dy).cuda())
+    torch.cuda.synchronize()
+    assert torch.allclose(yg.detach().cpu().float(), rms.detach().float(),
+                          rtol=5e-2, atol=5e-2)
+    assert torch.allclose(xg.grad.cpu().float(), x32.grad.float(),
+                          rtol=1e-1, atol=1e-1)
+    assert torch.allclose(wg.grad.cpu().float(), w32.grad.float(),
+                          rtol=1e-1, atol=R * 2e-2)
+
+
+# -------------------------------------------------------------------- RoPE
+def test_rope_roundtrip_and_ref(ext):
+    from scaffold import ops
+
+    torch.manual_seed(5)
+    B, H, S, D = 2, 4, 64, 128
+    x = torch.randn(B, H, S, D)
+    cos, sin = ops.rope_tables(S, D, 500000.0, "cpu")
+    ref = ops._rope_eager(x, cos, sin)
+
+    xg = _bf(x).cuda()
+    y = ext.rope_apply(xg, cos.cuda(), sin.cuda(), False)
+    torch.cuda.synchronize()
+    assert torch.allclose(y.cpu().float(), ref.float(), rtol=3e-2, atol=3e-2)
+    # inverse rotation undoes the forward
+    x2 = ext.rope_apply(y, cos.cuda(), sin.cuda(), True)
+    assert torch.allclose(x2.cpu().float(), x.float(), rtol=5e-2, atol=5e-2)
+
+
+# --------------------------------------------------- softmax cross-entropy
+@pytest.mark.parametrize("R,V", [(64, 1000), (32, 50257)])
+def test_softmax_xent(ext, R, V):
+    torch.manual_seed(6)
+    logits32 = (torch.randn(R, V) * 3).requires_grad_(True)
+    tgt = torch.randint(0, V, (R,))
+    loss32 = torch.nn.functional.cross_entropy(logits32, tgt)
+    loss32.backward()
+
+    from scaffold import ops
+
+    lg = _bf(logits32.detach()).cuda().requires_grad_(True)
+    lossg = ops.softmax_cross_entropy(lg, tgt.cuda())
+    lossg.backward()
+    torch.cuda.synchronize()
+    assert abs(lossg.item() - loss32.item()) < 0.05 * max(1.0, abs(loss32.item()))
+    assert torch.allclose(lg.grad.cpu().float(), logits32.grad.float(),
+                          rtol=1e-1, atol=1e-3)
+
+
+# --------------------------------------------------------------- attention
+@pytest.mark.parametrize("B,H,Hkv,S,D,causal", [
+    (2, 4, 4, 128, 64, True),
+    (1, 8, 2, 256, 128, True),   # GQA
+    (2, 2, 2, 96, 64, False),    # non-causal + ragged S
+])
+def test_attention_fwd_bwd(ext, B, H, Hkv, S, D, causal):
+    torch.manual_seed(7)
+    q32 = torch.randn(B, H, S, D, requires_grad=True)
+    k32 = torch.randn(B, Hkv, S, D, requires_grad=True)
+    v32 = torch.randn(B, Hkv, S, D, requires_grad=True)
+    rep = H // Hkv
+    kf = k32.repeat_interleave(rep, dim=1)
+    vf = v32.repeat_interleave(rep, dim=1)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q32, kf, vf, is_causal=causal)
+    do = torch.randn(B, H, S, D)
+    ref.backward(do)
+
+    from scaffold import ops
+
+    qg = _bf(q32.detach()).cuda().requires_grad_(True)
+    kg = _bf(k32.detach()).cuda().requires_grad_(True)
+    vg = _bf(v32.detach()).cuda().requires_grad_(True)
+    og = ops.attention(qg, kg, vg, causal=causal)
+    og.backward(_bf(do).cuda())
+    torch.cuda.synchronize()
+
+    assert torch.allclose(og.detach().cpu().float(), ref.detach().float(),
+                          rtol=5e-2, atol=5e-2), \
+        (og.detach().cpu().float() - ref.detach().float()).abs().max()
+    for name, got, want in [("dq", qg.grad, q32.grad), ("dk", kg.grad, k32.grad),
+                            ("dv", vg.grad, v32.grad)]:
+        g = got.cpu().float()
+        w = want.float()
+        err = (g - w).abs().max().item()
+        tol = 0.12 * max(1.0, w.abs().max().item())
+        assert err < tol, f"{name}: max err {err} vs tol {tol}"
