@@ -87,14 +87,25 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
     rank = ctx.rank
 
     named = _named_state_tensors(model, optimizer)
+    # dedupe aliased storages (fp32 params alias their optimizer master;
+    # tied embeddings alias lm_head) — safetensors forbids shared tensors
+    seen_ptr: Dict[int, str] = {}
+    unique, aliases = [], {}
+    for name, t in named:
+        ptr = t.untyped_storage().data_ptr()
+        if ptr in seen_ptr and t.numel() == t.untyped_storage().nbytes() // max(1, t.element_size()):
+            aliases[name] = seen_ptr[ptr]
+        else:
+            seen_ptr.setdefault(ptr, name)
+            unique.append((name, t))
     owner = layout.partition_by_size(
-        [(n, t.numel() * t.element_size()) for n, t in named], world
+        [(n, t.numel() * t.element_size()) for n, t in unique], world
     )
 
     stager = _PinnedStager(ctx.device) if async_d2h else None
     shard: Dict[str, torch.Tensor] = {}
     tensor_meta: Dict[str, dict] = {}
-    for name, t in named:
+    for name, t in unique:
         tensor_meta[name] = {
             "shard": owner[name],
             "shape": list(t.shape),
@@ -103,6 +114,8 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
         if owner[name] == rank:
             host = stager.to_host(t) if stager else t.detach().cpu()
             shard[name] = host.contiguous()
+    for name, target in aliases.items():
+        tensor_meta[name] = {"alias": target}
 
     # rank-local RNG + data state (every shard)
     shard[f"rng.torch.{rank}"] = torch.random.get_rng_state()
@@ -149,6 +162,8 @@ def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = 
 
     def fetch(name: str) -> torch.Tensor:
         info = m["tensors"][name]
+        if "alias" in info:
+            return fetch(info["alias"])
         return handles[info["shard"]].get_tensor(name)
 
     with torch.no_grad():

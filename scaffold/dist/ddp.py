@@ -65,21 +65,31 @@ def partition_buckets(
 
     Pure function — unit-tested on CPU (SURVEY §4 unit tier).
     """
-    buckets: List[List[torch.nn.Parameter]] = []
-    cur: List[torch.nn.Parameter] = []
-    cur_bytes = 0
-    cap = first_bucket_mb * (1 << 20)
+    # Buckets must be dtype-homogeneous (one flat comm buffer per bucket).
+    # Partition each dtype's params as its own stream (reverse order kept)
+    # instead of splitting at every dtype flip — a bf16 model with fp32
+    # BatchNorm params interleaved would otherwise shatter into ~2x-per-
+    # layer tiny buckets.  Readiness hooks make cross-stream ordering safe.
+    streams: Dict[torch.dtype, List[torch.nn.Parameter]] = {}
     for p in reversed(params):
-        p_bytes = p.numel() * p.element_size()
-        if cur and cur_bytes + p_bytes > cap:
+        streams.setdefault(p.dtype, []).append(p)
+
+    buckets: List[List[torch.nn.Parameter]] = []
+    for plist in streams.values():
+        cur: List[torch.nn.Parameter] = []
+        cur_bytes = 0
+        cap = first_bucket_mb * (1 << 20) if not buckets else bucket_cap_mb * (1 << 20)
+        for p in plist:
+            p_bytes = p.numel() * p.element_size()
+            if cur and cur_bytes + p_bytes > cap:
+                buckets.append(cur)
+                cur = []
+                cur_bytes = 0
+                cap = bucket_cap_mb * (1 << 20)
+            cur.append(p)
+            cur_bytes += p_bytes
+        if cur:
             buckets.append(cur)
-            cur = []
-            cur_bytes = 0
-            cap = bucket_cap_mb * (1 << 20)
-        cur.append(p)
-        cur_bytes += p_bytes
-    if cur:
-        buckets.append(cur)
     return buckets
 
 

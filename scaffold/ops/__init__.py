@@ -153,7 +153,7 @@ class _SoftmaxXent(torch.autograd.Function):
         t = targets.contiguous().view(-1)
         loss, lse = m.softmax_xent_fwd(logits2, t)
         ctx.save_for_backward(logits2, t, lse)
-        return loss
+        return loss.mean()  # mean reduction; bwd kernel folds the 1/R
 
     @staticmethod
     def backward(ctx, dloss):

@@ -34,10 +34,11 @@ class MasterAdamW:
         self.weight_decay = weight_decay
         self.fused = fused
         self.step_count = 0
-        self.mixed = any(p.dtype != torch.float32 for p in self.params)
-        # fp32 masters + moments
+        # fp32 masters + moments.  fp32 working params ALIAS their master
+        # (no copy-back pass); low-precision params get a separate master.
         self.master: List[torch.Tensor] = [
-            p.detach().to(torch.float32).clone() for p in self.params
+            p.data if p.dtype == torch.float32 else p.detach().to(torch.float32).clone()
+            for p in self.params
         ]
         self.exp_avg = [torch.zeros_like(m) for m in self.master]
         self.exp_avg_sq = [torch.zeros_like(m) for m in self.master]
@@ -53,27 +54,28 @@ class MasterAdamW:
     @torch.no_grad()
     def step(self, grad_scale: float = 1.0) -> None:
         self.step_count += 1
-        grads = []
-        masters, eas, eass, working = [], [], [], []
+        # one fused launch per homogeneous group: {bf16 working + bf16-out}
+        # and {fp32 working aliasing master, no out}
+        groups = {}  # (grad_dtype, needs_out) -> lists
         for i, p in enumerate(self.params):
             if p.grad is None:
                 continue
-            grads.append(p.grad)
-            masters.append(self.master[i])
-            eas.append(self.exp_avg[i])
-            eass.append(self.exp_avg_sq[i])
-            working.append(p.data)
-        bf16_out = working if self.mixed else None
-        ops.fused_adamw_step(
-            masters, grads, eas, eass, bf16_out,
-            lr=self.lr, beta1=self.beta1, beta2=self.beta2, eps=self.eps,
-            weight_decay=self.weight_decay, step=self.step_count,
-            grad_scale=grad_scale,
-        )
-        if not self.mixed:
-            # masters ARE authoritative; copy into working fp32 params
-            for w, m in zip(working, masters):
-                w.copy_(m)
+            needs_out = p.dtype != torch.float32
+            key = (p.grad.dtype, needs_out)
+            g = groups.setdefault(key, ([], [], [], [], []))
+            g[0].append(self.master[i])
+            g[1].append(p.grad)
+            g[2].append(self.exp_avg[i])
+            g[3].append(self.exp_avg_sq[i])
+            if needs_out:
+                g[4].append(p.data)
+        for (gdt, needs_out), (masters, grads, eas, eass, outs) in groups.items():
+            ops.fused_adamw_step(
+                masters, grads, eas, eass, outs if needs_out else None,
+                lr=self.lr, beta1=self.beta1, beta2=self.beta2, eps=self.eps,
+                weight_decay=self.weight_decay, step=self.step_count,
+                grad_scale=grad_scale,
+            )
 
     def state_dict(self) -> dict:
         return {

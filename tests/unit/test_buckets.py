@@ -37,3 +37,17 @@ def test_all_params_covered_once():
     buckets = partition_buckets(ps, bucket_cap_mb=0.01, first_bucket_mb=0.01)
     ids = [id(p) for b in buckets for p in b]
     assert sorted(ids) == sorted(id(p) for p in ps)
+
+
+def test_mixed_dtype_buckets():
+    import torch
+    ps = [torch.nn.Parameter(torch.zeros(100, dtype=torch.bfloat16)),
+          torch.nn.Parameter(torch.zeros(10, dtype=torch.float32)),
+          torch.nn.Parameter(torch.zeros(200, dtype=torch.bfloat16)),
+          torch.nn.Parameter(torch.zeros(20, dtype=torch.float32))]
+    buckets = partition_buckets(ps, bucket_cap_mb=1024, first_bucket_mb=1024)
+    for b in buckets:
+        dts = {p.dtype for p in b}
+        assert len(dts) == 1, "bucket mixes dtypes"
+    ids = sorted(id(p) for b in buckets for p in b)
+    assert ids == sorted(id(p) for p in ps)
