@@ -112,7 +112,7 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
             "dtype": str(t.dtype).replace("torch.", ""),
         }
         if owner[name] == rank:
-            host = stager.to_host(t) if stager else t.detach().cpu()
+            host = stager.to_host(t) if stager else t.detach().contiguous().cpu()
             shard[name] = host.contiguous()
     for name, target in aliases.items():
         tensor_meta[name] = {"alias": target}

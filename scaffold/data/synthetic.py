@@ -55,6 +55,8 @@ class SyntheticImageData(SyntheticBase):
             d.batch_size, d.channels, d.image_size, d.image_size,
             generator=self.gen, device=self.device, dtype=torch.float32,
         ).to(self.dtype)
+        if self.device.type == "cuda":
+            x = x.to(memory_format=torch.channels_last)
         y = torch.randint(
             0, d.num_classes, (d.batch_size,), generator=self.gen, device=self.device
         )

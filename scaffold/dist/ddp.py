@@ -130,7 +130,9 @@ class BucketedDDP(nn.Module):
             b = GradBucket(index=i, params=group, flat=flat)
             off = 0
             for p in group:
-                view = flat[off : off + p.numel()].view_as(p)
+                # as_strided matches the param's layout (channels_last conv
+                # weights are dense permutations, not C-contiguous)
+                view = flat[off : off + p.numel()].as_strided(p.shape, p.stride())
                 p.grad = view  # autograd accumulates in place into the bucket
                 b.views.append(view)
                 self._param_bucket[id(p)] = b

@@ -51,6 +51,13 @@ class Trainer:
         raw_model = raw_model.to(self.device)
         if self.dtype is not torch.float32:
             raw_model = self._to_mixed(raw_model)
+        self.channels_last = cfg.model.name == "resnet50" and self.ctx.on_gpu
+        if self.ctx.on_gpu:
+            # MIOpen: exhaustive algo find amortized over the run
+            torch.backends.cudnn.benchmark = True
+        if self.channels_last:
+            # NHWC is MIOpen's fast conv layout on CDNA
+            raw_model = raw_model.to(memory_format=torch.channels_last)
         self.raw_model = raw_model
 
         self.ddp = BucketedDDP(
