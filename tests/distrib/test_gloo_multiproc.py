@@ -49,7 +49,9 @@ def _worker(rank, world, file_store_path, result_q):
     grads = torch.cat([p.grad.flatten() * scale for p in t.raw_model.parameters()])
     params = torch.cat([p.detach().flatten() for p in t.raw_model.parameters()])
     if rank == 0:
-        result_q.put(("dist", grads, params))
+        # numpy arrays pickle by value; torch tensors over mp.Queue use fd
+        # passing and break if the worker exits before the parent reads
+        result_q.put(("dist", grads.numpy(), params.numpy()))
     dist.destroy_process_group()
 
 
@@ -67,6 +69,8 @@ def test_ddp_matches_large_batch(tmp_path):
     for p in procs:
         p.start()
     kind, dist_grads, dist_params = q.get(timeout=100)
+    dist_grads = torch.from_numpy(dist_grads)
+    dist_params = torch.from_numpy(dist_params)
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
