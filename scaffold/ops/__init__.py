@@ -153,6 +153,7 @@ class _SoftmaxXent(torch.autograd.Function):
         t = targets.contiguous().view(-1)
         loss, lse = m.softmax_xent_fwd(logits2, t)
         ctx.save_for_backward(logits2, t, lse)
+        ctx.logits_shape = logits.shape
         return loss.mean()  # mean reduction; bwd kernel folds the 1/R
 
     @staticmethod
@@ -160,7 +161,7 @@ class _SoftmaxXent(torch.autograd.Function):
         logits2, t, lse = ctx.saved_tensors
         m = require_ext()
         dlogits = m.softmax_xent_bwd(logits2, t, lse, float(dloss))
-        return dlogits.view(*t.shape, logits2.shape[-1]), None
+        return dlogits.view(ctx.logits_shape), None
 
 
 def softmax_cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:

@@ -1,0 +1,81 @@
+"""End-to-end GPU training smoke per model family (@gpu): a few steps
+through the full path (HIP ops + BucketedDDP ws=1 + fused AdamW), loss
+finite and decreasing for the small configs."""
+
+import pytest
+import torch
+
+from scaffold.config import load_config
+from scaffold.trainer import Trainer
+
+pytestmark = pytest.mark.gpu
+
+
+def _fit_steps(cfg, n):
+    t = Trainer(cfg)
+    it = iter(t.data)
+    losses = []
+    for _ in range(n):
+        losses.append(t._train_step(it, 1))
+    torch.cuda.synchronize()
+    return losses
+
+
+def test_resnet50_gpu_steps():
+    cfg = load_config(None, [
+        "model.name=resnet50", "data.kind=synthetic_image",
+        "data.batch_size=8", "dtype=bf16", "log.interval=1000",
+    ])
+    losses = _fit_steps(cfg, 3)
+    assert all(torch.isfinite(torch.tensor(losses))), losses
+
+
+def test_gpt2_tiny_gpu_loss_decreases():
+    cfg = load_config(None, [
+        "model.name=gpt2", "model.n_layers=2", "model.n_heads=4",
+        "model.d_model=256", "model.vocab_size=512", "model.max_seq_len=128",
+        "data.kind=synthetic_tokens", "data.batch_size=4", "data.seq_len=128",
+        "optim.lr=0.001", "dtype=bf16", "log.interval=1000",
+    ])
+    losses = _fit_steps(cfg, 15)
+    assert all(torch.isfinite(torch.tensor(losses))), losses
+    assert losses[-1] < losses[0], losses
+
+
+def test_llama_tiny_gpu_loss_decreases():
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=256", "model.d_ff=512",
+        "model.vocab_size=512", "model.max_seq_len=128",
+        "data.kind=synthetic_tokens", "data.batch_size=4", "data.seq_len=128",
+        "optim.lr=0.001", "dtype=bf16", "log.interval=1000",
+    ])
+    losses = _fit_steps(cfg, 15)
+    assert all(torch.isfinite(torch.tensor(losses))), losses
+    assert losses[-1] < losses[0], losses
+
+
+def test_ckpt_roundtrip_gpu(tmp_path):
+    import copy
+
+    cfg = load_config(None, [
+        "model.name=gpt2", "model.n_layers=2", "model.n_heads=4",
+        "model.d_model=256", "model.vocab_size=512", "model.max_seq_len=128",
+        "data.kind=synthetic_tokens", "data.batch_size=4", "data.seq_len=128",
+        "dtype=bf16", "log.interval=1000",
+    ])
+    cfg.ckpt.dir = str(tmp_path)
+    t = Trainer(cfg)
+    it = iter(t.data)
+    for _ in range(3):
+        t._train_step(it, 1)
+        t.step += 1
+    ck = t.save_checkpoint()
+
+    cfg2 = copy.deepcopy(cfg)
+    cfg2.ckpt.resume = ck
+    cfg2.seed = 7
+    t2 = Trainer(cfg2)
+    assert t2.step == 3
+    for a, b in zip(t.raw_model.parameters(), t2.raw_model.parameters()):
+        assert torch.equal(a.detach(), b.detach())
