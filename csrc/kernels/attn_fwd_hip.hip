@@ -155,6 +155,11 @@ void launch_fwd(const torch::Tensor& q, const torch::Tensor& k,
 
 }  // namespace
 
+// MFMA fast path (attn_fwd_mfma.hip): bf16, D in {64,128}
+void attn_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                   torch::Tensor o, torch::Tensor lse, int B, int H, int Hkv,
+                   int S, int D, double scale, bool causal);
+
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal) {
   CHECK_CUDA(q); CHECK_CONTIG(q); CHECK_CONTIG(k); CHECK_CONTIG(v);
@@ -167,10 +172,13 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
       torch::TensorOptions().dtype(torch::kFloat32).device(q.device()));
   const float scale = 1.0f / std::sqrt((float)D);
   auto stream = scaffold_stream();
-  if (q.scalar_type() == torch::kBFloat16)
+  if (q.scalar_type() == torch::kBFloat16 && (D == 64 || D == 128)) {
+    attn_fwd_mfma(q, k, v, o, lse, B, H, Hkv, S, D, scale, causal);
+  } else if (q.scalar_type() == torch::kBFloat16) {
     launch_fwd<short>(q, k, v, o, lse, B, H, Hkv, S, D, scale, causal, stream);
-  else
+  } else {
     launch_fwd<float>(q, k, v, o, lse, B, H, Hkv, S, D, scale, causal, stream);
+  }
   HIP_CHECK(hipGetLastError());
   return {o, lse};
 }

@@ -28,6 +28,7 @@ _SOURCES = [
     "kernels/softmax_xent.hip",
     "kernels/rmsnorm_rope.hip",
     "kernels/attn_fwd.hip",
+    "kernels/attn_fwd_mfma.hip",
     "kernels/attn_bwd.hip",
     "kernels/probe.hip",
 ]
