@@ -54,6 +54,7 @@ class DataConfig:
     seq_len: int = 1024
     num_classes: int = 1000
     vector_dim: int = 256
+    active_vocab: int = 0  # LM data: restrict token ids to [0,N) for learnable signal
     seed: int = 1234
 
 

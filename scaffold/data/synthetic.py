@@ -64,13 +64,20 @@ class SyntheticImageData(SyntheticBase):
 
 
 class SyntheticTokenData(SyntheticBase):
-    """Token streams for LM configs: (B,S) int64 inputs, (B,S) targets."""
+    """Token streams for LM configs: (B,S) int64 inputs, (B,S) targets.
+
+    data.active_vocab > 0 restricts sampling to the first N token ids so
+    a short training run has learnable signal (loss drops from ~ln(V)
+    toward ~ln(N)); 0 = uniform over the full vocab (pure throughput).
+    """
 
     def next_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
         d = self.cfg.data
         vocab = self.cfg.model.vocab_size
+        hi = d.active_vocab if d.active_vocab else vocab
         tokens = torch.randint(
-            0, vocab, (d.batch_size, d.seq_len + 1), generator=self.gen, device=self.device
+            0, min(hi, vocab), (d.batch_size, d.seq_len + 1), generator=self.gen,
+            device=self.device
         )
         return tokens[:, :-1].contiguous(), tokens[:, 1:].contiguous()
 

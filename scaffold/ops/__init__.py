@@ -93,6 +93,8 @@ class _Rope(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, cos, sin):
         m = require_ext()
+        cos = cos.float().contiguous()  # tables stay fp32 even if model
+        sin = sin.float().contiguous()  # buffers were cast to bf16
         y = m.rope_apply(x.contiguous(), cos, sin, False)
         ctx.save_for_backward(cos, sin)
         return y

@@ -35,11 +35,12 @@ def test_gpt2_tiny_gpu_loss_decreases():
         "model.name=gpt2", "model.n_layers=2", "model.n_heads=4",
         "model.d_model=256", "model.vocab_size=512", "model.max_seq_len=128",
         "data.kind=synthetic_tokens", "data.batch_size=4", "data.seq_len=128",
-        "optim.lr=0.001", "dtype=bf16", "log.interval=1000",
+        "data.active_vocab=16",
+        "optim.lr=0.003", "dtype=bf16", "log.interval=1000",
     ])
-    losses = _fit_steps(cfg, 15)
+    losses = _fit_steps(cfg, 25)
     assert all(torch.isfinite(torch.tensor(losses))), losses
-    assert losses[-1] < losses[0], losses
+    assert min(losses[-3:]) < losses[0] - 1.0, losses
 
 
 def test_llama_tiny_gpu_loss_decreases():
@@ -48,11 +49,12 @@ def test_llama_tiny_gpu_loss_decreases():
         "model.n_kv_heads=2", "model.d_model=256", "model.d_ff=512",
         "model.vocab_size=512", "model.max_seq_len=128",
         "data.kind=synthetic_tokens", "data.batch_size=4", "data.seq_len=128",
-        "optim.lr=0.001", "dtype=bf16", "log.interval=1000",
+        "data.active_vocab=16",
+        "optim.lr=0.003", "dtype=bf16", "log.interval=1000",
     ])
-    losses = _fit_steps(cfg, 15)
+    losses = _fit_steps(cfg, 25)
     assert all(torch.isfinite(torch.tensor(losses))), losses
-    assert losses[-1] < losses[0], losses
+    assert min(losses[-3:]) < losses[0] - 1.0, losses
 
 
 def test_ckpt_roundtrip_gpu(tmp_path):
