@@ -251,6 +251,13 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q,
 
 }  // namespace
 
+// MFMA fast path (attn_bwd_mfma.hip): bf16, D in {64,128}
+void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                   torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                   int B, int H, int Hkv, int S, int D, double scale,
+                   bool causal);
+
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
@@ -266,7 +273,21 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
       torch::TensorOptions().dtype(torch::kFloat32).device(q.device()));
   const float scale = 1.0f / std::sqrt((float)D);
   auto stream = scaffold_stream();
-  if (q.scalar_type() == torch::kBFloat16)
+  if (q.scalar_type() == torch::kBFloat16 && (D == 64 || D == 128)) {
+    // delta kernel from the tile path, then the MFMA dq/dkv kernels
+    const long rows = (long)B * H * S;
+    dim3 dblock(256), dgrid((unsigned)((rows + 3) / 4));
+    if (D == 64)
+      hipLaunchKernelGGL((attn_delta_kernel<short, 64>), dgrid, dblock, 0,
+                         stream, (const short*)dout.data_ptr(),
+                         (const short*)o.data_ptr(), delta.data_ptr<float>(), rows);
+    else
+      hipLaunchKernelGGL((attn_delta_kernel<short, 128>), dgrid, dblock, 0,
+                         stream, (const short*)dout.data_ptr(),
+                         (const short*)o.data_ptr(), delta.data_ptr<float>(), rows);
+    attn_bwd_mfma(dout, q, k, v, lse, delta, dq, dk, dv, B, H, Hkv, S, D,
+                  scale, causal);
+  } else if (q.scalar_type() == torch::kBFloat16)
     launch_bwd<short>(dout, q, k, v, o, lse, dq, dk, dv, delta, B, H, Hkv, S, D,
                       scale, causal, stream);
   else

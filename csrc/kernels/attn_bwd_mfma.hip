@@ -1,0 +1,302 @@
+// K4 — MFMA flash attention backward (bf16, 32x32x16 matrix cores).
+//
+// Recompute-based flash bwd on the fragment maps verified by probe.hip
+// (same maps as attn_fwd_mfma.hip).  delta = rowsum(dO*O) comes from the
+// wave kernel in attn_bwd.hip.
+//
+// dQ kernel  (wave = 32 q rows, loop kv tiles):
+//   S^T  = mfma(K, Q^T)            C: lane=q, regs=k   (as fwd)
+//   P^T  = exp(S^T*scale - lse[q])              lane-local
+//   dP^T = mfma(V, dO^T)           C: lane=q, regs=k
+//   dS^T = P^T*(dP^T - delta[q])*scale          lane-local
+//   dQ^T += mfma(K^T_cols, pack(dS^T))  C: lane=q, regs=d
+//
+// dKV kernel (wave = 32 kv rows, loop q tiles, loop GQA group):
+//   S    = mfma(Q, K^T)            C: lane=k, regs=q
+//   P    = exp(S*scale - lse[q_reg])   (lse/delta per-REG loads, L1-hot)
+//   dV^T += mfma(pack(P^T), dO_cols)   C: lane=d, regs=k
+//   dK^T += mfma(pack(dS^T), Q_cols)   C: lane=d, regs=k
+//   (pack() = quad pack + permlane32_swap, T12 — identical transform for
+//    a C-layout operand into an A/B fragment's 8-consecutive-k order.)
+
+#include "../common.h"
+
+namespace {
+
+typedef float f32x16_t __attribute__((ext_vector_type(16)));
+typedef short bf16x8_t __attribute__((ext_vector_type(8)));
+
+DEVINL unsigned pack_bf16x2(float lo, float hi) {
+  return (unsigned)(unsigned short)f2bf(lo) |
+         ((unsigned)(unsigned short)f2bf(hi) << 16);
+}
+
+// C-layout f32[16] (lane=col, regs=rows) -> two 8-k B/A fragments
+// (slice s covers rows 16s..16s+15 in the fragment's k order)
+DEVINL void pack_frags(const float p[16], unsigned out[2][4]) {
+  #pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    unsigned ax = pack_bf16x2(p[8 * s + 0], p[8 * s + 1]);
+    unsigned ay = pack_bf16x2(p[8 * s + 2], p[8 * s + 3]);
+    unsigned bx = pack_bf16x2(p[8 * s + 4], p[8 * s + 5]);
+    unsigned by = pack_bf16x2(p[8 * s + 6], p[8 * s + 7]);
+    auto rx = __builtin_amdgcn_permlane32_swap(ax, bx, false, false);
+    auto ry = __builtin_amdgcn_permlane32_swap(ay, by, false, false);
+    out[s][0] = rx[0];
+    out[s][1] = ry[0];
+    out[s][2] = rx[1];
+    out[s][3] = ry[1];
+  }
+}
+
+template <int D>
+__global__ void __launch_bounds__(256)
+attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
+                        const short* __restrict__ q,
+                        const short* __restrict__ k,
+                        const short* __restrict__ v,
+                        const float* __restrict__ lse,
+                        const float* __restrict__ delta,
+                        short* __restrict__ dq, int B, int H, int Hkv, int S,
+                        float scale, int causal) {
+  constexpr int NSL = D / 16;
+  constexpr int NDT = D / 32;
+  const int wid = threadIdx.x >> 6;
+  const int l = threadIdx.x & 63;
+  const int lq = l & 31;
+  const int half = l >> 5;
+
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int hk = h / (H / Hkv);
+  const int q0 = blockIdx.x * 128 + wid * 32;
+  if (q0 >= S) return;
+
+  const long qhead = (((long)b * H + h) * S) * (long)D;
+  const long kvhead = (((long)b * Hkv + hk) * S) * (long)D;
+  const long rhead = ((long)b * H + h) * S;
+
+  const int qrow = min(q0 + lq, S - 1);
+  bf16x8_t qf[NSL], dof[NSL];
+  #pragma unroll
+  for (int s = 0; s < NSL; ++s) {
+    qf[s] = *(const bf16x8_t*)(q + qhead + (long)qrow * D + 16 * s + 8 * half);
+    dof[s] = *(const bf16x8_t*)(dout + qhead + (long)qrow * D + 16 * s + 8 * half);
+  }
+  const float my_lse = lse[rhead + qrow];
+  const float my_delta = delta[rhead + qrow];
+
+  f32x16_t dqacc[NDT];
+  #pragma unroll
+  for (int t = 0; t < NDT; ++t) dqacc[t] = {};
+
+  const int kv_end = causal ? min(S, q0 + 32) : S;
+  for (int c0 = 0; c0 < kv_end; c0 += 32) {
+    const int krow = min(c0 + lq, S - 1);
+    f32x16_t sacc = {}, dpacc = {};
+    #pragma unroll
+    for (int s = 0; s < NSL; ++s) {
+      bf16x8_t kf = *(const bf16x8_t*)(k + kvhead + (long)krow * D + 16 * s + 8 * half);
+      bf16x8_t vf = *(const bf16x8_t*)(v + kvhead + (long)krow * D + 16 * s + 8 * half);
+      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
+      dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[s], dpacc, 0, 0, 0);
+    }
+
+    const bool need_mask = (causal && c0 + 31 > q0) || (c0 + 31 >= S);
+    float ds[16];
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krel = (r & 3) + 8 * (r >> 2) + 4 * half;
+      float sv = sacc[r] * scale;
+      if (need_mask) {
+        const int kg = c0 + krel;
+        if (kg >= S || (causal && kg > q0 + lq)) sv = -1e30f;
+      }
+      const float p = __expf(sv - my_lse);
+      ds[r] = p * (dpacc[r] - my_delta) * scale;
+    }
+
+    unsigned dsf[2][4];
+    pack_frags(ds, dsf);
+
+    // dQ^T += K^T . dS^T  per d-tile (A = K columns, uncoalesced v1)
+    #pragma unroll
+    for (int t = 0; t < NDT; ++t) {
+      #pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        short kf[8];
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int kg = min(c0 + 16 * s + 8 * half + j, S - 1);
+          kf[j] = k[kvhead + (long)kg * D + 32 * t + lq];
+        }
+        dqacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *(const bf16x8_t*)kf, *(const bf16x8_t*)dsf[s], dqacc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  if (q0 + lq < S) {
+    const long obase = qhead + (long)(q0 + lq) * D;
+    #pragma unroll
+    for (int t = 0; t < NDT; ++t)
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int d = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * half;
+        dq[obase + d] = f2bf(dqacc[t][r]);
+      }
+  }
+}
+
+template <int D>
+__global__ void __launch_bounds__(256)
+attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
+                         const short* __restrict__ q,
+                         const short* __restrict__ k,
+                         const short* __restrict__ v,
+                         const float* __restrict__ lse,
+                         const float* __restrict__ delta,
+                         short* __restrict__ dk, short* __restrict__ dv,
+                         int B, int H, int Hkv, int S, float scale,
+                         int causal) {
+  constexpr int NSL = D / 16;
+  constexpr int NDT = D / 32;
+  const int wid = threadIdx.x >> 6;
+  const int l = threadIdx.x & 63;
+  const int lk = l & 31;        // this lane's kv row (S-tile) / d col (acc)
+  const int half = l >> 5;
+
+  const int bhk = blockIdx.y;
+  const int b = bhk / Hkv, hk = bhk % Hkv;
+  const int group = H / Hkv;
+  const int k0 = blockIdx.x * 128 + wid * 32;
+  if (k0 >= S) return;
+
+  const long kvhead = (((long)b * Hkv + hk) * S) * (long)D;
+  const int krow = min(k0 + lk, S - 1);
+
+  // K^T B-fragments for S = Q.K^T (col=k=lane, kdim=d) — contiguous
+  bf16x8_t ktf[NSL];
+  #pragma unroll
+  for (int s = 0; s < NSL; ++s)
+    ktf[s] = *(const bf16x8_t*)(k + kvhead + (long)krow * D + 16 * s + 8 * half);
+
+  f32x16_t dvacc[NDT], dkacc[NDT];
+  #pragma unroll
+  for (int t = 0; t < NDT; ++t) { dvacc[t] = {}; dkacc[t] = {}; }
+
+  for (int g = 0; g < group; ++g) {
+    const int h = hk * group + g;
+    const long qhead = (((long)b * H + h) * S) * (long)D;
+    const long rhead = ((long)b * H + h) * S;
+    const int qstart = causal ? (k0 / 32) * 32 : 0;
+    for (int c0 = qstart; c0 < S; c0 += 32) {
+      // S = Q_tile . K^T  (C: lane=k, regs=q)
+      const int qrow = min(c0 + lk, S - 1);
+      f32x16_t sacc = {};
+      #pragma unroll
+      for (int s = 0; s < NSL; ++s) {
+        bf16x8_t qf = *(const bf16x8_t*)(q + qhead + (long)qrow * D + 16 * s + 8 * half);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
+      }
+
+      float p[16], ds[16];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
+        const int qg = c0 + qrel;
+        float sv = sacc[r] * scale;
+        const bool masked = qg >= S || (causal && (k0 + lk) > qg);
+        const int qc = min(qg, S - 1);
+        const float lse_r = lse[rhead + qc];    // L1-hot across lanes
+        p[r] = masked ? 0.f : __expf(sv - lse_r);
+      }
+
+      // dP = dO . V^T (C: lane=k, regs=q): A=dO rows (contig), B=V^T (contig)
+      f32x16_t dpacc = {};
+      #pragma unroll
+      for (int s = 0; s < NSL; ++s) {
+        bf16x8_t dof = *(const bf16x8_t*)(dout + qhead + (long)qrow * D + 16 * s + 8 * half);
+        bf16x8_t vf = *(const bf16x8_t*)(v + kvhead + (long)krow * D + 16 * s + 8 * half);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dpacc, 0, 0, 0);
+      }
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
+        const int qc = min(c0 + qrel, S - 1);
+        const float del_r = delta[rhead + qc];
+        ds[r] = p[r] * (dpacc[r] - del_r) * scale;
+      }
+
+      unsigned pf[2][4], dsf[2][4];
+      pack_frags(p, pf);
+      pack_frags(ds, dsf);
+
+      // dV^T += P^T . dO_cols ; dK^T += dS^T . Q_cols  (C: lane=d, regs=k)
+      #pragma unroll
+      for (int t = 0; t < NDT; ++t) {
+        #pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          short dof_c[8], qf_c[8];
+          #pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int qg = min(c0 + 16 * s + 8 * half + j, S - 1);
+            dof_c[j] = dout[qhead + (long)qg * D + 32 * t + lk];
+            qf_c[j] = q[qhead + (long)qg * D + 32 * t + lk];
+          }
+          dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *(const bf16x8_t*)pf[s], *(const bf16x8_t*)dof_c, dvacc[t], 0, 0, 0);
+          dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *(const bf16x8_t*)dsf[s], *(const bf16x8_t*)qf_c, dkacc[t], 0, 0, 0);
+        }
+      }
+    }
+  }
+
+  // epilogue: C layout lane=col=d(within tile), regs=rows=k
+  #pragma unroll
+  for (int t = 0; t < NDT; ++t)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krel = (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (k0 + krel < S) {
+        const long base = kvhead + (long)(k0 + krel) * D + 32 * t + lk;
+        dk[base] = f2bf(dkacc[t][r]);
+        dv[base] = f2bf(dvacc[t][r]);
+      }
+    }
+}
+
+}  // namespace
+
+void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                   torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+                   int B, int H, int Hkv, int S, int D, double scale,
+                   bool causal) {
+  auto stream = scaffold_stream();
+  dim3 block(256);
+  dim3 grid_q((S + 127) / 128, B * H);
+  dim3 grid_kv((S + 127) / 128, B * Hkv);
+  #define LAUNCH_D(DD)                                                         \
+    do {                                                                       \
+      hipLaunchKernelGGL((attn_bwd_dq_mfma_kernel<DD>), grid_q, block, 0,      \
+                         stream, (const short*)dout.data_ptr(),                \
+                         (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
+                         (const short*)v.data_ptr(), lse.data_ptr<float>(),    \
+                         delta.data_ptr<float>(), (short*)dq.data_ptr(),       \
+                         B, H, Hkv, S, (float)scale, causal ? 1 : 0);          \
+      hipLaunchKernelGGL((attn_bwd_dkv_mfma_kernel<DD>), grid_kv, block, 0,    \
+                         stream, (const short*)dout.data_ptr(),                \
+                         (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
+                         (const short*)v.data_ptr(), lse.data_ptr<float>(),    \
+                         delta.data_ptr<float>(), (short*)dk.data_ptr(),       \
+                         (short*)dv.data_ptr(), B, H, Hkv, S, (float)scale,    \
+                         causal ? 1 : 0);                                      \
+    } while (0)
+  if (D == 64) LAUNCH_D(64);
+  else if (D == 128) LAUNCH_D(128);
+  else TORCH_CHECK(false, "attn_bwd_mfma: unsupported D ", D);
+  #undef LAUNCH_D
+  HIP_CHECK(hipGetLastError());
+}

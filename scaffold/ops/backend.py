@@ -30,6 +30,7 @@ _SOURCES = [
     "kernels/attn_fwd.hip",
     "kernels/attn_fwd_mfma.hip",
     "kernels/attn_bwd.hip",
+    "kernels/attn_bwd_mfma.hip",
     "kernels/probe.hip",
 ]
 
