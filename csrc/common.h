@@ -96,6 +96,19 @@ DEVINL float block_reduce_max(float v, float* red) {
   return out;
 }
 
+// Element strides of a logical (B,H,S,D) tensor with contiguous D —
+// lets attention/RoPE consume permuted views (e.g. the (B,S,H,D)
+// physical layout of a QKV projection) without transpose copies.
+struct Strides4 {
+  long sb, sh, ss;
+};
+
+inline Strides4 strides_of(const torch::Tensor& t) {
+  TORCH_CHECK(t.dim() == 4 && t.stride(3) == 1,
+              "expected 4-D tensor with contiguous last dim");
+  return {t.stride(0), t.stride(1), t.stride(2)};
+}
+
 // Grid sizing guideline 11: cap grid, grid-stride the rest.
 DEVINL long grid_stride() { return (long)gridDim.x * blockDim.x; }
 

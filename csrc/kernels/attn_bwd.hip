@@ -252,45 +252,52 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q,
 
 // MFMA fast path (attn_bwd_mfma.hip): bf16, D in {64,128}
 void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
-                   torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
-                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
-                   int B, int H, int Hkv, int S, int D, double scale,
-                   bool causal);
+                   torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                   torch::Tensor delta, torch::Tensor dq, torch::Tensor dk,
+                   torch::Tensor dv, int B, int H, int Hkv, int S, int D,
+                   double scale, bool causal);
 
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
                                     bool causal) {
-  CHECK_CUDA(dout); CHECK_CONTIG(dout);
+  CHECK_CUDA(dout);
   const int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2),
             D = (int)q.size(3);
   const int Hkv = (int)k.size(1);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
   auto delta = torch::empty({(long)B * H * S},
       torch::TensorOptions().dtype(torch::kFloat32).device(q.device()));
   const float scale = 1.0f / std::sqrt((float)D);
   auto stream = scaffold_stream();
   if (q.scalar_type() == torch::kBFloat16 && (D == 64 || D == 128)) {
-    // delta kernel from the tile path, then the MFMA dq/dkv kernels
-    const long rows = (long)B * H * S;
-    dim3 dblock(256), dgrid((unsigned)((rows + 3) / 4));
-    if (D == 64)
-      hipLaunchKernelGGL((attn_delta_kernel<short, 64>), dgrid, dblock, 0,
-                         stream, (const short*)dout.data_ptr(),
-                         (const short*)o.data_ptr(), delta.data_ptr<float>(), rows);
-    else
-      hipLaunchKernelGGL((attn_delta_kernel<short, 128>), dgrid, dblock, 0,
-                         stream, (const short*)dout.data_ptr(),
-                         (const short*)o.data_ptr(), delta.data_ptr<float>(), rows);
-    attn_bwd_mfma(dout, q, k, v, lse, delta, dq, dk, dv, B, H, Hkv, S, D,
+    // strided MFMA path; grads produced in (B,S,H,D) physical layout so
+    // the upstream view backward (QKV un-transpose) is coalesced
+    auto doc = dout.stride(3) == 1 ? dout : dout.contiguous();
+    auto qc = q.stride(3) == 1 ? q : q.contiguous();
+    auto kc = k.stride(3) == 1 ? k : k.contiguous();
+    auto vc = v.stride(3) == 1 ? v : v.contiguous();
+    auto oc = o.stride(3) == 1 ? o : o.contiguous();
+    auto dq = torch::empty({B, S, H, D}, q.options()).permute({0, 2, 1, 3});
+    auto dk = torch::empty({B, S, Hkv, D}, q.options()).permute({0, 2, 1, 3});
+    auto dv = torch::empty({B, S, Hkv, D}, q.options()).permute({0, 2, 1, 3});
+    attn_bwd_mfma(doc, qc, kc, vc, oc, lse, delta, dq, dk, dv, B, H, Hkv, S, D,
                   scale, causal);
-  } else if (q.scalar_type() == torch::kBFloat16)
-    launch_bwd<short>(dout, q, k, v, o, lse, dq, dk, dv, delta, B, H, Hkv, S, D,
+    HIP_CHECK(hipGetLastError());
+    return {dq, dk, dv};
+  }
+  auto doc = dout.contiguous();
+  auto qc = q.contiguous();
+  auto kc = k.contiguous();
+  auto vc = v.contiguous();
+  auto oc = o.contiguous();
+  auto dq = torch::empty_like(qc);
+  auto dk = torch::empty_like(kc);
+  auto dv = torch::empty_like(vc);
+  if (q.scalar_type() == torch::kBFloat16)
+    launch_bwd<short>(doc, qc, kc, vc, oc, lse, dq, dk, dv, delta, B, H, Hkv, S, D,
                       scale, causal, stream);
   else
-    launch_bwd<float>(dout, q, k, v, o, lse, dq, dk, dv, delta, B, H, Hkv, S, D,
+    launch_bwd<float>(doc, qc, kc, vc, oc, lse, dq, dk, dv, delta, B, H, Hkv, S, D,
                       scale, causal, stream);
   HIP_CHECK(hipGetLastError());
   return {dq, dk, dv};

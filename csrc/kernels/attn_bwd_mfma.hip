@@ -49,6 +49,32 @@ DEVINL void pack_frags(const float p[16], unsigned out[2][4]) {
   }
 }
 
+// delta = rowsum(dO * O), stride-aware (one wave per row)
+template <int D>
+__global__ void __launch_bounds__(256)
+attn_delta_strided_kernel(const short* __restrict__ dout,
+                          const short* __restrict__ o,
+                          float* __restrict__ delta, int H, int S, long rows,
+                          Strides4 dos, Strides4 osr) {
+  const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= rows) return;
+  const int lane = threadIdx.x & 63;
+  const int s = (int)(row % S);
+  const int h = (int)((row / S) % H);
+  const long b = row / ((long)S * H);
+  const long dbase = b * dos.sb + (long)h * dos.sh + (long)s * dos.ss;
+  const long obase = b * osr.sb + (long)h * osr.sh + (long)s * osr.ss;
+  float acc = 0.f;
+  for (int d = lane * 8; d < D; d += 64 * 8) {
+    bf16x8_t dv = *(const bf16x8_t*)(dout + dbase + d);
+    bf16x8_t ov = *(const bf16x8_t*)(o + obase + d);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) acc += bf2f(dv[j]) * bf2f(ov[j]);
+  }
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) delta[row] = acc;
+}
+
 template <int D>
 __global__ void __launch_bounds__(256)
 attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
@@ -58,7 +84,8 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
                         const float* __restrict__ lse,
                         const float* __restrict__ delta,
                         short* __restrict__ dq, int B, int H, int Hkv, int S,
-                        float scale, int causal) {
+                        float scale, int causal, Strides4 dos, Strides4 qs,
+                        Strides4 ks, Strides4 vs, Strides4 dqs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
   const int wid = threadIdx.x >> 6;
@@ -72,16 +99,18 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   const int q0 = blockIdx.x * 128 + wid * 32;
   if (q0 >= S) return;
 
-  const long qhead = (((long)b * H + h) * S) * (long)D;
-  const long kvhead = (((long)b * Hkv + hk) * S) * (long)D;
+  const long qb = (long)b * qs.sb + (long)h * qs.sh;
+  const long dob = (long)b * dos.sb + (long)h * dos.sh;
+  const long kbb = (long)b * ks.sb + (long)hk * ks.sh;
+  const long vbb = (long)b * vs.sb + (long)hk * vs.sh;
   const long rhead = ((long)b * H + h) * S;
 
   const int qrow = min(q0 + lq, S - 1);
   bf16x8_t qf[NSL], dof[NSL];
   #pragma unroll
   for (int s = 0; s < NSL; ++s) {
-    qf[s] = *(const bf16x8_t*)(q + qhead + (long)qrow * D + 16 * s + 8 * half);
-    dof[s] = *(const bf16x8_t*)(dout + qhead + (long)qrow * D + 16 * s + 8 * half);
+    qf[s] = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
+    dof[s] = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
   }
   const float my_lse = lse[rhead + qrow];
   const float my_delta = delta[rhead + qrow];
@@ -96,8 +125,8 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
     f32x16_t sacc = {}, dpacc = {};
     #pragma unroll
     for (int s = 0; s < NSL; ++s) {
-      bf16x8_t kf = *(const bf16x8_t*)(k + kvhead + (long)krow * D + 16 * s + 8 * half);
-      bf16x8_t vf = *(const bf16x8_t*)(v + kvhead + (long)krow * D + 16 * s + 8 * half);
+      bf16x8_t kf = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
+      bf16x8_t vf = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
       sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
       dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[s], dpacc, 0, 0, 0);
     }
@@ -128,7 +157,7 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int kg = min(c0 + 16 * s + 8 * half + j, S - 1);
-          kf[j] = k[kvhead + (long)kg * D + 32 * t + lq];
+          kf[j] = k[kbb + (long)kg * ks.ss + 32 * t + lq];
         }
         dqacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *(const bf16x8_t*)kf, *(const bf16x8_t*)dsf[s], dqacc[t], 0, 0, 0);
@@ -137,7 +166,7 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   }
 
   if (q0 + lq < S) {
-    const long obase = qhead + (long)(q0 + lq) * D;
+    const long obase = (long)b * dqs.sb + (long)h * dqs.sh + (long)(q0 + lq) * dqs.ss;
     #pragma unroll
     for (int t = 0; t < NDT; ++t)
       #pragma unroll
@@ -158,7 +187,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          const float* __restrict__ delta,
                          short* __restrict__ dk, short* __restrict__ dv,
                          int B, int H, int Hkv, int S, float scale,
-                         int causal) {
+                         int causal, Strides4 dos, Strides4 qs, Strides4 ks,
+                         Strides4 vs, Strides4 dks, Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
   const int wid = threadIdx.x >> 6;
@@ -172,14 +202,15 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
   const int k0 = blockIdx.x * 128 + wid * 32;
   if (k0 >= S) return;
 
-  const long kvhead = (((long)b * Hkv + hk) * S) * (long)D;
+  const long kbb = (long)b * ks.sb + (long)hk * ks.sh;
+  const long vbb = (long)b * vs.sb + (long)hk * vs.sh;
   const int krow = min(k0 + lk, S - 1);
 
   // K^T B-fragments for S = Q.K^T (col=k=lane, kdim=d) — contiguous
   bf16x8_t ktf[NSL];
   #pragma unroll
   for (int s = 0; s < NSL; ++s)
-    ktf[s] = *(const bf16x8_t*)(k + kvhead + (long)krow * D + 16 * s + 8 * half);
+    ktf[s] = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
 
   f32x16_t dvacc[NDT], dkacc[NDT];
   #pragma unroll
@@ -187,7 +218,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
 
   for (int g = 0; g < group; ++g) {
     const int h = hk * group + g;
-    const long qhead = (((long)b * H + h) * S) * (long)D;
+    const long qb = (long)b * qs.sb + (long)h * qs.sh;
+    const long dob = (long)b * dos.sb + (long)h * dos.sh;
     const long rhead = ((long)b * H + h) * S;
     const int qstart = causal ? (k0 / 32) * 32 : 0;
     for (int c0 = qstart; c0 < S; c0 += 32) {
@@ -196,7 +228,7 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       f32x16_t sacc = {};
       #pragma unroll
       for (int s = 0; s < NSL; ++s) {
-        bf16x8_t qf = *(const bf16x8_t*)(q + qhead + (long)qrow * D + 16 * s + 8 * half);
+        bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
         sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
       }
 
@@ -216,8 +248,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       f32x16_t dpacc = {};
       #pragma unroll
       for (int s = 0; s < NSL; ++s) {
-        bf16x8_t dof = *(const bf16x8_t*)(dout + qhead + (long)qrow * D + 16 * s + 8 * half);
-        bf16x8_t vf = *(const bf16x8_t*)(v + kvhead + (long)krow * D + 16 * s + 8 * half);
+        bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
+        bf16x8_t vf = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
         dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dpacc, 0, 0, 0);
       }
       #pragma unroll
@@ -241,8 +273,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
           #pragma unroll
           for (int j = 0; j < 8; ++j) {
             const int qg = min(c0 + 16 * s + 8 * half + j, S - 1);
-            dof_c[j] = dout[qhead + (long)qg * D + 32 * t + lk];
-            qf_c[j] = q[qhead + (long)qg * D + 32 * t + lk];
+            dof_c[j] = dout[dob + (long)qg * dos.ss + 32 * t + lk];
+            qf_c[j] = q[qb + (long)qg * qs.ss + 32 * t + lk];
           }
           dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *(const bf16x8_t*)pf[s], *(const bf16x8_t*)dof_c, dvacc[t], 0, 0, 0);
@@ -260,9 +292,10 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
     for (int r = 0; r < 16; ++r) {
       const int krel = (r & 3) + 8 * (r >> 2) + 4 * half;
       if (k0 + krel < S) {
-        const long base = kvhead + (long)(k0 + krel) * D + 32 * t + lk;
-        dk[base] = f2bf(dkacc[t][r]);
-        dv[base] = f2bf(dvacc[t][r]);
+        dk[(long)b * dks.sb + (long)hk * dks.sh + (long)(k0 + krel) * dks.ss + 32 * t + lk]
+            = f2bf(dkacc[t][r]);
+        dv[(long)b * dvs.sb + (long)hk * dvs.sh + (long)(k0 + krel) * dvs.ss + 32 * t + lk]
+            = f2bf(dvacc[t][r]);
       }
     }
 }
@@ -270,14 +303,27 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
 }  // namespace
 
 void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
-                   torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
-                   torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
-                   int B, int H, int Hkv, int S, int D, double scale,
-                   bool causal) {
+                   torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                   torch::Tensor delta, torch::Tensor dq, torch::Tensor dk,
+                   torch::Tensor dv, int B, int H, int Hkv, int S, int D,
+                   double scale, bool causal) {
   auto stream = scaffold_stream();
   dim3 block(256);
   dim3 grid_q((S + 127) / 128, B * H);
   dim3 grid_kv((S + 127) / 128, B * Hkv);
+  const Strides4 dos = strides_of(dout), qs = strides_of(q), ks = strides_of(k),
+                 vs = strides_of(v), dqs = strides_of(dq), dks = strides_of(dk),
+                 dvs = strides_of(dv), osr = strides_of(o);
+  const long rows = (long)B * H * S;
+  dim3 dgrid((unsigned)((rows + 3) / 4));
+  if (D == 64)
+    hipLaunchKernelGGL((attn_delta_strided_kernel<64>), dgrid, block, 0, stream,
+                       (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
+                       delta.data_ptr<float>(), H, S, rows, dos, osr);
+  else
+    hipLaunchKernelGGL((attn_delta_strided_kernel<128>), dgrid, block, 0, stream,
+                       (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
+                       delta.data_ptr<float>(), H, S, rows, dos, osr);
   #define LAUNCH_D(DD)                                                         \
     do {                                                                       \
       hipLaunchKernelGGL((attn_bwd_dq_mfma_kernel<DD>), grid_q, block, 0,      \
@@ -285,14 +331,15 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                          (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
                          (const short*)v.data_ptr(), lse.data_ptr<float>(),    \
                          delta.data_ptr<float>(), (short*)dq.data_ptr(),       \
-                         B, H, Hkv, S, (float)scale, causal ? 1 : 0);          \
+                         B, H, Hkv, S, (float)scale, causal ? 1 : 0,           \
+                         dos, qs, ks, vs, dqs);                                \
       hipLaunchKernelGGL((attn_bwd_dkv_mfma_kernel<DD>), grid_kv, block, 0,    \
                          stream, (const short*)dout.data_ptr(),                \
                          (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
                          (const short*)v.data_ptr(), lse.data_ptr<float>(),    \
                          delta.data_ptr<float>(), (short*)dk.data_ptr(),       \
                          (short*)dv.data_ptr(), B, H, Hkv, S, (float)scale,    \
-                         causal ? 1 : 0);                                      \
+                         causal ? 1 : 0, dos, qs, ks, vs, dks, dvs);           \
     } while (0)
   if (D == 64) LAUNCH_D(64);
   else if (D == 128) LAUNCH_D(128);

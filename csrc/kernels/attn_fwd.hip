@@ -161,22 +161,34 @@ void attn_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal) {
-  CHECK_CUDA(q); CHECK_CONTIG(q); CHECK_CONTIG(k); CHECK_CONTIG(v);
+  CHECK_CUDA(q);
   const int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2),
             D = (int)q.size(3);
   const int Hkv = (int)k.size(1);
   TORCH_CHECK(H % Hkv == 0, "GQA: H must be a multiple of Hkv");
-  auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, S},
       torch::TensorOptions().dtype(torch::kFloat32).device(q.device()));
   const float scale = 1.0f / std::sqrt((float)D);
   auto stream = scaffold_stream();
   if (q.scalar_type() == torch::kBFloat16 && (D == 64 || D == 128)) {
-    attn_fwd_mfma(q, k, v, o, lse, B, H, Hkv, S, D, scale, causal);
-  } else if (q.scalar_type() == torch::kBFloat16) {
-    launch_fwd<short>(q, k, v, o, lse, B, H, Hkv, S, D, scale, causal, stream);
+    // strided MFMA path: consume permuted views directly; O written in
+    // (B,S,H,D) physical layout so the caller's transpose+reshape is free
+    auto qc = q.stride(3) == 1 ? q : q.contiguous();
+    auto kc = k.stride(3) == 1 ? k : k.contiguous();
+    auto vc = v.stride(3) == 1 ? v : v.contiguous();
+    auto o = torch::empty({B, S, H, D}, q.options()).permute({0, 2, 1, 3});
+    attn_fwd_mfma(qc, kc, vc, o, lse, B, H, Hkv, S, D, scale, causal);
+    HIP_CHECK(hipGetLastError());
+    return {o, lse};
+  }
+  auto qc = q.contiguous();
+  auto kc = k.contiguous();
+  auto vc = v.contiguous();
+  auto o = torch::empty_like(qc);
+  if (q.scalar_type() == torch::kBFloat16) {
+    launch_fwd<short>(qc, kc, vc, o, lse, B, H, Hkv, S, D, scale, causal, stream);
   } else {
-    launch_fwd<float>(q, k, v, o, lse, B, H, Hkv, S, D, scale, causal, stream);
+    launch_fwd<float>(qc, kc, vc, o, lse, B, H, Hkv, S, D, scale, causal, stream);
   }
   HIP_CHECK(hipGetLastError());
   return {o, lse};

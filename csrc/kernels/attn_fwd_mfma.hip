@@ -35,7 +35,8 @@ __global__ void __launch_bounds__(256)
 attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
                      const short* __restrict__ v, short* __restrict__ o,
                      float* __restrict__ lse, int B, int H, int Hkv, int S,
-                     float scale, int causal) {
+                     float scale, int causal, Strides4 qs, Strides4 ks,
+                     Strides4 vs, Strides4 os) {
   constexpr int NSL = D / 16;   // QK^T k-slices (over d)
   constexpr int NDT = D / 32;   // O^T d-tiles
   const int wid = threadIdx.x >> 6;
@@ -49,15 +50,16 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const int q0 = blockIdx.x * 128 + wid * 32;   // wave's q-tile origin
   if (q0 >= S) return;
 
-  const long qhead = (((long)b * H + h) * S) * (long)D;
-  const long kvhead = (((long)b * Hkv + hk) * S) * (long)D;
+  const long qb = (long)b * qs.sb + (long)h * qs.sh;
+  const long kb = (long)b * ks.sb + (long)hk * ks.sh;
+  const long vb = (long)b * vs.sb + (long)hk * vs.sh;
 
   // ---- load Q B-fragments once (row clamped at S-1 for pad rows)
   const int qrow = min(q0 + lq, S - 1);
   bf16x8_t qf[NSL];
   #pragma unroll
   for (int s = 0; s < NSL; ++s)
-    qf[s] = *(const bf16x8_t*)(q + qhead + (long)qrow * D + 16 * s + 8 * half);
+    qf[s] = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
 
   f32x16_t oacc[NDT];
   #pragma unroll
@@ -71,7 +73,7 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
     const int krow = min(c0 + lq, S - 1);
     #pragma unroll
     for (int s = 0; s < NSL; ++s) {
-      bf16x8_t kf = *(const bf16x8_t*)(k + kvhead + (long)krow * D + 16 * s + 8 * half);
+      bf16x8_t kf = *(const bf16x8_t*)(k + kb + (long)krow * ks.ss + 16 * s + 8 * half);
       sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
     }
 
@@ -135,7 +137,7 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
         #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int kg = min(c0 + 16 * s + 8 * half + j, S - 1);
-          vf[j] = v[kvhead + (long)kg * D + 32 * t + lq];
+          vf[j] = v[vb + (long)kg * vs.ss + 32 * t + lq];
         }
         oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             *(const bf16x8_t*)vf, *(const bf16x8_t*)pfrag[s], oacc[t], 0, 0, 0);
@@ -146,7 +148,7 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   // ---- epilogue: O = O^T / l, scatter store (lane's q row fixed)
   const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
   if (q0 + lq < S) {
-    const long obase = qhead + (long)(q0 + lq) * D;
+    const long obase = (long)b * os.sb + (long)h * os.sh + (long)(q0 + lq) * os.ss;
     #pragma unroll
     for (int t = 0; t < NDT; ++t)
       #pragma unroll
@@ -166,19 +168,17 @@ void attn_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                    int S, int D, double scale, bool causal) {
   auto stream = scaffold_stream();
   dim3 grid((S + 127) / 128, B * H), block(256);
-  if (D == 64)
-    hipLaunchKernelGGL((attn_fwd_mfma_kernel<64>), grid, block, 0, stream,
-                       (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                       (const short*)v.data_ptr(), (short*)o.data_ptr(),
-                       lse.data_ptr<float>(), B, H, Hkv, S, (float)scale,
-                       causal ? 1 : 0);
-  else if (D == 128)
-    hipLaunchKernelGGL((attn_fwd_mfma_kernel<128>), grid, block, 0, stream,
-                       (const short*)q.data_ptr(), (const short*)k.data_ptr(),
-                       (const short*)v.data_ptr(), (short*)o.data_ptr(),
-                       lse.data_ptr<float>(), B, H, Hkv, S, (float)scale,
-                       causal ? 1 : 0);
-  else
-    TORCH_CHECK(false, "attn_fwd_mfma: unsupported D ", D);
+  const Strides4 qs = strides_of(q), ks = strides_of(k), vs = strides_of(v),
+                 os = strides_of(o);
+  #define LAUNCH_F(DD)                                                         \
+    hipLaunchKernelGGL((attn_fwd_mfma_kernel<DD>), grid, block, 0, stream,     \
+                       (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
+                       (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
+                       lse.data_ptr<float>(), B, H, Hkv, S, (float)scale,      \
+                       causal ? 1 : 0, qs, ks, vs, os)
+  if (D == 64) LAUNCH_F(64);
+  else if (D == 128) LAUNCH_F(128);
+  else TORCH_CHECK(false, "attn_fwd_mfma: unsupported D ", D);
+  #undef LAUNCH_F
   HIP_CHECK(hipGetLastError());
 }

@@ -1,7 +1,9 @@
-// K2 — LayerNorm fwd/bwd: one workgroup per row, wave64
-// reductions, vectorized bf16x8 loads (G13).  Saves fp32 mean/rstd per
-// row for the backward.  dgamma/dbeta: two-stage — per-block partials
-// into a workspace, then a column-reduce kernel.
+// K2 — LayerNorm fwd/bwd.  One WAVE per row (4 rows per
+// 256-thread block), wave64 shuffle reductions, bf16x8 vector loads
+// (G13) when D % 8 == 0; scalar fallback otherwise.  dgamma/dbeta via
+// per-block partials + a parallel column-reduce (64 cols x 4 row-slices
+// per block) — the serial version was 22% of a GPT-2 step (see
+// profiles/gpt2_kernel_stats_r1_before_fixes.csv).
 
 #include "../common.h"
 
@@ -17,94 +19,191 @@ DEVINL void store_f(T* p, long i, float v);
 template <> DEVINL void store_f<short>(short* p, long i, float v) { p[i] = f2bf(v); }
 template <> DEVINL void store_f<float>(float* p, long i, float v) { p[i] = v; }
 
-template <typename T>
+// 8-wide vector access helpers: V=1 -> vectorized bf16x8/f32 path
+template <typename T, bool VEC>
+DEVINL void load8(const T* p, long i, float out[8]) {
+  if constexpr (VEC && std::is_same<T, short>::value) {
+    bf16x8 v = *(const bf16x8*)(p + i);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = bf2f(v[j]);
+  } else {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) out[j] = load_f(p, i + j);
+  }
+}
+
+template <typename T, bool VEC>
+DEVINL void store8(T* p, long i, const float in[8]) {
+  if constexpr (VEC && std::is_same<T, short>::value) {
+    bf16x8 v;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = f2bf(in[j]);
+    *(bf16x8*)(p + i) = v;
+  } else {
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) store_f(p, i + j, in[j]);
+  }
+}
+
+template <typename T, bool VEC>
 __global__ void __launch_bounds__(256)
 layernorm_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
                      const T* __restrict__ b, T* __restrict__ y,
                      float* __restrict__ mean_out, float* __restrict__ rstd_out,
-                     int D, float eps) {
-  __shared__ float red[8];
-  const long row = blockIdx.x;
+                     int D, float eps, long R) {
+  const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= R) return;
+  const int lane = threadIdx.x & 63;
   const T* xr = x + row * (long)D;
   T* yr = y + row * (long)D;
 
   float sum = 0.f, sumsq = 0.f;
-  for (int i = threadIdx.x; i < D; i += blockDim.x) {
-    float v = load_f(xr, i);
-    sum += v;
-    sumsq += v * v;
+  if constexpr (VEC) {
+    for (int i = lane * 8; i < D; i += 64 * 8) {
+      float v[8];
+      load8<T, true>(xr, i, v);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) { sum += v[j]; sumsq += v[j] * v[j]; }
+    }
+  } else {
+    for (int i = lane; i < D; i += 64) {
+      float v = load_f(xr, i);
+      sum += v;
+      sumsq += v * v;
+    }
   }
-  sum = block_reduce_sum(sum, red);
-  sumsq = block_reduce_sum(sumsq, red);
+  sum = wave_reduce_sum(sum);
+  sumsq = wave_reduce_sum(sumsq);
   const float mean = sum / D;
   const float var = sumsq / D - mean * mean;
   const float rstd = rsqrtf(fmaxf(var, 0.f) + eps);
-  if (threadIdx.x == 0) {
+  if (lane == 0) {
     mean_out[row] = mean;
     rstd_out[row] = rstd;
   }
-  for (int i = threadIdx.x; i < D; i += blockDim.x) {
-    float v = (load_f(xr, i) - mean) * rstd;
-    store_f(yr, i, v * load_f(w, i) + load_f(b, i));
+  if constexpr (VEC) {
+    for (int i = lane * 8; i < D; i += 64 * 8) {
+      float v[8], wv[8], bv[8];
+      load8<T, true>(xr, i, v);
+      load8<T, true>(w, i, wv);
+      load8<T, true>(b, i, bv);
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) v[j] = (v[j] - mean) * rstd * wv[j] + bv[j];
+      store8<T, true>(yr, i, v);
+    }
+  } else {
+    for (int i = lane; i < D; i += 64) {
+      float v = (load_f(xr, i) - mean) * rstd;
+      store_f(yr, i, v * load_f(w, i) + load_f(b, i));
+    }
   }
 }
 
 // dx = rstd * (dyw - mean(dyw) - xhat * mean(dyw * xhat)); dyw = dy * w
-template <typename T>
+// 4 waves per block, each wave owns whole rows; per-wave LDS column
+// accumulators merged at block end into fp32 partials.
+template <typename T, bool VEC>
 __global__ void __launch_bounds__(256)
 layernorm_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                         const T* __restrict__ w, const float* __restrict__ mean,
                         const float* __restrict__ rstd, T* __restrict__ dx,
                         float* __restrict__ dw_part, float* __restrict__ db_part,
                         int D, int rows_per_block, long R) {
-  __shared__ float red[8];
-  extern __shared__ float cols[];  // 2*D partials for dw/db
-  for (int i = threadIdx.x; i < 2 * D; i += blockDim.x) cols[i] = 0.f;
+  extern __shared__ float cols[];  // [4 waves][2][D]
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  float* mycols = cols + (size_t)wid * 2 * D;
+  for (int i = threadIdx.x; i < 4 * 2 * D; i += blockDim.x) cols[i] = 0.f;
   __syncthreads();
 
   const long row0 = (long)blockIdx.x * rows_per_block;
   const long row1 = min(row0 + rows_per_block, R);
-  for (long row = row0; row < row1; ++row) {
+  for (long row = row0 + wid; row < row1; row += 4) {
     const T* dyr = dy + row * (long)D;
     const T* xr = x + row * (long)D;
     T* dxr = dx + row * (long)D;
     const float mu = mean[row], rs = rstd[row];
 
     float s1 = 0.f, s2 = 0.f;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float xh = (load_f(xr, i) - mu) * rs;
-      float dyv = load_f(dyr, i);
-      float dyw = dyv * load_f(w, i);
-      s1 += dyw;
-      s2 += dyw * xh;
-      // column partials accumulate in LDS (no conflict: i strided by block)
-      cols[i] += dyv * xh;        // dgamma partial
-      cols[D + i] += dyv;         // dbeta partial
+    if constexpr (VEC) {
+      for (int i = lane * 8; i < D; i += 64 * 8) {
+        float xv[8], dyv[8], wv[8];
+        load8<T, true>(xr, i, xv);
+        load8<T, true>(dyr, i, dyv);
+        load8<T, true>(w, i, wv);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float xh = (xv[j] - mu) * rs;
+          const float dyw = dyv[j] * wv[j];
+          s1 += dyw;
+          s2 += dyw * xh;
+          mycols[i + j] += dyv[j] * xh;
+          mycols[D + i + j] += dyv[j];
+        }
+      }
+    } else {
+      for (int i = lane; i < D; i += 64) {
+        const float xh = (load_f(xr, i) - mu) * rs;
+        const float dyv = load_f(dyr, i);
+        const float dyw = dyv * load_f(w, i);
+        s1 += dyw;
+        s2 += dyw * xh;
+        mycols[i] += dyv * xh;
+        mycols[D + i] += dyv;
+      }
     }
-    s1 = block_reduce_sum(s1, red) / D;
-    s2 = block_reduce_sum(s2, red) / D;
-    for (int i = threadIdx.x; i < D; i += blockDim.x) {
-      float xh = (load_f(xr, i) - mu) * rs;
-      float dyw = load_f(dyr, i) * load_f(w, i);
-      store_f(dxr, i, rs * (dyw - s1 - xh * s2));
+    s1 = wave_reduce_sum(s1) / D;
+    s2 = wave_reduce_sum(s2) / D;
+    if constexpr (VEC) {
+      for (int i = lane * 8; i < D; i += 64 * 8) {
+        float xv[8], dyv[8], wv[8], o[8];
+        load8<T, true>(xr, i, xv);
+        load8<T, true>(dyr, i, dyv);
+        load8<T, true>(w, i, wv);
+        #pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const float xh = (xv[j] - mu) * rs;
+          o[j] = rs * (dyv[j] * wv[j] - s1 - xh * s2);
+        }
+        store8<T, true>(dxr, i, o);
+      }
+    } else {
+      for (int i = lane; i < D; i += 64) {
+        const float xh = (load_f(xr, i) - mu) * rs;
+        const float dyw = load_f(dyr, i) * load_f(w, i);
+        store_f(dxr, i, rs * (dyw - s1 - xh * s2));
+      }
     }
-    __syncthreads();
   }
+  __syncthreads();
   for (int i = threadIdx.x; i < D; i += blockDim.x) {
-    dw_part[(long)blockIdx.x * D + i] = cols[i];
-    db_part[(long)blockIdx.x * D + i] = cols[D + i];
+    float dwv = 0.f, dbv = 0.f;
+    #pragma unroll
+    for (int wv = 0; wv < 4; ++wv) {
+      dwv += cols[(size_t)wv * 2 * D + i];
+      dbv += cols[(size_t)wv * 2 * D + D + i];
+    }
+    dw_part[(long)blockIdx.x * D + i] = dwv;
+    db_part[(long)blockIdx.x * D + i] = dbv;
   }
 }
 
+// parallel column reduce: 64 cols x 4 row-slices per 256-thread block
 template <typename T>
 __global__ void __launch_bounds__(256)
 col_reduce_kernel(const float* __restrict__ part, T* __restrict__ out,
                   int nblocks, int D) {
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= D) return;
+  __shared__ float acc[4][64];
+  const int c = threadIdx.x & 63;
+  const int col = blockIdx.x * 64 + c;
+  const int slice = threadIdx.x >> 6;
   float s = 0.f;
-  for (int b = 0; b < nblocks; ++b) s += part[(long)b * D + i];
-  store_f(out, i, s);
+  if (col < D)
+    for (int b = slice; b < nblocks; b += 4) s += part[(long)b * D + col];
+  acc[slice][c] = s;
+  __syncthreads();
+  if (slice == 0 && col < D)
+    store_f(out, col, acc[0][c] + acc[1][c] + acc[2][c] + acc[3][c]);
 }
 
 }  // namespace
@@ -119,18 +218,22 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
   auto mean = torch::empty({R}, fopt);
   auto rstd = torch::empty({R}, fopt);
   auto stream = scaffold_stream();
-  dim3 grid((unsigned)R), block(256);
+  dim3 grid((unsigned)((R + 3) / 4)), block(256);
+  const bool vec = (D % 8) == 0;
+  #define LN_FWD(T, V, PX)                                                     \
+    hipLaunchKernelGGL((layernorm_fwd_kernel<T, V>), grid, block, 0, stream,   \
+                       (const T*)x.data_ptr(), (const T*)w.data_ptr(),         \
+                       (const T*)b.data_ptr(), (T*)y.data_ptr(),               \
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(), D,      \
+                       (float)eps, R)
   if (x.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL(layernorm_fwd_kernel<short>, grid, block, 0, stream,
-                       (const short*)x.data_ptr(), (const short*)w.data_ptr(),
-                       (const short*)b.data_ptr(), (short*)y.data_ptr(),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+    if (vec) LN_FWD(short, true, );
+    else LN_FWD(short, false, );
   } else {
-    hipLaunchKernelGGL(layernorm_fwd_kernel<float>, grid, block, 0, stream,
-                       x.data_ptr<float>(), w.data_ptr<float>(),
-                       b.data_ptr<float>(), y.data_ptr<float>(),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(), D, (float)eps);
+    if (vec) LN_FWD(float, true, );
+    else LN_FWD(float, false, );
   }
+  #undef LN_FWD
   HIP_CHECK(hipGetLastError());
   return {y, mean, rstd};
 }
@@ -147,42 +250,40 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto fopt = torch::TensorOptions().dtype(torch::kFloat32).device(x.device());
 
   const int target_blocks = 1024;  // >> 256 CUs
-  const int rows_per_block = (int)std::max<long>(1, (R + target_blocks - 1) / target_blocks);
+  const int rows_per_block = (int)std::max<long>(4, (R + target_blocks - 1) / target_blocks);
   const int nblocks = (int)((R + rows_per_block - 1) / rows_per_block);
   auto dw_part = torch::empty({nblocks, D}, fopt);
   auto db_part = torch::empty({nblocks, D}, fopt);
-  const size_t lds = (size_t)2 * D * sizeof(float);
+  const size_t lds = (size_t)4 * 2 * D * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024 - 1024, "layernorm_bwd: D too large for LDS path");
 
   auto stream = scaffold_stream();
   dim3 block(256);
+  const bool vec = (D % 8) == 0;
+  const int cr_grid = (D + 63) / 64;
+  #define LN_BWD(T, V)                                                         \
+    do {                                                                       \
+      hipLaunchKernelGGL((layernorm_bwd_dx_kernel<T, V>), dim3(nblocks), block,\
+                         lds, stream, (const T*)dy.data_ptr(),                 \
+                         (const T*)x.data_ptr(), (const T*)w.data_ptr(),       \
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
+                         (T*)dx.data_ptr(), dw_part.data_ptr<float>(),         \
+                         db_part.data_ptr<float>(), D, rows_per_block, R);     \
+      hipLaunchKernelGGL((col_reduce_kernel<T>), dim3(cr_grid), block, 0,      \
+                         stream, dw_part.data_ptr<float>(), (T*)dw.data_ptr(), \
+                         nblocks, D);                                          \
+      hipLaunchKernelGGL((col_reduce_kernel<T>), dim3(cr_grid), block, 0,      \
+                         stream, db_part.data_ptr<float>(), (T*)db.data_ptr(), \
+                         nblocks, D);                                          \
+    } while (0)
   if (x.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL(layernorm_bwd_dx_kernel<short>, dim3(nblocks), block, lds,
-                       stream, (const short*)dy.data_ptr(), (const short*)x.data_ptr(),
-                       (const short*)w.data_ptr(), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), (short*)dx.data_ptr(),
-                       dw_part.data_ptr<float>(), db_part.data_ptr<float>(),
-                       D, rows_per_block, R);
-    hipLaunchKernelGGL(col_reduce_kernel<short>, dim3((D + 255) / 256), block, 0,
-                       stream, dw_part.data_ptr<float>(), (short*)dw.data_ptr(),
-                       nblocks, D);
-    hipLaunchKernelGGL(col_reduce_kernel<short>, dim3((D + 255) / 256), block, 0,
-                       stream, db_part.data_ptr<float>(), (short*)db.data_ptr(),
-                       nblocks, D);
+    if (vec) LN_BWD(short, true);
+    else LN_BWD(short, false);
   } else {
-    hipLaunchKernelGGL(layernorm_bwd_dx_kernel<float>, dim3(nblocks), block, lds,
-                       stream, dy.data_ptr<float>(), x.data_ptr<float>(),
-                       w.data_ptr<float>(), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), dx.data_ptr<float>(),
-                       dw_part.data_ptr<float>(), db_part.data_ptr<float>(),
-                       D, rows_per_block, R);
-    hipLaunchKernelGGL(col_reduce_kernel<float>, dim3((D + 255) / 256), block, 0,
-                       stream, dw_part.data_ptr<float>(), dw.data_ptr<float>(),
-                       nblocks, D);
-    hipLaunchKernelGGL(col_reduce_kernel<float>, dim3((D + 255) / 256), block, 0,
-                       stream, db_part.data_ptr<float>(), db.data_ptr<float>(),
-                       nblocks, D);
+    if (vec) LN_BWD(float, true);
+    else LN_BWD(float, false);
   }
+  #undef LN_BWD
   HIP_CHECK(hipGetLastError());
   return {dx, dw, db};
 }

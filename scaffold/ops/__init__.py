@@ -95,7 +95,7 @@ class _Rope(torch.autograd.Function):
         m = require_ext()
         cos = cos.float().contiguous()  # tables stay fp32 even if model
         sin = sin.float().contiguous()  # buffers were cast to bf16
-        y = m.rope_apply(x.contiguous(), cos, sin, False)
+        y = m.rope_apply(x, cos, sin, False)
         ctx.save_for_backward(cos, sin)
         return y
 
@@ -103,7 +103,7 @@ class _Rope(torch.autograd.Function):
     def backward(ctx, dy):
         cos, sin = ctx.saved_tensors
         m = require_ext()
-        dx = m.rope_apply(dy.contiguous(), cos, sin, True)  # inverse rotation
+        dx = m.rope_apply(dy, cos, sin, True)  # inverse rotation
         return dx, None, None
 
 
@@ -182,7 +182,7 @@ class _Attention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, causal):
         m = require_ext()
-        o, lse = m.attn_fwd(q.contiguous(), k.contiguous(), v.contiguous(), causal)
+        o, lse = m.attn_fwd(q, k, v, causal)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal = causal
         return o
@@ -191,7 +191,7 @@ class _Attention(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         m = require_ext()
-        dq, dk, dv = m.attn_bwd(do.contiguous(), q, k, v, o, lse, ctx.causal)
+        dq, dk, dv = m.attn_bwd(do, q, k, v, o, lse, ctx.causal)
         return dq, dk, dv, None
 
 
