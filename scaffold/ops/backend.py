@@ -31,6 +31,7 @@ _SOURCES = [
     "kernels/attn_fwd_mfma.hip",
     "kernels/attn_bwd.hip",
     "kernels/attn_bwd_mfma.hip",
+    "kernels/gemm_bf16.hip",
     "kernels/probe.hip",
 ]
 
