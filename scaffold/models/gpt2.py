@@ -37,12 +37,16 @@ class CausalSelfAttention(nn.Module):
         self.qkv = nn.Linear(d_model, 3 * d_model)
         self.proj = nn.Linear(d_model, d_model)
 
-    def forward(self, x):
+    def forward(self, x, residual):
         B, S, C = x.shape
-        qkv = self.qkv(x).view(B, S, 3, self.n_heads, self.head_dim)
+        qkv = ops.linear(x, self.qkv.weight, self.qkv.bias)
+        qkv = qkv.view(B, S, 3, self.n_heads, self.head_dim)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # (B,H,S,D)
         o = ops.attention(q, k, v, causal=True)
-        return self.proj(o.transpose(1, 2).reshape(B, S, C))
+        # output projection with the residual add fused into the epilogue
+        return ops.linear_residual(
+            o.transpose(1, 2).reshape(B, S, C),
+            self.proj.weight, self.proj.bias, residual)
 
 
 class Block(nn.Module):
@@ -55,8 +59,11 @@ class Block(nn.Module):
         self.proj = nn.Linear(d_ff, d_model)
 
     def forward(self, x):
-        x = x + self.attn(self.ln1(x))
-        x = x + self.proj(F.gelu(self.fc(self.ln2(x)), approximate="tanh"))
+        # hot ops routed through scaffold.ops: K5 hand MFMA linears with
+        # fused bias+GELU (fc) and fused residual-add (projections)
+        x = self.attn(self.ln1(x), residual=x)
+        h = ops.linear_gelu(self.ln2(x), self.fc.weight, self.fc.bias)
+        x = ops.linear_residual(h, self.proj.weight, self.proj.bias, x)
         return x
 
 

@@ -36,15 +36,16 @@ class LlamaAttention(nn.Module):
         self.wv = nn.Linear(d_model, n_kv_heads * self.head_dim, bias=False)
         self.wo = nn.Linear(n_heads * self.head_dim, d_model, bias=False)
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, residual):
         B, S, _ = x.shape
-        q = self.wq(x).view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
-        k = self.wk(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
-        v = self.wv(x).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        q = ops.linear(x, self.wq.weight).view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
+        k = ops.linear(x, self.wk.weight).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        v = ops.linear(x, self.wv.weight).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
         q = ops.rope_apply(q, cos, sin)
         k = ops.rope_apply(k, cos, sin)
         o = ops.attention(q, k, v, causal=True)
-        return self.wo(o.transpose(1, 2).reshape(B, S, -1))
+        return ops.linear_residual(
+            o.transpose(1, 2).reshape(B, S, -1), self.wo.weight, None, residual)
 
 
 class LlamaBlock(nn.Module):
@@ -58,9 +59,10 @@ class LlamaBlock(nn.Module):
         self.w_down = nn.Linear(d_ff, d_model, bias=False)
 
     def forward(self, x, cos, sin):
-        x = x + self.attn(self.attn_norm(x), cos, sin)
+        x = self.attn(self.attn_norm(x), cos, sin, residual=x)
         h = self.mlp_norm(x)
-        x = x + self.w_down(F.silu(self.w_gate(h)) * self.w_up(h))
+        g = F.silu(ops.linear(h, self.w_gate.weight)) * ops.linear(h, self.w_up.weight)
+        x = ops.linear_residual(g, self.w_down.weight, None, x)
         return x
 
 

@@ -25,6 +25,9 @@ __all__ = [
     "rope_apply",
     "softmax_cross_entropy",
     "attention",
+    "linear",
+    "linear_gelu",
+    "linear_residual",
     "fused_adamw_step",
     "require_ext",
     "has_ext",
@@ -210,6 +213,109 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if q.is_cuda:
         return _Attention.apply(q, k, v, causal)
     return _attention_eager(q, k, v, causal)
+
+
+# ------------------------------------------------- K5 hand MFMA linear
+def _tile_ok(x: torch.Tensor, w: torch.Tensor) -> bool:
+    """The hand kernel covers the benched transformer-block shapes
+    (SURVEY §2.3 K5): M%128, N%128, K%64.  Other shapes (e.g. the
+    vocab-sized lm_head) take the hipBLASLt library-GEMM path, per
+    SURVEY §2.2 ("hipBLASLt/rocBLAS for general GEMM")."""
+    K = x.shape[-1]
+    M = x.numel() // K
+    return (x.dtype == torch.bfloat16 and w.dtype == torch.bfloat16
+            and M % 128 == 0 and w.shape[0] % 128 == 0 and K % 64 == 0)
+
+
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, b):
+        m = require_ext()
+        x = x.contiguous()
+        (y,) = m.linear_fwd(x, w, b, 0, None)
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        x2 = x.view(-1, x.shape[-1])
+        dx = (dy2 @ w).view(x.shape)
+        dw = dy2.t() @ x2
+        db = dy2.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+class _LinearGeluFn(torch.autograd.Function):
+    """fc + bias + tanh-GELU in ONE kernel (the GPT-2 MLP up-projection).
+    Saves the pre-activation for an exact GELU backward."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        m = require_ext()
+        x = x.contiguous()
+        y, pre = m.linear_fwd(x, w, b, 1, None)
+        ctx.save_for_backward(x, w, pre)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w, pre = ctx.saved_tensors
+        dpre = torch.ops.aten.gelu_backward(dy, pre, approximate="tanh")
+        dpre2 = dpre.contiguous().view(-1, dpre.shape[-1])
+        x2 = x.view(-1, x.shape[-1])
+        dx = (dpre2 @ w).view(x.shape)
+        dw = dpre2.t() @ x2
+        db = dpre2.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+class _LinearResidualFn(torch.autograd.Function):
+    """proj + bias + residual-add in ONE kernel (transformer residual
+    branches) — removes a full elementwise add pass over HBM."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, res):
+        m = require_ext()
+        x = x.contiguous()
+        (y,) = m.linear_fwd(x, w, b, 2, res.contiguous())
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        x2 = x.view(-1, x.shape[-1])
+        dx = (dy2 @ w).view(x.shape)
+        dw = dy2.t() @ x2
+        db = dy2.sum(0) if ctx.has_bias else None
+        return dx, dw, db, dy
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None
+           ) -> torch.Tensor:
+    if x.is_cuda and _tile_ok(x, w):
+        return _LinearFn.apply(x, w, b)
+    return F.linear(x, w, b)
+
+
+def linear_gelu(x: torch.Tensor, w: torch.Tensor,
+                b: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if x.is_cuda and _tile_ok(x, w):
+        return _LinearGeluFn.apply(x, w, b)
+    return F.gelu(F.linear(x, w, b), approximate="tanh")
+
+
+def linear_residual(x: torch.Tensor, w: torch.Tensor,
+                    b: Optional[torch.Tensor], res: torch.Tensor) -> torch.Tensor:
+    if x.is_cuda and _tile_ok(x, w):
+        return _LinearResidualFn.apply(x, w, b, res)
+    return F.linear(x, w, b) + res
 
 
 # ------------------------------------------------------------- fused AdamW

@@ -211,3 +211,80 @@ def test_attention_fwd_bwd(ext, B, H, Hkv, S, D, causal):
         err = (g - w).abs().max().item()
         tol = 0.12 * max(1.0, w.abs().max().item())
         assert err < tol, f"{name}: max err {err} vs tol {tol}"
+
+
+# ------------------------------------------------------- K5 MFMA linear
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (256, 384, 192), (512, 768, 320)])
+@pytest.mark.parametrize("epi", ["none", "gelu", "resid"])
+def test_linear_fwd_vs_fp32(ext, M, N, K, epi):
+    torch.manual_seed(11)
+    x = torch.randn(M, K)
+    w = torch.randn(N, K) * 0.05  # asymmetric, scaled like real weights
+    b = torch.randn(N)
+    res = torch.randn(M, N)
+    xb, wb, bb, rb = (_bf(t).cuda() for t in (x, w, b, res))
+
+    ref = _bf(x).float() @ _bf(w).float().t() + _bf(b).float()
+    if epi == "none":
+        (y,) = ext.linear_fwd(xb, wb, bb, 0, None)
+        want = ref
+    elif epi == "gelu":
+        y, pre = ext.linear_fwd(xb, wb, bb, 1, None)
+        assert torch.allclose(pre.cpu().float(), ref, rtol=3e-2, atol=3e-2)
+        want = torch.nn.functional.gelu(ref, approximate="tanh")
+    else:
+        (y,) = ext.linear_fwd(xb, wb, bb, 2, rb)
+        want = ref + _bf(res).float()
+    got = y.cpu().float()
+    err = (got - want).abs().max().item()
+    tol = 2e-2 * max(1.0, want.abs().max().item())
+    assert err < tol, f"{epi} M{M} N{N} K{K}: max err {err} vs tol {tol}"
+
+
+def test_linear_fwd_no_bias(ext):
+    torch.manual_seed(12)
+    x = _bf(torch.randn(128, 64)).cuda()
+    w = _bf(torch.randn(128, 64) * 0.05).cuda()
+    (y,) = ext.linear_fwd(x, w, None, 0, None)
+    ref = x.float() @ w.float().t()
+    assert torch.allclose(y.cpu().float(), ref.cpu(), rtol=2e-2, atol=2e-2)
+
+
+def test_linear_autograd_matches_eager():
+    """ops.linear/linear_gelu/linear_residual fwd+bwd vs bf16 eager."""
+    from scaffold import ops
+
+    torch.manual_seed(13)
+    M, N, K = 256, 256, 128
+    for fn, eager in [
+        (lambda x, w, b: ops.linear(x, w, b),
+         lambda x, w, b: torch.nn.functional.linear(x, w, b)),
+        (lambda x, w, b: ops.linear_gelu(x, w, b),
+         lambda x, w, b: torch.nn.functional.gelu(
+             torch.nn.functional.linear(x, w, b), approximate="tanh")),
+    ]:
+        x = _bf(torch.randn(M, K)).cuda().requires_grad_(True)
+        w = _bf(torch.randn(N, K) * 0.05).cuda().requires_grad_(True)
+        b = _bf(torch.randn(N)).cuda().requires_grad_(True)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        b2 = b.detach().clone().requires_grad_(True)
+        dy = _bf(torch.randn(M, N)).cuda()
+        y = fn(x, w, b); y.backward(dy)
+        ye = eager(x2, w2, b2); ye.backward(dy)
+        torch.cuda.synchronize()
+        for name, g, ge in [("y", y, ye), ("dx", x.grad, x2.grad),
+                            ("dw", w.grad, w2.grad), ("db", b.grad, b2.grad)]:
+            a, e = g.detach().float(), ge.detach().float()
+            err = (a - e).abs().max().item()
+            tol = 3e-2 * max(1.0, e.abs().max().item())
+            assert err < tol, f"{name}: {err} vs {tol}"
+
+    # residual variant
+    x = _bf(torch.randn(M, K)).cuda().requires_grad_(True)
+    w = _bf(torch.randn(N, K) * 0.05).cuda().requires_grad_(True)
+    r = _bf(torch.randn(M, N)).cuda().requires_grad_(True)
+    y = ops.linear_residual(x, w, None, r)
+    y.backward(_bf(torch.ones(M, N)).cuda())
+    torch.cuda.synchronize()
+    assert torch.allclose(r.grad.float(), torch.ones(M, N).cuda(), atol=0)
