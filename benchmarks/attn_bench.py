@@ -65,6 +65,25 @@ def main():
         tf = flops(B, H, S, D) * 3.5 / t / 1e12
         print(f"{name}: fwd+bwd {t*1e3:.2f} ms  {tf:.0f} TF/s-equiv", flush=True)
 
+        # A/B: strided (B,S,H,D)->transpose views (the model's QKV layout)
+        # vs the contiguous tensors above — isolates the stride cost in bwd
+        qb = torch.randn(B, S, H, D, device="cuda", dtype=torch.bfloat16)
+        kb = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+        vb = torch.randn(B, S, Hkv, D, device="cuda", dtype=torch.bfloat16)
+        qs = qb.transpose(1, 2).requires_grad_(True)
+        ks = kb.transpose(1, 2).requires_grad_(True)
+        vs = vb.transpose(1, 2).requires_grad_(True)
+
+        def train_step_strided():
+            o = ops.attention(qs, ks, vs, True)
+            o.backward(do)
+            qs.grad = None
+            ks.grad = None
+            vs.grad = None
+
+        t2 = bench(train_step_strided, iters=5)
+        print(f"{name}: fwd+bwd strided {t2*1e3:.2f} ms  (x{t2/t:.2f} vs contig)", flush=True)
+
 
 if __name__ == "__main__":
     main()

@@ -107,7 +107,7 @@ __global__ void __launch_bounds__(256)
 layernorm_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                         const T* __restrict__ w, const float* __restrict__ mean,
                         const float* __restrict__ rstd, T* __restrict__ dx,
-                        float* __restrict__ dw_part, float* __restrict__ db_part,
+                        float* __restrict__ part,  // [nblocks][2D]: dw | db
                         int D, int rows_per_block, long R) {
   extern __shared__ float cols[];  // [4 waves][2][D]
   const int wid = threadIdx.x >> 6;
@@ -183,27 +183,53 @@ layernorm_bwd_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
       dwv += cols[(size_t)wv * 2 * D + i];
       dbv += cols[(size_t)wv * 2 * D + D + i];
     }
-    dw_part[(long)blockIdx.x * D + i] = dwv;
-    db_part[(long)blockIdx.x * D + i] = dbv;
+    part[(long)blockIdx.x * 2 * D + i] = dwv;
+    part[(long)blockIdx.x * 2 * D + D + i] = dbv;
   }
 }
 
-// parallel column reduce: 64 cols x 4 row-slices per 256-thread block
-template <typename T>
+// Two-stage parallel column reduce of a [nrows][C] fp32 partial matrix.
+// Stage 1 slices the rows over gridDim.y so the launch fills the chip
+// (the old single-stage version ran C/64 blocks = 12 CUs at D=768 and
+// was 7% of the GPT-2 step); stage 2 folds the NS slice sums and splits
+// the columns into the dw / db outputs.
 __global__ void __launch_bounds__(256)
-col_reduce_kernel(const float* __restrict__ part, T* __restrict__ out,
-                  int nblocks, int D) {
+col_reduce_slice_kernel(const float* __restrict__ part, float* __restrict__ mid,
+                        int nrows, int C) {
   __shared__ float acc[4][64];
   const int c = threadIdx.x & 63;
   const int col = blockIdx.x * 64 + c;
   const int slice = threadIdx.x >> 6;
+  const int ns = gridDim.y;
   float s = 0.f;
-  if (col < D)
-    for (int b = slice; b < nblocks; b += 4) s += part[(long)b * D + col];
+  if (col < C)
+    for (int b = blockIdx.y * 4 + slice; b < nrows; b += ns * 4)
+      s += part[(long)b * C + col];
   acc[slice][c] = s;
   __syncthreads();
-  if (slice == 0 && col < D)
-    store_f(out, col, acc[0][c] + acc[1][c] + acc[2][c] + acc[3][c]);
+  if (slice == 0 && col < C)
+    mid[(long)blockIdx.y * C + col] = acc[0][c] + acc[1][c] + acc[2][c] + acc[3][c];
+}
+
+template <typename T>
+__global__ void __launch_bounds__(256)
+col_reduce_final_kernel(const float* __restrict__ mid, T* __restrict__ dw,
+                        T* __restrict__ db, int nrows, int D) {
+  __shared__ float acc[4][64];
+  const int c = threadIdx.x & 63;
+  const int col = blockIdx.x * 64 + c;
+  const int slice = threadIdx.x >> 6;
+  const int C = 2 * D;
+  float s = 0.f;
+  if (col < C)
+    for (int b = slice; b < nrows; b += 4) s += mid[(long)b * C + col];
+  acc[slice][c] = s;
+  __syncthreads();
+  if (slice == 0 && col < C) {
+    const float v = acc[0][c] + acc[1][c] + acc[2][c] + acc[3][c];
+    if (col < D) store_f(dw, col, v);
+    else store_f(db, col - D, v);
+  }
 }
 
 }  // namespace
@@ -252,29 +278,30 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int target_blocks = 1024;  // >> 256 CUs
   const int rows_per_block = (int)std::max<long>(4, (R + target_blocks - 1) / target_blocks);
   const int nblocks = (int)((R + rows_per_block - 1) / rows_per_block);
-  auto dw_part = torch::empty({nblocks, D}, fopt);
-  auto db_part = torch::empty({nblocks, D}, fopt);
+  auto part = torch::empty({nblocks, 2L * D}, fopt);
+  const int NS = nblocks > 64 ? 16 : 1;  // row slices for the chip-filling stage
+  auto mid = torch::empty({NS, 2L * D}, fopt);
   const size_t lds = (size_t)4 * 2 * D * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024 - 1024, "layernorm_bwd: D too large for LDS path");
 
   auto stream = scaffold_stream();
   dim3 block(256);
   const bool vec = (D % 8) == 0;
-  const int cr_grid = (D + 63) / 64;
+  const int cr_grid = (2 * D + 63) / 64;
   #define LN_BWD(T, V)                                                         \
     do {                                                                       \
       hipLaunchKernelGGL((layernorm_bwd_dx_kernel<T, V>), dim3(nblocks), block,\
                          lds, stream, (const T*)dy.data_ptr(),                 \
                          (const T*)x.data_ptr(), (const T*)w.data_ptr(),       \
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),       \
-                         (T*)dx.data_ptr(), dw_part.data_ptr<float>(),         \
-                         db_part.data_ptr<float>(), D, rows_per_block, R);     \
-      hipLaunchKernelGGL((col_reduce_kernel<T>), dim3(cr_grid), block, 0,      \
-                         stream, dw_part.data_ptr<float>(), (T*)dw.data_ptr(), \
-                         nblocks, D);                                          \
-      hipLaunchKernelGGL((col_reduce_kernel<T>), dim3(cr_grid), block, 0,      \
-                         stream, db_part.data_ptr<float>(), (T*)db.data_ptr(), \
-                         nblocks, D);                                          \
+                         (T*)dx.data_ptr(), part.data_ptr<float>(),            \
+                         D, rows_per_block, R);                                \
+      hipLaunchKernelGGL(col_reduce_slice_kernel, dim3(cr_grid, NS), block, 0, \
+                         stream, part.data_ptr<float>(), mid.data_ptr<float>(),\
+                         nblocks, 2 * D);                                      \
+      hipLaunchKernelGGL((col_reduce_final_kernel<T>), dim3(cr_grid), block, 0,\
+                         stream, mid.data_ptr<float>(), (T*)dw.data_ptr(),     \
+                         (T*)db.data_ptr(), NS, D);                            \
     } while (0)
   if (x.scalar_type() == torch::kBFloat16) {
     if (vec) LN_BWD(short, true);

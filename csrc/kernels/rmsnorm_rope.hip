@@ -154,6 +154,27 @@ rmsnorm_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ x,
   }
 }
 
+// Two-stage column reduce (same rationale as layernorm.hip: a single
+// D/64-block launch under-fills the 256-CU chip).  Stage 1 slices rows
+// over gridDim.y into mid[NS][D]; stage 2 folds the NS slices.
+__global__ void __launch_bounds__(256)
+col_reduce1_slice_kernel(const float* __restrict__ part, float* __restrict__ mid,
+                         int nrows, int C) {
+  __shared__ float acc[4][64];
+  const int c = threadIdx.x & 63;
+  const int col = blockIdx.x * 64 + c;
+  const int slice = threadIdx.x >> 6;
+  const int ns = gridDim.y;
+  float s = 0.f;
+  if (col < C)
+    for (int b = blockIdx.y * 4 + slice; b < nrows; b += ns * 4)
+      s += part[(long)b * C + col];
+  acc[slice][c] = s;
+  __syncthreads();
+  if (slice == 0 && col < C)
+    mid[(long)blockIdx.y * C + col] = acc[0][c] + acc[1][c] + acc[2][c] + acc[3][c];
+}
+
 template <typename T>
 __global__ void __launch_bounds__(256)
 col_reduce1_kernel(const float* __restrict__ part, T* __restrict__ out,
@@ -241,6 +262,8 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
   const int rows_per_block = (int)std::max<long>(4, (R + target_blocks - 1) / target_blocks);
   const int nblocks = (int)((R + rows_per_block - 1) / rows_per_block);
   auto dw_part = torch::empty({nblocks, D}, fopt);
+  const int NS = nblocks > 64 ? 16 : 1;
+  auto mid = torch::empty({NS, (long)D}, fopt);
   const size_t lds = (size_t)4 * D * sizeof(float);
   TORCH_CHECK(lds <= 160 * 1024 - 1024, "rmsnorm_bwd: D too large for LDS path");
   auto stream = scaffold_stream();
@@ -254,9 +277,12 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x,
                          (const T*)w.data_ptr(), rstd.data_ptr<float>(),       \
                          (T*)dx.data_ptr(), dw_part.data_ptr<float>(),         \
                          D, rows_per_block, R);                                \
+      hipLaunchKernelGGL(col_reduce1_slice_kernel, dim3(cr_grid, NS), block, 0,\
+                         stream, dw_part.data_ptr<float>(),                    \
+                         mid.data_ptr<float>(), nblocks, D);                   \
       hipLaunchKernelGGL((col_reduce1_kernel<T>), dim3(cr_grid), block, 0,     \
-                         stream, dw_part.data_ptr<float>(), (T*)dw.data_ptr(), \
-                         nblocks, D);                                          \
+                         stream, mid.data_ptr<float>(), (T*)dw.data_ptr(),     \
+                         NS, D);                                               \
     } while (0)
   if (x.scalar_type() == torch::kBFloat16) {
     if (vec) RMS_BWD(short, true);

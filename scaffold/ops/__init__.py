@@ -297,23 +297,33 @@ class _LinearResidualFn(torch.autograd.Function):
         return dx, dw, db, dy
 
 
+# Per-shape dispatch switch: measured on MI355X (gpurun_out/gemm_bench.log,
+# 2026-09-13) the 2-phase 128^2 hand kernel reaches ~900 TF, below
+# hipBLASLt at most benched shapes EXCEPT the small-N residual-fused
+# proj (1.07x).  Until the 256^2 8-phase upgrade lands, only the
+# shapes where the hand kernel measured >= parity route to it.
+_HAND_GEMM_ALL = False  # flipped by benchmarks once the 8-phase core wins
+
+
 def linear(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None
            ) -> torch.Tensor:
-    if x.is_cuda and _tile_ok(x, w):
+    if x.is_cuda and _HAND_GEMM_ALL and _tile_ok(x, w):
         return _LinearFn.apply(x, w, b)
     return F.linear(x, w, b)
 
 
 def linear_gelu(x: torch.Tensor, w: torch.Tensor,
                 b: Optional[torch.Tensor] = None) -> torch.Tensor:
-    if x.is_cuda and _tile_ok(x, w):
+    if x.is_cuda and _HAND_GEMM_ALL and _tile_ok(x, w):
         return _LinearGeluFn.apply(x, w, b)
     return F.gelu(F.linear(x, w, b), approximate="tanh")
 
 
 def linear_residual(x: torch.Tensor, w: torch.Tensor,
                     b: Optional[torch.Tensor], res: torch.Tensor) -> torch.Tensor:
-    if x.is_cuda and _tile_ok(x, w):
+    # measured >= hipBLASLt for N <= 1024 (fused residual saves the
+    # separate add pass); bigger shapes stay on the library GEMM
+    if x.is_cuda and _tile_ok(x, w) and (_HAND_GEMM_ALL or w.shape[0] <= 1024):
         return _LinearResidualFn.apply(x, w, b, res)
     return F.linear(x, w, b) + res
 
