@@ -109,6 +109,44 @@ inline Strides4 strides_of(const torch::Tensor& t) {
   return {t.stride(0), t.stride(1), t.stride(2)};
 }
 
+// ---- LDS transpose-read machinery (guide T10, §2 "ds_read_b64_tr_b16")
+//
+// The tr-image of a [KT rows][D cols] bf16 tile packs 4x16 subtiles
+// consecutively: element (k, d) lives at
+//   el = ((k>>2) * (D/16) + (d>>4)) * 64 + (k&3)*16 + (d&15)
+// Written coalesced (16-aligned 8-element chunks), read as columns with
+// ds_read_b64_tr_b16: lane byte addr 2*(subtile*64 + (d&15)) returns the
+// 4 bf16 of rows 4*(k>>2)..+3 at column d — a free 4x4 transpose.
+
+// async global->LDS copy, 16 B per lane (guide §5 common-mistake 1);
+// dest is wave-uniform base + lane*16, source is per-lane.
+DEVINL void glds16(const void* g, void* l) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)l, 16, 0, 0);
+}
+
+DEVINL unsigned lds_byte_off(const void* p) {
+  // generic -> LDS address space; uintptr of an AS(3) pointer is the
+  // LDS byte offset the DS instructions take.
+  return (unsigned)(uintptr_t)(__attribute__((address_space(3))) const void*)p;
+}
+
+// two tr reads (k-quads q4 and q4+1) -> one bf16x8 A/B fragment in k-order
+DEVINL bf16x8 tr_read_frag(unsigned addr_q0, unsigned addr_q1) {
+  unsigned long long a, b;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %3\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(a), "=&v"(b)
+      : "v"(addr_q0), "v"(addr_q1));
+  union { struct { unsigned long long lo, hi; } u; bf16x8 v; } r;
+  r.u.lo = a;
+  r.u.hi = b;
+  return r.v;
+}
+
 // Grid sizing guideline 11: cap grid, grid-stride the rest.
 DEVINL long grid_stride() { return (long)gridDim.x * blockDim.x; }
 

@@ -32,6 +32,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor o, torch::Tensor lse,
                                     bool causal);
 torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B, long which);
+torch::Tensor probe_tr(torch::Tensor offsets);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
                                       c10::optional<torch::Tensor> bias,
                                       long epilogue,
@@ -49,6 +50,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_fwd", &attn_fwd, "flash attention forward (K4)");
   m.def("attn_bwd", &attn_bwd, "flash attention backward (K4)");
   m.def("probe_mfma", &probe_mfma, "MFMA fragment-layout probe (tests)");
+  m.def("probe_tr", &probe_tr, "ds_read_b64_tr_b16 semantics probe (tests)");
   m.def("linear_fwd", &linear_fwd,
         "MFMA linear fwd, fused bias+GELU/residual epilogue (K5)");
 }

@@ -1,14 +1,19 @@
 // K4 — MFMA flash attention forward (bf16, CDNA4 32x32x16 matrix cores).
 //
 // Structure (cdna_hip_programming.md §B "fused attention prefill", adapted):
-//  * 4 waves / workgroup, each wave owns 32 q rows INDEPENDENTLY (no LDS,
-//    no barriers — K/V tiles are L2-resident at these S, and LDS staging
-//    of L2-fit data measured as pure overhead: guide common-mistake #7).
+//  * 4 waves / workgroup over a 128-row q block; wave w owns the
+//    INTERLEAVED rows q0 + 4*i + w (i=0..31) so every wave sees the same
+//    causal kv range — no idle tail waves at the block's barriers.
 //  * swapped QK^T: S^T = mfma(A=K_tile, B=Q^T) so each lane holds a full
 //    q-row slice of scores -> softmax is lane-local + one shfl_xor(32)
 //    to combine the two half-wave reg sets.  (guide §B "swapped QK^T")
 //  * P -> bf16 B-fragments via pack + permlane32_swap (T12: one swap
 //    fills two output words).
+//  * V tile cooperatively staged in LDS in the tr-image layout
+//    (common.h) by global_load_lds, then consumed as V^T A-fragments via
+//    ds_read_b64_tr_b16 (T10) — replaces the 32/64 scalar 2-B global
+//    column loads per tile of the v1 kernel, and makes the kernel
+//    insensitive to the (B,S,H,D)-transposed stride of the models' QKV.
 //  * O^T accumulated via mfma(A=V^T, B=P^T): C cols = q = lane&31, so
 //    the online-softmax rescale is a pure lane-local multiply.
 //
@@ -39,23 +44,27 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
                      Strides4 vs, Strides4 os) {
   constexpr int NSL = D / 16;   // QK^T k-slices (over d)
   constexpr int NDT = D / 32;   // O^T d-tiles
+  constexpr int NSLOT = 32 * D / 8 / 256;  // glds16 per thread per tile
+  __shared__ __attribute__((aligned(16))) short vtile[32 * D];  // tr image
+
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
-  const int lq = l & 31;        // this lane's q row (within wave tile)
+  const int lq = l & 31;
   const int half = l >> 5;
 
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int hk = h / (H / Hkv);
-  const int q0 = blockIdx.x * 128 + wid * 32;   // wave's q-tile origin
+  const int q0 = blockIdx.x * 128;
   if (q0 >= S) return;
 
   const long qb = (long)b * qs.sb + (long)h * qs.sh;
   const long kb = (long)b * ks.sb + (long)hk * ks.sh;
   const long vb = (long)b * vs.sb + (long)hk * vs.sh;
 
-  // ---- load Q B-fragments once (row clamped at S-1 for pad rows)
-  const int qrow = min(q0 + lq, S - 1);
+  // this lane's q row: interleaved assignment q0 + 4*lq + wid
+  const int qrow_g = q0 + 4 * lq + wid;
+  const int qrow = min(qrow_g, S - 1);
   bf16x8_t qf[NSL];
   #pragma unroll
   for (int s = 0; s < NSL; ++s)
@@ -66,8 +75,37 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   for (int t = 0; t < NDT; ++t) oacc[t] = {};
   float m_run = -1e30f, l_run = 0.f;
 
-  const int kv_end = causal ? min(S, q0 + 32) : S;
+  // tr-read addresses for the PV step (per d-tile t, k-quad pair at s)
+  // subtile(q4, g) at element (q4*(D/16)+g)*64; lane column = l&15
+  unsigned tr_addr[NDT][2][2];
+  {
+    const unsigned base = lds_byte_off(vtile);
+    #pragma unroll
+    for (int t = 0; t < NDT; ++t)
+      #pragma unroll
+      for (int s = 0; s < 2; ++s) {
+        const int g = 2 * t + (lq >> 4);
+        const int q4 = 4 * s + 2 * half;
+        tr_addr[t][s][0] = base + 2 * ((unsigned)(q4 * (D / 16) + g) * 64 + (l & 15));
+        tr_addr[t][s][1] = tr_addr[t][s][0] + 2 * 64 * (unsigned)(D / 16);
+      }
+  }
+
+  const int kv_end = causal ? min(S, q0 + 128) : S;
   for (int c0 = 0; c0 < kv_end; c0 += 32) {
+    // ---- cooperative V-tile stage into the tr image (glds16)
+    #pragma unroll
+    for (int it = 0; it < NSLOT; ++it) {
+      const int s8 = (it * 4 + wid) * 64 + l;       // 8-element chunk id
+      const int sub = s8 >> 3;
+      const int rem8 = s8 & 7;                      // (k&3)*2 + c8
+      const int krow = min(c0 + 4 * (sub / (D / 16)) + (rem8 >> 1), S - 1);
+      const int d0 = 16 * (sub % (D / 16)) + 8 * (rem8 & 1);
+      glds16(v + vb + (long)krow * vs.ss + d0,
+             vtile + (long)(it * 4 + wid) * 512);
+    }
+    __syncthreads();  // drains the glds (vmcnt 0) + barrier
+
     // ---- S^T = K_tile . Q^T  (32k x 32q)
     f32x16_t sacc = {};
     const int krow = min(c0 + lq, S - 1);
@@ -77,8 +115,8 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
       sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
     }
 
-    // ---- scale + mask (lane: col=q=lq; reg r: row=k idx)
-    const bool need_mask = (c0 + 31 > q0 && causal) || (c0 + 31 >= S);
+    // ---- scale + mask (lane: col=q; reg r: row=k idx)
+    const bool need_mask = (causal && c0 + 31 > qrow_g) || (c0 + 31 >= S);
     float p[16];
     float mtile = -1e30f;
     #pragma unroll
@@ -87,7 +125,7 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
       float sv = sacc[r] * scale;
       if (need_mask) {
         const int kg = c0 + krel;
-        if (kg >= S || (causal && kg > q0 + lq)) sv = -1e30f;
+        if (kg >= S || (causal && kg > qrow_g)) sv = -1e30f;
       }
       p[r] = sv;
       mtile = fmaxf(mtile, sv);
@@ -111,7 +149,6 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
       for (int r = 0; r < 16; ++r) oacc[t][r] *= alpha;
 
     // ---- P^T B-fragments: pack quads, permlane32_swap halves (T12)
-    // slice s (k=16s..16s+15): frag words = [a.x', a.y', b.x', b.y']
     unsigned pfrag[2][4];
     #pragma unroll
     for (int s = 0; s < 2; ++s) {
@@ -127,28 +164,23 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
       pfrag[s][3] = ry[1];
     }
 
-    // ---- O^T += V^T . P^T  per d-tile, 2 k-slices of 16
+    // ---- O^T += V^T . P^T per d-tile; V^T A-fragments by tr-read (T10)
     #pragma unroll
     for (int t = 0; t < NDT; ++t) {
       #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        // A = V^T: lane -> row=d=32t+lq, k=16s+8*half+j  (column read of V)
-        short vf[8];
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int kg = min(c0 + 16 * s + 8 * half + j, S - 1);
-          vf[j] = v[vb + (long)kg * vs.ss + 32 * t + lq];
-        }
+        bf16x8_t vf = tr_read_frag(tr_addr[t][s][0], tr_addr[t][s][1]);
         oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            *(const bf16x8_t*)vf, *(const bf16x8_t*)pfrag[s], oacc[t], 0, 0, 0);
+            vf, *(const bf16x8_t*)pfrag[s], oacc[t], 0, 0, 0);
       }
     }
+    __syncthreads();  // all waves done with vtile before restage
   }
 
   // ---- epilogue: O = O^T / l, scatter store (lane's q row fixed)
   const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
-  if (q0 + lq < S) {
-    const long obase = (long)b * os.sb + (long)h * os.sh + (long)(q0 + lq) * os.ss;
+  if (qrow_g < S) {
+    const long obase = (long)b * os.sb + (long)h * os.sh + (long)qrow_g * os.ss;
     #pragma unroll
     for (int t = 0; t < NDT; ++t)
       #pragma unroll
@@ -157,7 +189,7 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
         o[obase + d] = f2bf(oacc[t][r] * inv_l);
       }
     if (half == 0)
-      lse[((long)b * H + h) * S + q0 + lq] = m_run + __logf(fmaxf(l_run, 1e-30f));
+      lse[((long)b * H + h) * S + qrow_g] = m_run + __logf(fmaxf(l_run, 1e-30f));
   }
 }
 

@@ -39,12 +39,6 @@ typedef float f32x4_t __attribute__((ext_vector_type(4)));
 
 enum Epilogue { EPI_NONE = 0, EPI_GELU = 1, EPI_RESID = 2 };
 
-DEVINL void glds16(const void* g, void* l) {
-  __builtin_amdgcn_global_load_lds(
-      (const __attribute__((address_space(1))) void*)g,
-      (__attribute__((address_space(3))) void*)l, 16, 0, 0);
-}
-
 DEVINL float gelu_tanh(float v) {
   const float c = 0.7978845608028654f;  // sqrt(2/pi)
   return 0.5f * v * (1.0f + tanhf(c * (v + 0.044715f * v * v * v)));

@@ -55,7 +55,39 @@ __global__ void probe_mfma_16x16x32(const short* __restrict__ A,  // [16][32]
     C[((l >> 4) * 4 + r) * 16 + (l & 15)] = c[r];
 }
 
+// ds_read_b64_tr_b16 semantics probe: LDS holds bf16 elements with
+// value == element index (0..1023); each lane issues ONE tr read at
+// base + off[lane] bytes and dumps its 4 received elements.  A host
+// test infers the lane->element map from the dump (guide T10's wording
+// is ambiguous between a per-lane gather and a cross-lane exchange).
+__global__ void probe_tr_kernel(const int* __restrict__ off,
+                                short* __restrict__ out) {
+  __shared__ __attribute__((aligned(16))) short lds[1024];
+  const int l = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 1024; i += blockDim.x) lds[i] = (short)i;
+  __syncthreads();
+  const unsigned addr = lds_byte_off(lds) + (unsigned)off[l];
+  unsigned long long r;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=v"(r) : "v"(addr));
+  #pragma unroll
+  for (int j = 0; j < 4; ++j)
+    out[l * 4 + j] = (short)((r >> (16 * j)) & 0xffff);
+}
+
 }  // namespace
+
+torch::Tensor probe_tr(torch::Tensor offsets) {
+  CHECK_CUDA(offsets); CHECK_CONTIG(offsets);
+  TORCH_CHECK(offsets.scalar_type() == torch::kInt32 && offsets.numel() == 64);
+  auto out = torch::zeros({64, 4},
+      torch::TensorOptions().dtype(torch::kInt16).device(offsets.device()));
+  auto stream = scaffold_stream();
+  hipLaunchKernelGGL(probe_tr_kernel, dim3(1), dim3(64), 0, stream,
+                     offsets.data_ptr<int>(), (short*)out.data_ptr());
+  HIP_CHECK(hipGetLastError());
+  return out;
+}
 
 torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B, long which) {
   CHECK_CUDA(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
