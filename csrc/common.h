@@ -114,9 +114,17 @@ inline Strides4 strides_of(const torch::Tensor& t) {
 // The tr-image of a [KT rows][D cols] bf16 tile packs 4x16 subtiles
 // consecutively: element (k, d) lives at
 //   el = ((k>>2) * (D/16) + (d>>4)) * 64 + (k&3)*16 + (d&15)
-// Written coalesced (16-aligned 8-element chunks), read as columns with
-// ds_read_b64_tr_b16: lane byte addr 2*(subtile*64 + (d&15)) returns the
-// 4 bf16 of rows 4*(k>>2)..+3 at column d — a free 4x4 transpose.
+// written coalesced (16-aligned 8-element chunks).
+//
+// ds_read_b64_tr_b16 semantics — measured on hardware by probe_tr
+// (benchmarks/tr_probe.py, 2026-09-13), NOT the guide's ambiguous
+// wording: each lane reads 8 CONTIGUOUS bytes at its own 8-B-aligned
+// address; within each 16-lane group the hardware then redistributes so
+// lane l receives the bf16 at positions {16*j + (l&15), j=0..3} of the
+// group's 64 concatenated elements (lane p's 4 els sit at positions
+// 4p..4p+3).  To hand lane l column (l&15) of a [4 k-rows][16 d-cols]
+// row-major subtile, lane p=(l&15) must therefore read
+//   subtile_byte_base + (p>>2)*32 + (p&3)*8.
 
 // async global->LDS copy, 16 B per lane (guide §5 common-mistake 1);
 // dest is wave-uniform base + lane*16, source is per-lane.

@@ -2,20 +2,28 @@
 //
 // Recompute-based flash bwd on the fragment maps verified by probe.hip
 // (same maps as attn_fwd_mfma.hip).  delta = rowsum(dO*O) comes from the
-// wave kernel in attn_bwd.hip.
+// strided wave kernel below.
 //
-// dQ kernel  (wave = 32 q rows, loop kv tiles):
+// v2: the transpose-operand loads (K/Q/dO "column" fragments — 32-64
+// scalar 2-B global loads per tile per lane in v1, the top cost of the
+// GPT-2 step at 20.7%) now come from LDS tr-images staged cooperatively
+// with global_load_lds and read with ds_read_b64_tr_b16 (guide T10) —
+// 2 LDS instructions per fragment instead of 16 scalar loads.  Each
+// wave's rows are INTERLEAVED (row = base + 4*i + wid) so all 4 waves
+// share the same causal tile range at the block's staging barriers.
+//
+// dQ kernel  (wave = 32 interleaved q rows, loop kv tiles):
 //   S^T  = mfma(K, Q^T)            C: lane=q, regs=k   (as fwd)
 //   P^T  = exp(S^T*scale - lse[q])              lane-local
 //   dP^T = mfma(V, dO^T)           C: lane=q, regs=k
 //   dS^T = P^T*(dP^T - delta[q])*scale          lane-local
-//   dQ^T += mfma(K^T_cols, pack(dS^T))  C: lane=q, regs=d
+//   dQ^T += mfma(K^T_cols, pack(dS^T))  C: lane=q, regs=d  [K^T by tr-read]
 //
-// dKV kernel (wave = 32 kv rows, loop q tiles, loop GQA group):
+// dKV kernel (wave = 32 interleaved kv rows, loop q tiles, loop GQA group):
 //   S    = mfma(Q, K^T)            C: lane=k, regs=q
 //   P    = exp(S*scale - lse[q_reg])   (lse/delta per-REG loads, L1-hot)
-//   dV^T += mfma(pack(P^T), dO_cols)   C: lane=d, regs=k
-//   dK^T += mfma(pack(dS^T), Q_cols)   C: lane=d, regs=k
+//   dV^T += mfma(pack(P^T), dO_cols)   C: lane=d, regs=k  [dO by tr-read]
+//   dK^T += mfma(pack(dS^T), Q_cols)   C: lane=d, regs=k  [Q by tr-read]
 //   (pack() = quad pack + permlane32_swap, T12 — identical transform for
 //    a C-layout operand into an A/B fragment's 8-consecutive-k order.)
 
@@ -47,6 +55,42 @@ DEVINL void pack_frags(const float p[16], unsigned out[2][4]) {
     out[s][2] = rx[1];
     out[s][3] = ry[1];
   }
+}
+
+// cooperative glds16 stage of a 32-row global tile into a tr-image
+// (layout in common.h); rows clamped at S-1 (padded rows are masked out
+// of P downstream).
+template <int D>
+DEVINL void stage_tr(const short* __restrict__ src, long row_stride, int r0,
+                     int S, short* lds, int wid, int l) {
+  constexpr int NSLOT = 32 * D / 8 / 256;
+  #pragma unroll
+  for (int it = 0; it < NSLOT; ++it) {
+    const int s8 = (it * 4 + wid) * 64 + l;
+    const int sub = s8 >> 3;
+    const int rem8 = s8 & 7;
+    const int row = min(r0 + 4 * (sub / (D / 16)) + (rem8 >> 1), S - 1);
+    const int d0 = 16 * (sub % (D / 16)) + 8 * (rem8 & 1);
+    glds16(src + (long)row * row_stride + d0, lds + (long)(it * 4 + wid) * 512);
+  }
+}
+
+// per-lane tr-read addresses for the column fragments of a tr-image at
+// lds base: [t][s] -> fragment of rows 16s+8*half + 0..7 at col 32t+(l&31)
+template <int D, int NDT>
+DEVINL void tr_addrs(unsigned base, int l, unsigned addr[NDT][2][2]) {
+  const int half = l >> 5;
+  const int p = l & 15;  // lane's in-group read slot (see common.h)
+  #pragma unroll
+  for (int t = 0; t < NDT; ++t)
+    #pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      const int g = 2 * t + ((l & 31) >> 4);
+      const int q4 = 4 * s + 2 * half;
+      addr[t][s][0] = base + 128u * (unsigned)(q4 * (D / 16) + g)
+                      + (p >> 2) * 32 + (p & 3) * 8;
+      addr[t][s][1] = addr[t][s][0] + 128u * (unsigned)(D / 16);
+    }
 }
 
 // delta = rowsum(dO * O), stride-aware (one wave per row)
@@ -88,6 +132,8 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
                         Strides4 ks, Strides4 vs, Strides4 dqs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
+  __shared__ __attribute__((aligned(16))) short ktile[32 * D];  // tr image
+
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
   const int lq = l & 31;
@@ -96,7 +142,7 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int hk = h / (H / Hkv);
-  const int q0 = blockIdx.x * 128 + wid * 32;
+  const int q0 = blockIdx.x * 128;
   if (q0 >= S) return;
 
   const long qb = (long)b * qs.sb + (long)h * qs.sh;
@@ -105,7 +151,8 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   const long vbb = (long)b * vs.sb + (long)hk * vs.sh;
   const long rhead = ((long)b * H + h) * S;
 
-  const int qrow = min(q0 + lq, S - 1);
+  const int qrow_g = q0 + 4 * lq + wid;     // interleaved q row
+  const int qrow = min(qrow_g, S - 1);
   bf16x8_t qf[NSL], dof[NSL];
   #pragma unroll
   for (int s = 0; s < NSL; ++s) {
@@ -115,12 +162,18 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   const float my_lse = lse[rhead + qrow];
   const float my_delta = delta[rhead + qrow];
 
+  unsigned ktr[NDT][2][2];
+  tr_addrs<D, NDT>(lds_byte_off(ktile), l, ktr);
+
   f32x16_t dqacc[NDT];
   #pragma unroll
   for (int t = 0; t < NDT; ++t) dqacc[t] = {};
 
-  const int kv_end = causal ? min(S, q0 + 32) : S;
+  const int kv_end = causal ? min(S, q0 + 128) : S;
   for (int c0 = 0; c0 < kv_end; c0 += 32) {
+    stage_tr<D>(k + kbb, ks.ss, c0, S, ktile, wid, l);
+    __syncthreads();  // glds drained (vmcnt 0) + barrier
+
     const int krow = min(c0 + lq, S - 1);
     f32x16_t sacc = {}, dpacc = {};
     #pragma unroll
@@ -131,7 +184,7 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
       dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[s], dpacc, 0, 0, 0);
     }
 
-    const bool need_mask = (causal && c0 + 31 > q0) || (c0 + 31 >= S);
+    const bool need_mask = (causal && c0 + 31 > qrow_g) || (c0 + 31 >= S);
     float ds[16];
     #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -139,7 +192,7 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
       float sv = sacc[r] * scale;
       if (need_mask) {
         const int kg = c0 + krel;
-        if (kg >= S || (causal && kg > q0 + lq)) sv = -1e30f;
+        if (kg >= S || (causal && kg > qrow_g)) sv = -1e30f;
       }
       const float p = __expf(sv - my_lse);
       ds[r] = p * (dpacc[r] - my_delta) * scale;
@@ -148,25 +201,21 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
     unsigned dsf[2][4];
     pack_frags(ds, dsf);
 
-    // dQ^T += K^T . dS^T  per d-tile (A = K columns, uncoalesced v1)
+    // dQ^T += K^T . dS^T  per d-tile; K^T A-fragments by tr-read (T10)
     #pragma unroll
     for (int t = 0; t < NDT; ++t) {
       #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        short kf[8];
-        #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int kg = min(c0 + 16 * s + 8 * half + j, S - 1);
-          kf[j] = k[kbb + (long)kg * ks.ss + 32 * t + lq];
-        }
+        bf16x8_t kcf = tr_read_frag(ktr[t][s][0], ktr[t][s][1]);
         dqacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            *(const bf16x8_t*)kf, *(const bf16x8_t*)dsf[s], dqacc[t], 0, 0, 0);
+            kcf, *(const bf16x8_t*)dsf[s], dqacc[t], 0, 0, 0);
       }
     }
+    __syncthreads();  // all waves done with ktile before restage
   }
 
-  if (q0 + lq < S) {
-    const long obase = (long)b * dqs.sb + (long)h * dqs.sh + (long)(q0 + lq) * dqs.ss;
+  if (qrow_g < S) {
+    const long obase = (long)b * dqs.sb + (long)h * dqs.sh + (long)qrow_g * dqs.ss;
     #pragma unroll
     for (int t = 0; t < NDT; ++t)
       #pragma unroll
@@ -191,26 +240,37 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          Strides4 vs, Strides4 dks, Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
+  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 32 * D];  // Q | dO
+
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
-  const int lk = l & 31;        // this lane's kv row (S-tile) / d col (acc)
+  const int lk = l & 31;
   const int half = l >> 5;
 
   const int bhk = blockIdx.y;
   const int b = bhk / Hkv, hk = bhk % Hkv;
   const int group = H / Hkv;
-  const int k0 = blockIdx.x * 128 + wid * 32;
+  const int k0 = blockIdx.x * 128;
   if (k0 >= S) return;
 
   const long kbb = (long)b * ks.sb + (long)hk * ks.sh;
   const long vbb = (long)b * vs.sb + (long)hk * vs.sh;
-  const int krow = min(k0 + lk, S - 1);
+  const int krow_g = k0 + 4 * lk + wid;     // interleaved kv row
+  const int krow = min(krow_g, S - 1);
+
+  short* qtile = bwd_lds;
+  short* dotile = bwd_lds + 32 * D;
+  unsigned qtr[NDT][2][2], dotr[NDT][2][2];
+  tr_addrs<D, NDT>(lds_byte_off(qtile), l, qtr);
+  tr_addrs<D, NDT>(lds_byte_off(dotile), l, dotr);
 
   // K^T B-fragments for S = Q.K^T (col=k=lane, kdim=d) — contiguous
-  bf16x8_t ktf[NSL];
+  bf16x8_t ktf[NSL], vtf[NSL];
   #pragma unroll
-  for (int s = 0; s < NSL; ++s)
+  for (int s = 0; s < NSL; ++s) {
     ktf[s] = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
+    vtf[s] = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
+  }
 
   f32x16_t dvacc[NDT], dkacc[NDT];
   #pragma unroll
@@ -221,8 +281,12 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
     const long qb = (long)b * qs.sb + (long)h * qs.sh;
     const long dob = (long)b * dos.sb + (long)h * dos.sh;
     const long rhead = ((long)b * H + h) * S;
-    const int qstart = causal ? (k0 / 32) * 32 : 0;
+    const int qstart = causal ? k0 : 0;
     for (int c0 = qstart; c0 < S; c0 += 32) {
+      stage_tr<D>(q + qb, qs.ss, c0, S, qtile, wid, l);
+      stage_tr<D>(dout + dob, dos.ss, c0, S, dotile, wid, l);
+      __syncthreads();
+
       // S = Q_tile . K^T  (C: lane=k, regs=q)
       const int qrow = min(c0 + lk, S - 1);
       f32x16_t sacc = {};
@@ -238,7 +302,7 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
         const int qg = c0 + qrel;
         float sv = sacc[r] * scale;
-        const bool masked = qg >= S || (causal && (k0 + lk) > qg);
+        const bool masked = qg >= S || (causal && krow_g > qg);
         const int qc = min(qg, S - 1);
         const float lse_r = lse[rhead + qc];    // L1-hot across lanes
         p[r] = masked ? 0.f : __expf(sv - lse_r);
@@ -249,8 +313,7 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       #pragma unroll
       for (int s = 0; s < NSL; ++s) {
         bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
-        bf16x8_t vf = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
-        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vf, dpacc, 0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vtf[s], dpacc, 0, 0, 0);
       }
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -265,36 +328,34 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       pack_frags(ds, dsf);
 
       // dV^T += P^T . dO_cols ; dK^T += dS^T . Q_cols  (C: lane=d, regs=k)
+      // column B-fragments by tr-read from the staged tiles (T10)
       #pragma unroll
       for (int t = 0; t < NDT; ++t) {
         #pragma unroll
         for (int s = 0; s < 2; ++s) {
-          short dof_c[8], qf_c[8];
-          #pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            const int qg = min(c0 + 16 * s + 8 * half + j, S - 1);
-            dof_c[j] = dout[dob + (long)qg * dos.ss + 32 * t + lk];
-            qf_c[j] = q[qb + (long)qg * qs.ss + 32 * t + lk];
-          }
+          bf16x8_t dof_c = tr_read_frag(dotr[t][s][0], dotr[t][s][1]);
+          bf16x8_t qf_c = tr_read_frag(qtr[t][s][0], qtr[t][s][1]);
           dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)pf[s], *(const bf16x8_t*)dof_c, dvacc[t], 0, 0, 0);
+              *(const bf16x8_t*)pf[s], dof_c, dvacc[t], 0, 0, 0);
           dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)dsf[s], *(const bf16x8_t*)qf_c, dkacc[t], 0, 0, 0);
+              *(const bf16x8_t*)dsf[s], qf_c, dkacc[t], 0, 0, 0);
         }
       }
+      __syncthreads();  // all waves done with the tiles before restage
     }
   }
 
-  // epilogue: C layout lane=col=d(within tile), regs=rows=k
+  // epilogue: C layout lane=col=d(within tile), regs=rows=k (interleaved)
   #pragma unroll
   for (int t = 0; t < NDT; ++t)
     #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int krel = (r & 3) + 8 * (r >> 2) + 4 * half;
-      if (k0 + krel < S) {
-        dk[(long)b * dks.sb + (long)hk * dks.sh + (long)(k0 + krel) * dks.ss + 32 * t + lk]
+      const int kg = k0 + 4 * krel + wid;
+      if (kg < S) {
+        dk[(long)b * dks.sb + (long)hk * dks.sh + (long)kg * dks.ss + 32 * t + lk]
             = f2bf(dkacc[t][r]);
-        dv[(long)b * dvs.sb + (long)hk * dvs.sh + (long)(k0 + krel) * dvs.ss + 32 * t + lk]
+        dv[(long)b * dvs.sb + (long)hk * dvs.sh + (long)kg * dvs.ss + 32 * t + lk]
             = f2bf(dvacc[t][r]);
       }
     }

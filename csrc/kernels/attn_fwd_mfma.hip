@@ -77,17 +77,22 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
 
   // tr-read addresses for the PV step (per d-tile t, k-quad pair at s)
   // subtile(q4, g) at element (q4*(D/16)+g)*64; lane column = l&15
+  // per-lane read slot p: lane receives column (l&15) of each 4x16
+  // subtile when lane p reads subtile_base + (p>>2)*32 + (p&3)*8
+  // (hardware-measured semantics — common.h)
   unsigned tr_addr[NDT][2][2];
   {
     const unsigned base = lds_byte_off(vtile);
+    const int p = l & 15;
     #pragma unroll
     for (int t = 0; t < NDT; ++t)
       #pragma unroll
       for (int s = 0; s < 2; ++s) {
         const int g = 2 * t + (lq >> 4);
         const int q4 = 4 * s + 2 * half;
-        tr_addr[t][s][0] = base + 2 * ((unsigned)(q4 * (D / 16) + g) * 64 + (l & 15));
-        tr_addr[t][s][1] = tr_addr[t][s][0] + 2 * 64 * (unsigned)(D / 16);
+        tr_addr[t][s][0] = base + 128u * (unsigned)(q4 * (D / 16) + g)
+                           + (p >> 2) * 32 + (p & 3) * 8;
+        tr_addr[t][s][1] = tr_addr[t][s][0] + 128u * (unsigned)(D / 16);
       }
   }
 
