@@ -33,6 +33,12 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     bool causal);
 torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B, long which);
 torch::Tensor probe_tr(torch::Tensor offsets);
+torch::Tensor rccl_unique_id();
+long rccl_init(long rank, long world, torch::Tensor id_bytes);
+void rccl_all_reduce(long handle, torch::Tensor t);
+void rccl_join(long handle);
+double rccl_comm_time_ms(long handle);
+void rccl_destroy(long handle);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
                                       c10::optional<torch::Tensor> bias,
                                       long epilogue,
@@ -51,6 +57,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd, "flash attention backward (K4)");
   m.def("probe_mfma", &probe_mfma, "MFMA fragment-layout probe (tests)");
   m.def("probe_tr", &probe_tr, "ds_read_b64_tr_b16 semantics probe (tests)");
+  m.def("rccl_unique_id", &rccl_unique_id, "RCCL bootstrap id (rank 0)");
+  m.def("rccl_init", &rccl_init, "init direct-RCCL bucket communicator");
+  m.def("rccl_all_reduce", &rccl_all_reduce,
+        "bucket SUM all-reduce on the comm stream (async)");
+  m.def("rccl_join", &rccl_join, "compute stream waits comm stream");
+  m.def("rccl_comm_time_ms", &rccl_comm_time_ms,
+        "drain measured collective time (ms)");
+  m.def("rccl_destroy", &rccl_destroy, "destroy communicator");
   m.def("linear_fwd", &linear_fwd,
         "MFMA linear fwd, fused bias+GELU/residual epilogue (K5)");
 }

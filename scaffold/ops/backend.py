@@ -23,6 +23,7 @@ EXT_NAME = "scaffold_hip"
 
 _SOURCES = [
     "ext.cpp",
+    "comm/rccl_bucket.hip",
     "kernels/adamw.hip",
     "kernels/layernorm.hip",
     "kernels/softmax_xent.hip",
@@ -56,6 +57,7 @@ def build_extension(verbose: bool = True):
         build_directory=BUILD_DIR,
         extra_cflags=["-O3", "-std=c++17"],
         extra_cuda_cflags=["-O3", "-std=c++17", "--offload-arch=gfx950"],
+        extra_ldflags=["-L/opt/rocm/lib", "-lrccl"],
         verbose=verbose,
     )
     return ext
