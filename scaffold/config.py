@@ -76,6 +76,7 @@ class DistConfig:
     first_bucket_mb: float = 4  # small first bucket fires early in backward
     overlap: bool = True  # all-reduce overlapped with backward (C4)
     min_nchannels: int = 7  # RCCL channels >= xGMI links per GPU
+    comm: str = "auto"  # auto | rccl (direct C++ bucket path) | torch
     timeout_s: int = 600
 
 

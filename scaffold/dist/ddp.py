@@ -161,7 +161,12 @@ class BucketedDDP(nn.Module):
     def _launch(self, b: GradBucket) -> None:
         if b.launched:
             return
-        b.work = self.comm.all_reduce(b.flat, async_op=True)
+        if hasattr(self.comm, "all_reduce_bucket"):
+            # direct-RCCL path: enqueued on the dedicated comm stream,
+            # joined once in finish() (no per-bucket Work objects)
+            self.comm.all_reduce_bucket(b.flat)
+        else:
+            b.work = self.comm.all_reduce(b.flat, async_op=True)
         b.launched = True
 
     # -------------------------------------------------------------- API
@@ -177,6 +182,8 @@ class BucketedDDP(nn.Module):
         for b in self.buckets:
             if not b.launched:
                 self._launch(b)
+        if hasattr(self.comm, "join"):
+            self.comm.join()  # compute stream waits comm stream, no host sync
         for b in self.buckets:
             if b.work is not None:
                 b.work.wait()  # stream-ordered on RCCL; host wait on gloo

@@ -60,9 +60,11 @@ class Trainer:
             raw_model = raw_model.to(memory_format=torch.channels_last)
         self.raw_model = raw_model
 
+        from .dist.comm import make_comm
+
         self.ddp = BucketedDDP(
             raw_model,
-            comm=comm,
+            comm=comm if comm is not None else make_comm(cfg, self.ctx),
             bucket_cap_mb=cfg.dist.bucket_cap_mb,
             first_bucket_mb=cfg.dist.first_bucket_mb,
             overlap=cfg.dist.overlap,
