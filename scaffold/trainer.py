@@ -98,8 +98,15 @@ class Trainer:
             loss_val = self._train_step(batch_iter, accum)
             dt = time.perf_counter() - t0
             self.step += 1
+            extra = None
+            if (self.step % max(1, self.cfg.log.interval) == 0
+                    and hasattr(self.ddp.comm, "comm_time_ms")):
+                # measured per-bucket collective time on the comm stream
+                # (drains events — log cadence only, never per step)
+                extra = {"comm_ms": round(self.ddp.comm.comm_time_ms(), 3)}
             self.logger.log_step(
-                self.step, loss_val, self.cfg.data.batch_size * accum, dt
+                self.step, loss_val, self.cfg.data.batch_size * accum, dt,
+                extra=extra,
             )
             if (
                 self.cfg.ckpt.interval
