@@ -40,11 +40,11 @@ struct RcclComm {
 
   ~RcclComm() {
     for (auto& p : timers) {
-      hipEventDestroy(p.first);
-      hipEventDestroy(p.second);
+      (void)hipEventDestroy(p.first);
+      (void)hipEventDestroy(p.second);
     }
-    if (stream) hipStreamDestroy(stream);
-    if (comm) ncclCommDestroy(comm);
+    if (stream) (void)hipStreamDestroy(stream);
+    if (comm) (void)ncclCommDestroy(comm);
   }
 };
 

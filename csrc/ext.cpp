@@ -39,6 +39,10 @@ void rccl_all_reduce(long handle, torch::Tensor t);
 void rccl_join(long handle);
 double rccl_comm_time_ms(long handle);
 void rccl_destroy(long handle);
+void save_shard_pinned(std::string path, std::vector<std::string> names,
+                       std::vector<torch::Tensor> tensors,
+                       std::vector<std::string> meta_keys,
+                       std::vector<std::string> meta_vals);
 std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
                                       c10::optional<torch::Tensor> bias,
                                       long epilogue,
@@ -65,6 +69,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rccl_comm_time_ms", &rccl_comm_time_ms,
         "drain measured collective time (ms)");
   m.def("rccl_destroy", &rccl_destroy, "destroy communicator");
+  m.def("save_shard_pinned", &save_shard_pinned,
+        "stream tensors to a safetensors shard via pinned D2H (C6)");
   m.def("linear_fwd", &linear_fwd,
         "MFMA linear fwd, fused bias+GELU/residual epilogue (K5)");
 }

@@ -102,7 +102,6 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
         [(n, t.numel() * t.element_size()) for n, t in unique], world
     )
 
-    stager = _PinnedStager(ctx.device) if async_d2h else None
     shard: Dict[str, torch.Tensor] = {}
     tensor_meta: Dict[str, dict] = {}
     for name, t in unique:
@@ -112,8 +111,7 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
             "dtype": str(t.dtype).replace("torch.", ""),
         }
         if owner[name] == rank:
-            host = stager.to_host(t) if stager else t.detach().contiguous().cpu()
-            shard[name] = host.contiguous()
+            shard[name] = t.detach()
     for name, target in aliases.items():
         tensor_meta[name] = {"alias": target}
 
@@ -124,9 +122,29 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
     if data is not None:
         shard[f"rng.data.{rank}"] = data.state_dict()["gen_state"].cpu()
     sf = os.path.join(ckpt_dir, layout.shard_file(rank))
-    save_file(shard, sf, metadata={"rank": str(rank), "step": str(step)})
-    with open(sf, "rb") as fh:
-        os.fsync(fh.fileno())
+
+    from ..ops.backend import ext as _ext
+
+    native = _ext() if (ctx.on_gpu and async_d2h) else None
+    if native is not None:
+        # C6 native path: device tensors stream straight to the shard
+        # file through the pinned double-buffer on a side stream
+        # (csrc/io/pinned_stager.hip); file is safetensors-compatible.
+        names = list(shard.keys())
+        native.save_shard_pinned(
+            sf, names, [shard[n].contiguous() for n in names],
+            ["rank", "step"], [str(rank), str(step)],
+        )
+    else:
+        stager = _PinnedStager(ctx.device) if async_d2h else None
+        host_shard = {}
+        for name, t in shard.items():
+            host = stager.to_host(t) if (stager and t.is_cuda) \
+                else t.detach().contiguous().cpu()
+            host_shard[name] = host.contiguous()
+        save_file(host_shard, sf, metadata={"rank": str(rank), "step": str(step)})
+        with open(sf, "rb") as fh:
+            os.fsync(fh.fileno())
 
     # all shards durable before rank 0 commits the manifest
     if ctx.initialized:

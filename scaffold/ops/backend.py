@@ -24,6 +24,7 @@ EXT_NAME = "scaffold_hip"
 _SOURCES = [
     "ext.cpp",
     "comm/rccl_bucket.hip",
+    "io/pinned_stager.hip",
     "kernels/adamw.hip",
     "kernels/layernorm.hip",
     "kernels/softmax_xent.hip",
