@@ -151,6 +151,136 @@ __global__ __launch_bounds__(256) void linear_fwd_kernel(
   }
 }
 
+// ---- 256x256 8-wave phase-interleaved variant (toward the guide's
+// 8-phase class).  512 threads = 8 waves (2M x 4N), per-wave output
+// 128x64 = 8x4 fragments of 16x16x32 MFMA; BK=64; LDS = 2 x 64 KiB
+// double-buffered tiles (A 32K + B 32K each).
+//
+// Schedule per iteration (tile t, buffer cur):
+//   ph0..ph3 (quadrant qm=ph>>1, qn=ph&1):
+//     { issue 2 glds of tile t+1 into buf cur^1 ; ds_read the
+//       quadrant's A/B fragments ; raw s_barrier (phase-locks the 8
+//       waves -> T5 role-split) ; setprio(1) ; 16 MFMA ; setprio(0) }
+//   end: s_waitcnt vmcnt(0) ; raw s_barrier ; swap
+// Prefetch distance is one tile: tile t+1's 8 glds are spread over
+// iteration t's 4 phases (~64 MFMA of cover), and the single vmcnt
+// drain per K-tile amortizes 8x better than the 128^2 kernel's
+// per-K-step drain.  Gains over the 128^2 kernel: 2x arithmetic
+// intensity per staged byte, 8-wave role-split (setprio pays — T5),
+// one barrier per phase instead of two per K-step.  The guide's full
+// counted-vmcnt half-tile ring (never draining to 0 in the loop) is a
+// further step not taken here.
+template <int EPI, bool HAS_BIAS>
+__global__ void __launch_bounds__(512) linear_fwd256_kernel(
+    const short* __restrict__ X, const short* __restrict__ W,
+    const short* __restrict__ bias, const short* __restrict__ res,
+    short* __restrict__ Y, short* __restrict__ Pre, int M, int N, int K) {
+  __shared__ __attribute__((aligned(16))) short smem[2 * 2 * 256 * 64];
+
+  const int w = threadIdx.x >> 6;     // wave 0..7
+  const int l = threadIdx.x & 63;
+  const int wm = w >> 2, wn = w & 3;  // 2x4 wave grid
+
+  const unsigned nwg = gridDim.x;
+  const unsigned q8 = nwg / 8, r8 = nwg % 8;
+  const unsigned xcd = blockIdx.x % 8, idx = blockIdx.x / 8;
+  const unsigned wg =
+      (xcd < r8 ? xcd * (q8 + 1) : r8 * (q8 + 1) + (xcd - r8) * q8) + idx;
+  const int gn = N / 256;
+  const int m0 = (int)(wg / gn) * 256;
+  const int n0 = (int)(wg % gn) * 256;
+
+  const int nt = K / 64;
+  f32x4_t acc[8][4] = {};
+
+  // stage one phase-quarter (16 KiB) of tile t into buf: phase p covers
+  // combined slots [p*1024, p*1024+1024) of the 4096x16B tile image
+  // (A = slots 0..2047, B = 2048..4095), 2 glds per thread.
+  auto stage_q = [&](int buf, int t, int p) {
+    #pragma unroll
+    for (int g = 0; g < 2; ++g) {
+      const int slot = p * 1024 + g * 512 + w * 64 + l;
+      const int op = slot >> 11;            // 0 = A, 1 = B
+      const int c = slot & 2047;            // chunk within operand
+      const int row = c >> 3;
+      const int kb = (c & 7) ^ (row & 7);
+      const short* src = (op ? W + (long)n0 * K : X + (long)m0 * K);
+      glds16(src + (long)row * K + (long)t * 64 + kb * 8,
+             smem + (long)buf * (2 * 256 * 64) + (long)slot * 8);
+    }
+  };
+
+  stage_q(0, 0, 0); stage_q(0, 0, 1); stage_q(0, 0, 2); stage_q(0, 0, 3);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int cur = 0;
+  for (int t = 0; t < nt; ++t) {
+    const short* sA = smem + cur * (2 * 256 * 64);
+    const short* sB = sA + 256 * 64;
+    const bool prefetch = (t + 1 < nt);
+    #pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      const int qm = ph >> 1, qn = ph & 1;
+      if (prefetch) stage_q(cur ^ 1, t + 1, ph);
+      bf16x8_t af[4][2], bf[2][2];
+      #pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int kb = kk * 4 + (l >> 4);
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          const int row = wm * 128 + qm * 64 + mi * 16 + (l & 15);
+          af[mi][kk] = *(const bf16x8_t*)&sA[(row * 8 + (kb ^ (row & 7))) * 8];
+        }
+        #pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          const int col = wn * 64 + qn * 32 + ni * 16 + (l & 15);
+          bf[ni][kk] = *(const bf16x8_t*)&sB[(col * 8 + (kb ^ (col & 7))) * 8];
+        }
+      }
+      __builtin_amdgcn_s_barrier();   // phase-lock the 8 waves
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+        #pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+          #pragma unroll
+          for (int ni = 0; ni < 2; ++ni)
+            acc[qm * 4 + mi][qn * 2 + ni] =
+                __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    af[mi][kk], bf[ni][kk], acc[qm * 4 + mi][qn * 2 + ni],
+                    0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // next tile's 8 glds must have landed before its first ds_read
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+
+  // epilogue: same C/D map as the 128^2 kernel; acc is [8 m][4 n] f32x4
+  #pragma unroll
+  for (int mi = 0; mi < 8; ++mi)
+    #pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+      const int colg = n0 + wn * 64 + ni * 16 + (l & 15);
+      const float bv = HAS_BIAS ? bf2f(bias[colg]) : 0.0f;
+      const int rowg = m0 + wm * 128 + mi * 16 + (l >> 4) * 4;
+      #pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[mi][ni][r] + bv;
+        const long off = (long)(rowg + r) * N + colg;
+        if (EPI == EPI_GELU) {
+          Pre[off] = f2bf(v);
+          v = gelu_tanh(v);
+        } else if (EPI == EPI_RESID) {
+          v += bf2f(res[off]);
+        }
+        Y[off] = f2bf(v);
+      }
+    }
+}
+
 template <int EPI>
 void launch_epi(bool has_bias, dim3 grid, hipStream_t stream,
                 const short* X, const short* W, const short* bias,
@@ -160,6 +290,18 @@ void launch_epi(bool has_bias, dim3 grid, hipStream_t stream,
                        stream, X, W, bias, res, Y, Pre, M, N, K);
   else
     hipLaunchKernelGGL((linear_fwd_kernel<EPI, false>), grid, dim3(256), 0,
+                       stream, X, W, bias, res, Y, Pre, M, N, K);
+}
+
+template <int EPI>
+void launch_epi256(bool has_bias, dim3 grid, hipStream_t stream,
+                   const short* X, const short* W, const short* bias,
+                   const short* res, short* Y, short* Pre, int M, int N, int K) {
+  if (has_bias)
+    hipLaunchKernelGGL((linear_fwd256_kernel<EPI, true>), grid, dim3(512), 0,
+                       stream, X, W, bias, res, Y, Pre, M, N, K);
+  else
+    hipLaunchKernelGGL((linear_fwd256_kernel<EPI, false>), grid, dim3(512), 0,
                        stream, X, W, bias, res, Y, Pre, M, N, K);
 }
 
@@ -204,15 +346,28 @@ std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
     preptr = (short*)pre.data_ptr();
   }
 
-  dim3 grid((M / 128) * (N / 128));
   auto stream = scaffold_stream();
   const short* X = (const short*)x.data_ptr();
   const short* W = (const short*)w.data_ptr();
   short* Y = (short*)y.data_ptr();
+  // 256^2 8-wave kernel when the shape tiles it AND the grid still
+  // fills the chip; 128^2 otherwise
+  const bool big = (M % 256 == 0) && (N % 256 == 0) &&
+                   (M / 256) * (N / 256) >= 224;
+  dim3 grid(big ? (M / 256) * (N / 256) : (M / 128) * (N / 128));
   switch (epilogue) {
-    case 0: launch_epi<0>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K); break;
-    case 1: launch_epi<1>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K); break;
-    case 2: launch_epi<2>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K); break;
+    case 0:
+      if (big) launch_epi256<0>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K);
+      else launch_epi<0>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K);
+      break;
+    case 1:
+      if (big) launch_epi256<1>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K);
+      else launch_epi<1>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K);
+      break;
+    case 2:
+      if (big) launch_epi256<2>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K);
+      else launch_epi<2>(bptr != nullptr, grid, stream, X, W, bptr, rptr, Y, preptr, M, N, K);
+      break;
     default: TORCH_CHECK(false, "bad epilogue ", epilogue);
   }
   HIP_CHECK(hipGetLastError());
