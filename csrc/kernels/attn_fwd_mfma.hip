@@ -114,11 +114,13 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
     // ---- S^T = K_tile . Q^T  (32k x 32q)
     f32x16_t sacc = {};
     const int krow = min(c0 + lq, S - 1);
+    __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
     #pragma unroll
     for (int s = 0; s < NSL; ++s) {
       bf16x8_t kf = *(const bf16x8_t*)(k + kb + (long)krow * ks.ss + 16 * s + 8 * half);
       sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
     }
+    __builtin_amdgcn_s_setprio(0);
 
     // ---- scale + mask (lane: col=q; reg r: row=k idx)
     const bool need_mask = (causal && c0 + 31 > qrow_g) || (c0 + 31 >= S);
@@ -137,21 +139,33 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
     mtile = fmaxf(mtile, __shfl_xor(mtile, 32, 64));  // combine half-wave regs
 
-    // ---- online softmax update (lane-local; both halves identical)
-    const float m_new = fmaxf(m_run, mtile);
-    const float alpha = __expf(m_run - m_new);
-    m_run = m_new;
+    // ---- online softmax update (lane-local; both halves identical).
+    // T13 with THR=0 (exact numerics): once the running max stops
+    // growing — the common case after the first tiles — the whole wave
+    // skips the O-rescale pass.
     float psum = 0.f;
-    #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      p[r] = __expf(p[r] - m_new);
-      psum += p[r];
-    }
-    l_run = l_run * alpha + psum + __shfl_xor(psum, 32, 64);
-    #pragma unroll
-    for (int t = 0; t < NDT; ++t)
+    if (!__any(mtile > m_run)) {
       #pragma unroll
-      for (int r = 0; r < 16; ++r) oacc[t][r] *= alpha;
+      for (int r = 0; r < 16; ++r) {
+        p[r] = __expf(p[r] - m_run);
+        psum += p[r];
+      }
+      l_run += psum + __shfl_xor(psum, 32, 64);
+    } else {
+      const float m_new = fmaxf(m_run, mtile);
+      const float alpha = __expf(m_run - m_new);
+      m_run = m_new;
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        p[r] = __expf(p[r] - m_new);
+        psum += p[r];
+      }
+      l_run = l_run * alpha + psum + __shfl_xor(psum, 32, 64);
+      #pragma unroll
+      for (int t = 0; t < NDT; ++t)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) oacc[t][r] *= alpha;
+    }
 
     // ---- P^T B-fragments: pack quads, permlane32_swap halves (T12)
     unsigned pfrag[2][4];
@@ -170,6 +184,7 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
     }
 
     // ---- O^T += V^T . P^T per d-tile; V^T A-fragments by tr-read (T10)
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int t = 0; t < NDT; ++t) {
       #pragma unroll
@@ -179,6 +194,7 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
             vf, *(const bf16x8_t*)pfrag[s], oacc[t], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();  // all waves done with vtile before restage
   }
 
