@@ -1,21 +1,26 @@
 // K4 — MFMA flash attention forward (bf16, CDNA4 32x32x16 matrix cores).
 //
-// Structure (cdna_hip_programming.md §B "fused attention prefill", adapted):
+// Structure (cdna_hip_programming.md §B "fused attention prefill" ladder,
+// adapted to 32x32x16 fragments):
 //  * 4 waves / workgroup over a 128-row q block; wave w owns the
 //    INTERLEAVED rows q0 + 4*i + w (i=0..31) so every wave sees the same
 //    causal kv range — no idle tail waves at the block's barriers.
-//  * swapped QK^T: S^T = mfma(A=K_tile, B=Q^T) so each lane holds a full
-//    q-row slice of scores -> softmax is lane-local + one shfl_xor(32)
-//    to combine the two half-wave reg sets.  (guide §B "swapped QK^T")
-//  * P -> bf16 B-fragments via pack + permlane32_swap (T12: one swap
-//    fills two output words).
-//  * V tile cooperatively staged in LDS in the tr-image layout
-//    (common.h) by global_load_lds, then consumed as V^T A-fragments via
-//    ds_read_b64_tr_b16 (T10) — replaces the 32/64 scalar 2-B global
-//    column loads per tile of the v1 kernel, and makes the kernel
-//    insensitive to the (B,S,H,D)-transposed stride of the models' QKV.
-//  * O^T accumulated via mfma(A=V^T, B=P^T): C cols = q = lane&31, so
-//    the online-softmax rescale is a pure lane-local multiply.
+//  * KVBLK=64: K and V staged cooperatively in LDS per 64-row kv tile
+//    (one barrier pair per 64 rows; softmax max/rescale amortized over
+//    64 scores per lane).
+//  * K image: row-major with the 16-B chunk XOR swizzle
+//    (chunk ^= row&7, the T2 pattern — <=2-way on ds_read_b128) filled
+//    by global_load_lds with the inverse swizzle on the SOURCE address
+//    (rule 21; the XOR permutes chunks within one 128-B row = one
+//    cacheline, so global coalescing is preserved).
+//  * V image: tr-layout (common.h) consumed as V^T A-fragments via
+//    ds_read_b64_tr_b16 (T10) — hardware-measured semantics.
+//  * swapped QK^T: S^T = mfma(A=K, B=Q^T) so each lane holds full q-row
+//    score slices; softmax is lane-local + one shfl_xor(32).
+//  * P -> bf16 B-fragments via pack + permlane32_swap (T12).
+//  * T13 defer-max with THR=0 (exact): skip the O-rescale pass when the
+//    running max did not grow (wave-uniform branch).
+//  * T5 setprio(1) around the MFMA clusters.
 //
 // Fragment maps verified on hardware by tests/ops/test_gpu_ops.py::
 // test_mfma_layout_32x32x16 (probe.hip):
@@ -44,8 +49,12 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
                      Strides4 vs, Strides4 os) {
   constexpr int NSL = D / 16;   // QK^T k-slices (over d)
   constexpr int NDT = D / 32;   // O^T d-tiles
-  constexpr int NSLOT = 32 * D / 8 / 256;  // glds16 per thread per tile
-  __shared__ __attribute__((aligned(16))) short vtile[32 * D];  // tr image
+  constexpr int CH = D / 8;     // 16-B chunks per row
+  // K row-image (64 x D, chunk-XOR) + V tr-image (2 x 32 x D)
+  __shared__ __attribute__((aligned(16))) short kv_lds[2 * 64 * D];
+
+  short* kimg = kv_lds;
+  short* vimg = kv_lds + 64 * D;
 
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
@@ -75,91 +84,97 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   for (int t = 0; t < NDT; ++t) oacc[t] = {};
   float m_run = -1e30f, l_run = 0.f;
 
-  // tr-read addresses for the PV step (per d-tile t, k-quad pair at s)
-  // subtile(q4, g) at element (q4*(D/16)+g)*64; lane column = l&15
-  // per-lane read slot p: lane receives column (l&15) of each 4x16
-  // subtile when lane p reads subtile_base + (p>>2)*32 + (p&3)*8
-  // (hardware-measured semantics — common.h)
-  unsigned tr_addr[NDT][2][2];
-  {
-    const unsigned base = lds_byte_off(vtile);
-    const int p = l & 15;
-    #pragma unroll
-    for (int t = 0; t < NDT; ++t)
-      #pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const int g = 2 * t + (lq >> 4);
-        const int q4 = 4 * s + 2 * half;
-        tr_addr[t][s][0] = base + 128u * (unsigned)(q4 * (D / 16) + g)
-                           + (p >> 2) * 32 + (p & 3) * 8;
-        tr_addr[t][s][1] = tr_addr[t][s][0] + 128u * (unsigned)(D / 16);
-      }
-  }
+  // tr-read base for the PV step: per-(t,s,sub) offsets are constants
+  // added at the read site (1 VGPR instead of a 16-entry address array)
+  //   addr(t,s,q4half) = tr_base + 128*((4s+2*half)*(D/16) + 2t) [+128*(D/16) for q4+1]
+  const unsigned tr_base = lds_byte_off(vimg) + 128u * (lq >> 4)
+                           + ((l & 15) >> 2) * 32 + (l & 3) * 8
+                           + (half ? 128u * 2 * (D / 16) : 0);
 
   const int kv_end = causal ? min(S, q0 + 128) : S;
-  for (int c0 = 0; c0 < kv_end; c0 += 32) {
-    // ---- cooperative V-tile stage into the tr image (glds16)
-    #pragma unroll
-    for (int it = 0; it < NSLOT; ++it) {
-      const int s8 = (it * 4 + wid) * 64 + l;       // 8-element chunk id
-      const int sub = s8 >> 3;
-      const int rem8 = s8 & 7;                      // (k&3)*2 + c8
-      const int krow = min(c0 + 4 * (sub / (D / 16)) + (rem8 >> 1), S - 1);
-      const int d0 = 16 * (sub % (D / 16)) + 8 * (rem8 & 1);
-      glds16(v + vb + (long)krow * vs.ss + d0,
-             vtile + (long)(it * 4 + wid) * 512);
+  for (int c0 = 0; c0 < kv_end; c0 += 64) {
+    // ---- cooperative stage of the 64-row K and V tiles (glds16)
+    {
+      constexpr int KSLOT = 64 * D / 8 / 256;  // glds per thread per image
+      #pragma unroll
+      for (int it = 0; it < KSLOT; ++it) {
+        // K: row-image, chunk-XOR swizzled via the SOURCE address
+        const int s16 = (it * 4 + wid) * 64 + l;
+        const int krow = min(c0 + s16 / CH, S - 1);
+        const int kbb = (s16 % CH) ^ ((s16 / CH) & 7);
+        glds16(k + kb + (long)krow * ks.ss + kbb * 8,
+               kimg + (long)(it * 4 + wid) * 512);
+        // V: tr-image (two 32-row halves)
+        const int sub = s16 / (32 * D / 8);
+        const int s8 = s16 % (32 * D / 8);
+        const int subidx = s8 >> 3;
+        const int rem8 = s8 & 7;
+        const int vrow = min(c0 + sub * 32 + 4 * (subidx / (D / 16)) + (rem8 >> 1), S - 1);
+        const int d0 = 16 * (subidx % (D / 16)) + 8 * (rem8 & 1);
+        glds16(v + vb + (long)vrow * vs.ss + d0,
+               vimg + (long)(it * 4 + wid) * 512);
+      }
     }
     __syncthreads();  // drains the glds (vmcnt 0) + barrier
 
-    // ---- S^T = K_tile . Q^T  (32k x 32q)
-    f32x16_t sacc = {};
-    const int krow = min(c0 + lq, S - 1);
-    __builtin_amdgcn_s_setprio(1);  // T5: favor the MFMA cluster
+    // ---- S^T = K_tile . Q^T for both 32-row sub-tiles
+    f32x16_t sacc[2];
+    sacc[0] = {};
+    sacc[1] = {};
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
-    for (int s = 0; s < NSL; ++s) {
-      bf16x8_t kf = *(const bf16x8_t*)(k + kb + (long)krow * ks.ss + 16 * s + 8 * half);
-      sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
+    for (int sub = 0; sub < 2; ++sub) {
+      const int krow = sub * 32 + lq;
+      #pragma unroll
+      for (int s = 0; s < NSL; ++s) {
+        const int kbb = (2 * s + half) ^ (krow & 7);
+        bf16x8_t kf = *(const bf16x8_t*)(kimg + krow * D + kbb * 8);
+        sacc[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc[sub], 0, 0, 0);
+      }
     }
     __builtin_amdgcn_s_setprio(0);
 
-    // ---- scale + mask (lane: col=q; reg r: row=k idx)
-    const bool need_mask = (causal && c0 + 31 > qrow_g) || (c0 + 31 >= S);
-    float p[16];
+    // ---- scale + mask IN PLACE in sacc (register reuse: no separate
+    // p[] array), then exponentiate in place
+    const bool need_mask = (causal && c0 + 63 > qrow_g) || (c0 + 63 >= S);
     float mtile = -1e30f;
     #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int krel = (r & 3) + 8 * (r >> 2) + 4 * half;
-      float sv = sacc[r] * scale;
-      if (need_mask) {
-        const int kg = c0 + krel;
-        if (kg >= S || (causal && kg > qrow_g)) sv = -1e30f;
+    for (int sub = 0; sub < 2; ++sub)
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int krel = sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        float sv = sacc[sub][r] * scale;
+        if (need_mask) {
+          const int kg = c0 + krel;
+          if (kg >= S || (causal && kg > qrow_g)) sv = -1e30f;
+        }
+        sacc[sub][r] = sv;
+        mtile = fmaxf(mtile, sv);
       }
-      p[r] = sv;
-      mtile = fmaxf(mtile, sv);
-    }
-    mtile = fmaxf(mtile, __shfl_xor(mtile, 32, 64));  // combine half-wave regs
+    mtile = fmaxf(mtile, __shfl_xor(mtile, 32, 64));
 
-    // ---- online softmax update (lane-local; both halves identical).
-    // T13 with THR=0 (exact numerics): once the running max stops
-    // growing — the common case after the first tiles — the whole wave
-    // skips the O-rescale pass.
+    // ---- online softmax update; defer-max THR=0 (exact)
     float psum = 0.f;
     if (!__any(mtile > m_run)) {
       #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        p[r] = __expf(p[r] - m_run);
-        psum += p[r];
-      }
+      for (int sub = 0; sub < 2; ++sub)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          sacc[sub][r] = __expf(sacc[sub][r] - m_run);
+          psum += sacc[sub][r];
+        }
       l_run += psum + __shfl_xor(psum, 32, 64);
     } else {
       const float m_new = fmaxf(m_run, mtile);
       const float alpha = __expf(m_run - m_new);
       m_run = m_new;
       #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        p[r] = __expf(p[r] - m_new);
-        psum += p[r];
-      }
+      for (int sub = 0; sub < 2; ++sub)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          sacc[sub][r] = __expf(sacc[sub][r] - m_new);
+          psum += sacc[sub][r];
+        }
       l_run = l_run * alpha + psum + __shfl_xor(psum, 32, 64);
       #pragma unroll
       for (int t = 0; t < NDT; ++t)
@@ -167,35 +182,40 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
         for (int r = 0; r < 16; ++r) oacc[t][r] *= alpha;
     }
 
-    // ---- P^T B-fragments: pack quads, permlane32_swap halves (T12)
-    unsigned pfrag[2][4];
+    // ---- P^T B-fragments (T12) then O^T += V^T . P^T per sub-tile
     #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      unsigned ax = pack_bf16x2(p[8 * s + 0], p[8 * s + 1]);
-      unsigned ay = pack_bf16x2(p[8 * s + 2], p[8 * s + 3]);
-      unsigned bx = pack_bf16x2(p[8 * s + 4], p[8 * s + 5]);
-      unsigned by = pack_bf16x2(p[8 * s + 6], p[8 * s + 7]);
-      auto rx = __builtin_amdgcn_permlane32_swap(ax, bx, false, false);
-      auto ry = __builtin_amdgcn_permlane32_swap(ay, by, false, false);
-      pfrag[s][0] = rx[0];
-      pfrag[s][1] = ry[0];
-      pfrag[s][2] = rx[1];
-      pfrag[s][3] = ry[1];
-    }
-
-    // ---- O^T += V^T . P^T per d-tile; V^T A-fragments by tr-read (T10)
-    __builtin_amdgcn_s_setprio(1);
-    #pragma unroll
-    for (int t = 0; t < NDT; ++t) {
+    for (int sub = 0; sub < 2; ++sub) {
+      unsigned pfrag[2][4];
       #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        bf16x8_t vf = tr_read_frag(tr_addr[t][s][0], tr_addr[t][s][1]);
-        oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            vf, *(const bf16x8_t*)pfrag[s], oacc[t], 0, 0, 0);
+        unsigned ax = pack_bf16x2(sacc[sub][8 * s + 0], sacc[sub][8 * s + 1]);
+        unsigned ay = pack_bf16x2(sacc[sub][8 * s + 2], sacc[sub][8 * s + 3]);
+        unsigned bx = pack_bf16x2(sacc[sub][8 * s + 4], sacc[sub][8 * s + 5]);
+        unsigned by = pack_bf16x2(sacc[sub][8 * s + 6], sacc[sub][8 * s + 7]);
+        auto rx = __builtin_amdgcn_permlane32_swap(ax, bx, false, false);
+        auto ry = __builtin_amdgcn_permlane32_swap(ay, by, false, false);
+        pfrag[s][0] = rx[0];
+        pfrag[s][1] = ry[0];
+        pfrag[s][2] = rx[1];
+        pfrag[s][3] = ry[1];
       }
+      const unsigned sub_off = (unsigned)sub * (32u * D * 2);
+      __builtin_amdgcn_s_setprio(1);
+      #pragma unroll
+      for (int t = 0; t < NDT; ++t) {
+        #pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          // q4 = 4s (+2*half folded into tr_base)
+          const unsigned a0 = tr_base + sub_off
+                              + 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
+          bf16x8_t vf = tr_read_frag(a0, a0 + 128u * (D / 16));
+          oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              vf, *(const bf16x8_t*)pfrag[s], oacc[t], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_s_setprio(0);
     }
-    __builtin_amdgcn_s_setprio(0);
-    __syncthreads();  // all waves done with vtile before restage
+    __syncthreads();  // all waves done with the images before restage
   }
 
   // ---- epilogue: O = O^T / l, scatter store (lane's q row fixed)

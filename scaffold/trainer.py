@@ -93,6 +93,19 @@ class Trainer:
         total = steps if steps is not None else self.cfg.steps
         accum = max(1, self.cfg.grad_accum)
         batch_iter = iter(self.data)
+        prof = None
+        if self.cfg.log.profile_steps > 0:
+            # torch.profiler works on ROCm; kernel-level
+            # counters come from rocprofv3 runs instead (profiles/).
+            prof = torch.profiler.profile(
+                activities=[torch.profiler.ProfilerActivity.CPU,
+                            torch.profiler.ProfilerActivity.CUDA],
+                schedule=torch.profiler.schedule(
+                    wait=2, warmup=2, active=self.cfg.log.profile_steps),
+                on_trace_ready=lambda p: p.export_chrome_trace(
+                    f"trace_rank{self.ctx.rank}.json"),
+            )
+            prof.__enter__()
         while self.step < total:
             t0 = time.perf_counter()
             loss_val = self._train_step(batch_iter, accum)
@@ -108,12 +121,16 @@ class Trainer:
                 self.step, loss_val, self.cfg.data.batch_size * accum, dt,
                 extra=extra,
             )
+            if prof is not None:
+                prof.step()
             if (
                 self.cfg.ckpt.interval
                 and self.step % self.cfg.ckpt.interval == 0
                 and self.cfg.ckpt.dir
             ):
                 self.save_checkpoint()
+        if prof is not None:
+            prof.__exit__(None, None, None)
         if self.cfg.ckpt.save_final and self.cfg.ckpt.dir:
             self.save_checkpoint()
         return self
