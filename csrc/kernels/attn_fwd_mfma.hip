@@ -40,8 +40,12 @@ DEVINL unsigned pack_bf16x2(float lo, float hi) {
          ((unsigned)(unsigned short)f2bf(hi) << 16);
 }
 
+// second launch_bounds arg = waves per SIMD (guide §6 G1): force the
+// allocator to fit 2 waves/SIMD (<=256 regs incl. AGPRs) — at D=128 the
+// default allocation was 362 regs -> 1 wave/SIMD and every barrier
+// stalled the whole SIMD (measured 11% slower than the 2-wave build).
 template <int D>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(256, 2)
 attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
                      const short* __restrict__ v, short* __restrict__ o,
                      float* __restrict__ lse, int B, int H, int Hkv, int S,
@@ -74,10 +78,17 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   // this lane's q row: interleaved assignment q0 + 4*lq + wid
   const int qrow_g = q0 + 4 * lq + wid;
   const int qrow = min(qrow_g, S - 1);
-  bf16x8_t qf[NSL];
-  #pragma unroll
-  for (int s = 0; s < NSL; ++s)
-    qf[s] = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
+  // D=64: Q fragments pinned in registers.  D=128: reloaded from L1
+  // per tile (32 VGPRs saved buys 2 waves/SIMD with zero spill; the
+  // wave's 8-KiB q tile stays L1-resident across the kv loop).
+  constexpr bool QREG = (D <= 64);
+  const short* qrow_ptr = q + qb + (long)qrow * qs.ss + 8 * half;
+  bf16x8_t qf[QREG ? NSL : 1];
+  if constexpr (QREG) {
+    #pragma unroll
+    for (int s = 0; s < NSL; ++s)
+      qf[s] = *(const bf16x8_t*)(qrow_ptr + 16 * s);
+  }
 
   f32x16_t oacc[NDT];
   #pragma unroll
@@ -129,7 +140,9 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
       for (int s = 0; s < NSL; ++s) {
         const int kbb = (2 * s + half) ^ (krow & 7);
         bf16x8_t kf = *(const bf16x8_t*)(kimg + krow * D + kbb * 8);
-        sacc[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc[sub], 0, 0, 0);
+        const bf16x8_t qv =
+            QREG ? qf[s] : *(const bf16x8_t*)(qrow_ptr + 16 * s);
+        sacc[sub] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qv, sacc[sub], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
