@@ -39,6 +39,9 @@ void rccl_all_reduce(long handle, torch::Tensor t);
 void rccl_join(long handle);
 double rccl_comm_time_ms(long handle);
 void rccl_destroy(long handle);
+torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u);
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dh, torch::Tensor g,
+                                      torch::Tensor u);
 void save_shard_pinned(std::string path, std::vector<std::string> names,
                        std::vector<torch::Tensor> tensors,
                        std::vector<std::string> meta_keys,
@@ -69,6 +72,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rccl_comm_time_ms", &rccl_comm_time_ms,
         "drain measured collective time (ms)");
   m.def("rccl_destroy", &rccl_destroy, "destroy communicator");
+  m.def("swiglu_fwd", &swiglu_fwd, "fused silu(g)*u forward");
+  m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward (dg, du)");
   m.def("save_shard_pinned", &save_shard_pinned,
         "stream tensors to a safetensors shard via pinned D2H (C6)");
   m.def("linear_fwd", &linear_fwd,

@@ -61,7 +61,8 @@ class LlamaBlock(nn.Module):
     def forward(self, x, cos, sin):
         x = self.attn(self.attn_norm(x), cos, sin, residual=x)
         h = self.mlp_norm(x)
-        g = F.silu(ops.linear(h, self.w_gate.weight)) * ops.linear(h, self.w_up.weight)
+        g = ops.swiglu(ops.linear(h, self.w_gate.weight),
+                       ops.linear(h, self.w_up.weight))
         x = ops.linear_residual(g, self.w_down.weight, None, x)
         return x
 

@@ -28,6 +28,7 @@ __all__ = [
     "linear",
     "linear_gelu",
     "linear_residual",
+    "swiglu",
     "fused_adamw_step",
     "require_ext",
     "has_ext",
@@ -326,6 +327,32 @@ def linear_residual(x: torch.Tensor, w: torch.Tensor,
     if x.is_cuda and _tile_ok(x, w) and (_HAND_GEMM_ALL or w.shape[0] <= 1024):
         return _LinearResidualFn.apply(x, w, b, res)
     return F.linear(x, w, b) + res
+
+
+# ----------------------------------------------------------------- SwiGLU
+class _SwiGLU(torch.autograd.Function):
+    """Fused silu(gate)*up (Llama MLP) — one HBM pass each direction."""
+
+    @staticmethod
+    def forward(ctx, g, u):
+        m = require_ext()
+        g = g.contiguous()
+        u = u.contiguous()
+        ctx.save_for_backward(g, u)
+        return m.swiglu_fwd(g, u)
+
+    @staticmethod
+    def backward(ctx, dh):
+        g, u = ctx.saved_tensors
+        m = require_ext()
+        dg, du = m.swiglu_bwd(dh.contiguous(), g, u)
+        return dg, du
+
+
+def swiglu(g: torch.Tensor, u: torch.Tensor) -> torch.Tensor:
+    if g.is_cuda and g.dtype == torch.bfloat16 and g.numel() % 8 == 0:
+        return _SwiGLU.apply(g, u)
+    return F.silu(g) * u
 
 
 # ------------------------------------------------------------- fused AdamW

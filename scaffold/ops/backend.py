@@ -34,6 +34,7 @@ _SOURCES = [
     "kernels/attn_bwd.hip",
     "kernels/attn_bwd_mfma.hip",
     "kernels/gemm_bf16.hip",
+    "kernels/swiglu.hip",
     "kernels/probe.hip",
 ]
 

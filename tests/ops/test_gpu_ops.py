@@ -307,3 +307,26 @@ def test_linear_fwd_256_path(ext):
     want = torch.nn.functional.gelu(ref, approximate="tanh")
     err = (y2.float() - want).abs().max().item()
     assert err < tol, f"gelu: max err {err} vs tol {tol}"
+
+
+def test_swiglu_fwd_bwd(ext):
+    from scaffold import ops
+
+    torch.manual_seed(15)
+    g32 = torch.randn(64, 256, requires_grad=True)
+    u32 = torch.randn(64, 256, requires_grad=True)
+    ref = torch.nn.functional.silu(g32) * u32
+    dh = torch.randn(64, 256)
+    ref.backward(dh)
+
+    g = _bf(g32.detach()).cuda().requires_grad_(True)
+    u = _bf(u32.detach()).cuda().requires_grad_(True)
+    h = ops.swiglu(g, u)
+    h.backward(_bf(dh).cuda())
+    torch.cuda.synchronize()
+    for name, got, want in [("h", h.detach(), ref.detach()),
+                            ("dg", g.grad, g32.grad), ("du", u.grad, u32.grad)]:
+        a, e = got.cpu().float(), want.float()
+        err = (a - e).abs().max().item()
+        tol = 3e-2 * max(1.0, e.abs().max().item())
+        assert err < tol, f"{name}: {err} vs {tol}"
