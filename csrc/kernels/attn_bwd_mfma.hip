@@ -75,22 +75,17 @@ DEVINL void stage_tr(const short* __restrict__ src, long row_stride, int r0,
   }
 }
 
-// per-lane tr-read addresses for the column fragments of a tr-image at
-// lds base: [t][s] -> fragment of rows 16s+8*half + 0..7 at col 32t+(l&31)
-template <int D, int NDT>
-DEVINL void tr_addrs(unsigned base, int l, unsigned addr[NDT][2][2]) {
-  const int half = l >> 5;
-  const int p = l & 15;  // lane's in-group read slot (see common.h)
-  #pragma unroll
-  for (int t = 0; t < NDT; ++t)
-    #pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      const int g = 2 * t + ((l & 31) >> 4);
-      const int q4 = 4 * s + 2 * half;
-      addr[t][s][0] = base + 128u * (unsigned)(q4 * (D / 16) + g)
-                      + (p >> 2) * 32 + (p & 3) * 8;
-      addr[t][s][1] = addr[t][s][0] + 128u * (unsigned)(D / 16);
-    }
+// per-lane tr-read BASE for the column fragments of a tr-image: the
+// (t, s) fragment of rows 16s+8*half+0..7 at col 32t+(l&31) reads at
+//   base + 128*((4s)*(D/16) + 2t)   [+128*(D/16) for the second quad]
+// — one VGPR per image, constant offsets at the read sites (the
+// 16-entry address-array version cost 32 VGPRs in dkv and spilled).
+template <int D>
+DEVINL unsigned tr_col_base(const void* img, int l) {
+  const int p = l & 15;
+  return lds_byte_off(img) + 128u * ((l & 31) >> 4)
+         + (p >> 2) * 32 + (p & 3) * 8
+         + ((l >> 5) ? 128u * 2 * (D / 16) : 0);
 }
 
 // delta = rowsum(dO * O), stride-aware (one wave per row)
@@ -162,8 +157,7 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   const float my_lse = lse[rhead + qrow];
   const float my_delta = delta[rhead + qrow];
 
-  unsigned ktr[NDT][2][2];
-  tr_addrs<D, NDT>(lds_byte_off(ktile), l, ktr);
+  const unsigned ktr = tr_col_base<D>(ktile, l);
 
   f32x16_t dqacc[NDT];
   #pragma unroll
@@ -202,15 +196,18 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
     pack_frags(ds, dsf);
 
     // dQ^T += K^T . dS^T  per d-tile; K^T A-fragments by tr-read (T10)
+    __builtin_amdgcn_s_setprio(1);
     #pragma unroll
     for (int t = 0; t < NDT; ++t) {
       #pragma unroll
       for (int s = 0; s < 2; ++s) {
-        bf16x8_t kcf = tr_read_frag(ktr[t][s][0], ktr[t][s][1]);
+        const unsigned a0 = ktr + 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
+        bf16x8_t kcf = tr_read_frag(a0, a0 + 128u * (D / 16));
         dqacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
             kcf, *(const bf16x8_t*)dsf[s], dqacc[t], 0, 0, 0);
       }
     }
+    __builtin_amdgcn_s_setprio(0);
     __syncthreads();  // all waves done with ktile before restage
   }
 
@@ -227,7 +224,7 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
 }
 
 template <int D>
-__global__ void __launch_bounds__(256)
+__global__ void __launch_bounds__(256, 2)
 attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          const short* __restrict__ q,
                          const short* __restrict__ k,
@@ -260,16 +257,23 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
 
   short* qtile = bwd_lds;
   short* dotile = bwd_lds + 32 * D;
-  unsigned qtr[NDT][2][2], dotr[NDT][2][2];
-  tr_addrs<D, NDT>(lds_byte_off(qtile), l, qtr);
-  tr_addrs<D, NDT>(lds_byte_off(dotile), l, dotr);
+  const unsigned qtr = tr_col_base<D>(qtile, l);
+  const unsigned dotr = tr_col_base<D>(dotile, l);
 
-  // K^T B-fragments for S = Q.K^T (col=k=lane, kdim=d) — contiguous
-  bf16x8_t ktf[NSL], vtf[NSL];
-  #pragma unroll
-  for (int s = 0; s < NSL; ++s) {
-    ktf[s] = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
-    vtf[s] = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
+  // K^T/V^T B-fragments for S = Q.K^T / dP = dO.V^T (col=k=lane,
+  // kdim=d) — contiguous loads.  D=64: pinned in registers; D=128:
+  // reloaded from L1 per q-tile (64 VGPRs saved -> 2 waves/SIMD; the
+  // lane's K/V rows stay L1-resident across the whole q loop).
+  constexpr bool KVREG = (D <= 64);
+  const short* krow_ptr = k + kbb + (long)krow * ks.ss + 8 * half;
+  const short* vrow_ptr = v + vbb + (long)krow * vs.ss + 8 * half;
+  bf16x8_t ktf[KVREG ? NSL : 1], vtf[KVREG ? NSL : 1];
+  if constexpr (KVREG) {
+    #pragma unroll
+    for (int s = 0; s < NSL; ++s) {
+      ktf[s] = *(const bf16x8_t*)(krow_ptr + 16 * s);
+      vtf[s] = *(const bf16x8_t*)(vrow_ptr + 16 * s);
+    }
   }
 
   f32x16_t dvacc[NDT], dkacc[NDT];
@@ -290,13 +294,16 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       // S = Q_tile . K^T  (C: lane=k, regs=q)
       const int qrow = min(c0 + lk, S - 1);
       f32x16_t sacc = {};
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int s = 0; s < NSL; ++s) {
         bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
-        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
+        const bf16x8_t kv = KVREG ? ktf[s] : *(const bf16x8_t*)(krow_ptr + 16 * s);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
 
-      float p[16], ds[16];
+      // p overwrites sacc in place (register reuse)
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
@@ -305,42 +312,49 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         const bool masked = qg >= S || (causal && krow_g > qg);
         const int qc = min(qg, S - 1);
         const float lse_r = lse[rhead + qc];    // L1-hot across lanes
-        p[r] = masked ? 0.f : __expf(sv - lse_r);
+        sacc[r] = masked ? 0.f : __expf(sv - lse_r);
       }
 
       // dP = dO . V^T (C: lane=k, regs=q): A=dO rows (contig), B=V^T (contig)
       f32x16_t dpacc = {};
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int s = 0; s < NSL; ++s) {
         bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
-        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vtf[s], dpacc, 0, 0, 0);
+        const bf16x8_t vv = KVREG ? vtf[s] : *(const bf16x8_t*)(vrow_ptr + 16 * s);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
       }
+      __builtin_amdgcn_s_setprio(0);
+      // ds overwrites dpacc in place
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
         const int qc = min(c0 + qrel, S - 1);
         const float del_r = delta[rhead + qc];
-        ds[r] = p[r] * (dpacc[r] - del_r) * scale;
+        dpacc[r] = sacc[r] * (dpacc[r] - del_r) * scale;
       }
 
       unsigned pf[2][4], dsf[2][4];
-      pack_frags(p, pf);
-      pack_frags(ds, dsf);
+      pack_frags((const float*)&sacc, pf);
+      pack_frags((const float*)&dpacc, dsf);
 
       // dV^T += P^T . dO_cols ; dK^T += dS^T . Q_cols  (C: lane=d, regs=k)
       // column B-fragments by tr-read from the staged tiles (T10)
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int t = 0; t < NDT; ++t) {
         #pragma unroll
         for (int s = 0; s < 2; ++s) {
-          bf16x8_t dof_c = tr_read_frag(dotr[t][s][0], dotr[t][s][1]);
-          bf16x8_t qf_c = tr_read_frag(qtr[t][s][0], qtr[t][s][1]);
+          const unsigned off = 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
+          bf16x8_t dof_c = tr_read_frag(dotr + off, dotr + off + 128u * (D / 16));
+          bf16x8_t qf_c = tr_read_frag(qtr + off, qtr + off + 128u * (D / 16));
           dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *(const bf16x8_t*)pf[s], dof_c, dvacc[t], 0, 0, 0);
           dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
               *(const bf16x8_t*)dsf[s], qf_c, dkacc[t], 0, 0, 0);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
       __syncthreads();  // all waves done with the tiles before restage
     }
   }

@@ -322,9 +322,10 @@ def linear_gelu(x: torch.Tensor, w: torch.Tensor,
 
 def linear_residual(x: torch.Tensor, w: torch.Tensor,
                     b: Optional[torch.Tensor], res: torch.Tensor) -> torch.Tensor:
-    # measured >= hipBLASLt for N <= 1024 (fused residual saves the
-    # separate add pass); bigger shapes stay on the library GEMM
-    if x.is_cuda and _tile_ok(x, w) and (_HAND_GEMM_ALL or w.shape[0] <= 1024):
+    # measured >= hipBLASLt only for small N AND K (the square attn
+    # proj); bigger shapes stay on the library GEMM + add
+    if x.is_cuda and _tile_ok(x, w) and (
+            _HAND_GEMM_ALL or (w.shape[0] <= 1024 and w.shape[1] <= 1024)):
         return _LinearResidualFn.apply(x, w, b, res)
     return F.linear(x, w, b) + res
 

@@ -73,16 +73,21 @@ def init_distributed(cfg: TrainConfig) -> DistContext:
 
     backend = pick_backend(cfg)
     initialized = False
-    if world > 1 or "MASTER_ADDR" in os.environ:
-        if not dist.is_initialized():
-            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-            os.environ.setdefault("MASTER_PORT", "29500")
-            dist.init_process_group(
-                backend=backend,
-                rank=rank,
-                world_size=world,
-                timeout=datetime.timedelta(seconds=cfg.dist.timeout_s),
-            )
+    if dist.is_initialized():
+        # adopt an externally-initialized process group (tests, embedders)
+        rank = dist.get_rank()
+        world = dist.get_world_size()
+        backend = dist.get_backend()
+        initialized = True
+    elif world > 1 or "MASTER_ADDR" in os.environ:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        dist.init_process_group(
+            backend=backend,
+            rank=rank,
+            world_size=world,
+            timeout=datetime.timedelta(seconds=cfg.dist.timeout_s),
+        )
         initialized = True
         # fail-fast sanity collective
         t = torch.ones(1, device=device if backend == "nccl" else "cpu")
