@@ -127,7 +127,8 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
                         Strides4 ks, Strides4 vs, Strides4 dqs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
-  __shared__ __attribute__((aligned(16))) short ktile[32 * D];  // tr image
+  // double-buffered tr image: tile c0+32 streams while c0 computes
+  __shared__ __attribute__((aligned(16))) short ktile[2 * 32 * D];
 
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
@@ -157,16 +158,20 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   const float my_lse = lse[rhead + qrow];
   const float my_delta = delta[rhead + qrow];
 
-  const unsigned ktr = tr_col_base<D>(ktile, l);
+  const unsigned ktr0 = tr_col_base<D>(ktile, l);
 
   f32x16_t dqacc[NDT];
   #pragma unroll
   for (int t = 0; t < NDT; ++t) dqacc[t] = {};
 
   const int kv_end = causal ? min(S, q0 + 128) : S;
+  stage_tr<D>(k + kbb, ks.ss, 0, S, ktile, wid, l);
+  __syncthreads();
+  int cur = 0;
   for (int c0 = 0; c0 < kv_end; c0 += 32) {
-    stage_tr<D>(k + kbb, ks.ss, c0, S, ktile, wid, l);
-    __syncthreads();  // glds drained (vmcnt 0) + barrier
+    if (c0 + 32 < kv_end)
+      stage_tr<D>(k + kbb, ks.ss, c0 + 32, S, ktile + (cur ^ 1) * 32 * D, wid, l);
+    const unsigned ktr = ktr0 + (unsigned)cur * (32 * D * 2);
 
     const int krow = min(c0 + lq, S - 1);
     f32x16_t sacc = {}, dpacc = {};
@@ -208,7 +213,9 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
       }
     }
     __builtin_amdgcn_s_setprio(0);
-    __syncthreads();  // all waves done with ktile before restage
+    // drains the next tile's glds (hidden under this tile's MFMAs)
+    __syncthreads();
+    cur ^= 1;
   }
 
   if (qrow_g < S) {
@@ -237,7 +244,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          Strides4 vs, Strides4 dks, Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
-  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 32 * D];  // Q | dO
+  // [buf][Q tile | dO tile], double-buffered across q-tiles
+  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 2 * 32 * D];
 
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
@@ -255,10 +263,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
   const int krow_g = k0 + 4 * lk + wid;     // interleaved kv row
   const int krow = min(krow_g, S - 1);
 
-  short* qtile = bwd_lds;
-  short* dotile = bwd_lds + 32 * D;
-  const unsigned qtr = tr_col_base<D>(qtile, l);
-  const unsigned dotr = tr_col_base<D>(dotile, l);
+  const unsigned qtr0 = tr_col_base<D>(bwd_lds, l);
+  const unsigned dotr0 = qtr0 + (unsigned)(32 * D * 2);
 
   // K^T/V^T B-fragments for S = Q.K^T / dP = dO.V^T (col=k=lane,
   // kdim=d) — contiguous loads.  D=64: pinned in registers; D=128:
@@ -286,10 +292,18 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
     const long dob = (long)b * dos.sb + (long)h * dos.sh;
     const long rhead = ((long)b * H + h) * S;
     const int qstart = causal ? k0 : 0;
+    auto stage_qdo = [&](int buf, int c0) {
+      short* base = bwd_lds + buf * (2 * 32 * D);
+      stage_tr<D>(q + qb, qs.ss, c0, S, base, wid, l);
+      stage_tr<D>(dout + dob, dos.ss, c0, S, base + 32 * D, wid, l);
+    };
+    stage_qdo(0, qstart);
+    __syncthreads();
+    int cur = 0;
     for (int c0 = qstart; c0 < S; c0 += 32) {
-      stage_tr<D>(q + qb, qs.ss, c0, S, qtile, wid, l);
-      stage_tr<D>(dout + dob, dos.ss, c0, S, dotile, wid, l);
-      __syncthreads();
+      if (c0 + 32 < S) stage_qdo(cur ^ 1, c0 + 32);
+      const unsigned qtr = qtr0 + (unsigned)cur * (2 * 32 * D * 2);
+      const unsigned dotr = dotr0 + (unsigned)cur * (2 * 32 * D * 2);
 
       // S = Q_tile . K^T  (C: lane=k, regs=q)
       const int qrow = min(c0 + lk, S - 1);
@@ -355,7 +369,9 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         }
       }
       __builtin_amdgcn_s_setprio(0);
-      __syncthreads();  // all waves done with the tiles before restage
+      // drains the next q-tile's glds (hidden under this tile's MFMAs)
+      __syncthreads();
+      cur ^= 1;
     }
   }
 

@@ -54,11 +54,11 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   constexpr int NSL = D / 16;   // QK^T k-slices (over d)
   constexpr int NDT = D / 32;   // O^T d-tiles
   constexpr int CH = D / 8;     // 16-B chunks per row
-  // K row-image (64 x D, chunk-XOR) + V tr-image (2 x 32 x D)
-  __shared__ __attribute__((aligned(16))) short kv_lds[2 * 64 * D];
-
-  short* kimg = kv_lds;
-  short* vimg = kv_lds + 64 * D;
+  constexpr int BUF = 2 * 64 * D;  // K row-image + V tr-image per buffer
+  // double-buffered: tile c0+64 streams in while c0 computes, so the
+  // end-of-tile __syncthreads drains glds that issued a full compute
+  // phase earlier (HBM latency hidden, 2-phase minimum recipe)
+  __shared__ __attribute__((aligned(16))) short kv_lds[2 * BUF];
 
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
@@ -98,35 +98,43 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   // tr-read base for the PV step: per-(t,s,sub) offsets are constants
   // added at the read site (1 VGPR instead of a 16-entry address array)
   //   addr(t,s,q4half) = tr_base + 128*((4s+2*half)*(D/16) + 2t) [+128*(D/16) for q4+1]
-  const unsigned tr_base = lds_byte_off(vimg) + 128u * (lq >> 4)
-                           + ((l & 15) >> 2) * 32 + (l & 3) * 8
-                           + (half ? 128u * 2 * (D / 16) : 0);
+  const unsigned tr_base0 = lds_byte_off(kv_lds + 64 * D) + 128u * (lq >> 4)
+                            + ((l & 15) >> 2) * 32 + (l & 3) * 8
+                            + (half ? 128u * 2 * (D / 16) : 0);
+
+  // cooperative stage of the 64-row K and V tiles into buffer `buf`
+  auto stage = [&](int buf, int c0) {
+    constexpr int KSLOT = 64 * D / 8 / 256;  // glds per thread per image
+    short* kdst = kv_lds + buf * BUF;
+    short* vdst = kdst + 64 * D;
+    #pragma unroll
+    for (int it = 0; it < KSLOT; ++it) {
+      // K: row-image, chunk-XOR swizzled via the SOURCE address
+      const int s16 = (it * 4 + wid) * 64 + l;
+      const int krow = min(c0 + s16 / CH, S - 1);
+      const int kbb = (s16 % CH) ^ ((s16 / CH) & 7);
+      glds16(k + kb + (long)krow * ks.ss + kbb * 8,
+             kdst + (long)(it * 4 + wid) * 512);
+      // V: tr-image (two 32-row halves)
+      const int sub = s16 / (32 * D / 8);
+      const int s8 = s16 % (32 * D / 8);
+      const int subidx = s8 >> 3;
+      const int rem8 = s8 & 7;
+      const int vrow = min(c0 + sub * 32 + 4 * (subidx / (D / 16)) + (rem8 >> 1), S - 1);
+      const int d0 = 16 * (subidx % (D / 16)) + 8 * (rem8 & 1);
+      glds16(v + vb + (long)vrow * vs.ss + d0,
+             vdst + (long)(it * 4 + wid) * 512);
+    }
+  };
 
   const int kv_end = causal ? min(S, q0 + 128) : S;
+  stage(0, 0);
+  __syncthreads();
+  int cur = 0;
   for (int c0 = 0; c0 < kv_end; c0 += 64) {
-    // ---- cooperative stage of the 64-row K and V tiles (glds16)
-    {
-      constexpr int KSLOT = 64 * D / 8 / 256;  // glds per thread per image
-      #pragma unroll
-      for (int it = 0; it < KSLOT; ++it) {
-        // K: row-image, chunk-XOR swizzled via the SOURCE address
-        const int s16 = (it * 4 + wid) * 64 + l;
-        const int krow = min(c0 + s16 / CH, S - 1);
-        const int kbb = (s16 % CH) ^ ((s16 / CH) & 7);
-        glds16(k + kb + (long)krow * ks.ss + kbb * 8,
-               kimg + (long)(it * 4 + wid) * 512);
-        // V: tr-image (two 32-row halves)
-        const int sub = s16 / (32 * D / 8);
-        const int s8 = s16 % (32 * D / 8);
-        const int subidx = s8 >> 3;
-        const int rem8 = s8 & 7;
-        const int vrow = min(c0 + sub * 32 + 4 * (subidx / (D / 16)) + (rem8 >> 1), S - 1);
-        const int d0 = 16 * (subidx % (D / 16)) + 8 * (rem8 & 1);
-        glds16(v + vb + (long)vrow * vs.ss + d0,
-               vimg + (long)(it * 4 + wid) * 512);
-      }
-    }
-    __syncthreads();  // drains the glds (vmcnt 0) + barrier
+    if (c0 + 64 < kv_end) stage(cur ^ 1, c0 + 64);
+    const short* kimg = kv_lds + cur * BUF;
+    const unsigned tr_base = tr_base0 + (unsigned)cur * (BUF * 2);
 
     // ---- S^T = K_tile . Q^T for both 32-row sub-tiles
     f32x16_t sacc[2];
@@ -228,7 +236,10 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
       }
       __builtin_amdgcn_s_setprio(0);
     }
-    __syncthreads();  // all waves done with the images before restage
+    // drains the NEXT tile's glds (issued before compute — latency
+    // hidden under this tile's MFMAs) + read-write barrier
+    __syncthreads();
+    cur ^= 1;
   }
 
   // ---- epilogue: O = O^T / l, scatter store (lane's q row fixed)
