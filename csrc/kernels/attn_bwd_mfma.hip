@@ -244,8 +244,9 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          Strides4 vs, Strides4 dks, Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
-  // [buf][Q tile | dO tile], double-buffered across q-tiles
-  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 2 * 32 * D];
+  // [Q tile | dO tile] — single-buffered: the double-buffer variant
+  // pushed D=128 to 148 B/lane of spill and measured net-slower
+  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 32 * D];
 
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
@@ -292,18 +293,12 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
     const long dob = (long)b * dos.sb + (long)h * dos.sh;
     const long rhead = ((long)b * H + h) * S;
     const int qstart = causal ? k0 : 0;
-    auto stage_qdo = [&](int buf, int c0) {
-      short* base = bwd_lds + buf * (2 * 32 * D);
-      stage_tr<D>(q + qb, qs.ss, c0, S, base, wid, l);
-      stage_tr<D>(dout + dob, dos.ss, c0, S, base + 32 * D, wid, l);
-    };
-    stage_qdo(0, qstart);
-    __syncthreads();
-    int cur = 0;
     for (int c0 = qstart; c0 < S; c0 += 32) {
-      if (c0 + 32 < S) stage_qdo(cur ^ 1, c0 + 32);
-      const unsigned qtr = qtr0 + (unsigned)cur * (2 * 32 * D * 2);
-      const unsigned dotr = dotr0 + (unsigned)cur * (2 * 32 * D * 2);
+      stage_tr<D>(q + qb, qs.ss, c0, S, bwd_lds, wid, l);
+      stage_tr<D>(dout + dob, dos.ss, c0, S, bwd_lds + 32 * D, wid, l);
+      __syncthreads();
+      const unsigned qtr = qtr0;
+      const unsigned dotr = dotr0;
 
       // S = Q_tile . K^T  (C: lane=k, regs=q)
       const int qrow = min(c0 + lk, S - 1);
@@ -369,9 +364,7 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         }
       }
       __builtin_amdgcn_s_setprio(0);
-      // drains the next q-tile's glds (hidden under this tile's MFMAs)
-      __syncthreads();
-      cur ^= 1;
+      __syncthreads();  // all waves done with the tiles before restage
     }
   }
 
