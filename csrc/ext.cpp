@@ -40,6 +40,8 @@ void rccl_join(long handle);
 double rccl_comm_time_ms(long handle);
 void rccl_destroy(long handle);
 torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u);
+torch::Tensor swiglu_packed_fwd(torch::Tensor gu);
+torch::Tensor swiglu_packed_bwd(torch::Tensor dh, torch::Tensor gu);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dh, torch::Tensor g,
                                       torch::Tensor u);
 void save_shard_pinned(std::string path, std::vector<std::string> names,
@@ -73,6 +75,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "drain measured collective time (ms)");
   m.def("rccl_destroy", &rccl_destroy, "destroy communicator");
   m.def("swiglu_fwd", &swiglu_fwd, "fused silu(g)*u forward");
+  m.def("swiglu_packed_fwd", &swiglu_packed_fwd,
+        "fused SwiGLU on a packed gate|up buffer");
+  m.def("swiglu_packed_bwd", &swiglu_packed_bwd,
+        "packed SwiGLU backward (dgu)");
   m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward (dg, du)");
   m.def("save_shard_pinned", &save_shard_pinned,
         "stream tensors to a safetensors shard via pinned D2H (C6)");

@@ -26,21 +26,25 @@ class RMSNorm(nn.Module):
 
 
 class LlamaAttention(nn.Module):
+    """GQA attention with ONE fused QKV projection (q|k|v heads packed
+    along the output dim -> a single large GEMM instead of three)."""
+
     def __init__(self, d_model: int, n_heads: int, n_kv_heads: int):
         super().__init__()
         self.n_heads = n_heads
         self.n_kv = n_kv_heads
         self.head_dim = d_model // n_heads
-        self.wq = nn.Linear(d_model, n_heads * self.head_dim, bias=False)
-        self.wk = nn.Linear(d_model, n_kv_heads * self.head_dim, bias=False)
-        self.wv = nn.Linear(d_model, n_kv_heads * self.head_dim, bias=False)
+        self.wqkv = nn.Linear(
+            d_model, (n_heads + 2 * n_kv_heads) * self.head_dim, bias=False)
         self.wo = nn.Linear(n_heads * self.head_dim, d_model, bias=False)
 
     def forward(self, x, cos, sin, residual):
         B, S, _ = x.shape
-        q = ops.linear(x, self.wq.weight).view(B, S, self.n_heads, self.head_dim).transpose(1, 2)
-        k = ops.linear(x, self.wk.weight).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
-        v = ops.linear(x, self.wv.weight).view(B, S, self.n_kv, self.head_dim).transpose(1, 2)
+        qkv = ops.linear(x, self.wqkv.weight)
+        hv = qkv.view(B, S, self.n_heads + 2 * self.n_kv, self.head_dim)
+        q = hv[:, :, :self.n_heads].transpose(1, 2)          # (B,H,S,D) strided
+        k = hv[:, :, self.n_heads:self.n_heads + self.n_kv].transpose(1, 2)
+        v = hv[:, :, self.n_heads + self.n_kv:].transpose(1, 2)
         q = ops.rope_apply(q, cos, sin)
         k = ops.rope_apply(k, cos, sin)
         o = ops.attention(q, k, v, causal=True)
@@ -54,15 +58,15 @@ class LlamaBlock(nn.Module):
         self.attn_norm = RMSNorm(d_model, eps)
         self.attn = LlamaAttention(d_model, n_heads, n_kv_heads)
         self.mlp_norm = RMSNorm(d_model, eps)
-        self.w_gate = nn.Linear(d_model, d_ff, bias=False)
-        self.w_up = nn.Linear(d_model, d_ff, bias=False)
+        # gate and up fused into one [2*d_ff, d] GEMM; the packed SwiGLU
+        # kernel consumes the gate|up halves in place
+        self.w_gateup = nn.Linear(d_model, 2 * d_ff, bias=False)
         self.w_down = nn.Linear(d_ff, d_model, bias=False)
 
     def forward(self, x, cos, sin):
         x = self.attn(self.attn_norm(x), cos, sin, residual=x)
         h = self.mlp_norm(x)
-        g = ops.swiglu(ops.linear(h, self.w_gate.weight),
-                       ops.linear(h, self.w_up.weight))
+        g = ops.swiglu_packed(ops.linear(h, self.w_gateup.weight))
         x = ops.linear_residual(g, self.w_down.weight, None, x)
         return x
 

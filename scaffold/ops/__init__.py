@@ -29,6 +29,7 @@ __all__ = [
     "linear_gelu",
     "linear_residual",
     "swiglu",
+    "swiglu_packed",
     "fused_adamw_step",
     "require_ext",
     "has_ext",
@@ -354,6 +355,33 @@ def swiglu(g: torch.Tensor, u: torch.Tensor) -> torch.Tensor:
     if g.is_cuda and g.dtype == torch.bfloat16 and g.numel() % 8 == 0:
         return _SwiGLU.apply(g, u)
     return F.silu(g) * u
+
+
+class _SwiGLUPacked(torch.autograd.Function):
+    """SwiGLU over ONE packed [.., 2F] gate|up buffer (the fused
+    gate+up GEMM output) — bwd emits one packed dgu for a single
+    weight-grad GEMM."""
+
+    @staticmethod
+    def forward(ctx, gu):
+        m = require_ext()
+        gu = gu.contiguous()
+        ctx.save_for_backward(gu)
+        return m.swiglu_packed_fwd(gu)
+
+    @staticmethod
+    def backward(ctx, dh):
+        (gu,) = ctx.saved_tensors
+        m = require_ext()
+        return m.swiglu_packed_bwd(dh.contiguous(), gu)
+
+
+def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
+    """h = silu(gu[..., :F]) * gu[..., F:] where F = gu.shape[-1]//2."""
+    if gu.is_cuda and gu.dtype == torch.bfloat16 and gu.shape[-1] % 16 == 0:
+        return _SwiGLUPacked.apply(gu)
+    F_ = gu.shape[-1] // 2
+    return F.silu(gu[..., :F_]) * gu[..., F_:]
 
 
 # ------------------------------------------------------------- fused AdamW
