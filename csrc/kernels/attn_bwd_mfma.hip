@@ -312,6 +312,13 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       }
       __builtin_amdgcn_s_setprio(0);
 
+      // per-tile q-row constants loaded ONCE per wave (lane j holds row
+      // c0+j's values) and broadcast per-reg via shfl — replaces 32
+      // scalar L1 loads per lane per tile with 2 loads + shuffles
+      const int qrow_l = min(c0 + lq, S - 1);
+      const float lse_lane = lse[rhead + qrow_l];
+      const float del_lane = delta[rhead + qrow_l];
+
       // p overwrites sacc in place (register reuse)
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -319,8 +326,7 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         const int qg = c0 + qrel;
         float sv = sacc[r] * scale;
         const bool masked = qg >= S || (causal && krow_g > qg);
-        const int qc = min(qg, S - 1);
-        const float lse_r = lse[rhead + qc];    // L1-hot across lanes
+        const float lse_r = __shfl(lse_lane, qrel, 64);
         sacc[r] = masked ? 0.f : __expf(sv - lse_r);
       }
 
@@ -338,8 +344,7 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
-        const int qc = min(c0 + qrel, S - 1);
-        const float del_r = delta[rhead + qc];
+        const float del_r = __shfl(del_lane, qrel, 64);
         dpacc[r] = sacc[r] * (dpacc[r] - del_r) * scale;
       }
 
