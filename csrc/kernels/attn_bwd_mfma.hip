@@ -315,7 +315,7 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       // per-tile q-row constants loaded ONCE per wave (lane j holds row
       // c0+j's values) and broadcast per-reg via shfl — replaces 32
       // scalar L1 loads per lane per tile with 2 loads + shuffles
-      const int qrow_l = min(c0 + lq, S - 1);
+      const int qrow_l = min(c0 + lk, S - 1);
       const float lse_lane = lse[rhead + qrow_l];
       const float del_lane = delta[rhead + qrow_l];
 
