@@ -42,6 +42,7 @@ void rccl_destroy(long handle);
 torch::Tensor swiglu_fwd(torch::Tensor g, torch::Tensor u);
 torch::Tensor swiglu_packed_fwd(torch::Tensor gu);
 torch::Tensor swiglu_packed_bwd(torch::Tensor dh, torch::Tensor gu);
+torch::Tensor colsum(torch::Tensor x);
 std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dh, torch::Tensor g,
                                       torch::Tensor u);
 void save_shard_pinned(std::string path, std::vector<std::string> names,
@@ -79,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused SwiGLU on a packed gate|up buffer");
   m.def("swiglu_packed_bwd", &swiglu_packed_bwd,
         "packed SwiGLU backward (dgu)");
+  m.def("colsum", &colsum, "bf16 column sum (bias gradients)");
   m.def("swiglu_bwd", &swiglu_bwd, "fused SwiGLU backward (dg, du)");
   m.def("save_shard_pinned", &save_shard_pinned,
         "stream tensors to a safetensors shard via pinned D2H (C6)");

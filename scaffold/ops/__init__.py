@@ -217,6 +217,13 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return _attention_eager(q, k, v, causal)
 
 
+def _colsum(dy2: torch.Tensor) -> torch.Tensor:
+    m = ext()
+    if m is not None and dy2.is_cuda and dy2.dtype == torch.bfloat16:
+        return m.colsum(dy2)
+    return dy2.sum(0)
+
+
 # ------------------------------------------------- K5 hand MFMA linear
 def _tile_ok(x: torch.Tensor, w: torch.Tensor) -> bool:
     """The hand kernel covers the benched transformer-block shapes
@@ -246,7 +253,7 @@ class _LinearFn(torch.autograd.Function):
         x2 = x.view(-1, x.shape[-1])
         dx = (dy2 @ w).view(x.shape)
         dw = dy2.t() @ x2
-        db = dy2.sum(0) if ctx.has_bias else None
+        db = _colsum(dy2) if ctx.has_bias else None
         return dx, dw, db
 
 
@@ -271,7 +278,7 @@ class _LinearGeluFn(torch.autograd.Function):
         x2 = x.view(-1, x.shape[-1])
         dx = (dpre2 @ w).view(x.shape)
         dw = dpre2.t() @ x2
-        db = dpre2.sum(0) if ctx.has_bias else None
+        db = _colsum(dpre2) if ctx.has_bias else None
         return dx, dw, db
 
 
@@ -295,7 +302,7 @@ class _LinearResidualFn(torch.autograd.Function):
         x2 = x.view(-1, x.shape[-1])
         dx = (dy2 @ w).view(x.shape)
         dw = dy2.t() @ x2
-        db = dy2.sum(0) if ctx.has_bias else None
+        db = _colsum(dy2) if ctx.has_bias else None
         return dx, dw, db, dy
 
 
@@ -307,10 +314,40 @@ class _LinearResidualFn(torch.autograd.Function):
 _HAND_GEMM_ALL = False  # flipped by benchmarks once the 8-phase core wins
 
 
+class _LinearLib(torch.autograd.Function):
+    """Library-GEMM (hipBLASLt) forward with an explicit backward:
+    rocBLAS matmuls for dX/dW plus the hand colsum kernel for the bias
+    gradient (torch's generic reduce measured ~1 TB/s there)."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return F.linear(x, w, b)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        x2 = x.contiguous().view(-1, x.shape[-1])
+        dx = (dy2 @ w).view(x.shape)
+        dw = dy2.t() @ x2
+        db = None
+        if ctx.has_bias:
+            m = ext()
+            if m is not None and dy2.dtype == torch.bfloat16:
+                db = m.colsum(dy2)
+            else:
+                db = dy2.sum(0)
+        return dx, dw, db
+
+
 def linear(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None
            ) -> torch.Tensor:
     if x.is_cuda and _HAND_GEMM_ALL and _tile_ok(x, w):
         return _LinearFn.apply(x, w, b)
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _LinearLib.apply(x, w, b)
     return F.linear(x, w, b)
 
 
@@ -318,6 +355,8 @@ def linear_gelu(x: torch.Tensor, w: torch.Tensor,
                 b: Optional[torch.Tensor] = None) -> torch.Tensor:
     if x.is_cuda and _HAND_GEMM_ALL and _tile_ok(x, w):
         return _LinearGeluFn.apply(x, w, b)
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return F.gelu(_LinearLib.apply(x, w, b), approximate="tanh")
     return F.gelu(F.linear(x, w, b), approximate="tanh")
 
 

@@ -35,6 +35,7 @@ _SOURCES = [
     "kernels/attn_bwd_mfma.hip",
     "kernels/gemm_bf16.hip",
     "kernels/swiglu.hip",
+    "kernels/colsum.hip",
     "kernels/probe.hip",
 ]
 
