@@ -155,6 +155,40 @@ DEVINL bf16x8 tr_read_frag(unsigned addr_q0, unsigned addr_q1) {
   return r.v;
 }
 
+// FOUR fragments (8 tr reads) behind ONE lgkmcnt wait: per-fragment
+// waits serialized every consumer MFMA behind a full LDS round trip
+// (measured MfmaUtil 6-10% on the attention kernels); batching the
+// reads keeps the LDS pipe full.  §5.7 form (i): loads + wait in one
+// asm statement, early-clobber outputs.
+struct bf16x8_quad { bf16x8 f[4]; };
+
+DEVINL bf16x8_quad tr_read_quad(unsigned a0, unsigned a1, unsigned b0,
+                                unsigned b1, unsigned c0, unsigned c1,
+                                unsigned d0, unsigned d1) {
+  unsigned long long r0, r1, r2, r3, r4, r5, r6, r7;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8\n\t"
+      "ds_read_b64_tr_b16 %1, %9\n\t"
+      "ds_read_b64_tr_b16 %2, %10\n\t"
+      "ds_read_b64_tr_b16 %3, %11\n\t"
+      "ds_read_b64_tr_b16 %4, %12\n\t"
+      "ds_read_b64_tr_b16 %5, %13\n\t"
+      "ds_read_b64_tr_b16 %6, %14\n\t"
+      "ds_read_b64_tr_b16 %7, %15\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(r0), "=&v"(r1), "=&v"(r2), "=&v"(r3),
+        "=&v"(r4), "=&v"(r5), "=&v"(r6), "=&v"(r7)
+      : "v"(a0), "v"(a1), "v"(b0), "v"(b1),
+        "v"(c0), "v"(c1), "v"(d0), "v"(d1));
+  union { struct { unsigned long long lo, hi; } u; bf16x8 v; } t;
+  bf16x8_quad q;
+  t.u.lo = r0; t.u.hi = r1; q.f[0] = t.v;
+  t.u.lo = r2; t.u.hi = r3; q.f[1] = t.v;
+  t.u.lo = r4; t.u.hi = r5; q.f[2] = t.v;
+  t.u.lo = r6; t.u.hi = r7; q.f[3] = t.v;
+  return q;
+}
+
 // Grid sizing guideline 11: cap grid, grid-stride the rest.
 DEVINL long grid_stride() { return (long)gridDim.x * blockDim.x; }
 

@@ -200,17 +200,27 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
     unsigned dsf[2][4];
     pack_frags(ds, dsf);
 
-    // dQ^T += K^T . dS^T  per d-tile; K^T A-fragments by tr-read (T10)
+    // dQ^T += K^T . dS^T  per d-tile; K^T A-fragments by batched
+    // tr-read (one LDS wait per 4 fragments — common.h)
     __builtin_amdgcn_s_setprio(1);
     #pragma unroll
-    for (int t = 0; t < NDT; ++t) {
-      #pragma unroll
-      for (int s = 0; s < 2; ++s) {
-        const unsigned a0 = ktr + 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
-        bf16x8_t kcf = tr_read_frag(a0, a0 + 128u * (D / 16));
-        dqacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kcf, *(const bf16x8_t*)dsf[s], dqacc[t], 0, 0, 0);
-      }
+    for (int t2 = 0; t2 < NDT; t2 += 2) {
+      const unsigned qstep = 128u * (D / 16);
+      auto ka = [&](int t, int s) {
+        return ktr + 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
+      };
+      bf16x8_quad kq = tr_read_quad(
+          ka(t2, 0), ka(t2, 0) + qstep, ka(t2, 1), ka(t2, 1) + qstep,
+          ka(t2 + 1, 0), ka(t2 + 1, 0) + qstep,
+          ka(t2 + 1, 1), ka(t2 + 1, 1) + qstep);
+      dqacc[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          kq.f[0], *(const bf16x8_t*)dsf[0], dqacc[t2], 0, 0, 0);
+      dqacc[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          kq.f[1], *(const bf16x8_t*)dsf[1], dqacc[t2], 0, 0, 0);
+      dqacc[t2 + 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          kq.f[2], *(const bf16x8_t*)dsf[0], dqacc[t2 + 1], 0, 0, 0);
+      dqacc[t2 + 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          kq.f[3], *(const bf16x8_t*)dsf[1], dqacc[t2 + 1], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
     // drains the next tile's glds (hidden under this tile's MFMAs)
@@ -353,20 +363,25 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       pack_frags((const float*)&dpacc, dsf);
 
       // dV^T += P^T . dO_cols ; dK^T += dS^T . Q_cols  (C: lane=d, regs=k)
-      // column B-fragments by tr-read from the staged tiles (T10)
+      // column B-fragments by batched tr-read: one LDS wait per d-tile
+      // pair covers both tensors' 2 k-slices (common.h)
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int t = 0; t < NDT; ++t) {
-        #pragma unroll
-        for (int s = 0; s < 2; ++s) {
-          const unsigned off = 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
-          bf16x8_t dof_c = tr_read_frag(dotr + off, dotr + off + 128u * (D / 16));
-          bf16x8_t qf_c = tr_read_frag(qtr + off, qtr + off + 128u * (D / 16));
-          dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)pf[s], dof_c, dvacc[t], 0, 0, 0);
-          dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)dsf[s], qf_c, dkacc[t], 0, 0, 0);
-        }
+        const unsigned qstep = 128u * (D / 16);
+        const unsigned o0 = 128u * (unsigned)(2 * t);
+        const unsigned o1 = o0 + 128u * (unsigned)(4 * (D / 16));  // s=1
+        bf16x8_quad cq = tr_read_quad(
+            dotr + o0, dotr + o0 + qstep, dotr + o1, dotr + o1 + qstep,
+            qtr + o0, qtr + o0 + qstep, qtr + o1, qtr + o1 + qstep);
+        dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *(const bf16x8_t*)pf[0], cq.f[0], dvacc[t], 0, 0, 0);
+        dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *(const bf16x8_t*)pf[1], cq.f[1], dvacc[t], 0, 0, 0);
+        dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *(const bf16x8_t*)dsf[0], cq.f[2], dkacc[t], 0, 0, 0);
+        dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *(const bf16x8_t*)dsf[1], cq.f[3], dkacc[t], 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
       __syncthreads();  // all waves done with the tiles before restage

@@ -221,18 +221,27 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
         pfrag[s][3] = ry[1];
       }
       const unsigned sub_off = (unsigned)sub * (32u * D * 2);
+      // batch 2 d-tiles x 2 k-slices of V^T fragments per LDS wait
+      // (q4 = 4s, +2*half folded into tr_base)
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
-      for (int t = 0; t < NDT; ++t) {
-        #pragma unroll
-        for (int s = 0; s < 2; ++s) {
-          // q4 = 4s (+2*half folded into tr_base)
-          const unsigned a0 = tr_base + sub_off
-                              + 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
-          bf16x8_t vf = tr_read_frag(a0, a0 + 128u * (D / 16));
-          oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              vf, *(const bf16x8_t*)pfrag[s], oacc[t], 0, 0, 0);
-        }
+      for (int t2 = 0; t2 < NDT; t2 += 2) {
+        auto addr = [&](int t, int s) {
+          return tr_base + sub_off + 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
+        };
+        const unsigned qstep = 128u * (D / 16);
+        bf16x8_quad vq = tr_read_quad(
+            addr(t2, 0), addr(t2, 0) + qstep, addr(t2, 1), addr(t2, 1) + qstep,
+            addr(t2 + 1, 0), addr(t2 + 1, 0) + qstep,
+            addr(t2 + 1, 1), addr(t2 + 1, 1) + qstep);
+        oacc[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            vq.f[0], *(const bf16x8_t*)pfrag[0], oacc[t2], 0, 0, 0);
+        oacc[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            vq.f[1], *(const bf16x8_t*)pfrag[1], oacc[t2], 0, 0, 0);
+        oacc[t2 + 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            vq.f[2], *(const bf16x8_t*)pfrag[0], oacc[t2 + 1], 0, 0, 0);
+        oacc[t2 + 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            vq.f[3], *(const bf16x8_t*)pfrag[1], oacc[t2 + 1], 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
     }
