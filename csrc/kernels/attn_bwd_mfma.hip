@@ -103,12 +103,21 @@ attn_delta_strided_kernel(const short* __restrict__ dout,
   const long b = row / ((long)S * H);
   const long dbase = b * dos.sb + (long)h * dos.sh + (long)s * dos.ss;
   const long obase = b * osr.sb + (long)h * osr.sh + (long)s * osr.ss;
+  // pairwise bf16->f32 (shl for lo, mask for hi: 1 VALU/elem — this
+  // kernel measured 98% VALUBusy with the scalar 2-3 op/elem unpack)
   float acc = 0.f;
   for (int d = lane * 8; d < D; d += 64 * 8) {
-    bf16x8_t dv = *(const bf16x8_t*)(dout + dbase + d);
-    bf16x8_t ov = *(const bf16x8_t*)(o + obase + d);
+    typedef unsigned uint4_t __attribute__((ext_vector_type(4)));
+    uint4_t dv = *(const uint4_t*)(dout + dbase + d);
+    uint4_t ov = *(const uint4_t*)(o + obase + d);
     #pragma unroll
-    for (int j = 0; j < 8; ++j) acc += bf2f(dv[j]) * bf2f(ov[j]);
+    for (int j = 0; j < 4; ++j) {
+      union { unsigned i; float f; } dl, dh, ol, oh;
+      dl.i = dv[j] << 16; dh.i = dv[j] & 0xffff0000u;
+      ol.i = ov[j] << 16; oh.i = ov[j] & 0xffff0000u;
+      acc = fmaf(dl.f, ol.f, acc);
+      acc = fmaf(dh.f, oh.f, acc);
+    }
   }
   acc = wave_reduce_sum(acc);
   if (lane == 0) delta[row] = acc;
