@@ -116,7 +116,7 @@ def main() -> int:
                 "model": args.model,
                 "global_batch": global_batch,
                 "seq_len": cfg.data.seq_len if is_lm else None,
-                "image_size": cfg.data.image_size if not is_lm else None,
+                "image_size": cfg.data.image_size if args.model == "resnet50" else None,
                 "parallelism": f"dp{n_gpus}",
                 "bucket_cap_mb": cfg.dist.bucket_cap_mb,
                 **extra_cfg,
