@@ -319,24 +319,28 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       const unsigned qtr = qtr0;
       const unsigned dotr = dotr0;
 
-      // S = Q_tile . K^T  (C: lane=k, regs=q)
+      // S = Q_tile . K^T and dP = dO . V^T in ONE loop: the two
+      // accumulator chains are independent, so interleaving them hides
+      // the dependent-MFMA latency that serialized two separate loops
       const int qrow = min(c0 + lk, S - 1);
-      f32x16_t sacc = {};
+      f32x16_t sacc = {}, dpacc = {};
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int s = 0; s < NSL; ++s) {
         bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
+        bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
         const bf16x8_t kv = KVREG ? ktf[s] : *(const bf16x8_t*)(krow_ptr + 16 * s);
+        const bf16x8_t vv = KVREG ? vtf[s] : *(const bf16x8_t*)(vrow_ptr + 16 * s);
         sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
 
       // per-tile q-row constants loaded ONCE per wave (lane j holds row
       // c0+j's values) and broadcast per-reg via shfl — replaces 32
       // scalar L1 loads per lane per tile with 2 loads + shuffles
-      const int qrow_l = min(c0 + lk, S - 1);
-      const float lse_lane = lse[rhead + qrow_l];
-      const float del_lane = delta[rhead + qrow_l];
+      const float lse_lane = lse[rhead + qrow];
+      const float del_lane = delta[rhead + qrow];
 
       // p overwrites sacc in place (register reuse)
       #pragma unroll
@@ -348,17 +352,6 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         const float lse_r = __shfl(lse_lane, qrel, 64);
         sacc[r] = masked ? 0.f : __expf(sv - lse_r);
       }
-
-      // dP = dO . V^T (C: lane=k, regs=q): A=dO rows (contig), B=V^T (contig)
-      f32x16_t dpacc = {};
-      __builtin_amdgcn_s_setprio(1);
-      #pragma unroll
-      for (int s = 0; s < NSL; ++s) {
-        bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
-        const bf16x8_t vv = KVREG ? vtf[s] : *(const bf16x8_t*)(vrow_ptr + 16 * s);
-        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
-      }
-      __builtin_amdgcn_s_setprio(0);
       // ds overwrites dpacc in place
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
