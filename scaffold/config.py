@@ -106,6 +106,7 @@ class TrainConfig:
     log: LogConfig = field(default_factory=LogConfig)
     steps: int = 100
     grad_accum: int = 1
+    cudnn_benchmark: bool = True  # MIOpen exhaustive find (cached in miopen_db)
     seed: int = 0
     dtype: str = "bf16"  # compute dtype: bf16 | fp32
     device: str = "auto"  # auto | cuda | cpu

@@ -53,8 +53,11 @@ class Trainer:
             raw_model = self._to_mixed(raw_model)
         self.channels_last = cfg.model.name == "resnet50" and self.ctx.on_gpu
         if self.ctx.on_gpu:
-            # MIOpen: exhaustive algo find amortized over the run
-            torch.backends.cudnn.benchmark = True
+            # MIOpen: exhaustive algo find; results persist in the
+            # in-tree miopen_db so later runs skip the search.  Set
+            # cudnn_benchmark=false for heuristic (immediate) algo
+            # choice when the search cost cannot be amortized.
+            torch.backends.cudnn.benchmark = bool(cfg.cudnn_benchmark)
         if self.channels_last:
             # NHWC is MIOpen's fast conv layout on CDNA
             raw_model = raw_model.to(memory_format=torch.channels_last)
