@@ -38,6 +38,7 @@ def main() -> int:
     ap.add_argument("--batch-size", type=int, default=0, help="per-rank; 0=config default")
     ap.add_argument("--bucket-mb", type=float, default=0.0, help="bucket sweep override")
     ap.add_argument("--config", type=str, default="")
+    ap.add_argument("overrides", nargs="*", help="section.key=value overrides")
     args = ap.parse_args()
 
     cfg_file = args.config or {
@@ -47,7 +48,7 @@ def main() -> int:
         "mlp": "configs/mlp_cpu.json",
     }[args.model]
     here = os.path.dirname(os.path.abspath(__file__))
-    cfg = load_config(os.path.join(here, cfg_file), [])
+    cfg = load_config(os.path.join(here, cfg_file), args.overrides)
     if args.batch_size:
         cfg.data.batch_size = args.batch_size
     if args.bucket_mb:
