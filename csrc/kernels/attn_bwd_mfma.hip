@@ -319,20 +319,35 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       const unsigned qtr = qtr0;
       const unsigned dotr = dotr0;
 
-      // S = Q_tile . K^T and dP = dO . V^T in ONE loop: the two
-      // accumulator chains are independent, so interleaving them hides
-      // the dependent-MFMA latency that serialized two separate loops
+      // S = Q_tile . K^T and dP = dO . V^T: at D<=64 (fragments pinned
+      // in registers) the two independent accumulator chains interleave
+      // in ONE loop to hide dependent-MFMA latency (measured -6% on the
+      // GPT-2 bwd); at D=128 the doubled live operands spilled and
+      // measured slower, so the chains stay in two loops there.
       const int qrow = min(c0 + lk, S - 1);
       f32x16_t sacc = {}, dpacc = {};
       __builtin_amdgcn_s_setprio(1);
-      #pragma unroll
-      for (int s = 0; s < NSL; ++s) {
-        bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
-        bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
-        const bf16x8_t kv = KVREG ? ktf[s] : *(const bf16x8_t*)(krow_ptr + 16 * s);
-        const bf16x8_t vv = KVREG ? vtf[s] : *(const bf16x8_t*)(vrow_ptr + 16 * s);
-        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
-        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
+      if constexpr (KVREG) {
+        #pragma unroll
+        for (int s = 0; s < NSL; ++s) {
+          bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
+          bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
+          sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
+          dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vtf[s], dpacc, 0, 0, 0);
+        }
+      } else {
+        #pragma unroll
+        for (int s = 0; s < NSL; ++s) {
+          bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
+          const bf16x8_t kv = *(const bf16x8_t*)(krow_ptr + 16 * s);
+          sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
+        }
+        #pragma unroll
+        for (int s = 0; s < NSL; ++s) {
+          bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
+          const bf16x8_t vv = *(const bf16x8_t*)(vrow_ptr + 16 * s);
+          dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
+        }
       }
       __builtin_amdgcn_s_setprio(0);
 
