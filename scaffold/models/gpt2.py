@@ -102,5 +102,5 @@ class GPT2(nn.Module):
         x = self.wte(tokens) + self.wpe(pos)
         for blk in self.blocks:
             x = blk(x)
-        logits = self.lm_head(self.ln_f(x))
-        return ops.softmax_cross_entropy(logits, targets)
+        # fused chunked lm_head+CE: the [T, V] logits never round-trip HBM
+        return ops.lm_head_ce(self.ln_f(x), self.lm_head.weight, targets)

@@ -100,5 +100,5 @@ class Llama(nn.Module):
         x = self.embed(tokens)
         for blk in self.blocks:
             x = blk(x, cos, sin)
-        logits = self.lm_head(self.norm_f(x))
-        return ops.softmax_cross_entropy(logits, targets)
+        # fused chunked lm_head+CE (ops.lm_head_ce)
+        return ops.lm_head_ce(self.norm_f(x), self.lm_head.weight, targets)

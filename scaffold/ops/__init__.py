@@ -30,6 +30,7 @@ __all__ = [
     "linear_residual",
     "swiglu",
     "swiglu_packed",
+    "lm_head_ce",
     "fused_adamw_step",
     "require_ext",
     "has_ext",
@@ -368,6 +369,62 @@ def linear_residual(x: torch.Tensor, w: torch.Tensor,
             _HAND_GEMM_ALL or (w.shape[0] <= 1024 and w.shape[1] <= 1024)):
         return _LinearResidualFn.apply(x, w, b, res)
     return F.linear(x, w, b) + res
+
+
+# ----------------------------------------------- fused lm_head + CE
+class _LmHeadCE(torch.autograd.Function):
+    """Chunked fused final projection + softmax-cross-entropy.
+
+    The full [T, V] logits matrix (1.65 GB for GPT-2, 4 GB for Llama at
+    the bench shapes) never round-trips HBM: each row chunk's logits are
+    produced (hipBLASLt), reduced to loss and dlogits (K3 kernels), and
+    immediately consumed into dh and an fp32 dW accumulator while still
+    cache-resident.  Since the logits are not kept, both grads are
+    computed in forward (per-chunk bwd kernel scale R_c/T folds the mean
+    over ALL T rows) and scaled by the incoming dloss in backward.
+    """
+
+    CHUNK = 2048
+
+    @staticmethod
+    def forward(ctx, h2, w, targets):
+        m = require_ext()
+        T = h2.shape[0]
+        dh = torch.empty_like(h2)
+        dw32 = torch.zeros_like(w, dtype=torch.float32)
+        loss_sum = None
+        for r0 in range(0, T, _LmHeadCE.CHUNK):
+            hc = h2[r0:r0 + _LmHeadCE.CHUNK]
+            tc = targets[r0:r0 + _LmHeadCE.CHUNK]
+            logits = hc @ w.t()
+            loss_c, lse = m.softmax_xent_fwd(logits, tc)
+            s = loss_c.sum()
+            loss_sum = s if loss_sum is None else loss_sum + s
+            dlogits = m.softmax_xent_bwd(logits, tc, lse, hc.shape[0] / T)
+            dh[r0:r0 + _LmHeadCE.CHUNK] = dlogits @ w
+            dw32 += (dlogits.t() @ hc).float()
+        ctx.saved_dh = dh
+        ctx.saved_dw32 = dw32
+        ctx.w_dtype = w.dtype
+        return loss_sum / T
+
+    @staticmethod
+    def backward(ctx, dloss):
+        dh = ctx.saved_dh * dloss
+        dw = (ctx.saved_dw32 * dloss.float()).to(ctx.w_dtype)
+        ctx.saved_dh = None
+        ctx.saved_dw32 = None
+        return dh, dw, None
+
+
+def lm_head_ce(x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor
+               ) -> torch.Tensor:
+    """mean CE of (x @ w^T) against targets; x (..., D), targets (...)."""
+    x2 = x.contiguous().view(-1, x.shape[-1])
+    t = targets.contiguous().view(-1)
+    if x.is_cuda:
+        return _LmHeadCE.apply(x2, w, t)
+    return F.cross_entropy(x2.float() @ w.float().t(), t)
 
 
 # ----------------------------------------------------------------- SwiGLU

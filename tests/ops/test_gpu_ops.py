@@ -381,3 +381,33 @@ def test_kernel_determinism(ext):
     y1, *_ = ext.layernorm_fwd(x, w, b, 1e-5)
     y2, *_ = ext.layernorm_fwd(x, w, b, 1e-5)
     assert torch.equal(y1, y2)
+
+
+def test_lm_head_ce_matches_eager():
+    from scaffold import ops
+
+    torch.manual_seed(18)
+    T, D, V = 512, 128, 1024
+    h32 = torch.randn(T, D, requires_grad=True)
+    w32 = torch.randn(V, D, requires_grad=True) * 0.05
+    t = torch.randint(0, V, (T,))
+    ref = torch.nn.functional.cross_entropy(h32 @ w32.t(), t)
+    ref.backward()
+
+    # chunk smaller than T so the loop is exercised
+    old = ops._LmHeadCE.CHUNK
+    ops._LmHeadCE.CHUNK = 128
+    try:
+        h = _bf(h32.detach()).cuda().requires_grad_(True)
+        w = _bf(w32.detach()).cuda().requires_grad_(True)
+        loss = ops.lm_head_ce(h, w, t.cuda())
+        loss.backward()
+        torch.cuda.synchronize()
+    finally:
+        ops._LmHeadCE.CHUNK = old
+    assert abs(loss.item() - ref.item()) < 0.03 * max(1.0, abs(ref.item()))
+    for name, g, ge in [("dh", h.grad, h32.grad), ("dw", w.grad, w32.grad)]:
+        a, e = g.cpu().float(), ge.float()
+        err = (a - e).abs().max().item()
+        tol = 5e-2 * max(0.05, e.abs().max().item())
+        assert err < tol, f"{name}: {err} vs {tol}"
