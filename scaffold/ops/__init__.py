@@ -387,8 +387,10 @@ class _LmHeadCE(torch.autograd.Function):
     CHUNK = 2048
 
     @staticmethod
-    def forward(ctx, h2, w, targets):
+    def forward(ctx, x, w, targets):
         m = require_ext()
+        h2 = x.contiguous().view(-1, x.shape[-1])
+        ctx.x_shape = x.shape
         T = h2.shape[0]
         dh = torch.empty_like(h2)
         dw32 = torch.zeros_like(w, dtype=torch.float32)
@@ -410,7 +412,7 @@ class _LmHeadCE(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dloss):
-        dh = ctx.saved_dh * dloss
+        dh = (ctx.saved_dh * dloss).view(ctx.x_shape)
         dw = (ctx.saved_dw32 * dloss.float()).to(ctx.w_dtype)
         ctx.saved_dh = None
         ctx.saved_dw32 = None
@@ -420,11 +422,11 @@ class _LmHeadCE(torch.autograd.Function):
 def lm_head_ce(x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor
                ) -> torch.Tensor:
     """mean CE of (x @ w^T) against targets; x (..., D), targets (...)."""
-    x2 = x.contiguous().view(-1, x.shape[-1])
     t = targets.contiguous().view(-1)
     if x.is_cuda:
-        return _LmHeadCE.apply(x2, w, t)
-    return F.cross_entropy(x2.float() @ w.float().t(), t)
+        return _LmHeadCE.apply(x, w, t)
+    return F.cross_entropy(
+        x.contiguous().view(-1, x.shape[-1]).float() @ w.float().t(), t)
 
 
 # ----------------------------------------------------------------- SwiGLU
