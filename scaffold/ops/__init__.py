@@ -419,12 +419,20 @@ class _LmHeadCE(torch.autograd.Function):
         return dh, dw, None
 
 
+import os as _os_env
+
+_FUSED_CE = _os_env.environ.get("SCAFFOLD_FUSED_CE", "1") != "0"
+
+
 def lm_head_ce(x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor
                ) -> torch.Tensor:
     """mean CE of (x @ w^T) against targets; x (..., D), targets (...)."""
     t = targets.contiguous().view(-1)
-    if x.is_cuda:
+    if x.is_cuda and _FUSED_CE:
         return _LmHeadCE.apply(x, w, t)
+    if x.is_cuda:
+        # unfused: full logits through blaslt + K3 CE kernels
+        return softmax_cross_entropy(linear(x, w), t)
     return F.cross_entropy(
         x.contiguous().view(-1, x.shape[-1]).float() @ w.float().t(), t)
 
