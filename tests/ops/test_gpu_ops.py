@@ -394,13 +394,14 @@ def test_lm_head_ce_matches_eager():
     ref = torch.nn.functional.cross_entropy(h32 @ w32.t(), t)
     ref.backward()
 
-    # chunk smaller than T so the loop is exercised
+    # exercise the CHUNKED Function explicitly (the default dispatch is
+    # the unfused path), with a chunk smaller than T to cover the loop
     old = ops._LmHeadCE.CHUNK
     ops._LmHeadCE.CHUNK = 128
     try:
         h = _bf(h32.detach()).cuda().requires_grad_(True)
         w = _bf(w32.detach()).cuda().requires_grad_(True)
-        loss = ops.lm_head_ce(h, w, t.cuda())
+        loss = ops._LmHeadCE.apply(h, w, t.cuda())
         loss.backward()
         torch.cuda.synchronize()
     finally:
