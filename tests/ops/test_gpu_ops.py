@@ -389,7 +389,7 @@ def test_lm_head_ce_matches_eager():
     torch.manual_seed(18)
     T, D, V = 512, 128, 1024
     h32 = torch.randn(T, D, requires_grad=True)
-    w32 = torch.randn(V, D, requires_grad=True) * 0.05
+    w32 = (torch.randn(V, D) * 0.05).requires_grad_(True)
     t = torch.randint(0, V, (T,))
     ref = torch.nn.functional.cross_entropy(h32 @ w32.t(), t)
     ref.backward()
