@@ -65,8 +65,11 @@ class OptimConfig:
     betas: List[float] = field(default_factory=lambda: [0.9, 0.95])
     eps: float = 1e-8
     weight_decay: float = 0.01
-    grad_clip: float = 0.0  # 0 = off
+    grad_clip: float = 0.0  # global-norm clip; 0 = off
     fused: bool = True  # use the HIP fused multi-tensor kernel on GPU
+    schedule: str = "const"  # const | cosine | linear
+    warmup_steps: int = 0
+    min_lr_ratio: float = 0.1  # floor for cosine/linear decay
 
 
 @dataclass
@@ -94,6 +97,8 @@ class LogConfig:
     interval: int = 10
     metrics_file: str = ""  # rank-0 JSONL; "" = disabled
     profile_steps: int = 0  # wrap N steps in torch.profiler if >0
+    eval_interval: int = 0  # run evaluate() every N steps; 0 = off
+    eval_steps: int = 10
 
 
 @dataclass

@@ -126,6 +126,11 @@ class Trainer:
             )
             if prof is not None:
                 prof.step()
+            if (self.cfg.log.eval_interval
+                    and self.step % self.cfg.log.eval_interval == 0):
+                ev = self.evaluate(self.cfg.log.eval_steps)
+                if self.ctx.is_main:
+                    print(f"step {self.step:6d} | eval loss {ev:8.4f}", flush=True)
             if (
                 self.cfg.ckpt.interval
                 and self.step % self.cfg.ckpt.interval == 0
@@ -152,7 +157,21 @@ class Trainer:
                 loss.backward()
                 self.ddp.finish()
             loss_val += loss.detach().float().item() / accum
-        self.optimizer.step(grad_scale=self.ddp.grad_scale / accum)
+
+        from .schedule import lr_at
+
+        self.optimizer.lr = lr_at(self.cfg, self.step, self.cfg.steps)
+        scale = self.ddp.grad_scale / accum
+        if self.cfg.optim.grad_clip > 0:
+            # global-norm clip FOLDED into the fused step's grad_scale:
+            # norm of the averaged grads = ||flat buckets|| * scale, so
+            # the clip costs a few _foreach_norm launches and ZERO extra
+            # full-gradient passes.
+            norms = torch._foreach_norm([b.flat for b in self.ddp.buckets])
+            gnorm = float(torch.linalg.vector_norm(torch.stack(norms))) * scale
+            if gnorm > self.cfg.optim.grad_clip:
+                scale *= self.cfg.optim.grad_clip / (gnorm + 1e-12)
+        self.optimizer.step(grad_scale=scale)
         return loss_val
 
     # -------------------------------------------------------------- eval
