@@ -93,3 +93,59 @@ def test_ddp_matches_large_batch(tmp_path):
     assert torch.allclose(ref_grads, dist_grads, rtol=1e-5, atol=1e-7), (
         (ref_grads - dist_grads).abs().max().item()
     )
+
+
+def _worker_accum(rank, world, file_store_path, result_q):
+    from scaffold.trainer import Trainer
+
+    store = dist.FileStore(file_store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    cfg = _cfg(batch_size=4)
+    cfg.grad_accum = 2
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
+                      backend="gloo", device=torch.device("cpu"), initialized=True)
+    t = Trainer(cfg, ctx=ctx)
+
+    # 4 deterministic micro-batches = one global batch of 16
+    g = torch.Generator().manual_seed(77)
+    X = torch.randn(world * 2 * 4, 64, generator=g)
+    Y = torch.randn(world * 2 * 4, 8, generator=g)
+    micros = [(X[i * 4:(i + 1) * 4], Y[i * 4:(i + 1) * 4])
+              for i in range(world * 2)]
+    mine = iter([micros[rank], micros[world + rank]])
+
+    t._train_step(mine, accum=2)
+    params = torch.cat([p.detach().flatten() for p in t.raw_model.parameters()])
+    if rank == 0:
+        result_q.put(params.numpy())
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_grad_accum_matches_large_batch(tmp_path):
+    """no_sync micro-steps + one reduced step == single-process batch 16."""
+    from scaffold.trainer import Trainer
+
+    ctxmp = mp.get_context("spawn")
+    q = ctxmp.Queue()
+    procs = [
+        ctxmp.Process(target=_worker_accum, args=(r, WORLD, str(tmp_path / "fs"), q))
+        for r in range(WORLD)
+    ]
+    for p in procs:
+        p.start()
+    dist_params = torch.from_numpy(q.get(timeout=100))
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    cfg = _cfg(batch_size=WORLD * 2 * 4)
+    t = Trainer(cfg)
+    g = torch.Generator().manual_seed(77)
+    X = torch.randn(WORLD * 2 * 4, 64, generator=g)
+    Y = torch.randn(WORLD * 2 * 4, 8, generator=g)
+    t._train_step(iter([(X, Y)]), accum=1)
+    ref_params = torch.cat([p.detach().flatten() for p in t.raw_model.parameters()])
+    assert torch.allclose(ref_params, dist_params, rtol=1e-5, atol=1e-6), (
+        (ref_params - dist_params).abs().max().item()
+    )
