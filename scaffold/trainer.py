@@ -15,6 +15,7 @@ from typing import Optional
 import torch
 
 from .config import TrainConfig, config_hash
+from .schedule import lr_at
 from .data.synthetic import build_data
 from .dist.comm import Comm
 from .dist.ddp import BucketedDDP
@@ -157,8 +158,6 @@ class Trainer:
                 loss.backward()
                 self.ddp.finish()
             loss_val += loss.detach().float().item() / accum
-
-        from .schedule import lr_at
 
         self.optimizer.lr = lr_at(self.cfg, self.step, self.cfg.steps)
         scale = self.ddp.grad_scale / accum
