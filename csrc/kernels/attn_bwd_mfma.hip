@@ -178,8 +178,6 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   __syncthreads();
   int cur = 0;
   for (int c0 = 0; c0 < kv_end; c0 += 32) {
-    if (c0 + 32 < kv_end)
-      stage_tr<D>(k + kbb, ks.ss, c0 + 32, S, ktile + (cur ^ 1) * 32 * D, wid, l);
     const unsigned ktr = ktr0 + (unsigned)cur * (32 * D * 2);
 
     const int krow = min(c0 + lq, S - 1);
@@ -191,6 +189,11 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
       sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
       dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[s], dpacc, 0, 0, 0);
     }
+
+    // next tile's glds issued AFTER the block with ordinary global
+    // loads (kf/vf rows) — see the fwd kernel's trap-(b) note
+    if (c0 + 32 < kv_end)
+      stage_tr<D>(k + kbb, ks.ss, c0 + 32, S, ktile + (cur ^ 1) * 32 * D, wid, l);
 
     const bool need_mask = (causal && c0 + 31 > qrow_g) || (c0 + 31 >= S);
     float ds[16];
@@ -263,9 +266,13 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          Strides4 vs, Strides4 dks, Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
-  // [Q tile | dO tile] — single-buffered: the double-buffer variant
-  // pushed D=128 to 148 B/lane of spill and measured net-slower
-  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 32 * D];
+  // [buf][Q tile | dO tile]; double-buffered at D<=64 (stage of tile
+  // t+1 issued after the global-load block, drained under the dV/dK
+  // MFMAs); single-buffered at D=128 where the extra live state
+  // spilled (148 B/lane) and measured net-slower
+  constexpr bool KVREG_ = (D <= 64);
+  constexpr int NBUF = KVREG_ ? 2 : 1;
+  __shared__ __attribute__((aligned(16))) short bwd_lds[NBUF * 2 * 32 * D];
 
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
@@ -312,12 +319,23 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
     const long dob = (long)b * dos.sb + (long)h * dos.sh;
     const long rhead = ((long)b * H + h) * S;
     const int qstart = causal ? k0 : 0;
-    for (int c0 = qstart; c0 < S; c0 += 32) {
-      stage_tr<D>(q + qb, qs.ss, c0, S, bwd_lds, wid, l);
-      stage_tr<D>(dout + dob, dos.ss, c0, S, bwd_lds + 32 * D, wid, l);
+    auto stage_qdo = [&](int buf, int c0s) {
+      short* base = bwd_lds + buf * (2 * 32 * D);
+      stage_tr<D>(q + qb, qs.ss, c0s, S, base, wid, l);
+      stage_tr<D>(dout + dob, dos.ss, c0s, S, base + 32 * D, wid, l);
+    };
+    if (KVREG_) {
+      stage_qdo(0, qstart);
       __syncthreads();
-      const unsigned qtr = qtr0;
-      const unsigned dotr = dotr0;
+    }
+    int cur = 0;
+    for (int c0 = qstart; c0 < S; c0 += 32) {
+      if (!KVREG_) {
+        stage_qdo(0, c0);
+        __syncthreads();
+      }
+      const unsigned qtr = qtr0 + (unsigned)cur * (2 * 32 * D * 2);
+      const unsigned dotr = dotr0 + (unsigned)cur * (2 * 32 * D * 2);
 
       // S = Q_tile . K^T and dP = dO . V^T: at D<=64 (fragments pinned
       // in registers) the two independent accumulator chains interleave
@@ -350,6 +368,10 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         }
       }
       __builtin_amdgcn_s_setprio(0);
+
+      // D<=64: next q-tile's glds issued after the global-load block
+      // (trap-b note in the fwd kernel); drained under the dV/dK MFMAs
+      if (KVREG_ && c0 + 32 < S) stage_qdo(cur ^ 1, c0 + 32);
 
       // per-tile q-row constants loaded ONCE per wave (lane j holds row
       // c0+j's values) and broadcast per-reg via shfl — replaces 32
@@ -401,7 +423,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
             *(const bf16x8_t*)dsf[1], cq.f[3], dkacc[t], 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
-      __syncthreads();  // all waves done with the tiles before restage
+      __syncthreads();  // tiles consumed; (KVREG_) next tile's glds drained
+      if (KVREG_) cur ^= 1;
     }
   }
 

@@ -132,7 +132,6 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   __syncthreads();
   int cur = 0;
   for (int c0 = 0; c0 < kv_end; c0 += 64) {
-    if (c0 + 64 < kv_end) stage(cur ^ 1, c0 + 64);
     const short* kimg = kv_lds + cur * BUF;
     const unsigned tr_base = tr_base0 + (unsigned)cur * (BUF * 2);
 
@@ -154,6 +153,14 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
       }
     }
     __builtin_amdgcn_s_setprio(0);
+
+    // issue the NEXT tile's glds AFTER the QK^T block: any ordinary
+    // VGPR-destination load inside that block (the D=128 per-tile Q
+    // reload) makes hipcc wait vmcnt(0) — which would drain a prefetch
+    // issued before it with zero compute cover (guide §5 trap b).
+    // Issued here, the drain at the end-of-tile barrier is covered by
+    // softmax + pack + PV.
+    if (c0 + 64 < kv_end) stage(cur ^ 1, c0 + 64);
 
     // ---- scale + mask IN PLACE in sacc (register reuse: no separate
     // p[] array), then exponentiate in place
