@@ -334,8 +334,10 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         stage_qdo(0, c0);
         __syncthreads();
       }
-      const unsigned qtr = qtr0 + (unsigned)cur * (2 * 32 * D * 2);
-      const unsigned dotr = dotr0 + (unsigned)cur * (2 * 32 * D * 2);
+      const unsigned bufoff =
+          KVREG_ ? (unsigned)cur * (2 * 32 * D * 2) : 0u;  // folds at D=128
+      const unsigned qtr = qtr0 + bufoff;
+      const unsigned dotr = dotr0 + bufoff;
 
       // S = Q_tile . K^T and dP = dO . V^T: at D<=64 (fragments pinned
       // in registers) the two independent accumulator chains interleave
@@ -375,9 +377,14 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
 
       // per-tile q-row constants loaded ONCE per wave (lane j holds row
       // c0+j's values) and broadcast per-reg via shfl — replaces 32
-      // scalar L1 loads per lane per tile with 2 loads + shuffles
-      const float lse_lane = lse[rhead + qrow];
-      const float del_lane = delta[rhead + qrow];
+      // scalar L1 loads per lane per tile with 2 loads + shuffles.
+      // D=128 keeps the L1 loads: the two extra live scalars tipped the
+      // register budget into deeper spill there.
+      float lse_lane = 0.f, del_lane = 0.f;
+      if constexpr (KVREG) {
+        lse_lane = lse[rhead + qrow];
+        del_lane = delta[rhead + qrow];
+      }
 
       // p overwrites sacc in place (register reuse)
       #pragma unroll
@@ -386,14 +393,16 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         const int qg = c0 + qrel;
         float sv = sacc[r] * scale;
         const bool masked = qg >= S || (causal && krow_g > qg);
-        const float lse_r = __shfl(lse_lane, qrel, 64);
+        const float lse_r = KVREG ? __shfl(lse_lane, qrel, 64)
+                                  : lse[rhead + min(qg, S - 1)];
         sacc[r] = masked ? 0.f : __expf(sv - lse_r);
       }
       // ds overwrites dpacc in place
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
-        const float del_r = __shfl(del_lane, qrel, 64);
+        const float del_r = KVREG ? __shfl(del_lane, qrel, 64)
+                                  : delta[rhead + min(c0 + qrel, S - 1)];
         dpacc[r] = sacc[r] * (dpacc[r] - del_r) * scale;
       }
 
