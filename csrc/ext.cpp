@@ -3,6 +3,16 @@
 
 #include <vector>
 
+torch::Tensor adamw_build_meta(std::vector<torch::Tensor> masters,
+                               std::vector<torch::Tensor> grads,
+                               std::vector<torch::Tensor> exp_avgs,
+                               std::vector<torch::Tensor> exp_avg_sqs,
+                               std::vector<torch::Tensor> outs);
+void fused_adamw_run(torch::Tensor meta_dev, long n, long total_vec4,
+                     double lr, double beta1, double beta2, double eps,
+                     double weight_decay, long step, double grad_scale,
+                     bool grad_bf16, bool has_out,
+                     c10::optional<torch::Tensor> step_dev);
 void fused_adamw(std::vector<torch::Tensor> masters,
                  std::vector<torch::Tensor> grads,
                  std::vector<torch::Tensor> exp_avgs,
@@ -56,6 +66,10 @@ std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (K1)");
+  m.def("adamw_build_meta", &adamw_build_meta,
+        "one-time device meta table for the cached AdamW path");
+  m.def("fused_adamw_run", &fused_adamw_run,
+        "AdamW step on a prebuilt meta (graph-capturable; optional device step)");
   m.def("layernorm_fwd", &layernorm_fwd, "LayerNorm forward (K2)");
   m.def("layernorm_bwd", &layernorm_bwd, "LayerNorm backward (K2)");
   m.def("softmax_xent_fwd", &softmax_xent_fwd, "softmax cross-entropy fwd (K3)");

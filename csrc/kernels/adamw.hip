@@ -21,10 +21,20 @@ constexpr int MAX_TENSORS = 4096;
 // pointer blocks: master[n], grad[n], m[n], v[n], out[n], numel[n]
 __global__ void __launch_bounds__(256)
 adamw_mt_kernel(const long* __restrict__ meta, int n, long total_vec4,
-                AdamwArgs a, int grad_bf16, int has_out) {
+                AdamwArgs a, int grad_bf16, int has_out,
+                const int* __restrict__ step_dev) {
   __shared__ long s_starts[MAX_TENSORS + 1];
   for (int i = threadIdx.x; i <= n; i += blockDim.x) s_starts[i] = meta[i];
   __syncthreads();
+
+  if (step_dev != nullptr) {
+    // hipGraph-replay path: the step lives on DEVICE (incremented by a
+    // captured op), so bias corrections stay exact under replay where
+    // host-baked args would freeze at the capture step.
+    const float s = (float)*step_dev;
+    a.bc1_inv = 1.0f / (1.0f - __powf(a.beta1, s));
+    a.bc2_inv = 1.0f / (1.0f - __powf(a.beta2, s));
+  }
 
   const long* p_master = meta + (n + 1);
   const long* p_grad = p_master + n;
@@ -104,18 +114,18 @@ adamw_mt_kernel(const long* __restrict__ meta, int n, long total_vec4,
 
 }  // namespace
 
-void fused_adamw(std::vector<torch::Tensor> masters,
-                 std::vector<torch::Tensor> grads,
-                 std::vector<torch::Tensor> exp_avgs,
-                 std::vector<torch::Tensor> exp_avg_sqs,
-                 std::vector<torch::Tensor> outs,  // bf16 working copies or empty
-                 double lr, double beta1, double beta2, double eps,
-                 double weight_decay, long step, double grad_scale) {
+// Build the device meta table ONCE (pointers are stable across steps:
+// grads are bucket views, masters/moments never reallocate) so the hot
+// step — and a hipGraph capture of it — carries no H2D copy.
+torch::Tensor adamw_build_meta(std::vector<torch::Tensor> masters,
+                               std::vector<torch::Tensor> grads,
+                               std::vector<torch::Tensor> exp_avgs,
+                               std::vector<torch::Tensor> exp_avg_sqs,
+                               std::vector<torch::Tensor> outs) {
   const int n = (int)masters.size();
   TORCH_CHECK(n > 0 && n <= MAX_TENSORS, "fused_adamw: 1..4096 tensors, got ", n);
   const bool has_out = !outs.empty();
   TORCH_CHECK(!has_out || (int)outs.size() == n);
-  const bool grad_bf16 = grads[0].scalar_type() == torch::kBFloat16;
 
   std::vector<long> meta;
   meta.reserve((size_t)n * 7 + 1);
@@ -138,8 +148,14 @@ void fused_adamw(std::vector<torch::Tensor> masters,
 
   auto opts = torch::TensorOptions().dtype(torch::kInt64);
   auto meta_cpu = torch::from_blob(meta.data(), {(long)meta.size()}, opts).clone();
-  auto meta_dev = meta_cpu.to(masters[0].device(), /*non_blocking=*/true);
+  return meta_cpu.to(masters[0].device());
+}
 
+void fused_adamw_run(torch::Tensor meta_dev, long n, long total_vec4,
+                     double lr, double beta1, double beta2, double eps,
+                     double weight_decay, long step, double grad_scale,
+                     bool grad_bf16, bool has_out,
+                     c10::optional<torch::Tensor> step_dev) {
   AdamwArgs a;
   a.lr = (float)lr;
   a.beta1 = (float)beta1;
@@ -150,11 +166,33 @@ void fused_adamw(std::vector<torch::Tensor> masters,
   a.bc2_inv = (float)(1.0 / (1.0 - std::pow(beta2, (double)step)));
   a.grad_scale = (float)grad_scale;
 
+  const int* sd = nullptr;
+  if (step_dev.has_value()) {
+    TORCH_CHECK(step_dev->scalar_type() == torch::kInt32 && step_dev->is_cuda());
+    sd = step_dev->data_ptr<int>();
+  }
   const int block = 256;
-  const int grid = capped_blocks(cum, block, 2048);
+  const int grid = capped_blocks(total_vec4, block, 2048);
   auto stream = scaffold_stream();
   hipLaunchKernelGGL(adamw_mt_kernel, dim3(grid), dim3(block), 0, stream,
-                     meta_dev.data_ptr<long>(), n, cum, a,
-                     grad_bf16 ? 1 : 0, has_out ? 1 : 0);
+                     meta_dev.data_ptr<long>(), (int)n, total_vec4, a,
+                     grad_bf16 ? 1 : 0, has_out ? 1 : 0, sd);
   HIP_CHECK(hipGetLastError());
+}
+
+void fused_adamw(std::vector<torch::Tensor> masters,
+                 std::vector<torch::Tensor> grads,
+                 std::vector<torch::Tensor> exp_avgs,
+                 std::vector<torch::Tensor> exp_avg_sqs,
+                 std::vector<torch::Tensor> outs,  // bf16 working copies or empty
+                 double lr, double beta1, double beta2, double eps,
+                 double weight_decay, long step, double grad_scale) {
+  const int n = (int)masters.size();
+  TORCH_CHECK(n > 0, "fused_adamw: empty tensor list");
+  long cum = 0;
+  for (int i = 0; i < n; ++i) cum += (masters[i].numel() + 3) / 4;
+  auto meta_dev = adamw_build_meta(masters, grads, exp_avgs, exp_avg_sqs, outs);
+  fused_adamw_run(meta_dev, n, cum, lr, beta1, beta2, eps, weight_decay, step,
+                  grad_scale, grads[0].scalar_type() == torch::kBFloat16,
+                  !outs.empty(), c10::nullopt);
 }

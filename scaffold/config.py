@@ -112,6 +112,7 @@ class TrainConfig:
     steps: int = 100
     grad_accum: int = 1
     cudnn_benchmark: bool = True  # MIOpen exhaustive find (cached in miopen_db)
+    hip_graph: bool = False  # capture the train step in a hipGraph and replay
     seed: int = 0
     dtype: str = "bf16"  # compute dtype: bf16 | fp32
     device: str = "auto"  # auto | cuda | cpu

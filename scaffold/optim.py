@@ -34,6 +34,8 @@ class MasterAdamW:
         self.weight_decay = weight_decay
         self.fused = fused
         self.step_count = 0
+        self._cached = None  # prebuilt (meta_dev, n, total, bf16, out) per group
+        self.step_dev = None  # optional device step (hipGraph replay path)
         # fp32 masters + moments.  fp32 working params ALIAS their master
         # (no copy-back pass); low-precision params get a separate master.
         self.master: List[torch.Tensor] = [
@@ -52,8 +54,7 @@ class MasterAdamW:
                     p.grad.zero_()
 
     @torch.no_grad()
-    def step(self, grad_scale: float = 1.0) -> None:
-        self.step_count += 1
+    def _build_groups(self):
         # one fused launch per homogeneous group: {bf16 working + bf16-out}
         # and {fp32 working aliasing master, no out}
         groups = {}  # (grad_dtype, needs_out) -> lists
@@ -69,7 +70,34 @@ class MasterAdamW:
             g[3].append(self.exp_avg_sq[i])
             if needs_out:
                 g[4].append(p.data)
-        for (gdt, needs_out), (masters, grads, eas, eass, outs) in groups.items():
+        return groups
+
+    def step(self, grad_scale: float = 1.0) -> None:
+        self.step_count += 1
+        on_gpu = self.params[0].is_cuda and ops.has_ext()
+        if on_gpu:
+            if self._cached is None:
+                # tensor pointers are stable across steps (grads are
+                # DDP bucket views; masters/moments never reallocate):
+                # build the device meta table ONCE — the per-step call
+                # then carries no H2D copy and is hipGraph-capturable
+                m = ops.require_ext()
+                cached = []
+                for (gdt, needs_out), (ms, gs, eas, eass, outs) in self._build_groups().items():
+                    meta = m.adamw_build_meta(ms, gs, eas, eass, outs if needs_out else [])
+                    total = sum((t.numel() + 3) // 4 for t in ms)
+                    cached.append((meta, len(ms), total,
+                                   gdt == torch.bfloat16, needs_out))
+                self._cached = cached
+            m = ops.require_ext()
+            for meta, n, total, gbf16, has_out in self._cached:
+                m.fused_adamw_run(
+                    meta, n, total, self.lr, self.beta1, self.beta2, self.eps,
+                    self.weight_decay, self.step_count, grad_scale,
+                    gbf16, has_out, self.step_dev,
+                )
+            return
+        for (gdt, needs_out), (masters, grads, eas, eass, outs) in self._build_groups().items():
             ops.fused_adamw_step(
                 masters, grads, eas, eass, outs if needs_out else None,
                 lr=self.lr, beta1=self.beta1, beta2=self.beta2, eps=self.eps,
