@@ -63,8 +63,14 @@ def main() -> int:
     batch_iter = iter(t.data)
     accum = max(1, cfg.grad_accum)
 
-    def one_step():
-        t._train_step(batch_iter, accum)
+    if t._graph_ok():
+        t._capture_step(batch_iter)
+
+        def one_step():
+            t._train_step_graph()
+    else:
+        def one_step():
+            t._train_step(batch_iter, accum)
 
     for _ in range(args.warmup):
         one_step()
