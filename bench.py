@@ -63,9 +63,15 @@ def main() -> int:
     batch_iter = iter(t.data)
     accum = max(1, cfg.grad_accum)
 
+    use_graph = False
     if t._graph_ok():
-        t._capture_step(batch_iter)
+        try:
+            t._capture_step(batch_iter)
+            use_graph = True
+        except Exception as e:  # noqa: BLE001 — e.g. non-capturable RNG op
+            print(f"[bench] hipGraph capture failed ({e}); eager", flush=True)
 
+    if use_graph:
         def one_step():
             t._train_step_graph()
     else:
