@@ -178,6 +178,7 @@ def test_softmax_xent(ext, R, V):
     (2, 4, 4, 128, 64, True),
     (1, 8, 2, 256, 128, True),   # GQA
     (2, 2, 2, 96, 64, False),    # non-causal + ragged S
+    (1, 4, 2, 200, 128, True),   # causal + ragged S + GQA at D=128
 ])
 def test_attention_fwd_bwd(ext, B, H, Hkv, S, D, causal):
     torch.manual_seed(7)
