@@ -46,6 +46,7 @@ class GradBucket:
     pending: int = 0
     work: object = None
     launched: bool = False
+    ready: bool = False
 
     @property
     def numel(self) -> int:
@@ -140,6 +141,8 @@ class BucketedDDP(nn.Module):
             b.pending = len(group)
             self.buckets.append(b)
 
+        self._next_launch = 0  # index-ordered collective issue cursor
+
         # 3) readiness hooks
         self._hooks = [
             p.register_post_accumulate_grad_hook(self._make_hook(p))
@@ -154,9 +157,26 @@ class BucketedDDP(nn.Module):
             b = self._param_bucket[id(p)]
             b.pending -= 1
             if b.pending == 0 and self.overlap:
-                self._launch(b)
+                b.ready = True
+                self._launch_ready_in_order()
 
         return hook
+
+    def _launch_ready_in_order(self) -> None:
+        """Launch ready buckets strictly by INDEX.  Collectives on one
+        communicator must be issued in the same order on every rank or
+        NCCL-class backends can mismatch/deadlock; autograd readiness
+        order is not guaranteed identical across ranks, so a bucket
+        that becomes ready early waits for its predecessors (same rule
+        as torch DDP's reducer).  Buckets are partitioned in reverse
+        parameter order, so index order ~= expected readiness order and
+        the wait is normally empty."""
+        while self._next_launch < len(self.buckets):
+            b = self.buckets[self._next_launch]
+            if not b.ready:
+                return
+            self._launch(b)
+            self._next_launch += 1
 
     def _launch(self, b: GradBucket) -> None:
         if b.launched:
@@ -195,6 +215,8 @@ class BucketedDDP(nn.Module):
             b.pending = len(b.params)
             b.work = None
             b.launched = False
+            b.ready = False
+        self._next_launch = 0
 
     def zero_grad(self) -> None:
         """Zero the flat buffers (cheap contiguous memset per bucket)."""

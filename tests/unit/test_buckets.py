@@ -51,3 +51,34 @@ def test_mixed_dtype_buckets():
         assert len(dts) == 1, "bucket mixes dtypes"
     ids = sorted(id(p) for b in buckets for p in b)
     assert ids == sorted(id(p) for p in ps)
+
+
+def test_launch_order_is_index_order_despite_ragged_readiness():
+    """Collectives must issue in bucket-index order on every rank even
+    when autograd readiness arrives out of order (NCCL-class backends
+    deadlock on cross-rank order mismatch)."""
+    import torch.nn as nn
+
+    from scaffold.dist.ddp import BucketedDDP
+    from scaffold.dist.fake_pg import RecordingComm
+
+    model = nn.Sequential(*[nn.Linear(16, 16) for _ in range(6)])
+    comm = RecordingComm(world_size=2)
+    ddp = BucketedDDP(model, comm=comm, bucket_cap_mb=0.002, first_bucket_mb=0.001)
+    assert len(ddp.buckets) >= 3
+
+    # adversarial: mark the LAST bucket's params ready first
+    order = list(reversed(range(len(ddp.buckets))))
+    for bi in order:
+        b = ddp.buckets[bi]
+        for p in b.params:
+            # simulate the post-accumulate hook firing
+            b.pending -= 1
+            if b.pending == 0:
+                b.ready = True
+                ddp._launch_ready_in_order()
+    ddp.finish()
+    reduces = [r for r in comm.records if r.op == "all_reduce"]
+    sizes = [r.numel for r in reduces]
+    expected = [b.flat.numel() for b in ddp.buckets]
+    assert sizes == expected, (sizes, expected)
