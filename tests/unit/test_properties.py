@@ -1,0 +1,91 @@
+"""Property-based unit tier (hypothesis — SURVEY §4)."""
+
+import hypothesis.strategies as st
+import torch
+from hypothesis import given, settings
+
+from scaffold.config import TrainConfig, config_from_dict, config_to_dict, load_config
+from scaffold.dist.ddp import partition_buckets
+from scaffold.schedule import lr_at
+
+
+class _P:
+    """Stand-in parameter: numel/element_size/dtype like nn.Parameter."""
+
+    def __init__(self, numel, dtype):
+        self._n = numel
+        self.dtype = dtype
+
+    def numel(self):
+        return self._n
+
+    def element_size(self):
+        return 2 if self.dtype == torch.bfloat16 else 4
+
+
+@settings(max_examples=200, deadline=None)
+@given(
+    sizes=st.lists(st.integers(1, 5_000_000), min_size=1, max_size=64),
+    dtypes=st.lists(st.sampled_from([torch.bfloat16, torch.float32]),
+                    min_size=1, max_size=64),
+    cap=st.floats(0.01, 64),
+    first=st.floats(0.001, 8),
+)
+def test_partition_covers_every_param_once(sizes, dtypes, cap, first):
+    params = [_P(n, dtypes[i % len(dtypes)]) for i, n in enumerate(sizes)]
+    buckets = partition_buckets(params, cap, first)
+    flat = [p for b in buckets for p in b]
+    assert sorted(map(id, flat)) == sorted(map(id, params))  # exact cover
+    for b in buckets:
+        assert len({p.dtype for p in b}) == 1  # dtype-homogeneous
+        # no bucket except possibly singletons exceeds the cap
+        nbytes = sum(p.numel() * p.element_size() for p in b)
+        if len(b) > 1:
+            assert nbytes <= max(cap, first) * (1 << 20) + max(
+                p.numel() * p.element_size() for p in b)
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    lr=st.floats(1e-6, 1.0),
+    warmup=st.integers(0, 50),
+    total=st.integers(51, 500),
+    ratio=st.floats(0.0, 1.0),
+    sched=st.sampled_from(["const", "cosine", "linear"]),
+)
+def test_schedule_bounds(lr, warmup, total, ratio, sched):
+    cfg = load_config(None, [
+        f"optim.lr={lr}", f"optim.warmup_steps={warmup}",
+        f"optim.schedule={sched}", f"optim.min_lr_ratio={ratio}",
+        f"steps={total}",
+    ])
+    floor = 0.0 if sched == "const" else lr * ratio
+    for s in range(0, total, max(1, total // 37)):
+        v = lr_at(cfg, s, total)
+        assert 0.0 <= v <= lr * (1 + 1e-9)
+        if s >= warmup:
+            assert v >= min(floor, lr) - 1e-12
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    n_layers=st.integers(1, 48),
+    bsz=st.integers(1, 512),
+    steps=st.integers(1, 10000),
+    clip=st.floats(0, 10),
+)
+def test_config_roundtrip(n_layers, bsz, steps, clip):
+    cfg = config_from_dict({
+        "model": {"name": "gpt2", "n_layers": n_layers},
+        "data": {"batch_size": bsz},
+        "optim": {"grad_clip": clip},
+        "steps": steps,
+    })
+    d = config_to_dict(cfg)
+    cfg2 = config_from_dict({k: v for k, v in d.items()
+                             if k in ("model", "data", "optim", "dist",
+                                      "ckpt", "log") or not isinstance(v, dict)})
+    assert cfg2.model.n_layers == n_layers
+    assert cfg2.data.batch_size == bsz
+    assert cfg2.steps == steps
+    assert abs(cfg2.optim.grad_clip - clip) < 1e-12
