@@ -60,3 +60,19 @@ def test_grad_accum_equivalence(cpu_cfg):
     for a, b in zip(ga, gb):
         # mse_loss mean over batch: half-batch grads average to full-batch grad
         assert torch.allclose(a, b / 2, rtol=1e-4, atol=1e-5)
+
+
+def test_evaluate_and_eval_cadence(cpu_cfg, capsys):
+    import copy
+
+    cfg = copy.deepcopy(cpu_cfg)
+    cfg.steps = 4
+    cfg.log.eval_interval = 2
+    cfg.log.eval_steps = 3
+    t = Trainer(cfg)
+    t.fit()
+    ev = t.evaluate(eval_steps=3)
+    assert torch.isfinite(torch.tensor(ev))
+    assert t.raw_model.training  # evaluate() restores train mode
+    out = capsys.readouterr().out
+    assert "eval loss" in out
