@@ -65,3 +65,31 @@ def test_hash_changes_with_config():
     b = TrainConfig()
     b.optim.lr = 123.0
     assert config_hash(a) != config_hash(b)
+
+
+def test_bool_and_float_coercion():
+    cfg = load_config(None, [
+        "dist.overlap=false", "ckpt.async_d2h=0", "hip_graph=true",
+        "optim.lr=2.5e-4", "dist.bucket_cap_mb=12.5",
+    ])
+    assert cfg.dist.overlap is False
+    assert cfg.ckpt.async_d2h is False
+    assert cfg.hip_graph is True
+    assert abs(cfg.optim.lr - 2.5e-4) < 1e-12
+    assert cfg.dist.bucket_cap_mb == 12.5
+
+
+def test_latest_checkpoint_picks_newest_complete(tmp_path):
+    import json
+    import os
+
+    from scaffold.ckpt import layout
+
+    for step, complete in [(2, True), (10, True), (30, False)]:
+        d = tmp_path / f"step_{step:08d}"
+        d.mkdir()
+        if complete:
+            (d / "manifest.json").write_text(json.dumps({"step": step}))
+    # newest COMPLETE is step 10 (step 30 lacks a manifest = mid-crash)
+    got = layout.latest_checkpoint(str(tmp_path))
+    assert got.endswith("step_00000010"), got
