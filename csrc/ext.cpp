@@ -40,7 +40,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    bool causal);
+                                    bool causal,
+                                    c10::optional<torch::Tensor> dlse);
 torch::Tensor probe_mfma(torch::Tensor A, torch::Tensor B, long which);
 torch::Tensor probe_tr(torch::Tensor offsets);
 torch::Tensor rccl_unique_id();
@@ -78,7 +79,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward (K7)");
   m.def("rope_apply", &rope_apply, "RoPE apply/inverse (K7)");
   m.def("attn_fwd", &attn_fwd, "flash attention forward (K4)");
-  m.def("attn_bwd", &attn_bwd, "flash attention backward (K4)");
+  m.def("attn_bwd", &attn_bwd, "flash attention backward (K4)",
+        py::arg("dout"), py::arg("q"), py::arg("k"), py::arg("v"),
+        py::arg("o"), py::arg("lse"), py::arg("causal"),
+        py::arg("dlse") = py::none());
   m.def("probe_mfma", &probe_mfma, "MFMA fragment-layout probe (tests)");
   m.def("probe_tr", &probe_tr, "ds_read_b64_tr_b16 semantics probe (tests)");
   m.def("rccl_unique_id", &rccl_unique_id, "RCCL bootstrap id (rank 0)");

@@ -25,6 +25,7 @@ constexpr int QC = 64;
 template <typename T, int D>
 __global__ void __launch_bounds__(256)
 attn_delta_kernel(const T* __restrict__ dout, const T* __restrict__ o,
+                  const float* __restrict__ dlse,
                   float* __restrict__ delta, long rows) {
   // one wave per row: rowsum(do * o) over D
   const long row = (long)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
@@ -35,7 +36,7 @@ attn_delta_kernel(const T* __restrict__ dout, const T* __restrict__ o,
   for (int d = lane; d < D; d += 64)
     acc += load_f(dout, base + d) * load_f(o, base + d);
   acc = wave_reduce_sum(acc);
-  if (lane == 0) delta[row] = acc;
+  if (lane == 0) delta[row] = acc - (dlse ? dlse[row] : 0.f);
 }
 
 template <typename T, int D>
@@ -218,7 +219,8 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q,
                 const torch::Tensor& o, const torch::Tensor& lse,
                 torch::Tensor& dq, torch::Tensor& dk, torch::Tensor& dv,
                 torch::Tensor& delta, int B, int H, int Hkv, int S, int D,
-                float scale, bool causal, hipStream_t stream) {
+                float scale, bool causal, hipStream_t stream,
+                const float* dlse_ptr) {
   const long rows = (long)B * H * S;
   dim3 dblock(256), dgrid((unsigned)((rows + 3) / 4));
   dim3 block(256);
@@ -228,7 +230,7 @@ void launch_bwd(const torch::Tensor& dout, const torch::Tensor& q,
     do {                                                                        \
       hipLaunchKernelGGL((attn_delta_kernel<T, DD>), dgrid, dblock, 0, stream,  \
                          (const T*)dout.data_ptr(), (const T*)o.data_ptr(),     \
-                         delta.data_ptr<float>(), rows);                        \
+                         dlse_ptr, delta.data_ptr<float>(), rows);                        \
       hipLaunchKernelGGL((attn_bwd_dq_kernel<T, DD>), grid_q, block, 0, stream, \
                          (const T*)dout.data_ptr(), (const T*)q.data_ptr(),     \
                          (const T*)k.data_ptr(), (const T*)v.data_ptr(),        \
@@ -255,12 +257,21 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                    torch::Tensor v, torch::Tensor o, torch::Tensor lse,
                    torch::Tensor delta, torch::Tensor dq, torch::Tensor dk,
                    torch::Tensor dv, int B, int H, int Hkv, int S, int D,
-                   double scale, bool causal);
+                   double scale, bool causal, const float* dlse_ptr);
 
 std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
                                     torch::Tensor k, torch::Tensor v,
                                     torch::Tensor o, torch::Tensor lse,
-                                    bool causal) {
+                                    bool causal,
+                                    c10::optional<torch::Tensor> dlse) {
+  const float* dlse_ptr = nullptr;
+  torch::Tensor dlse_c;
+  if (dlse.has_value()) {
+    dlse_c = dlse->contiguous();
+    TORCH_CHECK(dlse_c.scalar_type() == torch::kFloat32 &&
+                dlse_c.numel() == lse.numel(), "attn_bwd: bad dlse");
+    dlse_ptr = dlse_c.data_ptr<float>();
+  }
   CHECK_CUDA(dout);
   const int B = (int)q.size(0), H = (int)q.size(1), S = (int)q.size(2),
             D = (int)q.size(3);
@@ -281,7 +292,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
     auto dk = torch::empty({B, S, Hkv, D}, q.options()).permute({0, 2, 1, 3});
     auto dv = torch::empty({B, S, Hkv, D}, q.options()).permute({0, 2, 1, 3});
     attn_bwd_mfma(doc, qc, kc, vc, oc, lse, delta, dq, dk, dv, B, H, Hkv, S, D,
-                  scale, causal);
+                  scale, causal, dlse_ptr);
     HIP_CHECK(hipGetLastError());
     return {dq, dk, dv};
   }
@@ -295,10 +306,10 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor dout, torch::Tensor q,
   auto dv = torch::empty_like(vc);
   if (q.scalar_type() == torch::kBFloat16)
     launch_bwd<short>(doc, qc, kc, vc, oc, lse, dq, dk, dv, delta, B, H, Hkv, S, D,
-                      scale, causal, stream);
+                      scale, causal, stream, dlse_ptr);
   else
     launch_bwd<float>(doc, qc, kc, vc, oc, lse, dq, dk, dv, delta, B, H, Hkv, S, D,
-                      scale, causal, stream);
+                      scale, causal, stream, dlse_ptr);
   HIP_CHECK(hipGetLastError());
   return {dq, dk, dv};
 }

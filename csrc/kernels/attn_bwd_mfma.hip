@@ -88,11 +88,16 @@ DEVINL unsigned tr_col_base(const void* img, int l) {
          + ((l >> 5) ? 128u * 2 * (D / 16) : 0);
 }
 
-// delta = rowsum(dO * O), stride-aware (one wave per row)
+// delta = rowsum(dO * O) - dlse (one wave per row, stride-aware).
+// dlse (may be null) is the gradient of the attention's logsumexp
+// output — context-parallel partial-attention merges differentiate
+// through lse (scaffold/dist/cp.py); folding it here makes the dq/dkv
+// kernels' (dP - delta) term exact with zero extra passes.
 template <int D>
 __global__ void __launch_bounds__(256)
 attn_delta_strided_kernel(const short* __restrict__ dout,
                           const short* __restrict__ o,
+                          const float* __restrict__ dlse,
                           float* __restrict__ delta, int H, int S, long rows,
                           Strides4 dos, Strides4 osr) {
   const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
@@ -120,7 +125,7 @@ attn_delta_strided_kernel(const short* __restrict__ dout,
     }
   }
   acc = wave_reduce_sum(acc);
-  if (lane == 0) delta[row] = acc;
+  if (lane == 0) delta[row] = acc - (dlse ? dlse[row] : 0.f);
 }
 
 template <int D>
@@ -459,7 +464,7 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                    torch::Tensor v, torch::Tensor o, torch::Tensor lse,
                    torch::Tensor delta, torch::Tensor dq, torch::Tensor dk,
                    torch::Tensor dv, int B, int H, int Hkv, int S, int D,
-                   double scale, bool causal) {
+                   double scale, bool causal, const float* dlse_ptr) {
   auto stream = scaffold_stream();
   dim3 block(256);
   dim3 grid_q((S + 127) / 128, B * H);
@@ -472,11 +477,11 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   if (D == 64)
     hipLaunchKernelGGL((attn_delta_strided_kernel<64>), dgrid, block, 0, stream,
                        (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
-                       delta.data_ptr<float>(), H, S, rows, dos, osr);
+                       dlse_ptr, delta.data_ptr<float>(), H, S, rows, dos, osr);
   else
     hipLaunchKernelGGL((attn_delta_strided_kernel<128>), dgrid, block, 0, stream,
                        (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
-                       delta.data_ptr<float>(), H, S, rows, dos, osr);
+                       dlse_ptr, delta.data_ptr<float>(), H, S, rows, dos, osr);
   #define LAUNCH_D(DD)                                                         \
     do {                                                                       \
       hipLaunchKernelGGL((attn_bwd_dq_mfma_kernel<DD>), grid_q, block, 0,      \

@@ -1,0 +1,66 @@
+"""Context-parallel attention over a sequence-sharded batch
+(the attention op is a single boundary whose (o, lse)
+online-softmax state is exactly what a context-parallel wrapper needs —
+this module composes AROUND ops.attention_lse without touching the HIP
+kernel).
+
+Sharding model: the sequence is split evenly over the CP group's ranks;
+rank r holds q_r, k_r, v_r = shard r of (B, H/ Hkv, S_total, D).  Each
+rank computes its queries' attention against every kv shard it may
+attend (j < r: full; j == r: causal; j > r: skipped under causality)
+and merges the partials EXACTLY through their logsumexp:
+
+    lse = logsumexp_j(lse_j)        O = sum_j exp(lse_j - lse) * O_j
+
+The kv shards are exchanged with the DIFFERENTIABLE all_gather from
+torch.distributed.nn.functional, so backward reduce-scatters dk/dv to
+their owner ranks automatically; d(lse_j) flows into each partial's
+backward through attention_lse (see the dlse plumbing note in
+scaffold/ops).  This is the all-gather form of ring attention (the
+ring-pass form changes only the schedule of the same partial merges).
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+
+from .. import ops
+
+
+def _merge(partials: List[torch.Tensor], lses: List[torch.Tensor]) -> torch.Tensor:
+    """Exact lse-weighted merge of per-shard attention partials."""
+    if len(partials) == 1:
+        return partials[0]
+    lse_all = torch.stack(lses, dim=0)                     # (J, B, H, S)
+    lse_tot = torch.logsumexp(lse_all, dim=0)              # (B, H, S)
+    w = torch.exp(lse_all - lse_tot.unsqueeze(0))          # (J, B, H, S)
+    out = torch.zeros_like(partials[0], dtype=torch.float32)
+    for j, oj in enumerate(partials):
+        out = out + w[j].unsqueeze(-1) * oj.float()
+    return out.to(partials[0].dtype)
+
+
+def cp_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                 rank: int, world: int, causal: bool = True,
+                 group=None) -> torch.Tensor:
+    """Attention for THIS rank's query shard against the group's full
+    (sharded) kv sequence.  q/k/v: (B, H|Hkv, S_local, D)."""
+    if world <= 1:
+        return ops.attention(q, k, v, causal)
+
+    from torch.distributed import nn as dist_nn
+
+    k_shards = list(dist_nn.functional.all_gather(k, group=group))
+    v_shards = list(dist_nn.functional.all_gather(v, group=group))
+
+    partials, lses = [], []
+    last = rank if causal else world - 1
+    for j in range(0, last + 1):
+        o_j, lse_j = ops.attention_lse(
+            q, k_shards[j], v_shards[j],
+            causal=(causal and j == rank))
+        partials.append(o_j)
+        lses.append(lse_j)
+    return _merge(partials, lses)

@@ -218,6 +218,59 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return _attention_eager(q, k, v, causal)
 
 
+class _AttentionLse(torch.autograd.Function):
+    """Attention exposing the per-row logsumexp as a DIFFERENTIABLE
+    second output — the building block of the context-parallel partial
+    merge (scaffold/dist/cp.py).  d(lse) folds into the backward's
+    delta term (delta' = rowsum(dO*O) - dlse) with zero extra kernels:
+    dS gains  p * dlse_row  exactly (d lse / d s_k = p_k)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        m = require_ext()
+        o, lse = m.attn_fwd(q, k, v, causal)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        return o, lse
+
+    @staticmethod
+    def backward(ctx, do, dlse):
+        q, k, v, o, lse = ctx.saved_tensors
+        m = require_ext()
+        dq, dk, dv = m.attn_bwd(do, q, k, v, o, lse, ctx.causal,
+                                dlse.contiguous().float())
+        return dq, dk, dv, None
+
+
+def _eager_attn_lse(q, k, v, causal):
+    """fp32 reference attention returning (o, lse); fully
+    autograd-differentiable (the CPU path of attention_lse and the
+    ground truth of the CP tests)."""
+    Hq, Hkv = q.shape[1], k.shape[1]
+    if Hkv != Hq:
+        rep = Hq // Hkv
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    scale = 1.0 / (q.shape[-1] ** 0.5)
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    if causal:
+        S_, Sk = s.shape[-2], s.shape[-1]
+        mask = torch.ones(S_, Sk, dtype=torch.bool, device=s.device).triu(1)
+        s = s.masked_fill(mask, float("-inf"))
+    lse = torch.logsumexp(s, dim=-1)
+    p = torch.exp(s - lse.unsqueeze(-1))
+    o = (p @ v.float()).to(q.dtype)
+    return o, lse
+
+
+def attention_lse(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                  causal: bool = True):
+    """(o, lse) with gradients through BOTH outputs."""
+    if q.is_cuda:
+        return _AttentionLse.apply(q, k, v, causal)
+    return _eager_attn_lse(q, k, v, causal)
+
+
 def _colsum(dy2: torch.Tensor) -> torch.Tensor:
     m = ext()
     if m is not None and dy2.is_cuda and dy2.dtype == torch.bfloat16:
