@@ -413,3 +413,31 @@ def test_lm_head_ce_matches_eager():
         err = (a - e).abs().max().item()
         tol = 5e-2 * max(0.05, e.abs().max().item())
         assert err < tol, f"{name}: {err} vs {tol}"
+
+
+def test_attention_lse_dlse_backward(ext):
+    """attention_lse: gradients through BOTH outputs (o AND lse) vs the
+    fp32 eager reference — the context-parallel merge's building block
+    (dlse folded into the backward's delta term)."""
+    from scaffold import ops
+
+    torch.manual_seed(3)
+    q32 = torch.randn(1, 2, 96, 64)
+    k32 = torch.randn(1, 2, 96, 64)
+    v32 = torch.randn(1, 2, 96, 64)
+    do = torch.randn(1, 2, 96, 64)
+    dl = torch.randn(1, 2, 96) * 0.1
+    qe, ke, ve = [t.clone().requires_grad_(True) for t in (q32, k32, v32)]
+    oe, le = ops._eager_attn_lse(qe, ke, ve, True)
+    torch.autograd.backward([oe, le], [do, dl])
+
+    qb, kb, vb = [t.to(torch.bfloat16).cuda().requires_grad_(True)
+                  for t in (q32, k32, v32)]
+    ob, lb = ops.attention_lse(qb, kb, vb, True)
+    torch.autograd.backward([ob, lb], [do.to(torch.bfloat16).cuda(), dl.cuda()])
+    torch.cuda.synchronize()
+    for n, a, b in [("o", ob, oe), ("lse", lb, le), ("dq", qb.grad, qe.grad),
+                    ("dk", kb.grad, ke.grad), ("dv", vb.grad, ve.grad)]:
+        err = (a.detach().cpu().float() - b.detach().float()).abs().max().item()
+        tol = 0.08 * max(1.0, b.detach().float().abs().max().item())
+        assert err < tol, (n, err, tol)
