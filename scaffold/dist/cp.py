@@ -64,3 +64,38 @@ def cp_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
         partials.append(o_j)
         lses.append(lse_j)
     return _merge(partials, lses)
+
+
+def ulysses_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      rank: int, world: int, causal: bool = True,
+                      group=None) -> torch.Tensor:
+    """Ulysses-style sequence parallelism: all-to-all
+    swaps the sharding axis from SEQUENCE to HEADS, each rank runs
+    ordinary full-sequence attention on H/world (and Hkv/world kv)
+    heads, and the inverse all-to-all restores sequence sharding.
+    Exact by construction; requires world | H and world | Hkv.
+    q/k/v: (B, H|Hkv, S_local, D)."""
+    if world <= 1:
+        return ops.attention(q, k, v, causal)
+    assert q.shape[1] % world == 0 and k.shape[1] % world == 0, \
+        "ulysses: world must divide both H and Hkv"
+
+    from torch.distributed import nn as dist_nn
+
+    def seq_to_heads(t):
+        # shard heads, gather sequence: rank r keeps head chunk r of
+        # every rank's sequence shard, concatenated in rank (=sequence)
+        # order
+        chunks = list(t.chunk(world, dim=1))
+        outs = [torch.empty_like(c) for c in chunks]
+        outs = dist_nn.functional.all_to_all(outs, chunks, group=group)
+        return torch.cat(list(outs), dim=2)
+
+    def heads_to_seq(t):
+        chunks = list(t.chunk(world, dim=2))
+        outs = [torch.empty_like(c) for c in chunks]
+        outs = dist_nn.functional.all_to_all(outs, chunks, group=group)
+        return torch.cat(list(outs), dim=1)
+
+    o = ops.attention(seq_to_heads(q), seq_to_heads(k), seq_to_heads(v), causal)
+    return heads_to_seq(o)
