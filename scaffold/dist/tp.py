@@ -1,0 +1,149 @@
+"""Tensor-parallel (Megatron-style) linear primitives (SURVEY §2.4/§5.7
+additive layer).
+
+Sharding model over a TP group of `world` ranks:
+
+  ColumnParallelLinear  W: (N, K) sharded along N -> rank r holds
+      W_r: (N/world, K); y_r = x W_r^T (+ b_r).  The full activation is
+      the concatenation over ranks; with `gather=False` the shard stays
+      local (the usual case — the next op is row-parallel).
+  RowParallelLinear     W: (N, K) sharded along K -> rank r holds
+      W_r: (N, K/world); partial = x_r W_r^T, summed over ranks with a
+      DIFFERENTIABLE all_reduce (backward is an identity broadcast).
+
+A column-parallel -> row-parallel pair (the transformer MLP / attention
+projection pattern) needs exactly ONE all_reduce per forward and one in
+backward, with weight gradients staying rank-local.
+
+The matmuls go through scaffold.ops.linear, so on GPU the per-shape
+dispatch (hand MFMA kernel vs hipBLASLt) applies unchanged.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+
+
+def _initialized() -> bool:
+    import torch.distributed as dist
+
+    return dist.is_available() and dist.is_initialized()
+
+
+class _CopyToTp(torch.autograd.Function):
+    """Megatron's `f`: identity forward, all-reduce backward — wraps the
+    REPLICATED input of a column-parallel linear so dx sums the shards'
+    contributions (the loss itself is computed redundantly per rank, so
+    activation grads must be reduced exactly once, here)."""
+
+    @staticmethod
+    def forward(ctx, t, group):
+        ctx.group = group
+        return t
+
+    @staticmethod
+    def backward(ctx, g):
+        import torch.distributed as dist
+
+        if _initialized():
+            g = g.contiguous()
+            dist.all_reduce(g, group=ctx.group)
+        return g, None
+
+
+class _ReduceFromTp(torch.autograd.Function):
+    """Megatron's `g`: all-reduce forward, identity backward — sums the
+    row-parallel partials; downstream grads are already replicated."""
+
+    @staticmethod
+    def forward(ctx, t, group):
+        import torch.distributed as dist
+
+        if _initialized():
+            t = t.contiguous()
+            dist.all_reduce(t, group=group)
+        return t
+
+    @staticmethod
+    def backward(ctx, g):
+        return g, None
+
+
+class ColumnParallelLinear(nn.Module):
+    """y_r = x @ W_r^T + b_r with W sharded along the OUTPUT dim."""
+
+    def __init__(self, in_features: int, out_features: int, world: int,
+                 rank: int, bias: bool = True):
+        super().__init__()
+        assert out_features % world == 0, "TP: world must divide out_features"
+        self.in_features = in_features
+        self.out_local = out_features // world
+        self.world = world
+        self.rank = rank
+        self.weight = nn.Parameter(torch.empty(self.out_local, in_features))
+        self.bias = nn.Parameter(torch.zeros(self.out_local)) if bias else None
+        nn.init.normal_(self.weight, std=0.02)
+
+    @torch.no_grad()
+    def load_full(self, w_full: torch.Tensor, b_full: Optional[torch.Tensor] = None):
+        r0 = self.rank * self.out_local
+        self.weight.copy_(w_full[r0:r0 + self.out_local])
+        if self.bias is not None and b_full is not None:
+            self.bias.copy_(b_full[r0:r0 + self.out_local])
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = _CopyToTp.apply(x, None)
+        return ops.linear(x, self.weight, self.bias)
+
+
+class RowParallelLinear(nn.Module):
+    """y = sum_r x_r @ W_r^T (+ b on rank-agnostic add) with W sharded
+    along the INPUT dim; the sum is a differentiable all_reduce."""
+
+    def __init__(self, in_features: int, out_features: int, world: int,
+                 rank: int, bias: bool = True, group=None):
+        super().__init__()
+        assert in_features % world == 0, "TP: world must divide in_features"
+        self.in_local = in_features // world
+        self.world = world
+        self.rank = rank
+        self.group = group
+        self.weight = nn.Parameter(torch.empty(out_features, self.in_local))
+        self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
+        nn.init.normal_(self.weight, std=0.02)
+
+    @torch.no_grad()
+    def load_full(self, w_full: torch.Tensor, b_full: Optional[torch.Tensor] = None):
+        c0 = self.rank * self.in_local
+        self.weight.copy_(w_full[:, c0:c0 + self.in_local])
+        if self.bias is not None and b_full is not None:
+            self.bias.copy_(b_full)
+
+    def forward(self, x_local: torch.Tensor) -> torch.Tensor:
+        partial = ops.linear(x_local, self.weight, None)
+        out = _ReduceFromTp.apply(partial, self.group)
+        if self.bias is not None:
+            out = out + self.bias
+        return out
+
+
+class TpMLP(nn.Module):
+    """Column-parallel up-projection + activation + row-parallel
+    down-projection: ONE all_reduce per forward (the transformer-MLP TP
+    pattern).  act: "gelu" (GPT-2) or "silu-glu" would compose the same
+    way with a sharded gate|up."""
+
+    def __init__(self, d_model: int, d_ff: int, world: int, rank: int,
+                 group=None):
+        super().__init__()
+        self.fc = ColumnParallelLinear(d_model, d_ff, world, rank)
+        self.proj = RowParallelLinear(d_ff, d_model, world, rank, group=group)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        h = torch.nn.functional.gelu(self.fc(x), approximate="tanh")
+        return self.proj(h)
