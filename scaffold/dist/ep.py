@@ -1,0 +1,82 @@
+"""Expert parallelism (SURVEY §2.4 additive layer): a top-1-routed MoE
+feed-forward with the EXPERTS sharded across the EP group.
+
+Data flow (dense-dispatch formulation, exact and fully differentiable):
+  1. all_gather the group's token shards (differentiable: backward
+     reduce-scatters dx to the owner ranks);
+  2. every rank evaluates the replicated gate over ALL tokens and runs
+     only ITS experts on the tokens routed to them, scattering the
+     gate-weighted outputs into a zero buffer;
+  3. a differentiable all_reduce sums the per-rank partial outputs
+     (each token was computed on exactly one rank), and each rank keeps
+     its own token slice.
+Expert weight gradients stay rank-local (each rank owns its experts);
+gate gradients are averaged by the caller's DDP like any replicated
+parameter.  The production dispatch would exchange only the routed
+tokens with an all_to_all — same math, sparser transport (NOTES.md).
+"""
+
+from __future__ import annotations
+
+from typing import List
+
+import torch
+import torch.nn as nn
+
+
+class ExpertParallelMoE(nn.Module):
+    """Top-1 MoE FFN; `n_experts` must be divisible by `world`."""
+
+    def __init__(self, d_model: int, d_ff: int, n_experts: int, world: int,
+                 rank: int, group=None):
+        super().__init__()
+        assert n_experts % world == 0
+        self.world = world
+        self.rank = rank
+        self.group = group
+        self.n_experts = n_experts
+        self.e_local = n_experts // world
+        self.e0 = rank * self.e_local
+        self.gate = nn.Linear(d_model, n_experts, bias=False)  # replicated
+        self.w_up: nn.ParameterList = nn.ParameterList(
+            nn.Parameter(torch.randn(d_ff, d_model) * 0.02)
+            for _ in range(self.e_local))
+        self.w_down: nn.ParameterList = nn.ParameterList(
+            nn.Parameter(torch.randn(d_model, d_ff) * 0.02)
+            for _ in range(self.e_local))
+
+    def _expert(self, i_local: int, x: torch.Tensor) -> torch.Tensor:
+        h = torch.nn.functional.gelu(x @ self.w_up[i_local].t(),
+                                     approximate="tanh")
+        return h @ self.w_down[i_local].t()
+
+    def forward(self, x_local: torch.Tensor) -> torch.Tensor:
+        import torch.distributed as dist
+
+        T = x_local.shape[0]
+        if self.world > 1 and dist.is_initialized():
+            from torch.distributed import nn as dist_nn
+
+            x_all = torch.cat(
+                list(dist_nn.functional.all_gather(x_local, group=self.group)),
+                dim=0)
+        else:
+            x_all = x_local
+
+        probs = torch.softmax(self.gate(x_all), dim=-1)      # (T_all, E)
+        top_p, top_e = probs.max(dim=-1)                     # top-1 routing
+
+        y_partial = torch.zeros_like(x_all)
+        for i in range(self.e_local):
+            sel = (top_e == self.e0 + i).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            out = self._expert(i, x_all[sel]) * top_p[sel].unsqueeze(-1)
+            y_partial = y_partial.index_add(0, sel, out)
+
+        if self.world > 1 and dist.is_initialized():
+            from torch.distributed import nn as dist_nn
+
+            y_all = dist_nn.functional.all_reduce(y_partial, group=self.group)
+            return y_all[self.rank * T:(self.rank + 1) * T]
+        return y_partial
