@@ -1,0 +1,139 @@
+"""Autoregressive generation with per-layer KV caches (the serving face
+of the scaffold).
+
+Prefill runs the prompt through the normal block math (the flash
+attention op on GPU) while recording K/V; each decode step then attends
+the single new query against the cache with plain library ops (the
+per-step work is a handful of GEMV-shaped products — the library path
+per SURVEY §2.2; a fused decode kernel is a round-2 item).  Supports
+both model families (GPT-2 LayerNorm/GELU, Llama RMSNorm/RoPE/GQA/
+SwiGLU).  Greedy (temperature=0) or temperature sampling.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import ops
+from .models.gpt2 import GPT2
+from .models.llama import Llama
+
+
+def _sdpa(q, k, v, causal):
+    """(B,H,Sq,D) x (B,Hkv,Skv,D) library attention with GQA expand."""
+    Hq, Hkv = q.shape[1], k.shape[1]
+    if Hkv != Hq:
+        rep = Hq // Hkv
+        k = k.repeat_interleave(rep, dim=1)
+        v = v.repeat_interleave(rep, dim=1)
+    scale = 1.0 / math.sqrt(q.shape[-1])
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    if causal and q.shape[2] > 1:
+        mask = torch.ones(q.shape[2], k.shape[2], dtype=torch.bool,
+                          device=q.device).triu(k.shape[2] - q.shape[2] + 1)
+        s = s.masked_fill(mask, float("-inf"))
+    p = torch.softmax(s, dim=-1)
+    return (p @ v.float()).to(q.dtype)
+
+
+class KVCache:
+    def __init__(self, n_layers: int):
+        self.k: List[Optional[torch.Tensor]] = [None] * n_layers
+        self.v: List[Optional[torch.Tensor]] = [None] * n_layers
+
+    def append(self, i: int, k: torch.Tensor, v: torch.Tensor
+               ) -> Tuple[torch.Tensor, torch.Tensor]:
+        self.k[i] = k if self.k[i] is None else torch.cat([self.k[i], k], dim=2)
+        self.v[i] = v if self.v[i] is None else torch.cat([self.v[i], v], dim=2)
+        return self.k[i], self.v[i]
+
+    @property
+    def seq_len(self) -> int:
+        return 0 if self.k[0] is None else self.k[0].shape[2]
+
+
+class Generator:
+    """Greedy / temperature decoding for the scaffold's LM families."""
+
+    def __init__(self, model: torch.nn.Module):
+        self.m = model
+        self.is_llama = isinstance(model, Llama)
+        assert self.is_llama or isinstance(model, GPT2), type(model)
+
+    # ---- per-family block math with external KV ------------------------
+    def _gpt2_block(self, blk, x, cache: KVCache, i: int, pos0: int):
+        m = blk.attn
+        B, S, C = x.shape
+        h = ops.layer_norm(x, blk.ln1.weight, blk.ln1.bias, blk.ln1.eps)
+        qkv = F.linear(h, m.qkv.weight, m.qkv.bias).view(
+            B, S, 3, m.n_heads, m.head_dim)
+        q, k, v = (qkv[:, :, j].transpose(1, 2) for j in range(3))
+        k, v = cache.append(i, k, v)
+        o = _sdpa(q, k, v, causal=True)
+        x = x + F.linear(o.transpose(1, 2).reshape(B, S, C),
+                         m.proj.weight, m.proj.bias)
+        h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+        h = F.gelu(F.linear(h, blk.fc.weight, blk.fc.bias), approximate="tanh")
+        return x + F.linear(h, blk.proj.weight, blk.proj.bias)
+
+    def _llama_block(self, blk, x, cache: KVCache, i: int, pos0: int):
+        m = blk.attn
+        B, S, _ = x.shape
+        h = ops.rms_norm(x, blk.attn_norm.weight, blk.attn_norm.eps)
+        qkv = F.linear(h, m.wqkv.weight).view(
+            B, S, m.n_heads + 2 * m.n_kv, m.head_dim)
+        q = qkv[:, :, :m.n_heads].transpose(1, 2)
+        k = qkv[:, :, m.n_heads:m.n_heads + m.n_kv].transpose(1, 2)
+        v = qkv[:, :, m.n_heads + m.n_kv:].transpose(1, 2)
+        cos = self.m.rope_cos[pos0:pos0 + S].float()
+        sin = self.m.rope_sin[pos0:pos0 + S].float()
+        q = ops._rope_eager(q, cos, sin) if not q.is_cuda else ops.rope_apply(q, cos, sin)
+        k = ops._rope_eager(k, cos, sin) if not k.is_cuda else ops.rope_apply(k, cos, sin)
+        k, v = cache.append(i, k, v)
+        o = _sdpa(q, k, v, causal=True)
+        x = x + F.linear(o.transpose(1, 2).reshape(B, S, -1), m.wo.weight)
+        h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
+        gu = F.linear(h, blk.w_gateup.weight)
+        Fh = gu.shape[-1] // 2
+        g = F.silu(gu[..., :Fh]) * gu[..., Fh:]
+        return x + F.linear(g, blk.w_down.weight)
+
+    def _forward_tokens(self, tokens: torch.Tensor, cache: KVCache,
+                        pos0: int) -> torch.Tensor:
+        """tokens (B, S) at positions [pos0, pos0+S) -> logits (B, V)."""
+        m = self.m
+        if self.is_llama:
+            x = m.embed(tokens)
+            for i, blk in enumerate(m.blocks):
+                x = self._llama_block(blk, x, cache, i, pos0)
+            x = ops.rms_norm(x, m.norm_f.weight, m.norm_f.eps)
+        else:
+            pos = torch.arange(pos0, pos0 + tokens.shape[1], device=tokens.device)
+            x = m.wte(tokens) + m.wpe(pos)
+            for i, blk in enumerate(m.blocks):
+                x = self._gpt2_block(blk, x, cache, i, pos0)
+            x = ops.layer_norm(x, m.ln_f.weight, m.ln_f.bias, m.ln_f.eps)
+        return F.linear(x[:, -1], m.lm_head.weight)
+
+    @torch.no_grad()
+    def generate(self, prompt: torch.Tensor, max_new_tokens: int,
+                 temperature: float = 0.0,
+                 generator: Optional[torch.Generator] = None) -> torch.Tensor:
+        """prompt (B, S0) int64 -> (B, S0 + max_new_tokens)."""
+        n_layers = len(self.m.blocks)
+        cache = KVCache(n_layers)
+        out = prompt
+        logits = self._forward_tokens(prompt, cache, 0)   # prefill
+        for _ in range(max_new_tokens):
+            if temperature > 0:
+                probs = torch.softmax(logits.float() / temperature, dim=-1)
+                nxt = torch.multinomial(probs, 1, generator=generator)
+            else:
+                nxt = logits.argmax(dim=-1, keepdim=True)
+            out = torch.cat([out, nxt], dim=1)
+            logits = self._forward_tokens(nxt, cache, cache.seq_len)
+        return out

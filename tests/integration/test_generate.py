@@ -1,0 +1,73 @@
+"""KV-cached decoding equivalence: greedy generation with the cache
+matches recompute-from-scratch generation (the model's training forward
+repurposed for logits), for both LM families (fp32, CPU)."""
+
+import torch
+import torch.nn.functional as F
+
+from scaffold.config import load_config
+from scaffold.generate import Generator
+from scaffold.models import build_model
+
+
+def _tiny(name, extra=()):
+    cfg = load_config(None, [
+        f"model.name={name}", "model.n_layers=2", "model.n_heads=4",
+        "model.d_model=64", "model.vocab_size=128", "model.max_seq_len=64",
+        "dtype=fp32", "device=cpu", *extra,
+    ])
+    torch.manual_seed(9)
+    return build_model(cfg)
+
+
+def _ref_greedy(gen: Generator, prompt, n):
+    """no-cache reference: full forward from scratch every step."""
+    from scaffold.generate import KVCache
+
+    out = prompt
+    for _ in range(n):
+        cache = KVCache(len(gen.m.blocks))
+        logits = gen._forward_tokens(out, cache, 0)
+        out = torch.cat([out, logits.argmax(-1, keepdim=True)], dim=1)
+    return out
+
+
+def test_gpt2_greedy_cache_matches_recompute():
+    m = _tiny("gpt2")
+    g = Generator(m)
+    prompt = torch.randint(0, 128, (2, 7))
+    got = g.generate(prompt, max_new_tokens=8)
+    want = _ref_greedy(g, prompt, 8)
+    assert torch.equal(got, want)
+
+
+def test_llama_greedy_cache_matches_recompute():
+    m = _tiny("llama", ["model.n_kv_heads=2", "model.d_ff=128"])
+    g = Generator(m)
+    prompt = torch.randint(0, 128, (2, 5))
+    got = g.generate(prompt, max_new_tokens=8)
+    want = _ref_greedy(g, prompt, 8)
+    assert torch.equal(got, want)
+
+
+def test_decode_logits_match_training_forward():
+    """the decode path's block math == the training path's block math:
+    per-position logits from Generator._forward_tokens equal the
+    training forward's logits (via its loss on a delta probe... here
+    directly: rebuild logits with the model's own modules)."""
+    m = _tiny("gpt2")
+    g = Generator(m)
+    tokens = torch.randint(0, 128, (1, 9))
+    from scaffold.generate import KVCache
+
+    logits_dec = g._forward_tokens(tokens, KVCache(len(m.blocks)), 0)
+    # training-path logits for the LAST position: run model internals
+    pos = torch.arange(9)
+    x = m.wte(tokens) + m.wpe(pos)
+    for blk in m.blocks:
+        x = blk(x)
+    from scaffold import ops
+
+    x = ops.layer_norm(x, m.ln_f.weight, m.ln_f.bias, m.ln_f.eps)
+    logits_train = F.linear(x[:, -1], m.lm_head.weight)
+    assert torch.allclose(logits_dec, logits_train, atol=1e-5)
