@@ -18,7 +18,6 @@ tokens with an all_to_all — same math, sparser transport (NOTES.md).
 
 from __future__ import annotations
 
-from typing import List
 
 import torch
 import torch.nn as nn

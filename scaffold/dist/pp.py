@@ -18,7 +18,7 @@ works on gloo (CPU tests) and RCCL alike.
 
 from __future__ import annotations
 
-from typing import Callable, List
+from typing import List
 
 import torch
 import torch.distributed as dist

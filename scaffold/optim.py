@@ -8,7 +8,7 @@ also refreshes the bf16 working copy in the same pass (K8 folded in).
 
 from __future__ import annotations
 
-from typing import Iterable, List, Optional
+from typing import Iterable, List
 
 import torch
 
@@ -53,7 +53,6 @@ class MasterAdamW:
                 else:
                     p.grad.zero_()
 
-    @torch.no_grad()
     def _build_groups(self):
         # one fused launch per homogeneous group: {bf16 working + bf16-out}
         # and {fp32 working aliasing master, no out}
@@ -72,6 +71,7 @@ class MasterAdamW:
                 g[4].append(p.data)
         return groups
 
+    @torch.no_grad()
     def step(self, grad_scale: float = 1.0) -> None:
         self.step_count += 1
         on_gpu = self.params[0].is_cuda and ops.has_ext()
