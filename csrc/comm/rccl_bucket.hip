@@ -36,13 +36,29 @@ struct RcclComm {
   hipStream_t stream = nullptr;
   std::vector<std::pair<hipEvent_t, hipEvent_t>> timers;  // per-bucket
   size_t timers_used = 0;
+  // pooled ordering events (hot path: one per all_reduce + one per join;
+  // created once, reused — per-call create/destroy churned the runtime)
+  std::vector<hipEvent_t> order_pool;
+  size_t order_next = 0;
   int rank = -1, world = 0;
+
+  hipEvent_t next_order_event() {
+    if (order_pool.empty()) {
+      order_pool.resize(64, nullptr);
+      for (auto& e : order_pool)
+        HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    }
+    hipEvent_t e = order_pool[order_next];
+    order_next = (order_next + 1) % order_pool.size();
+    return e;
+  }
 
   ~RcclComm() {
     for (auto& p : timers) {
       (void)hipEventDestroy(p.first);
       (void)hipEventDestroy(p.second);
     }
+    for (auto& e : order_pool) (void)hipEventDestroy(e);
     if (stream) (void)hipStreamDestroy(stream);
     if (comm) (void)ncclCommDestroy(comm);
   }
@@ -103,11 +119,9 @@ void rccl_all_reduce(long handle, torch::Tensor t) {
   CHECK_CUDA(t);
   CHECK_CONTIG(t);
   hipStream_t cur = scaffold_stream();
-  hipEvent_t ready;
-  HIP_CHECK(hipEventCreateWithFlags(&ready, hipEventDisableTiming));
+  hipEvent_t ready = c.next_order_event();
   HIP_CHECK(hipEventRecord(ready, cur));
   HIP_CHECK(hipStreamWaitEvent(c.stream, ready, 0));
-  HIP_CHECK(hipEventDestroy(ready));
   auto& tm = next_timer(c);
   HIP_CHECK(hipEventRecord(tm.first, c.stream));
   RCCL_CHECK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
@@ -119,11 +133,9 @@ void rccl_all_reduce(long handle, torch::Tensor t) {
 void rccl_join(long handle) {
   auto& c = *g_comms.at(handle);
   hipStream_t cur = scaffold_stream();
-  hipEvent_t done;
-  HIP_CHECK(hipEventCreateWithFlags(&done, hipEventDisableTiming));
+  hipEvent_t done = c.next_order_event();
   HIP_CHECK(hipEventRecord(done, c.stream));
   HIP_CHECK(hipStreamWaitEvent(cur, done, 0));
-  HIP_CHECK(hipEventDestroy(done));
 }
 
 // total measured collective time (ms) since last call; host-syncs the

@@ -97,6 +97,15 @@ void save_shard_pinned(std::string path, std::vector<std::string> names,
   PinnedBuf bufs[2];
   if (any_cuda) {
     HIP_CHECK(hipStreamCreateWithFlags(&side, hipStreamNonBlocking));
+    // Order the D2H copies after everything already enqueued on the
+    // compute stream (the optimizer step that produced this state may
+    // still be in flight) — otherwise shards can capture torn
+    // mid-update param/moment bytes.
+    hipEvent_t ordered;
+    HIP_CHECK(hipEventCreateWithFlags(&ordered, hipEventDisableTiming));
+    HIP_CHECK(hipEventRecord(ordered, scaffold_stream()));
+    HIP_CHECK(hipStreamWaitEvent(side, ordered, 0));
+    HIP_CHECK(hipEventDestroy(ordered));
     for (auto& b : bufs) {
       HIP_CHECK(hipHostMalloc(&b.host, BUF, hipHostMallocDefault));
       HIP_CHECK(hipEventCreateWithFlags(&b.ev, hipEventDisableTiming));

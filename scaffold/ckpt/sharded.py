@@ -58,6 +58,9 @@ class _PinnedStager:
         """Return a CPU copy of t; async through the pinned ring on GPU."""
         if not self.on_gpu or not t.is_cuda:
             return t.detach().cpu()
+        # order after in-flight compute (e.g. the optimizer step that
+        # produced this state) before copying on the side stream
+        self.stream.wait_stream(torch.cuda.current_stream())
         flat = t.detach().contiguous().view(torch.uint8).view(-1)
         n = flat.numel()
         if n > self.buf_bytes:

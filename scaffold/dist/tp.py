@@ -78,13 +78,14 @@ class ColumnParallelLinear(nn.Module):
     """y_r = x @ W_r^T + b_r with W sharded along the OUTPUT dim."""
 
     def __init__(self, in_features: int, out_features: int, world: int,
-                 rank: int, bias: bool = True):
+                 rank: int, bias: bool = True, group=None):
         super().__init__()
         assert out_features % world == 0, "TP: world must divide out_features"
         self.in_features = in_features
         self.out_local = out_features // world
         self.world = world
         self.rank = rank
+        self.group = group
         self.weight = nn.Parameter(torch.empty(self.out_local, in_features))
         self.bias = nn.Parameter(torch.zeros(self.out_local)) if bias else None
         nn.init.normal_(self.weight, std=0.02)
@@ -97,7 +98,10 @@ class ColumnParallelLinear(nn.Module):
             self.bias.copy_(b_full[r0:r0 + self.out_local])
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = _CopyToTp.apply(x, None)
+        # backward dx all-reduce must span the TP group, not WORLD —
+        # group=None here would silently sum dx across data-parallel
+        # replicas in a hybrid TP x DP layout
+        x = _CopyToTp.apply(x, self.group)
         return ops.linear(x, self.weight, self.bias)
 
 
@@ -141,7 +145,7 @@ class TpMLP(nn.Module):
     def __init__(self, d_model: int, d_ff: int, world: int, rank: int,
                  group=None):
         super().__init__()
-        self.fc = ColumnParallelLinear(d_model, d_ff, world, rank)
+        self.fc = ColumnParallelLinear(d_model, d_ff, world, rank, group=group)
         self.proj = RowParallelLinear(d_ff, d_model, world, rank, group=group)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

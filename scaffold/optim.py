@@ -35,6 +35,7 @@ class MasterAdamW:
         self.fused = fused
         self.step_count = 0
         self._cached = None  # prebuilt (meta_dev, n, total, bf16, out) per group
+        self._cached_grad_ids = None  # grad data_ptrs the cache was built for
         self.step_dev = None  # optional device step (hipGraph replay path)
         # fp32 masters + moments.  fp32 working params ALIAS their master
         # (no copy-back pass); low-precision params get a separate master.
@@ -76,11 +77,23 @@ class MasterAdamW:
         self.step_count += 1
         on_gpu = self.params[0].is_cuda and ops.has_ext()
         if on_gpu:
+            # the device meta table caches raw grad POINTERS; rebuild if
+            # any grad tensor identity changed since it was built (e.g.
+            # zero_grad(set_to_none=True) re-materialized grads, or a
+            # param had no grad at the first step) — stale pointers
+            # would silently update from freed memory
+            grad_ids = tuple(
+                p.grad.data_ptr() if p.grad is not None else 0
+                for p in self.params
+            )
+            if self._cached is not None and grad_ids != self._cached_grad_ids:
+                self._cached = None
             if self._cached is None:
                 # tensor pointers are stable across steps (grads are
                 # DDP bucket views; masters/moments never reallocate):
                 # build the device meta table ONCE — the per-step call
                 # then carries no H2D copy and is hipGraph-capturable
+                self._cached_grad_ids = grad_ids
                 m = ops.require_ext()
                 cached = []
                 for (gdt, needs_out), (ms, gs, eas, eass, outs) in self._build_groups().items():

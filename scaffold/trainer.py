@@ -240,12 +240,19 @@ class Trainer:
     def evaluate(self, eval_steps: int = 10) -> float:
         self.raw_model.eval()
         total = 0.0
+        # Eval must not perturb the training data stream: snapshot the
+        # generator state and restore it after drawing eval batches, so
+        # training with eval cadence on sees the same data as with it off.
+        saved_gen = self.data.state_dict()
         it = iter(self.data)
         for _ in range(eval_steps):
             x, y = next(it)
             total += self.model.module(x, y).float().item()
+        self.data.load_state_dict(saved_gen)
         self.raw_model.train()
-        t = torch.tensor([total / eval_steps])
+        # Reduction tensor lives on the PG's device: the GPU process
+        # group is RCCL ("nccl"), which cannot reduce CPU tensors.
+        t = torch.tensor([total / eval_steps], device=self.device)
         self.ddp.comm.all_reduce(t)
         return t.item() / max(1, self.ddp.comm.world_size)
 

@@ -95,6 +95,56 @@ def test_ddp_matches_large_batch(tmp_path):
     )
 
 
+def _worker_full_fit(rank, world, file_store_path, ckpt_dir, result_q):
+    """The FULL fit loop at ws=2: eval cadence + checkpoint cadence +
+    metric logging — the paths that only fire on real multi-rank runs
+    (evaluate()'s all_reduce had never executed at ws>1)."""
+    from scaffold.trainer import Trainer
+
+    store = dist.FileStore(file_store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    cfg = _cfg(batch_size=8)
+    cfg.steps = 6
+    cfg.log.interval = 2
+    cfg.log.eval_interval = 3
+    cfg.log.eval_steps = 2
+    cfg.ckpt.interval = 3
+    cfg.ckpt.dir = ckpt_dir
+    cfg.ckpt.save_final = True
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
+                      backend="gloo", device=torch.device("cpu"), initialized=True)
+    t = Trainer(cfg, ctx=ctx)
+    t.fit()
+    ev = t.evaluate(eval_steps=2)
+    if rank == 0:
+        result_q.put(("ok", float(ev), t.step))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_full_fit_eval_ckpt_ws2(tmp_path):
+    ctxmp = mp.get_context("spawn")
+    q = ctxmp.Queue()
+    ckpt_dir = str(tmp_path / "ckpt")
+    procs = [
+        ctxmp.Process(target=_worker_full_fit,
+                      args=(r, WORLD, str(tmp_path / "fs_full"), ckpt_dir, q))
+        for r in range(WORLD)
+    ]
+    for p in procs:
+        p.start()
+    kind, ev, step = q.get(timeout=150)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert kind == "ok"
+    assert torch.isfinite(torch.tensor(ev))
+    assert step == 6
+    # both cadence checkpoint and final checkpoint committed
+    assert os.path.isdir(os.path.join(ckpt_dir, "step_00000003"))
+    assert os.path.isdir(os.path.join(ckpt_dir, "step_00000006"))
+
+
 def _worker_accum(rank, world, file_store_path, result_q):
     from scaffold.trainer import Trainer
 

@@ -43,6 +43,59 @@ def _worker(rank, world, store_path, q_ret):
     dist.destroy_process_group()
 
 
+def _worker_hybrid(rank, world, store_path, q_ret):
+    """TP degree 1 inside a ws=2 world (pure-DP corner of hybrid TP x DP):
+    each rank runs an unsharded TpMLP over its own single-rank TP group.
+    dx must equal the single-process dx — the r1 bug (group=None in
+    _CopyToTp) all-reduced dx over WORLD and doubled it."""
+    from scaffold.dist.tp import TpMLP
+
+    store = dist.FileStore(store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    # every rank must create every group (new_group is collective)
+    groups = [dist.new_group([r]) for r in range(world)]
+    w1, b1, w2, b2, x, dy = _full_weights()
+    m = TpMLP(DM, DF, 1, 0, group=groups[rank])
+    m.fc.load_full(w1, b1)
+    m.proj.load_full(w2, b2)
+    xl = x.clone().requires_grad_(True)
+    y = m(xl)
+    y.backward(dy)
+    q_ret.put((rank, xl.grad.numpy(), m.fc.weight.grad.numpy()))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_tp_subgroup_hybrid_dp(tmp_path):
+    ctxmp = mp.get_context("spawn")
+    ret = ctxmp.Queue()
+    procs = [ctxmp.Process(target=_worker_hybrid,
+                           args=(r, WORLD, str(tmp_path / "fs2"), ret))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    shards = {}
+    for _ in range(WORLD):
+        r, dx, dw1 = ret.get(timeout=120)
+        shards[r] = (torch.from_numpy(dx), torch.from_numpy(dw1))
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    w1, b1, w2, b2, x, dy = _full_weights()
+    w1 = w1.requires_grad_(True)
+    x = x.requires_grad_(True)
+    y_ref = torch.nn.functional.gelu(
+        torch.nn.functional.linear(x, w1, b1), approximate="tanh") @ w2.t() + b2
+    y_ref.backward(dy)
+    for r in range(WORLD):
+        dx, dw1 = shards[r]
+        assert torch.allclose(dx, x.grad, atol=2e-5), (
+            "dx wrong under TP subgroup (was it reduced over WORLD?)")
+        assert torch.allclose(dw1, w1.grad, atol=2e-5)
+
+
 @pytest.mark.timeout(180)
 def test_tp_mlp_matches_full(tmp_path):
     ctxmp = mp.get_context("spawn")

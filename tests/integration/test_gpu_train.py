@@ -113,6 +113,41 @@ def test_rccl_direct_bucket_ws1(tmp_path):
         dist.destroy_process_group()
 
 
+def test_fused_adamw_cache_invalidation():
+    """The fused step caches device grad POINTERS; zero_grad(set_to_none)
+    re-materializes grads, so the cache must rebuild (stale
+    pointers silently updated from freed memory)."""
+    from scaffold.optim import MasterAdamW
+
+    torch.manual_seed(0)
+    lin_gpu = torch.nn.Linear(64, 64).to("cuda", torch.bfloat16)
+    lin_cpu = torch.nn.Linear(64, 64)
+    with torch.no_grad():
+        lin_cpu.weight.copy_(lin_gpu.weight.float())
+        lin_cpu.bias.copy_(lin_gpu.bias.float())
+    og = MasterAdamW(lin_gpu.parameters(), lr=1e-2)
+    oc = MasterAdamW(lin_cpu.parameters(), lr=1e-2)
+
+    for it in range(3):
+        g_w = torch.randn(64, 64)
+        g_b = torch.randn(64)
+        # set_to_none between steps: grad tensors get NEW storage each time
+        for p in lin_gpu.parameters():
+            p.grad = None
+        for p in lin_cpu.parameters():
+            p.grad = None
+        lin_gpu.weight.grad = g_w.to("cuda", torch.bfloat16)
+        lin_gpu.bias.grad = g_b.to("cuda", torch.bfloat16)
+        lin_cpu.weight.grad = g_w.to(torch.bfloat16).float()
+        lin_cpu.bias.grad = g_b.to(torch.bfloat16).float()
+        og.step()
+        oc.step()
+    torch.cuda.synchronize()
+    for mg, mc in zip(og.master, oc.master):
+        assert torch.allclose(mg.cpu(), mc, rtol=2e-3, atol=2e-3), (
+            (mg.cpu() - mc).abs().max().item())
+
+
 def test_hip_graph_step_replay():
     """hipGraph-captured train step: loss decreases across replays and
     the device step counter keeps AdamW bias corrections exact."""

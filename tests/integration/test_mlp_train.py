@@ -62,6 +62,25 @@ def test_grad_accum_equivalence(cpu_cfg):
         assert torch.allclose(a, b / 2, rtol=1e-4, atol=1e-5)
 
 
+def test_eval_does_not_perturb_training_stream(cpu_cfg):
+    """Eval cadence must not change the training data stream: evaluate()
+    snapshots/restores the data generator, so final params match a run
+    with eval off (eval drew from the training RNG)."""
+    import copy
+
+    cfg_on = copy.deepcopy(cpu_cfg)
+    cfg_on.steps = 6
+    cfg_on.log.eval_interval = 2
+    cfg_on.log.eval_steps = 2
+    cfg_off = copy.deepcopy(cfg_on)
+    cfg_off.log.eval_interval = 0
+
+    pa = Trainer(cfg_on).fit().raw_model.parameters()
+    pb = Trainer(cfg_off).fit().raw_model.parameters()
+    for a, b in zip(pa, pb):
+        assert torch.equal(a, b), "eval cadence perturbed the training stream"
+
+
 def test_evaluate_and_eval_cadence(cpu_cfg, capsys):
     import copy
 
