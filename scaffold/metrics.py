@@ -52,6 +52,68 @@ class MetricLogger:
             self.fh.close()
 
 
+class StepBreakdown:
+    """Per-phase step-time breakdown: data / fwd / bwd /
+    opt spans measured with stream-ordered CUDA events (device time, no
+    host syncs on the step path), host clock on CPU.  ``mark(phase)``
+    stamps the end of a phase; ``drain()`` — call at log cadence only —
+    syncs the newest retained event and returns mean ms per phase.
+
+    Only the most recent ``keep`` steps are retained so an arbitrarily
+    long log interval cannot accumulate events without bound."""
+
+    def __init__(self, device: torch.device, keep: int = 8):
+        self.on_gpu = device.type == "cuda"
+        self._steps = deque(maxlen=keep)
+        self._cur = None
+
+    def _stamp(self):
+        if self.on_gpu:
+            ev = torch.cuda.Event(enable_timing=True)
+            ev.record()
+            return ev
+        return time.perf_counter()
+
+    def start_step(self):
+        self._cur = [("start", self._stamp())]
+
+    def mark(self, phase: str):
+        if self._cur is not None:
+            self._cur.append((phase, self._stamp()))
+
+    def end_step(self):
+        if self._cur is not None and len(self._cur) > 1:
+            self._steps.append(self._cur)
+        self._cur = None
+
+    def drain(self) -> Dict[str, float]:
+        if not self._steps:
+            return {}
+        agg: Dict[str, float] = {}
+        n = len(self._steps)
+        for marks in self._steps:
+            if self.on_gpu:
+                marks[-1][1].synchronize()  # newest event; log cadence only
+            for (_, a), (phase, b) in zip(marks, marks[1:]):
+                ms = a.elapsed_time(b) if self.on_gpu else (b - a) * 1000
+                agg[phase] = agg.get(phase, 0.0) + ms
+        self._steps.clear()
+        return {f"{k}_ms": round(v / n, 3) for k, v in agg.items()}
+
+
+def hbm_stats(device: torch.device) -> Dict[str, float]:
+    """HBM-in-use metrics; the C12 288 GB sizing becomes a
+    measured fact in the log rather than docstring arithmetic."""
+    if device.type != "cuda":
+        return {}
+    gib = 1 << 30
+    return {
+        "hbm_alloc_gb": round(torch.cuda.memory_allocated(device) / gib, 3),
+        "hbm_peak_gb": round(torch.cuda.max_memory_allocated(device) / gib, 3),
+        "hbm_reserved_gb": round(torch.cuda.memory_reserved(device) / gib, 3),
+    }
+
+
 class CudaTimer:
     """Event-based GPU span timer; host-clock fallback on CPU."""
 

@@ -19,7 +19,7 @@ from .schedule import lr_at
 from .data.synthetic import build_data
 from .dist.comm import Comm
 from .dist.ddp import BucketedDDP
-from .metrics import MetricLogger
+from .metrics import MetricLogger, StepBreakdown, hbm_stats
 from .models import build_model
 from .optim import build_optimizer, MasterAdamW
 from .runtime import DistContext, init_distributed, teardown
@@ -79,6 +79,7 @@ class Trainer:
             cfg, self.ctx.rank, self.device, self.dtype
         )
         self.logger = MetricLogger(self.ctx, cfg)
+        self.breakdown = StepBreakdown(self.device)
         self.step = 0
         self.cfg_hash = config_hash(cfg)
         if cfg.ckpt.resume:
@@ -126,11 +127,15 @@ class Trainer:
             dt = time.perf_counter() - t0
             self.step += 1
             extra = None
-            if (self.step % max(1, self.cfg.log.interval) == 0
-                    and hasattr(self.ddp.comm, "comm_time_ms")):
-                # measured per-bucket collective time on the comm stream
-                # (drains events — log cadence only, never per step)
-                extra = {"comm_ms": round(self.ddp.comm.comm_time_ms(), 3)}
+            if self.step % max(1, self.cfg.log.interval) == 0:
+                # measured breakdowns drain events — log cadence only,
+                # never per step (§5.5)
+                extra = {}
+                if hasattr(self.ddp.comm, "comm_time_ms"):
+                    # per-bucket collective time on the comm stream
+                    extra["comm_ms"] = round(self.ddp.comm.comm_time_ms(), 3)
+                extra.update(self.breakdown.drain())
+                extra.update(hbm_stats(self.device))
             self.logger.log_step(
                 self.step, loss_val, self.cfg.data.batch_size * accum, dt,
                 extra=extra,
@@ -207,18 +212,24 @@ class Trainer:
         return float(self._graph_loss.detach().float().item())
 
     def _train_step(self, batch_iter, accum: int) -> float:
+        bd = self.breakdown
+        bd.start_step()
         self.ddp.zero_grad()
         loss_val = 0.0
         for micro in range(accum):
             x, y = next(batch_iter)
+            bd.mark("data")
             if micro < accum - 1:
                 with self.ddp.no_sync():
                     loss = self.model(x, y)
+                    bd.mark("fwd")
                     loss.backward()
             else:
                 loss = self.model(x, y)
+                bd.mark("fwd")
                 loss.backward()
                 self.ddp.finish()
+            bd.mark("bwd")
             loss_val += loss.detach().float().item() / accum
 
         self.optimizer.lr = lr_at(self.cfg, self.step, self.cfg.steps)
@@ -233,6 +244,8 @@ class Trainer:
             if gnorm > self.cfg.optim.grad_clip:
                 scale *= self.cfg.optim.grad_clip / (gnorm + 1e-12)
         self.optimizer.step(grad_scale=scale)
+        bd.mark("opt")
+        bd.end_step()
         return loss_val
 
     # -------------------------------------------------------------- eval
