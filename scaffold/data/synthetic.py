@@ -24,7 +24,17 @@ class SyntheticBase:
         self.rank = rank
         self.device = device
         self.dtype = dtype
-        self.gen = torch.Generator(device=device)
+        if device.type == "cuda":
+            # the DEFAULT cuda generator: it is the only one registered
+            # with hipGraph capture, so in-graph batch generation (randn
+            # included) replays with graph-safe philox state; a custom
+            # Generator here made ResNet capture fail in round 1
+            # (NOTES.md).  Resume-exactness is unchanged: state_dict
+            # snapshots this generator's state like any other.
+            idx = device.index if device.index is not None else torch.cuda.current_device()
+            self.gen = torch.cuda.default_generators[idx]
+        else:
+            self.gen = torch.Generator(device=device)
         self.reset()
 
     def reset(self, step: int = 0) -> None:
