@@ -154,10 +154,27 @@ void launch_fwd(const torch::Tensor& q, const torch::Tensor& k,
 
 }  // namespace
 
-// MFMA fast path (attn_fwd_mfma.hip): bf16, D in {64,128}
+// MFMA fast paths: 4-wave/128-row (attn_fwd_mfma.hip) and 8-wave/256-row
+// (attn_fwd_8w.hip); bf16, D in {64,128}
 void attn_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                    torch::Tensor o, torch::Tensor lse, int B, int H, int Hkv,
                    int S, int D, double scale, bool causal);
+void attn_fwd_8w(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                 torch::Tensor o, torch::Tensor lse, int B, int H, int Hkv,
+                 int S, int D, double scale, bool causal);
+
+namespace {
+// SCAFFOLD_ATTN_FWD = 4w | 8w | auto (default).  auto: the 8-wave
+// kernel for q blocks it can fill (S >= 256), 4-wave otherwise.
+// Read per call (cheap vs the kernel) so tests can exercise both
+// kernels in one process.
+int attn_fwd_mode() {
+  const char* e = getenv("SCAFFOLD_ATTN_FWD");
+  if (e && strcmp(e, "4w") == 0) return 4;
+  if (e && strcmp(e, "8w") == 0) return 8;
+  return 0;
+}
+}  // namespace
 
 std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
                                     torch::Tensor v, bool causal) {
@@ -177,7 +194,12 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor q, torch::Tensor k,
     auto kc = k.stride(3) == 1 ? k : k.contiguous();
     auto vc = v.stride(3) == 1 ? v : v.contiguous();
     auto o = torch::empty({B, S, H, D}, q.options()).permute({0, 2, 1, 3});
-    attn_fwd_mfma(qc, kc, vc, o, lse, B, H, Hkv, S, D, scale, causal);
+    const int mode = attn_fwd_mode();
+    const bool use8w = (mode == 8) || (mode == 0 && S >= 256);
+    if (use8w)
+      attn_fwd_8w(qc, kc, vc, o, lse, B, H, Hkv, S, D, scale, causal);
+    else
+      attn_fwd_mfma(qc, kc, vc, o, lse, B, H, Hkv, S, D, scale, causal);
     HIP_CHECK(hipGetLastError());
     return {o, lse};
   }

@@ -31,6 +31,7 @@ _SOURCES = [
     "kernels/rmsnorm_rope.hip",
     "kernels/attn_fwd.hip",
     "kernels/attn_fwd_mfma.hip",
+    "kernels/attn_fwd_8w.hip",
     "kernels/attn_bwd.hip",
     "kernels/attn_bwd_mfma.hip",
     "kernels/gemm_bf16.hip",

@@ -176,9 +176,11 @@ def test_softmax_xent(ext, R, V):
 # --------------------------------------------------------------- attention
 @pytest.mark.parametrize("B,H,Hkv,S,D,causal", [
     (2, 4, 4, 128, 64, True),
-    (1, 8, 2, 256, 128, True),   # GQA
+    (1, 8, 2, 256, 128, True),   # GQA (8-wave fwd kernel in auto mode)
     (2, 2, 2, 96, 64, False),    # non-causal + ragged S
     (1, 4, 2, 200, 128, True),   # causal + ragged S + GQA at D=128
+    (1, 4, 4, 512, 64, True),    # 8-wave fwd at D=64, 2 q-blocks
+    (1, 4, 2, 384, 128, False),  # 8-wave fwd, ragged S (384 = 1.5 blocks)
 ])
 def test_attention_fwd_bwd(ext, B, H, Hkv, S, D, causal):
     torch.manual_seed(7)
