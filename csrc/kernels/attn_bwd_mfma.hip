@@ -460,6 +460,27 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
 
 }  // namespace
 
+// 8-wave variants (attn_bwd_8w.hip)
+void attn_bwd_dq_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                    torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                    torch::Tensor dq, int B, int H, int Hkv, int S, int D,
+                    double scale, bool causal);
+void attn_bwd_dkv_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                     torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                     torch::Tensor dk, torch::Tensor dv, int B, int H,
+                     int Hkv, int S, int D, double scale, bool causal);
+
+namespace {
+// SCAFFOLD_ATTN_BWD = 4w | 8w | auto (default: 8w for S >= 256).
+// Read per call so tests can exercise both kernels in one process.
+int attn_bwd_mode() {
+  const char* e = getenv("SCAFFOLD_ATTN_BWD");
+  if (e && strcmp(e, "4w") == 0) return 4;
+  if (e && strcmp(e, "8w") == 0) return 8;
+  return 0;
+}
+}  // namespace
+
 void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                    torch::Tensor v, torch::Tensor o, torch::Tensor lse,
                    torch::Tensor delta, torch::Tensor dq, torch::Tensor dk,
@@ -482,6 +503,15 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
     hipLaunchKernelGGL((attn_delta_strided_kernel<128>), dgrid, block, 0, stream,
                        (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
                        dlse_ptr, delta.data_ptr<float>(), H, S, rows, dos, osr);
+  const int mode = attn_bwd_mode();
+  if ((mode == 8) || (mode == 0 && S >= 256)) {
+    attn_bwd_dq_8w(dout, q, k, v, lse, delta, dq, B, H, Hkv, S, D, scale,
+                   causal);
+    attn_bwd_dkv_8w(dout, q, k, v, lse, delta, dk, dv, B, H, Hkv, S, D,
+                    scale, causal);
+    HIP_CHECK(hipGetLastError());
+    return;
+  }
   #define LAUNCH_D(DD)                                                         \
     do {                                                                       \
       hipLaunchKernelGGL((attn_bwd_dq_mfma_kernel<DD>), grid_q, block, 0,      \

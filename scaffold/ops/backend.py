@@ -34,6 +34,7 @@ _SOURCES = [
     "kernels/attn_fwd_8w.hip",
     "kernels/attn_bwd.hip",
     "kernels/attn_bwd_mfma.hip",
+    "kernels/attn_bwd_8w.hip",
     "kernels/gemm_bf16.hip",
     "kernels/swiglu.hip",
     "kernels/colsum.hip",
