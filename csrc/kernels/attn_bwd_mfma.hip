@@ -504,7 +504,9 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                        (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
                        dlse_ptr, delta.data_ptr<float>(), H, S, rows, dos, osr);
   const int mode = attn_bwd_mode();
-  if ((mode == 8) || (mode == 0 && S >= 256)) {
+  // auto: 8-wave at D=64 (measured ~even to +2%); 4-wave at D=128 where
+  // the 8-wave dkv's 36 B/lane spill measured net-slower (NOTES.md r2)
+  if ((mode == 8) || (mode == 0 && S >= 256 && D <= 64)) {
     attn_bwd_dq_8w(dout, q, k, v, lse, delta, dq, B, H, Hkv, S, D, scale,
                    causal);
     attn_bwd_dkv_8w(dout, q, k, v, lse, delta, dk, dv, B, H, Hkv, S, D,
