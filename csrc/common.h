@@ -189,6 +189,29 @@ DEVINL bf16x8_quad tr_read_quad(unsigned a0, unsigned a1, unsigned b0,
   return q;
 }
 
+// TWO fragments (4 tr reads) behind one wait: half the forced-live
+// output registers of tr_read_quad (16 -> 8) for kernels at the VGPR
+// budget edge, still batched enough to keep the LDS pipe busy.
+struct bf16x8_pair { bf16x8 f[2]; };
+
+DEVINL bf16x8_pair tr_read_pair(unsigned a0, unsigned a1, unsigned b0,
+                                unsigned b1) {
+  unsigned long long r0, r1, r2, r3;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %4\n\t"
+      "ds_read_b64_tr_b16 %1, %5\n\t"
+      "ds_read_b64_tr_b16 %2, %6\n\t"
+      "ds_read_b64_tr_b16 %3, %7\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(r0), "=&v"(r1), "=&v"(r2), "=&v"(r3)
+      : "v"(a0), "v"(a1), "v"(b0), "v"(b1));
+  union { struct { unsigned long long lo, hi; } u; bf16x8 v; } t;
+  bf16x8_pair p;
+  t.u.lo = r0; t.u.hi = r1; p.f[0] = t.v;
+  t.u.lo = r2; t.u.hi = r3; p.f[1] = t.v;
+  return p;
+}
+
 // Grid sizing guideline 11: cap grid, grid-stride the rest.
 DEVINL long grid_stride() { return (long)gridDim.x * blockDim.x; }
 

@@ -124,18 +124,18 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
 
   const int qrow_g = q0 + 8 * lq + wid;     // interleaved q row
   const int qrow = min(qrow_g, S - 1);
-  // Q/dO fragments: D=64 pinned (32 VGPRs); D=128 reloaded from L1 per
-  // sub-tile (the lane's two 256-B rows stay L1-resident)
-  constexpr bool QREG = (D <= 64);
+  // Q/dO fragments: D=64 pins both (32 VGPRs); D=128 pins Q and
+  // reloads dO from L1 per sub-tile — reloading BOTH measured dq 53%
+  // slower (16 loads/lane/sub on the QK^T critical path), pinning both
+  // spills 179 regs at the 2-wave/SIMD budget.
+  constexpr bool DOREG = (D <= 64);
   const short* qrow_ptr = q + qb + (long)qrow * qs.ss + 8 * half;
   const short* dorow_ptr = dout + dob + (long)qrow * dos.ss + 8 * half;
-  bf16x8_t qf[QREG ? NSL : 1], dof[QREG ? NSL : 1];
-  if constexpr (QREG) {
-    #pragma unroll
-    for (int s = 0; s < NSL; ++s) {
-      qf[s] = *(const bf16x8_t*)(qrow_ptr + 16 * s);
-      dof[s] = *(const bf16x8_t*)(dorow_ptr + 16 * s);
-    }
+  bf16x8_t qf[NSL], dof[DOREG ? NSL : 1];
+  #pragma unroll
+  for (int s = 0; s < NSL; ++s) {
+    qf[s] = *(const bf16x8_t*)(qrow_ptr + 16 * s);
+    if constexpr (DOREG) dof[s] = *(const bf16x8_t*)(dorow_ptr + 16 * s);
   }
   const float my_lse2 = lse[rhead + qrow] * LOG2E;
   const float my_delta = delta[rhead + qrow];
@@ -164,13 +164,15 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
       const int krow = min(cs + lq, S - 1);
       f32x16_t sacc = {}, dpacc = {};
       __builtin_amdgcn_s_setprio(1);
-      #pragma unroll
+      // unroll capped at D=128: a full unroll buffers all 8 k-slices'
+      // kf+vf loads (64 VGPRs) on top of the pinned Q/dO + accumulators
+      #pragma unroll (D <= 64 ? NSL : 4)
       for (int s = 0; s < NSL; ++s) {
         bf16x8_t kf = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
         bf16x8_t vf = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
-        const bf16x8_t qv = QREG ? qf[s] : *(const bf16x8_t*)(qrow_ptr + 16 * s);
-        const bf16x8_t dov = QREG ? dof[s] : *(const bf16x8_t*)(dorow_ptr + 16 * s);
-        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qv, sacc, 0, 0, 0);
+        const bf16x8_t dov =
+            DOREG ? dof[s] : *(const bf16x8_t*)(dorow_ptr + 16 * s);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
         dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dov, dpacc, 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
@@ -199,26 +201,22 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
       unsigned dsf[2][4];
       pack_frags(ds, dsf);
 
-      // dQ^T += K^T . dS^T per d-tile (K^T A-fragments by batched tr-read)
+      // dQ^T += K^T . dS^T per d-tile (K^T A-fragments by PAIR-batched
+      // tr-read — the quad form's 16 forced-live outputs pushed the
+      // pinned-Q/dO register budget over at D=128)
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
-      for (int t2 = 0; t2 < NDT; t2 += 2) {
+      for (int t = 0; t < NDT; ++t) {
         const unsigned qstep = 128u * (D / 16);
-        auto ka = [&](int t, int s) {
+        auto ka = [&](int s) {
           return ktr + 128u * (unsigned)((4 * s) * (D / 16) + 2 * t);
         };
-        bf16x8_quad kq = tr_read_quad(
-            ka(t2, 0), ka(t2, 0) + qstep, ka(t2, 1), ka(t2, 1) + qstep,
-            ka(t2 + 1, 0), ka(t2 + 1, 0) + qstep,
-            ka(t2 + 1, 1), ka(t2 + 1, 1) + qstep);
-        dqacc[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kq.f[0], *(const bf16x8_t*)dsf[0], dqacc[t2], 0, 0, 0);
-        dqacc[t2] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kq.f[1], *(const bf16x8_t*)dsf[1], dqacc[t2], 0, 0, 0);
-        dqacc[t2 + 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kq.f[2], *(const bf16x8_t*)dsf[0], dqacc[t2 + 1], 0, 0, 0);
-        dqacc[t2 + 1] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-            kq.f[3], *(const bf16x8_t*)dsf[1], dqacc[t2 + 1], 0, 0, 0);
+        bf16x8_pair kq = tr_read_pair(ka(0), ka(0) + qstep,
+                                      ka(1), ka(1) + qstep);
+        dqacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kq.f[0], *(const bf16x8_t*)dsf[0], dqacc[t], 0, 0, 0);
+        dqacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            kq.f[1], *(const bf16x8_t*)dsf[1], dqacc[t], 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
     }
@@ -383,25 +381,27 @@ attn_bwd_dkv_8w_kernel(const short* __restrict__ dout,
         pack_frags((const float*)&sacc, pf);
         pack_frags((const float*)&dpacc, dsf);
 
-        // dV^T += P^T . dO_cols ; dK^T += dS^T . Q_cols — one batched
-        // tr-read covers both tensors' fragments per d-tile
+        // dV^T += P^T . dO_cols ; dK^T += dS^T . Q_cols — PAIR-batched
+        // tr-reads (4 per wait): the quad form's 16 forced-live output
+        // regs were half the dkv spill at the D=128 budget edge
         __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int t = 0; t < NDT; ++t) {
           const unsigned qstep = 128u * (D / 16);
           const unsigned o0 = 128u * (unsigned)(2 * t);
           const unsigned o1 = o0 + 128u * (unsigned)(4 * (D / 16));  // s=1
-          bf16x8_quad cq = tr_read_quad(
-              dotr + o0, dotr + o0 + qstep, dotr + o1, dotr + o1 + qstep,
+          bf16x8_pair dop = tr_read_pair(
+              dotr + o0, dotr + o0 + qstep, dotr + o1, dotr + o1 + qstep);
+          dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *(const bf16x8_t*)pf[0], dop.f[0], dvacc[t], 0, 0, 0);
+          dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *(const bf16x8_t*)pf[1], dop.f[1], dvacc[t], 0, 0, 0);
+          bf16x8_pair qp = tr_read_pair(
               qtr + o0, qtr + o0 + qstep, qtr + o1, qtr + o1 + qstep);
-          dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)pf[0], cq.f[0], dvacc[t], 0, 0, 0);
-          dvacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)pf[1], cq.f[1], dvacc[t], 0, 0, 0);
           dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)dsf[0], cq.f[2], dkacc[t], 0, 0, 0);
+              *(const bf16x8_t*)dsf[0], qp.f[0], dkacc[t], 0, 0, 0);
           dkacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-              *(const bf16x8_t*)dsf[1], cq.f[3], dkacc[t], 0, 0, 0);
+              *(const bf16x8_t*)dsf[1], qp.f[1], dkacc[t], 0, 0, 0);
         }
         __builtin_amdgcn_s_setprio(0);
       }
