@@ -164,9 +164,12 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
       const int krow = min(cs + lq, S - 1);
       f32x16_t sacc = {}, dpacc = {};
       __builtin_amdgcn_s_setprio(1);
-      // unroll capped at D=128: a full unroll buffers all 8 k-slices'
-      // kf+vf loads (64 VGPRs) on top of the pinned Q/dO + accumulators
-      #pragma unroll (D <= 64 ? NSL : 4)
+      // FULL unroll (a partial '#pragma unroll N' leaves a rolled loop
+      // whose qf[s] becomes a per-element compare-select tree — rule
+      // 20's trap in select form, measured 14x slower); the mid-loop
+      // memory-clobber asm splits the scheduling region so the
+      // compiler cannot buffer all 8 k-slices' loads at once
+      #pragma unroll
       for (int s = 0; s < NSL; ++s) {
         bf16x8_t kf = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
         bf16x8_t vf = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
@@ -174,6 +177,9 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
             DOREG ? dof[s] : *(const bf16x8_t*)(dorow_ptr + 16 * s);
         sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
         dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dov, dpacc, 0, 0, 0);
+        if constexpr (D > 64) {
+          if (s == NSL / 2 - 1) asm volatile("" ::: "memory");
+        }
       }
       __builtin_amdgcn_s_setprio(0);
 
