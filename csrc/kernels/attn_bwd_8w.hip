@@ -124,11 +124,10 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
 
   const int qrow_g = q0 + 8 * lq + wid;     // interleaved q row
   const int qrow = min(qrow_g, S - 1);
-  // Q/dO fragments: D=64 pins both (32 VGPRs); D=128 pins Q and
-  // reloads dO from L1 per sub-tile — reloading BOTH measured dq 53%
-  // slower (16 loads/lane/sub on the QK^T critical path), pinning both
-  // spills 179 regs at the 2-wave/SIMD budget.
-  constexpr bool DOREG = (D <= 64);
+  // Q/dO fragments pinned in registers (64 VGPRs at D=128): reloading
+  // either per sub-tile measured dq 14-53% slower; with the mid-loop
+  // scheduling split below the allocation stays at 198 regs, 0 spill.
+  constexpr bool DOREG = true;
   const short* qrow_ptr = q + qb + (long)qrow * qs.ss + 8 * half;
   const short* dorow_ptr = dout + dob + (long)qrow * dos.ss + 8 * half;
   bf16x8_t qf[NSL], dof[DOREG ? NSL : 1];
