@@ -151,25 +151,34 @@ __global__ __launch_bounds__(256) void linear_fwd_kernel(
   }
 }
 
-// ---- 256x256 8-wave phase-interleaved variant (toward the guide's
-// 8-phase class).  512 threads = 8 waves (2M x 4N), per-wave output
-// 128x64 = 8x4 fragments of 16x16x32 MFMA; BK=64; LDS = 2 x 64 KiB
+// ---- 256x256 8-wave counted-vmcnt variant (the guide's 8-phase
+// class).  512 threads = 8 waves (2M x 4N), per-wave output 128x64 =
+// 8x4 fragments of 16x16x32 MFMA; BK=64; LDS = 2 x 64 KiB
 // double-buffered tiles (A 32K + B 32K each).
 //
-// Schedule per iteration (tile t, buffer cur):
-//   ph0..ph3 (quadrant qm=ph>>1, qn=ph&1):
-//     { issue 2 glds of tile t+1 into buf cur^1 ; ds_read the
-//       quadrant's A/B fragments ; raw s_barrier (phase-locks the 8
-//       waves -> T5 role-split) ; setprio(1) ; 16 MFMA ; setprio(0) }
-//   end: s_waitcnt vmcnt(0) ; raw s_barrier ; swap
-// Prefetch distance is one tile: tile t+1's 8 glds are spread over
-// iteration t's 4 phases (~64 MFMA of cover), and the single vmcnt
-// drain per K-tile amortizes 8x better than the 128^2 kernel's
-// per-K-step drain.  Gains over the 128^2 kernel: 2x arithmetic
-// intensity per staged byte, 8-wave role-split (setprio pays — T5),
-// one barrier per phase instead of two per K-step.  The guide's full
-// counted-vmcnt half-tile ring (never draining to 0 in the loop) is a
-// further step not taken here.
+// The r1 4-phase version drained vmcnt(0) once per K-tile — the guide
+// measures counted-vs-drain0 at +38%@4k / +73%@8k (T3/T4), so this
+// version NEVER drains in the loop.  Key idea: each wave stages
+// exactly the data IT will read (its A-half rows and B-quarter cols),
+// split so that the pair of glds issued in phase p carries the data
+// quadrant p of the NEXT tile needs.  Then a per-wave counted
+// s_waitcnt before each phase barrier covers the cross-wave hand-off:
+// every wave passes its own wait before the barrier that precedes the
+// first read of what it staged.
+//
+// Phase p (quadrant qm=p>>1, qn=p&1) of tile t:
+//   { issue 2 glds of tile t+1's quadrant-p data into buf cur^1 ;
+//     s_waitcnt vmcnt(W[p]) ; raw s_barrier ; ds_read the quadrant's
+//     A/B fragments ; setprio(1) ; 16 MFMA ; setprio(0) }
+// W = [4,4,6,8]: derived from the latest-issuing contributor of each
+// quadrant's data (A quadrant qm staged in phases 2qm,2qm+1; B
+// quadrant qn in phases qn,2+qn of the previous iteration).  The last
+// tile issues nothing, so its waits fall to [2,0,0,0].
+template <int N_>
+DEVINL void waitcnt_vm() {
+  asm volatile("s_waitcnt vmcnt(%0)" ::"n"(N_) : "memory");
+}
+
 template <int EPI, bool HAS_BIAS>
 __global__ void __launch_bounds__(512) linear_fwd256_kernel(
     const short* __restrict__ X, const short* __restrict__ W,
@@ -193,24 +202,34 @@ __global__ void __launch_bounds__(512) linear_fwd256_kernel(
   const int nt = K / 64;
   f32x4_t acc[8][4] = {};
 
-  // stage one phase-quarter (16 KiB) of tile t into buf: phase p covers
-  // combined slots [p*1024, p*1024+1024) of the 4096x16B tile image
-  // (A = slots 0..2047, B = 2048..4095), 2 glds per thread.
-  auto stage_q = [&](int buf, int t, int p) {
-    #pragma unroll
-    for (int g = 0; g < 2; ++g) {
-      const int slot = p * 1024 + g * 512 + w * 64 + l;
-      const int op = slot >> 11;            // 0 = A, 1 = B
-      const int c = slot & 2047;            // chunk within operand
-      const int row = c >> 3;
-      const int kb = (c & 7) ^ (row & 7);
-      const short* src = (op ? W + (long)n0 * K : X + (long)m0 * K);
-      glds16(src + (long)row * K + (long)t * 64 + kb * 8,
-             smem + (long)buf * (2 * 256 * 64) + (long)slot * 8);
+  // Stage this wave's 2 glds for phase p of tile t into buf:
+  //  A: 8 rows of the quadrant-qm A-half this wave reads
+  //     (local rows wm*128 + qm*64 + (wn*2+qn)*8 + lane>>3)
+  //  B: 8 N-rows of the quadrant-qn B-quarter this wave reads
+  //     (local rows wn*64 + qn*32 + (wm*2+qm)*8 + lane>>3)
+  // Both lane-linear in LDS (slot = row*8 + (l&7) = rowbase*8 + l);
+  // the chunk XOR-swizzle rides on the SOURCE address (rule 21).
+  auto stage_pair = [&](int buf, int t, int p) {
+    const int qm = p >> 1, qn = p & 1;
+    short* base = smem + (long)buf * (2 * 256 * 64);
+    {
+      const int rowl = wm * 128 + qm * 64 + (wn * 2 + qn) * 8 + (l >> 3);
+      const int kb = (l & 7) ^ (rowl & 7);
+      // dest slot = rowbase*8 + l (lane-linear); *8 shorts per slot
+      glds16(X + (long)(m0 + rowl) * K + (long)t * 64 + kb * 8,
+             base + (long)((rowl - (l >> 3)) * 8 + l) * 8);
+    }
+    {
+      const int rowl = wn * 64 + qn * 32 + (wm * 2 + qm) * 8 + (l >> 3);
+      const int kb = (l & 7) ^ (rowl & 7);
+      glds16(W + (long)(n0 + rowl) * K + (long)t * 64 + kb * 8,
+             base + 256 * 64 + (long)((rowl - (l >> 3)) * 8 + l) * 8);
     }
   };
 
-  stage_q(0, 0, 0); stage_q(0, 0, 1); stage_q(0, 0, 2); stage_q(0, 0, 3);
+  // prologue: tile 0 fully staged, drained once
+  stage_pair(0, 0, 0); stage_pair(0, 0, 1);
+  stage_pair(0, 0, 2); stage_pair(0, 0, 3);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
@@ -222,7 +241,19 @@ __global__ void __launch_bounds__(512) linear_fwd256_kernel(
     #pragma unroll
     for (int ph = 0; ph < 4; ++ph) {
       const int qm = ph >> 1, qn = ph & 1;
-      if (prefetch) stage_q(cur ^ 1, t + 1, ph);
+      if (prefetch) {
+        stage_pair(cur ^ 1, t + 1, ph);
+        // counted wait: tile t's quadrant-ph data landed, up to W[ph]
+        // of the newest prefetch loads still in flight (never 0)
+        if (ph == 0) waitcnt_vm<4>();
+        else if (ph == 1) waitcnt_vm<4>();
+        else if (ph == 2) waitcnt_vm<6>();
+        else waitcnt_vm<8>();
+      } else {
+        if (ph == 0) waitcnt_vm<2>();
+        else waitcnt_vm<0>();
+      }
+      __builtin_amdgcn_s_barrier();   // all waves' waits done -> reads safe
       bf16x8_t af[4][2], bf[2][2];
       #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
@@ -238,7 +269,6 @@ __global__ void __launch_bounds__(512) linear_fwd256_kernel(
           bf[ni][kk] = *(const bf16x8_t*)&sB[(col * 8 + (kb ^ (col & 7))) * 8];
         }
       }
-      __builtin_amdgcn_s_barrier();   // phase-lock the 8 waves
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
@@ -252,9 +282,6 @@ __global__ void __launch_bounds__(512) linear_fwd256_kernel(
                     0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
     }
-    // next tile's 8 glds must have landed before its first ds_read
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __builtin_amdgcn_s_barrier();
     cur ^= 1;
   }
 

@@ -217,6 +217,35 @@ def test_attention_fwd_bwd(ext, B, H, Hkv, S, D, causal):
 
 
 # ------------------------------------------------------- K5 MFMA linear
+@pytest.mark.parametrize("epi", ["none", "gelu", "resid"])
+def test_linear_fwd256_vs_fp32(ext, epi):
+    """The 256^2 counted-vmcnt kernel needs >=224 blocks to dispatch;
+    M=4096 x N=3584 = 224 tiles exercises it (K=192 covers the last-tile
+    wait path: 3 K-tiles)."""
+    M, N, K = 4096, 3584, 192
+    torch.manual_seed(12)
+    x = torch.randn(M, K)
+    w = torch.randn(N, K) * 0.05
+    b = torch.randn(N)
+    res = torch.randn(M, N)
+    xb, wb, bb, rb = (_bf(t).cuda() for t in (x, w, b, res))
+    ref = _bf(x).float() @ _bf(w).float().t() + _bf(b).float()
+    if epi == "none":
+        (y,) = ext.linear_fwd(xb, wb, bb, 0, None)
+        want = ref
+    elif epi == "gelu":
+        y, pre = ext.linear_fwd(xb, wb, bb, 1, None)
+        assert torch.allclose(pre.cpu().float(), ref, rtol=3e-2, atol=3e-2)
+        want = torch.nn.functional.gelu(ref, approximate="tanh")
+    else:
+        (y,) = ext.linear_fwd(xb, wb, bb, 2, rb)
+        want = ref + _bf(res).float()
+    got = y.cpu().float()
+    torch.cuda.synchronize()
+    assert torch.allclose(got, want, rtol=3e-2, atol=3e-2), \
+        (got - want).abs().max().item()
+
+
 @pytest.mark.parametrize("M,N,K", [(128, 128, 64), (256, 384, 192), (512, 768, 320)])
 @pytest.mark.parametrize("epi", ["none", "gelu", "resid"])
 def test_linear_fwd_vs_fp32(ext, M, N, K, epi):
