@@ -214,14 +214,14 @@ __global__ void __launch_bounds__(512) linear_fwd256_kernel(
     short* base = smem + (long)buf * (2 * 256 * 64);
     {
       const int rowl = wm * 128 + qm * 64 + (wn * 2 + qn) * 8 + (l >> 3);
-      const int kb = (l & 7) ^ (rowl & 7);
+      const int kb = (l & 7) ^ ((rowl >> 1) & 7);
       // dest slot = rowbase*8 + l (lane-linear); *8 shorts per slot
       glds16(X + (long)(m0 + rowl) * K + (long)t * 64 + kb * 8,
              base + (long)((rowl - (l >> 3)) * 8 + l) * 8);
     }
     {
       const int rowl = wn * 64 + qn * 32 + (wm * 2 + qm) * 8 + (l >> 3);
-      const int kb = (l & 7) ^ (rowl & 7);
+      const int kb = (l & 7) ^ ((rowl >> 1) & 7);
       glds16(W + (long)(n0 + rowl) * K + (long)t * 64 + kb * 8,
              base + 256 * 64 + (long)((rowl - (l >> 3)) * 8 + l) * 8);
     }
@@ -239,12 +239,12 @@ __global__ void __launch_bounds__(512) linear_fwd256_kernel(
     #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const int row = wm * 128 + qm * 64 + mi * 16 + (l & 15);
-      af[mi] = *(const bf16x8_t*)&sA_[(row * 8 + (kb ^ (row & 7))) * 8];
+      af[mi] = *(const bf16x8_t*)&sA_[(row * 8 + (kb ^ ((row >> 1) & 7))) * 8];
     }
     #pragma unroll
     for (int ni = 0; ni < 2; ++ni) {
       const int col = wn * 64 + qn * 32 + ni * 16 + (l & 15);
-      bf[ni] = *(const bf16x8_t*)&sB_[(col * 8 + (kb ^ (col & 7))) * 8];
+      bf[ni] = *(const bf16x8_t*)&sB_[(col * 8 + (kb ^ ((col >> 1) & 7))) * 8];
     }
   };
   auto mfma_kk = [&](bf16x8_t (&af)[4], bf16x8_t (&bf)[2], int qm, int qn) {
