@@ -277,17 +277,21 @@ __global__ void __launch_bounds__(512) linear_fwd256_kernel(
     #pragma unroll
     for (int ph = 0; ph < 4; ++ph) {
       const int qm = ph >> 1, qn = ph & 1;
-      if (prefetch) {
-        stage_pair(cur ^ 1, t + 1, ph);
-        // counted wait covering the NEXT phase's read target: never 0
-        if (ph == 0) waitcnt_vm<2>();
-        else if (ph == 1) waitcnt_vm<4>();
-        else if (ph == 2) waitcnt_vm<6>();
-        else waitcnt_vm<2>();
-      } else {
-        if (ph < 3) waitcnt_vm<0>();
-      }
+      // Burst-issue ALL of tile t+1 at ph0 (buf cur^1 is free once the
+      // ph0 barrier passes): every load then has 2-4 phases (>HBM
+      // latency) to land, so both waits below are near-free.  The
+      // per-phase spread tried first gave the last-issued pair only
+      // 1-2 phases of cover and parked waves 33% of the time
+      // (SQ_WAIT_ANY, profiles/pmc_gemm).
+      if (ph == 0) waitcnt_vm<0>();          // tile t fully landed
+      else if (ph == 3 && prefetch) waitcnt_vm<2>();  // t+1 q0 for lookahead
       __builtin_amdgcn_s_barrier();   // all waves' waits done -> reads safe
+      if (ph == 0 && prefetch) {
+        stage_pair(cur ^ 1, t + 1, 0);
+        stage_pair(cur ^ 1, t + 1, 1);
+        stage_pair(cur ^ 1, t + 1, 2);
+        stage_pair(cur ^ 1, t + 1, 3);
+      }
       const int nqm = ((ph + 1) & 3) >> 1, nqn = (ph + 1) & 1;
       // kk=0: read (ph,kk1) into B; MFMA (ph,kk0) from A
       read_kk(afB, bfB, sA, sB, qm, qn, 1);
