@@ -46,6 +46,63 @@ def _worker(rank, world, store_path, ckpt_dir, q):
     dist.destroy_process_group()
 
 
+def _worker_stream(rank, world, store_path, ckpt_dir, q):
+    """Phase 2: a FRESH ws=2 world resumes the ws=2 checkpoint — the
+    same-world STREAMING load path (own-shard read + owner broadcasts)."""
+    from scaffold.trainer import Trainer
+    from scaffold.ckpt import layout
+
+    store = dist.FileStore(store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
+                      backend="gloo", device=torch.device("cpu"), initialized=True)
+    path = layout.latest_checkpoint(ckpt_dir)
+    cfg = _cfg(ckpt_dir, resume=path)
+    cfg.seed = 777  # different init: the load must overwrite everything
+    t = Trainer(cfg, ctx=ctx)
+    params = torch.cat([p.detach().flatten() for p in t.raw_model.parameters()])
+    moms = torch.cat([m.flatten() for m in t.optimizer.exp_avg])
+    q.put((rank, t.step, params.numpy(), moms.numpy()))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(240)
+def test_ws2_streaming_resume_matches(tmp_path):
+    """ws=2 save -> ws=2 resume through the streaming path: every rank's
+    full state bit-matches the saved rank-0 state."""
+    ctxmp = mp.get_context("spawn")
+    q = ctxmp.Queue()
+    procs = [
+        ctxmp.Process(target=_worker,
+                      args=(r, WORLD, str(tmp_path / "fs1"), str(tmp_path / "ck"), q))
+        for r in range(WORLD)
+    ]
+    for p in procs:
+        p.start()
+    _path, params_saved, moms_saved = q.get(timeout=120)
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+
+    q2 = ctxmp.Queue()
+    procs = [
+        ctxmp.Process(target=_worker_stream,
+                      args=(r, WORLD, str(tmp_path / "fs2"), str(tmp_path / "ck"), q2))
+        for r in range(WORLD)
+    ]
+    for p in procs:
+        p.start()
+    results = [q2.get(timeout=120) for _ in range(WORLD)]
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    for rank, step, params, moms in results:
+        assert step == 3
+        assert torch.equal(torch.from_numpy(params), torch.from_numpy(params_saved)), rank
+        assert torch.equal(torch.from_numpy(moms), torch.from_numpy(moms_saved)), rank
+
+
 @pytest.mark.timeout(180)
 def test_ws2_checkpoint_loads_into_ws1(tmp_path):
     from scaffold.trainer import Trainer
