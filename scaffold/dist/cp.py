@@ -66,6 +66,80 @@ def cp_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return _merge(partials, lses)
 
 
+class _RingShift(torch.autograd.Function):
+    """Differentiable ring shift: out_r = in_{(r-1) mod world}.
+
+    Forward sends this rank's tensor to (r+1) and receives from (r-1);
+    backward reverses the ring (grad of the received tensor returns to
+    its sender).  Point-to-point send/recv — exactly the xGMI transport
+    shape (7 p2p links per GPU); at any moment only ONE extra kv shard
+    is live per rank instead of the all-gather form's world shards, and
+    the next shard's transfer overlaps this shard's attention partial.
+    """
+
+    @staticmethod
+    def forward(ctx, t, rank, world, group):
+        import torch.distributed as dist
+
+        ctx.ring = (rank, world, group)
+        nxt = (rank + 1) % world
+        prv = (rank - 1 + world) % world
+        out = torch.empty_like(t)
+        send = t.contiguous()
+        req = dist.isend(send, dst=nxt, group=group)
+        dist.recv(out, src=prv, group=group)
+        req.wait()
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        import torch.distributed as dist
+
+        rank, world, group = ctx.ring
+        nxt = (rank + 1) % world
+        prv = (rank - 1 + world) % world
+        gin = torch.empty_like(g)
+        req = dist.isend(g.contiguous(), dst=prv, group=group)
+        dist.recv(gin, src=nxt, group=group)
+        req.wait()
+        return gin, None, None, None
+
+
+def cp_attention_ring(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                      rank: int, world: int, causal: bool = True,
+                      group=None) -> torch.Tensor:
+    """Ring-pass schedule of the same lse-merge as cp_attention: kv
+    shards travel around the ring one hop per step while each rank
+    accumulates its partials — peak extra memory is ONE kv shard (vs
+    world shards for the all-gather form) and each hop's transfer
+    overlaps the previous shard's attention compute.  Math identical
+    (exact lse merge); dk/dv return to their owners through the
+    reversed ring in backward."""
+    if world <= 1:
+        return ops.attention(q, k, v, causal)
+
+    kv = torch.stack([k, v], dim=0)
+    partials, lses = [], []
+    for s in range(world):
+        j = (rank - s + world) % world
+        if (not causal) or j <= rank:
+            o_j, lse_j = ops.attention_lse(
+                q, kv[0], kv[1], causal=(causal and j == rank))
+            partials.append(o_j)
+            lses.append(lse_j)
+        if s < world - 1:
+            kv = _RingShift.apply(kv, rank, world, group)
+    out = _merge(partials, lses)
+    if causal:
+        # Tie the final kv into EVERY rank's graph (zero contribution):
+        # a rank that skipped late shards under causality would never
+        # fire the shift chain's backward, while its peers block on the
+        # matching p2p sends of dk/dv — the ring must run symmetrically
+        # in backward on all ranks.
+        out = out + (kv.flatten()[0] * 0).to(out.dtype)
+    return out
+
+
 def ulysses_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                       rank: int, world: int, causal: bool = True,
                       group=None) -> torch.Tensor:

@@ -27,12 +27,14 @@ class ExpertParallelMoE(nn.Module):
     """Top-1 MoE FFN; `n_experts` must be divisible by `world`."""
 
     def __init__(self, d_model: int, d_ff: int, n_experts: int, world: int,
-                 rank: int, group=None):
+                 rank: int, group=None, dispatch: str = "alltoall"):
         super().__init__()
         assert n_experts % world == 0
+        assert dispatch in ("alltoall", "dense")
         self.world = world
         self.rank = rank
         self.group = group
+        self.dispatch = dispatch
         self.n_experts = n_experts
         self.e_local = n_experts // world
         self.e0 = rank * self.e_local
@@ -50,6 +52,67 @@ class ExpertParallelMoE(nn.Module):
         return h @ self.w_down[i_local].t()
 
     def forward(self, x_local: torch.Tensor) -> torch.Tensor:
+        import torch.distributed as dist
+
+        if (self.dispatch == "alltoall" and self.world > 1
+                and dist.is_initialized()):
+            return self._forward_alltoall(x_local)
+        return self._forward_dense(x_local)
+
+    def _forward_alltoall(self, x_local: torch.Tensor) -> torch.Tensor:
+        """Sparse dispatch: exchange only the ROUTED tokens (SURVEY
+        §2.4 / NOTES r1 planned form).  Transport per rank is its own T
+        tokens out + back, vs the dense form's (world-1)*T all_gather +
+        world*T all_reduce — and it uses all_to_all, the collective
+        xGMI serves with world-1 concurrent p2p links.
+
+        Gating runs on LOCAL tokens; the gate weight travels nowhere
+        (the prob multiplies the RETURNED expert output, keeping the
+        gate gradient rank-local).  Expert ids ride a separate non-grad
+        all_to_all with the same splits."""
+        import torch.distributed as dist
+        from torch.distributed import nn as dist_nn
+
+        probs = torch.softmax(self.gate(x_local), dim=-1)
+        top_p, top_e = probs.max(dim=-1)                     # (T,)
+        dest = top_e // self.e_local                         # owner rank
+
+        order = torch.argsort(dest, stable=True)
+        inv = torch.argsort(order)
+        x_s = x_local[order]
+        e_s = top_e[order]
+
+        send_counts = torch.bincount(dest, minlength=self.world)
+        recv_counts = torch.empty_like(send_counts)
+        dist.all_to_all_single(recv_counts, send_counts, group=self.group)
+        send_l = send_counts.tolist()
+        recv_l = recv_counts.tolist()
+
+        n_recv = int(sum(recv_l))
+        x_recv = torch.empty(n_recv, x_local.shape[1], dtype=x_local.dtype,
+                             device=x_local.device)
+        x_recv = dist_nn.functional.all_to_all_single(
+            x_recv, x_s.contiguous(), output_split_sizes=recv_l,
+            input_split_sizes=send_l, group=self.group)
+        e_recv = torch.empty(n_recv, dtype=e_s.dtype, device=e_s.device)
+        dist.all_to_all_single(e_recv, e_s.contiguous(),
+                               output_split_sizes=recv_l,
+                               input_split_sizes=send_l, group=self.group)
+
+        y_recv = torch.zeros_like(x_recv)
+        for i in range(self.e_local):
+            sel = (e_recv == self.e0 + i).nonzero(as_tuple=True)[0]
+            if sel.numel() == 0:
+                continue
+            y_recv = y_recv.index_add(0, sel, self._expert(i, x_recv[sel]))
+
+        y_back = torch.empty_like(x_s)
+        y_back = dist_nn.functional.all_to_all_single(
+            y_back, y_recv.contiguous(), output_split_sizes=send_l,
+            input_split_sizes=recv_l, group=self.group)
+        return top_p.unsqueeze(-1) * y_back[inv]
+
+    def _forward_dense(self, x_local: torch.Tensor) -> torch.Tensor:
         import torch.distributed as dist
 
         T = x_local.shape[0]

@@ -23,8 +23,8 @@ def _make_inputs():
     return q, k, v, do
 
 
-def _worker(rank, world, store_path, causal, q_ret):
-    from scaffold.dist.cp import cp_attention
+def _worker(rank, world, store_path, causal, q_ret, variant="allgather"):
+    from scaffold.dist.cp import cp_attention, cp_attention_ring
 
     store = dist.FileStore(store_path, world)
     dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
@@ -33,7 +33,8 @@ def _worker(rank, world, store_path, causal, q_ret):
     qs = q[:, :, rank * sl:(rank + 1) * sl].clone().requires_grad_(True)
     ks = k[:, :, rank * sl:(rank + 1) * sl].clone().requires_grad_(True)
     vs = v[:, :, rank * sl:(rank + 1) * sl].clone().requires_grad_(True)
-    o = cp_attention(qs, ks, vs, rank, world, causal=causal)
+    fn = cp_attention_ring if variant == "ring" else cp_attention
+    o = fn(qs, ks, vs, rank, world, causal=causal)
     o.backward(do[:, :, rank * sl:(rank + 1) * sl])
     q_ret.put((rank, o.detach().numpy(), qs.grad.numpy(),
                ks.grad.numpy(), vs.grad.numpy()))
@@ -42,13 +43,15 @@ def _worker(rank, world, store_path, causal, q_ret):
 
 
 @pytest.mark.timeout(180)
+@pytest.mark.parametrize("variant", ["allgather", "ring"])
 @pytest.mark.parametrize("causal", [True, False])
-def test_cp_matches_full_attention(tmp_path, causal):
+def test_cp_matches_full_attention(tmp_path, causal, variant):
     ctxmp = mp.get_context("spawn")
     ret = ctxmp.Queue()
     procs = [
         ctxmp.Process(target=_worker,
-                      args=(r, WORLD, str(tmp_path / f"fs{causal}"), causal, ret))
+                      args=(r, WORLD, str(tmp_path / f"fs{causal}{variant}"),
+                            causal, ret, variant))
         for r in range(WORLD)
     ]
     for p in procs:

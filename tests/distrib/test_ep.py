@@ -27,12 +27,12 @@ def _seed_weights(moe, base):
     return X, dy
 
 
-def _worker(rank, world, store_path, q_ret):
+def _worker(rank, world, store_path, q_ret, dispatch="dense"):
     from scaffold.dist.ep import ExpertParallelMoE
 
     store = dist.FileStore(store_path, world)
     dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
-    moe = ExpertParallelMoE(DM, DF, E, world, rank)
+    moe = ExpertParallelMoE(DM, DF, E, world, rank, dispatch=dispatch)
     X, dy = _seed_weights(moe, rank * (E // world))
     xl = X[rank * T:(rank + 1) * T].clone().requires_grad_(True)
     y = moe(xl)
@@ -46,12 +46,14 @@ def _worker(rank, world, store_path, q_ret):
 
 
 @pytest.mark.timeout(180)
-def test_ep_moe_matches_full(tmp_path):
+@pytest.mark.parametrize("dispatch", ["dense", "alltoall"])
+def test_ep_moe_matches_full(tmp_path, dispatch):
     from scaffold.dist.ep import ExpertParallelMoE
 
     ctxmp = mp.get_context("spawn")
     ret = ctxmp.Queue()
-    procs = [ctxmp.Process(target=_worker, args=(r, WORLD, str(tmp_path / "fs"), ret))
+    procs = [ctxmp.Process(target=_worker,
+                           args=(r, WORLD, str(tmp_path / f"fs{dispatch}"), ret, dispatch))
              for r in range(WORLD)]
     for p in procs:
         p.start()
