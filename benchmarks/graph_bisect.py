@@ -42,3 +42,12 @@ try_capture("fwd_bwd_static", fwd_bwd)
 def opt_only():
     t.optimizer.step(grad_scale=1.0)
 try_capture("opt", opt_only)
+
+# second attempts: benchmark=False (heuristic algo mode) and a retry
+# (first failure may have initialized whatever rejects capture)
+torch.backends.cudnn.benchmark = False
+for _ in range(2):
+    t._train_step(it, 1)
+torch.cuda.synchronize()
+try_capture("fwd_bwd_heuristic", fwd_bwd)
+try_capture("fwd_bwd_retry", fwd_bwd)
