@@ -64,6 +64,17 @@ std::vector<torch::Tensor> linear_fwd(torch::Tensor x, torch::Tensor w,
                                       c10::optional<torch::Tensor> bias,
                                       long epilogue,
                                       c10::optional<torch::Tensor> residual);
+std::vector<torch::Tensor> bn_fwd(torch::Tensor x,
+                                  c10::optional<torch::Tensor> res,
+                                  torch::Tensor weight, torch::Tensor bias,
+                                  c10::optional<torch::Tensor> running_mean,
+                                  c10::optional<torch::Tensor> running_var,
+                                  double momentum, double eps, bool relu);
+std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
+                                  c10::optional<torch::Tensor> res,
+                                  torch::Tensor weight, torch::Tensor bias,
+                                  torch::Tensor mean, torch::Tensor rstd,
+                                  bool relu);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_adamw", &fused_adamw, "fused multi-tensor AdamW (K1)");
@@ -104,4 +115,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "stream tensors to a safetensors shard via pinned D2H (C6)");
   m.def("linear_fwd", &linear_fwd,
         "MFMA linear fwd, fused bias+GELU/residual epilogue (K5)");
+  m.def("bn_fwd", &bn_fwd,
+        "fused bf16 NHWC BatchNorm fwd (+ReLU, +residual)",
+        py::arg("x"), py::arg("res"), py::arg("weight"), py::arg("bias"),
+        py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
+        py::arg("eps"), py::arg("relu"));
+  m.def("bn_bwd", &bn_bwd, "fused bf16 NHWC BatchNorm bwd",
+        py::arg("dy"), py::arg("x"), py::arg("res"), py::arg("weight"),
+        py::arg("bias"), py::arg("mean"), py::arg("rstd"), py::arg("relu"));
 }

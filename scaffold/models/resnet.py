@@ -18,14 +18,28 @@ from .. import ops
 
 
 class Fp32BatchNorm(nn.BatchNorm2d):
-    """BatchNorm computed in fp32 regardless of activation dtype.
+    """Fused BatchNorm (+optional ReLU, +optional residual add).
 
-    Trainer._to_mixed keeps BN params/running-stats fp32; this wrapper
-    casts the bf16 activation up and back so the module works under the
-    scaffold's explicit-bf16 regime (no autocast)."""
+    fp32 parameters/stats regardless of activation dtype (Trainer's
+    _to_mixed keeps BatchNorm2d subclasses fp32).  On GPU in training
+    mode this runs the fused bf16-NHWC HIP kernels (stats + apply in 3
+    activation passes, ReLU and the block's residual add folded in) —
+    MIOpen's fp32-only BN plus its cast copies plus the separate
+    add/relu passes were ~57% of the r1 headline step
+    (profiles/resnet_kstats_r2.txt).  CPU and eval mode use the fp32
+    eager reference."""
 
-    def forward(self, x):
-        return super().forward(x.float()).to(x.dtype)
+    def __init__(self, ch: int, relu: bool = False):
+        super().__init__(ch)
+        self.relu = relu
+
+    def forward(self, x, res=None):
+        if self.training and self.num_batches_tracked is not None:
+            self.num_batches_tracked += 1
+        return ops.batch_norm_act(
+            x, res, self.weight, self.bias, self.running_mean,
+            self.running_var, self.momentum, self.eps, self.relu,
+            self.training)
 
 
 class Bottleneck(nn.Module):
@@ -35,22 +49,21 @@ class Bottleneck(nn.Module):
         super().__init__()
         out_ch = width * self.expansion
         self.conv1 = nn.Conv2d(in_ch, width, 1, bias=False)
-        self.bn1 = Fp32BatchNorm(width)
+        self.bn1 = Fp32BatchNorm(width, relu=True)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1, bias=False)
-        self.bn2 = Fp32BatchNorm(width)
+        self.bn2 = Fp32BatchNorm(width, relu=True)
         self.conv3 = nn.Conv2d(width, out_ch, 1, bias=False)
-        self.bn3 = Fp32BatchNorm(out_ch)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn3 = Fp32BatchNorm(out_ch, relu=True)  # fused add+relu
         self.downsample = downsample
 
     def forward(self, x):
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
         if self.downsample is not None:
             identity = self.downsample(x)
-        return self.relu(out + identity)
+        # bn3 folds the residual add AND the final relu into its apply
+        return self.bn3(self.conv3(out), res=identity)
 
 
 class ResNet50(nn.Module):
@@ -61,8 +74,7 @@ class ResNet50(nn.Module):
         num_classes = cfg.model.num_classes
         self.in_ch = 64
         self.conv1 = nn.Conv2d(3, 64, 7, stride=2, padding=3, bias=False)
-        self.bn1 = Fp32BatchNorm(64)
-        self.relu = nn.ReLU(inplace=True)
+        self.bn1 = Fp32BatchNorm(64, relu=True)
         self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
         self.layer1 = self._make_layer(64, self.layers[0], stride=1)
         self.layer2 = self._make_layer(128, self.layers[1], stride=2)
@@ -97,7 +109,7 @@ class ResNet50(nn.Module):
         return nn.Sequential(*layers)
 
     def features(self, x):
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(self.bn1(self.conv1(x)))
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         return torch.flatten(self.avgpool(x), 1)
 

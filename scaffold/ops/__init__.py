@@ -32,6 +32,7 @@ __all__ = [
     "swiglu_packed",
     "lm_head_ce",
     "fused_adamw_step",
+    "batch_norm_act",
     "require_ext",
     "has_ext",
 ]
@@ -592,3 +593,63 @@ def fused_adamw_step(
         p.addcdiv_(ea / bc1, denom, value=-lr)
         if params_bf16 is not None and params_bf16:
             params_bf16[i].copy_(p.to(params_bf16[i].dtype))
+
+# ------------------------------------------------- fused BatchNorm(+ReLU)
+class _BatchNormAct(torch.autograd.Function):
+    """Training-mode bf16 NHWC BatchNorm with fused ReLU and optional
+    fused residual add (csrc/kernels/batchnorm.hip).  Replaces MIOpen's
+    fp32-only BN + the bf16<->fp32 cast copies + separate add/relu
+    passes that together were ~57%% of the r1 ResNet step
+    (profiles/resnet_kstats_r2.txt)."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, bias, running_mean, running_var,
+                momentum, eps, relu):
+        m = require_ext()
+        y, mean, rstd = m.bn_fwd(x, res, weight, bias, running_mean,
+                                 running_var, momentum, eps, relu)
+        ctx.save_for_backward(x, weight, bias, mean, rstd,
+                              res if res is not None else x.new_empty(0))
+        ctx.relu = relu
+        ctx.has_res = res is not None
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, bias, mean, rstd, res = ctx.saved_tensors
+        m = require_ext()
+        out = m.bn_bwd(dy, x, res if ctx.has_res else None, weight, bias,
+                       mean, rstd, ctx.relu)
+        if ctx.has_res:
+            dx, dres, dw, db = out
+        else:
+            (dx, dw, db), dres = out, None
+        return dx, dres, dw, db, None, None, None, None, None
+
+
+def _bn_act_eager(x, res, weight, bias, running_mean, running_var,
+                  momentum, eps, relu, training):
+    """CPU/eval reference: fp32 BN + add + relu."""
+    y = F.batch_norm(x.float(), running_mean, running_var, weight, bias,
+                     training, momentum, eps)
+    if res is not None:
+        y = y + res.float()
+    if relu:
+        y = F.relu(y)
+    return y.to(x.dtype)
+
+
+def batch_norm_act(x: torch.Tensor, res: Optional[torch.Tensor],
+                   weight: torch.Tensor, bias: torch.Tensor,
+                   running_mean: torch.Tensor, running_var: torch.Tensor,
+                   momentum: float, eps: float, relu: bool,
+                   training: bool) -> torch.Tensor:
+    if x.is_cuda and training:
+        return _BatchNormAct.apply(
+            x.contiguous(memory_format=torch.channels_last),
+            None if res is None else res.contiguous(memory_format=torch.channels_last),
+            weight, bias, running_mean, running_var, momentum, eps, relu)
+    # CPU path and eval mode (running-stat normalize): eager reference
+    return _bn_act_eager(x, res, weight, bias, running_mean, running_var,
+                         momentum, eps, relu, training)
+

@@ -38,6 +38,7 @@ _SOURCES = [
     "kernels/gemm_bf16.hip",
     "kernels/swiglu.hip",
     "kernels/colsum.hip",
+    "kernels/batchnorm.hip",
     "kernels/probe.hip",
 ]
 

@@ -472,3 +472,54 @@ def test_attention_lse_dlse_backward(ext):
         err = (a.detach().cpu().float() - b.detach().float()).abs().max().item()
         tol = 0.08 * max(1.0, b.detach().float().abs().max().item())
         assert err < tol, (n, err, tol)
+
+# ----------------------------------------------------- fused BatchNorm
+@pytest.mark.parametrize("C,relu,res", [
+    (64, True, False), (128, True, True), (256, False, False),
+])
+def test_batch_norm_act_vs_fp32(ext, C, relu, res):
+    torch.manual_seed(21)
+    N, H, W = 4, 14, 14
+    x32 = torch.randn(N, C, H, W) * 2 + 0.5
+    r32 = torch.randn(N, C, H, W) if res else None
+    w32 = (torch.rand(C) + 0.5).requires_grad_(True)
+    b32 = torch.randn(C).requires_grad_(True)
+    rm = torch.zeros(C)
+    rv = torch.ones(C)
+    mom, eps = 0.1, 1e-5
+
+    xg = _bf(x32).cuda().to(memory_format=torch.channels_last).requires_grad_(True)
+    rg = (_bf(r32).cuda().to(memory_format=torch.channels_last).requires_grad_(True)
+          if res else None)
+    wg = w32.detach().clone().cuda().requires_grad_(True)
+    bg = b32.detach().clone().cuda().requires_grad_(True)
+    rmg, rvg = rm.clone().cuda(), rv.clone().cuda()
+
+    from scaffold import ops
+
+    y = ops.batch_norm_act(xg, rg, wg, bg, rmg, rvg, mom, eps, relu, True)
+    dy = torch.randn(N, C, H, W)
+    y.backward(_bf(dy).cuda().to(memory_format=torch.channels_last))
+    torch.cuda.synchronize()
+
+    # fp32 reference on the bf16-quantized inputs
+    xr = _bf(x32).float().requires_grad_(True)
+    rr = _bf(r32).float().requires_grad_(True) if res else None
+    ref = torch.nn.functional.batch_norm(xr, rm, rv, w32, b32, True, mom, eps)
+    if res:
+        ref = ref + rr
+    if relu:
+        ref = torch.nn.functional.relu(ref)
+    ref.backward(_bf(dy).float())
+
+    assert torch.allclose(y.detach().cpu().float(), ref.detach(),
+                          rtol=3e-2, atol=3e-2),         (y.detach().cpu().float() - ref.detach()).abs().max()
+    assert torch.allclose(rmg.cpu(), rm, rtol=1e-3, atol=1e-4)
+    assert torch.allclose(rvg.cpu(), rv, rtol=1e-3, atol=1e-3)
+    for name, got, want in [("dx", xg.grad, xr.grad), ("dw", wg.grad, w32.grad),
+                            ("db", bg.grad, b32.grad)] +                            ([("dres", rg.grad, rr.grad)] if res else []):
+        g = got.cpu().float()
+        err = (g - want).abs().max().item()
+        tol = 3e-2 * max(1.0, want.abs().max().item())
+        assert err < tol, f"{name}: max err {err} vs tol {tol}"
+
