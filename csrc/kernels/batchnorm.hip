@@ -119,13 +119,13 @@ bn_stats_final_kernel(const float* __restrict__ mid, float* __restrict__ mean,
                       float* __restrict__ rstd, float* __restrict__ rmean,
                       float* __restrict__ rvar, int ns, long R, int C,
                       float momentum, float eps) {
-  __shared__ float acc[2][4][64];
-  const int cl = (int)threadIdx.x & 63;
-  const int g = (int)threadIdx.x >> 6;
-  const int c = blockIdx.x * 64 + cl;
+  __shared__ float acc[2][16][16];
+  const int cl = (int)threadIdx.x & 15;
+  const int g = (int)threadIdx.x >> 4;
+  const int c = blockIdx.x * 16 + cl;
   float s = 0.f, q = 0.f;
   if (c < C)
-    for (int i = g; i < ns; i += 4) {
+    for (int i = g; i < ns; i += 16) {
       s += mid[((long)i * 2) * C + c];
       q += mid[((long)i * 2 + 1) * C + c];
     }
@@ -133,8 +133,12 @@ bn_stats_final_kernel(const float* __restrict__ mid, float* __restrict__ mean,
   acc[1][g][cl] = q;
   __syncthreads();
   if (g || c >= C) return;
-  s = acc[0][0][cl] + acc[0][1][cl] + acc[0][2][cl] + acc[0][3][cl];
-  q = acc[1][0][cl] + acc[1][1][cl] + acc[1][2][cl] + acc[1][3][cl];
+  s = 0.f; q = 0.f;
+  #pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    s += acc[0][j][cl];
+    q += acc[1][j][cl];
+  }
   const float mu = s / (float)R;
   float var = q / (float)R - mu * mu;
   var = fmaxf(var, 0.f);
@@ -263,13 +267,13 @@ bn_bwd_reduce_kernel(const short* __restrict__ dy, const short* __restrict__ x,
 __global__ void __launch_bounds__(256)
 bn_bwd_final_kernel(const float* __restrict__ mid, float* __restrict__ dw,
                     float* __restrict__ db, int ns, int C) {
-  __shared__ float acc[2][4][64];
-  const int cl = (int)threadIdx.x & 63;
-  const int g = (int)threadIdx.x >> 6;
-  const int c = blockIdx.x * 64 + cl;
+  __shared__ float acc[2][16][16];
+  const int cl = (int)threadIdx.x & 15;
+  const int g = (int)threadIdx.x >> 4;
+  const int c = blockIdx.x * 16 + cl;
   float sd = 0.f, sx = 0.f;
   if (c < C)
-    for (int i = g; i < ns; i += 4) {
+    for (int i = g; i < ns; i += 16) {
       sd += mid[((long)i * 2) * C + c];
       sx += mid[((long)i * 2 + 1) * C + c];
     }
@@ -277,8 +281,14 @@ bn_bwd_final_kernel(const float* __restrict__ mid, float* __restrict__ dw,
   acc[1][g][cl] = sx;
   __syncthreads();
   if (g || c >= C) return;
-  db[c] = acc[0][0][cl] + acc[0][1][cl] + acc[0][2][cl] + acc[0][3][cl];
-  dw[c] = acc[1][0][cl] + acc[1][1][cl] + acc[1][2][cl] + acc[1][3][cl];
+  float td = 0.f, tx = 0.f;
+  #pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    td += acc[0][j][cl];
+    tx += acc[1][j][cl];
+  }
+  db[c] = td;
+  dw[c] = tx;
 }
 
 // ---- bwd dx (+ dres): dx = gamma*rstd*(dy_eff - db/R - xhat*dw/R)
@@ -346,8 +356,8 @@ int slices(long R, int C) {
   const int CV = C / 8;
   const int cvb = CV >= 256 ? (CV + 255) / 256 : 1;
   const int G = CV >= 256 ? 1 : 256 / CV;
-  long want = 8192 / ((long)G * cvb);
-  if (want > 1024) want = 1024;
+  long want = 4096 / ((long)G * cvb);
+  if (want > 512) want = 512;
   const long max_useful = (R + G - 1) / G;
   if (want > max_useful) want = max_useful;
   return (int)std::max<long>(1, want);
@@ -376,7 +386,7 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x,
   auto mid = torch::empty({ns, 2, (long)C}, fopt);
   hipLaunchKernelGGL(bn_stats_kernel, row_grid(R, C, ns), dim3(256), 0, stream,
                      (const short*)x.data_ptr(), mid.data_ptr<float>(), R, C);
-  hipLaunchKernelGGL(bn_stats_final_kernel, dim3((C + 63) / 64), dim3(256),
+  hipLaunchKernelGGL(bn_stats_final_kernel, dim3((C + 15) / 16), dim3(256),
                      0, stream, mid.data_ptr<float>(), mean.data_ptr<float>(),
                      rstd.data_ptr<float>(),
                      running_mean ? running_mean->data_ptr<float>() : nullptr,
@@ -440,7 +450,7 @@ std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
   else if (rp) RED(false, true);
   else RED(false, false);
   #undef RED
-  hipLaunchKernelGGL(bn_bwd_final_kernel, dim3((C + 63) / 64), dim3(256), 0,
+  hipLaunchKernelGGL(bn_bwd_final_kernel, dim3((C + 15) / 16), dim3(256), 0,
                      stream, mid.data_ptr<float>(), dw.data_ptr<float>(),
                      db.data_ptr<float>(), ns, C);
   #define DXK(RL, RS)                                                          \
