@@ -523,3 +523,14 @@ def test_batch_norm_act_vs_fp32(ext, C, relu, res):
         tol = 3e-2 * max(1.0, want.abs().max().item())
         assert err < tol, f"{name}: max err {err} vs tol {tol}"
 
+# ------------------------------------------------------------- colsum
+@pytest.mark.parametrize("R,C", [(16384, 768), (1000, 333), (64, 2304)])
+def test_colsum_vs_fp32(ext, R, C):
+    torch.manual_seed(31)
+    x = torch.randn(R, C)
+    got = ext.colsum(_bf(x).cuda()).cpu().float()
+    want = _bf(x).float().sum(dim=0)
+    torch.cuda.synchronize()
+    assert torch.allclose(got, want, rtol=1e-2, atol=R * 2e-3), \
+        (got - want).abs().max().item()
+
