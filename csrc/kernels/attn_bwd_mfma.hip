@@ -34,6 +34,17 @@ namespace {
 typedef float f32x16_t __attribute__((ext_vector_type(16)));
 typedef short bf16x8_t __attribute__((ext_vector_type(8)));
 
+// v_cvt_pk_bf16_f32: ONE VALU op per packed pair (guide T12; the r1
+// manual round-to-nearest-even bit sequence was ~7 VALU per pair).
+// The asm form constrains the register allocator: the dkv D=128 kernel
+// (at the 256-reg edge) spilled 264 B/lane with it, so dkv keeps the
+// manual sequence (template flag); dq uses cvt_pk (0 spill there).
+DEVINL unsigned pack_bf16x2_cvt(float lo, float hi) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
 DEVINL unsigned pack_bf16x2(float lo, float hi) {
   return (unsigned)(unsigned short)f2bf(lo) |
          ((unsigned)(unsigned short)f2bf(hi) << 16);
@@ -41,13 +52,18 @@ DEVINL unsigned pack_bf16x2(float lo, float hi) {
 
 // C-layout f32[16] (lane=col, regs=rows) -> two 8-k B/A fragments
 // (slice s covers rows 16s..16s+15 in the fragment's k order)
+template <bool CVT = false>
 DEVINL void pack_frags(const float p[16], unsigned out[2][4]) {
   #pragma unroll
   for (int s = 0; s < 2; ++s) {
-    unsigned ax = pack_bf16x2(p[8 * s + 0], p[8 * s + 1]);
-    unsigned ay = pack_bf16x2(p[8 * s + 2], p[8 * s + 3]);
-    unsigned bx = pack_bf16x2(p[8 * s + 4], p[8 * s + 5]);
-    unsigned by = pack_bf16x2(p[8 * s + 6], p[8 * s + 7]);
+    unsigned ax = CVT ? pack_bf16x2_cvt(p[8 * s + 0], p[8 * s + 1])
+                      : pack_bf16x2(p[8 * s + 0], p[8 * s + 1]);
+    unsigned ay = CVT ? pack_bf16x2_cvt(p[8 * s + 2], p[8 * s + 3])
+                      : pack_bf16x2(p[8 * s + 2], p[8 * s + 3]);
+    unsigned bx = CVT ? pack_bf16x2_cvt(p[8 * s + 4], p[8 * s + 5])
+                      : pack_bf16x2(p[8 * s + 4], p[8 * s + 5]);
+    unsigned by = CVT ? pack_bf16x2_cvt(p[8 * s + 6], p[8 * s + 7])
+                      : pack_bf16x2(p[8 * s + 6], p[8 * s + 7]);
     auto rx = __builtin_amdgcn_permlane32_swap(ax, bx, false, false);
     auto ry = __builtin_amdgcn_permlane32_swap(ay, by, false, false);
     out[s][0] = rx[0];
@@ -169,8 +185,11 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
     qf[s] = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
     dof[s] = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
   }
-  const float my_lse = lse[rhead + qrow];
+  // exp2-fold: softmax in the base-2 domain (v_exp_f32 IS exp2 — one
+  // VALU multiply per score saved); lse converted per row, not element
+  const float my_lse2 = lse[rhead + qrow] * 1.4426950408889634f;
   const float my_delta = delta[rhead + qrow];
+  const float scale2 = scale * 1.4426950408889634f;
 
   const unsigned ktr0 = tr_col_base<D>(ktile, l);
 
@@ -205,12 +224,12 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
     #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int krel = (r & 3) + 8 * (r >> 2) + 4 * half;
-      float sv = sacc[r] * scale;
+      float sv = sacc[r] * scale2;
       if (need_mask) {
         const int kg = c0 + krel;
         if (kg >= S || (causal && kg > qrow_g)) sv = -1e30f;
       }
-      const float p = __expf(sv - my_lse);
+      const float p = __builtin_amdgcn_exp2f(sv - my_lse2);
       ds[r] = p * (dpacc[r] - my_delta) * scale;
     }
 
