@@ -212,6 +212,28 @@ DEVINL bf16x8_pair tr_read_pair(unsigned a0, unsigned a1, unsigned b0,
   return p;
 }
 
+// XCD-affinity decomposition of a 1-D launch (guide T1): the dispatcher
+// places block b on XCD b%8, so mapping every tile-block of one
+// (batch, head) to ids ≡ bh (mod 8) co-locates them on ONE XCD — the
+// K/V (fwd) or Q/dO (bwd) windows they all stage are then fetched from
+// HBM once per chip instead of once per XCD (8x staging amplification
+// measured as the dominant stall class of the bwd kernels).  Falls back
+// to the identity map when 8 does not divide the bh count (speed-only
+// either way — correctness never depends on placement).
+DEVINL void xcd_tile_map(int nbx, int* bx, int* bh) {
+  const unsigned id = blockIdx.x;
+  const unsigned ubx = (unsigned)nbx;
+  const unsigned gy = gridDim.x / ubx;
+  if (gy % 8 == 0) {
+    const unsigned xcd = id % 8, idx = id / 8;
+    *bh = (int)(xcd + 8 * (idx / ubx));
+    *bx = (int)(idx % ubx);
+  } else {
+    *bx = (int)(id % ubx);
+    *bh = (int)(id / ubx);
+  }
+}
+
 // Grid sizing guideline 11: cap grid, grid-stride the rest.
 DEVINL long grid_stride() { return (long)gridDim.x * blockDim.x; }
 

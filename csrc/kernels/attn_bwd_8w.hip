@@ -97,8 +97,8 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
                       const float* __restrict__ lse,
                       const float* __restrict__ delta,
                       short* __restrict__ dq, int B, int H, int Hkv, int S,
-                      float scale, int causal, Strides4 dos, Strides4 qs,
-                      Strides4 ks, Strides4 vs, Strides4 dqs) {
+                      float scale, int causal, int nbx, Strides4 dos,
+                      Strides4 qs, Strides4 ks, Strides4 vs, Strides4 dqs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
   // double-buffered 64-row K tr-image (2 x 32-row images per buffer)
@@ -110,10 +110,11 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
   const int lq = l & 31;
   const int half = l >> 5;
 
-  const int bh = blockIdx.y;
+  int bx, bh;
+  xcd_tile_map(nbx, &bx, &bh);
   const int b = bh / H, h = bh % H;
   const int hk = h / (H / Hkv);
-  const int q0 = blockIdx.x * 256;
+  const int q0 = bx * 256;
   if (q0 >= S) return;
 
   const long qb = (long)b * qs.sb + (long)h * qs.sh;
@@ -253,8 +254,8 @@ attn_bwd_dkv_8w_kernel(const short* __restrict__ dout,
                        const float* __restrict__ delta,
                        short* __restrict__ dk, short* __restrict__ dv,
                        int B, int H, int Hkv, int S, float scale,
-                       int causal, Strides4 dos, Strides4 qs, Strides4 ks,
-                       Strides4 vs, Strides4 dks, Strides4 dvs) {
+                       int causal, int nbx, Strides4 dos, Strides4 qs,
+                       Strides4 ks, Strides4 vs, Strides4 dks, Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
   // [buf][Q 64-row tr pair | dO 64-row tr pair], double-buffered.
@@ -268,10 +269,11 @@ attn_bwd_dkv_8w_kernel(const short* __restrict__ dout,
   const int lk = l & 31;
   const int half = l >> 5;
 
-  const int bhk = blockIdx.y;
+  int bx, bhk;
+  xcd_tile_map(nbx, &bx, &bhk);
   const int b = bhk / Hkv, hk = bhk % Hkv;
   const int group = H / Hkv;
-  const int k0 = blockIdx.x * 256;
+  const int k0 = bx * 256;
   if (k0 >= S) return;
 
   const long kbb = (long)b * ks.sb + (long)hk * ks.sh;
@@ -438,7 +440,8 @@ void attn_bwd_dq_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                     torch::Tensor dq, int B, int H, int Hkv, int S, int D,
                     double scale, bool causal) {
   auto stream = scaffold_stream();
-  dim3 grid((S + 255) / 256, B * H), block(512);
+  const int nbx = (S + 255) / 256;
+  dim3 grid(nbx * B * H), block(512);
   const Strides4 dos = strides_of(dout), qs = strides_of(q), ks = strides_of(k),
                  vs = strides_of(v), dqs = strides_of(dq);
   #define LAUNCH_DQ8(DD)                                                       \
@@ -447,7 +450,7 @@ void attn_bwd_dq_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                        (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
                        (const short*)v.data_ptr(), lse.data_ptr<float>(),      \
                        delta.data_ptr<float>(), (short*)dq.data_ptr(),         \
-                       B, H, Hkv, S, (float)scale, causal ? 1 : 0,             \
+                       B, H, Hkv, S, (float)scale, causal ? 1 : 0, nbx,        \
                        dos, qs, ks, vs, dqs)
   if (D == 64) LAUNCH_DQ8(64);
   else if (D == 128) LAUNCH_DQ8(128);
@@ -461,7 +464,8 @@ void attn_bwd_dkv_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                      torch::Tensor dk, torch::Tensor dv, int B, int H,
                      int Hkv, int S, int D, double scale, bool causal) {
   auto stream = scaffold_stream();
-  dim3 grid((S + 255) / 256, B * Hkv), block(512);
+  const int nbx = (S + 255) / 256;
+  dim3 grid(nbx * B * Hkv), block(512);
   const Strides4 dos = strides_of(dout), qs = strides_of(q), ks = strides_of(k),
                  vs = strides_of(v), dks = strides_of(dk), dvs = strides_of(dv);
   #define LAUNCH_DKV8(DD)                                                      \
@@ -471,7 +475,7 @@ void attn_bwd_dkv_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                        (const short*)v.data_ptr(), lse.data_ptr<float>(),      \
                        delta.data_ptr<float>(), (short*)dk.data_ptr(),         \
                        (short*)dv.data_ptr(), B, H, Hkv, S, (float)scale,      \
-                       causal ? 1 : 0, dos, qs, ks, vs, dks, dvs)
+                       causal ? 1 : 0, nbx, dos, qs, ks, vs, dks, dvs)
   if (D == 64) LAUNCH_DKV8(64);
   else if (D == 128) LAUNCH_DKV8(128);
   else TORCH_CHECK(false, "attn_bwd_dkv_8w: unsupported D ", D);

@@ -153,8 +153,8 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
                         const float* __restrict__ lse,
                         const float* __restrict__ delta,
                         short* __restrict__ dq, int B, int H, int Hkv, int S,
-                        float scale, int causal, Strides4 dos, Strides4 qs,
-                        Strides4 ks, Strides4 vs, Strides4 dqs) {
+                        float scale, int causal, int nbx, Strides4 dos,
+                        Strides4 qs, Strides4 ks, Strides4 vs, Strides4 dqs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
   // double-buffered tr image: tile c0+32 streams while c0 computes
@@ -165,10 +165,11 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
   const int lq = l & 31;
   const int half = l >> 5;
 
-  const int bh = blockIdx.y;
+  int bx, bh;
+  xcd_tile_map(nbx, &bx, &bh);   // co-locate one head's tile-blocks per XCD
   const int b = bh / H, h = bh % H;
   const int hk = h / (H / Hkv);
-  const int q0 = blockIdx.x * 128;
+  const int q0 = bx * 128;
   if (q0 >= S) return;
 
   const long qb = (long)b * qs.sb + (long)h * qs.sh;
@@ -286,8 +287,9 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          const float* __restrict__ delta,
                          short* __restrict__ dk, short* __restrict__ dv,
                          int B, int H, int Hkv, int S, float scale,
-                         int causal, Strides4 dos, Strides4 qs, Strides4 ks,
-                         Strides4 vs, Strides4 dks, Strides4 dvs) {
+                         int causal, int nbx, Strides4 dos, Strides4 qs,
+                         Strides4 ks, Strides4 vs, Strides4 dks,
+                         Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
   // [buf][Q tile | dO tile]; double-buffered at D<=64 (stage of tile
@@ -303,10 +305,11 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
   const int lk = l & 31;
   const int half = l >> 5;
 
-  const int bhk = blockIdx.y;
+  int bx, bhk;
+  xcd_tile_map(nbx, &bx, &bhk);
   const int b = bhk / Hkv, hk = bhk % Hkv;
   const int group = H / Hkv;
-  const int k0 = blockIdx.x * 128;
+  const int k0 = bx * 128;
   if (k0 >= S) return;
 
   const long kbb = (long)b * ks.sb + (long)hk * ks.sh;
@@ -507,8 +510,9 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                    double scale, bool causal, const float* dlse_ptr) {
   auto stream = scaffold_stream();
   dim3 block(256);
-  dim3 grid_q((S + 127) / 128, B * H);
-  dim3 grid_kv((S + 127) / 128, B * Hkv);
+  const int nbx4 = (S + 127) / 128;
+  dim3 grid_q(nbx4 * B * H);
+  dim3 grid_kv(nbx4 * B * Hkv);
   const Strides4 dos = strides_of(dout), qs = strides_of(q), ks = strides_of(k),
                  vs = strides_of(v), dqs = strides_of(dq), dks = strides_of(dk),
                  dvs = strides_of(dv), osr = strides_of(o);
@@ -540,7 +544,7 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                          (const short*)q.data_ptr(), (const short*)k.data_ptr(),\
                          (const short*)v.data_ptr(), lse.data_ptr<float>(),    \
                          delta.data_ptr<float>(), (short*)dq.data_ptr(),       \
-                         B, H, Hkv, S, (float)scale, causal ? 1 : 0,           \
+                         B, H, Hkv, S, (float)scale, causal ? 1 : 0, nbx4,     \
                          dos, qs, ks, vs, dqs);                                \
       hipLaunchKernelGGL((attn_bwd_dkv_mfma_kernel<DD>), grid_kv, block, 0,    \
                          stream, (const short*)dout.data_ptr(),                \
@@ -548,7 +552,7 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                          (const short*)v.data_ptr(), lse.data_ptr<float>(),    \
                          delta.data_ptr<float>(), (short*)dk.data_ptr(),       \
                          (short*)dv.data_ptr(), B, H, Hkv, S, (float)scale,    \
-                         causal ? 1 : 0, dos, qs, ks, vs, dks, dvs);           \
+                         causal ? 1 : 0, nbx4, dos, qs, ks, vs, dks, dvs);     \
     } while (0)
   if (D == 64) LAUNCH_D(64);
   else if (D == 128) LAUNCH_D(128);

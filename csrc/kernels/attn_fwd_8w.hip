@@ -52,8 +52,8 @@ __global__ void __launch_bounds__(512, 2)
 attn_fwd_8w_kernel(const short* __restrict__ q, const short* __restrict__ k,
                    const short* __restrict__ v, short* __restrict__ o,
                    float* __restrict__ lse, int B, int H, int Hkv, int S,
-                   float scale2, int causal, Strides4 qs, Strides4 ks,
-                   Strides4 vs, Strides4 os) {
+                   float scale2, int causal, int nbx, Strides4 qs,
+                   Strides4 ks, Strides4 vs, Strides4 os) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
   constexpr int CH = D / 8;          // 16-B chunks per row
@@ -65,10 +65,11 @@ attn_fwd_8w_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const int lq = l & 31;
   const int half = l >> 5;
 
-  const int bh = blockIdx.y;
+  int bx, bh;
+  xcd_tile_map(nbx, &bx, &bh);   // co-locate one head's tile-blocks per XCD
   const int b = bh / H, h = bh % H;
   const int hk = h / (H / Hkv);
-  const int q0 = blockIdx.x * 256;
+  const int q0 = bx * 256;
   if (q0 >= S) return;
 
   const long qb = (long)b * qs.sb + (long)h * qs.sh;
@@ -334,7 +335,8 @@ void attn_fwd_8w(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  torch::Tensor o, torch::Tensor lse, int B, int H, int Hkv,
                  int S, int D, double scale, bool causal) {
   auto stream = scaffold_stream();
-  dim3 grid((S + 255) / 256, B * H), block(512);
+  const int nbx = (S + 255) / 256;
+  dim3 grid(nbx * B * H), block(512);
   const Strides4 qs = strides_of(q), ks = strides_of(k), vs = strides_of(v),
                  os = strides_of(o);
   // exp2-fold: scale * log2(e) — softmax runs in the base-2 domain
@@ -344,7 +346,7 @@ void attn_fwd_8w(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
                        (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
                        lse.data_ptr<float>(), B, H, Hkv, S, scale2,            \
-                       causal ? 1 : 0, qs, ks, vs, os)
+                       causal ? 1 : 0, nbx, qs, ks, vs, os)
   if (D == 64) LAUNCH_F8(64);
   else if (D == 128) LAUNCH_F8(128);
   else TORCH_CHECK(false, "attn_fwd_8w: unsupported D ", D);

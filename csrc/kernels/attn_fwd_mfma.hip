@@ -49,8 +49,8 @@ __global__ void __launch_bounds__(256, 2)
 attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
                      const short* __restrict__ v, short* __restrict__ o,
                      float* __restrict__ lse, int B, int H, int Hkv, int S,
-                     float scale, int causal, Strides4 qs, Strides4 ks,
-                     Strides4 vs, Strides4 os) {
+                     float scale, int causal, int nbx, Strides4 qs,
+                     Strides4 ks, Strides4 vs, Strides4 os) {
   constexpr int NSL = D / 16;   // QK^T k-slices (over d)
   constexpr int NDT = D / 32;   // O^T d-tiles
   constexpr int CH = D / 8;     // 16-B chunks per row
@@ -65,10 +65,11 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const int lq = l & 31;
   const int half = l >> 5;
 
-  const int bh = blockIdx.y;
+  int bx, bh;
+  xcd_tile_map(nbx, &bx, &bh);   // co-locate one head's tile-blocks per XCD
   const int b = bh / H, h = bh % H;
   const int hk = h / (H / Hkv);
-  const int q0 = blockIdx.x * 128;
+  const int q0 = bx * 128;
   if (q0 >= S) return;
 
   const long qb = (long)b * qs.sb + (long)h * qs.sh;
@@ -280,7 +281,8 @@ void attn_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                    torch::Tensor o, torch::Tensor lse, int B, int H, int Hkv,
                    int S, int D, double scale, bool causal) {
   auto stream = scaffold_stream();
-  dim3 grid((S + 127) / 128, B * H), block(256);
+  const int nbx = (S + 127) / 128;
+  dim3 grid(nbx * B * H), block(256);
   const Strides4 qs = strides_of(q), ks = strides_of(k), vs = strides_of(v),
                  os = strides_of(o);
   #define LAUNCH_F(DD)                                                         \
@@ -288,7 +290,7 @@ void attn_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                        (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
                        (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
                        lse.data_ptr<float>(), B, H, Hkv, S, (float)scale,      \
-                       causal ? 1 : 0, qs, ks, vs, os)
+                       causal ? 1 : 0, nbx, qs, ks, vs, os)
   if (D == 64) LAUNCH_F(64);
   else if (D == 128) LAUNCH_F(128);
   else TORCH_CHECK(false, "attn_fwd_mfma: unsupported D ", D);
