@@ -292,13 +292,14 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                          Strides4 dvs) {
   constexpr int NSL = D / 16;
   constexpr int NDT = D / 32;
-  // [buf][Q tile | dO tile]; double-buffered at D<=64 (stage of tile
-  // t+1 issued after the global-load block, drained under the dV/dK
-  // MFMAs); single-buffered at D=128 where the extra live state
-  // spilled (148 B/lane) and measured net-slower
-  constexpr bool KVREG_ = (D <= 64);
-  constexpr int NBUF = KVREG_ ? 2 : 1;
-  __shared__ __attribute__((aligned(16))) short bwd_lds[NBUF * 2 * 32 * D];
+  // [buf][Q tile | dO tile], double-buffered at BOTH head dims: tile
+  // t+1's glds issue after this tile's LAST ordinary global load (the
+  // lse/delta lane loads moved above the issue point — an ordinary
+  // load beside an in-flight glds drains vmcnt(0), guide trap b, which
+  // was also why the r1 D=128 double-buffer attempt measured
+  // net-slower) and drain under the dV/dK MFMAs.  PMC r2: the
+  // synchronous D=128 staging parked waves 73% of the time.
+  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 2 * 32 * D];
 
   const int wid = threadIdx.x >> 6;
   const int l = threadIdx.x & 63;
@@ -351,18 +352,11 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       stage_tr<D>(q + qb, qs.ss, c0s, S, base, wid, l);
       stage_tr<D>(dout + dob, dos.ss, c0s, S, base + 32 * D, wid, l);
     };
-    if (KVREG_) {
-      stage_qdo(0, qstart);
-      __syncthreads();
-    }
+    stage_qdo(0, qstart);
+    __syncthreads();
     int cur = 0;
     for (int c0 = qstart; c0 < S; c0 += 32) {
-      if (!KVREG_) {
-        stage_qdo(0, c0);
-        __syncthreads();
-      }
-      const unsigned bufoff =
-          KVREG_ ? (unsigned)cur * (2 * 32 * D * 2) : 0u;  // folds at D=128
+      const unsigned bufoff = (unsigned)cur * (2 * 32 * D * 2);
       const unsigned qtr = qtr0 + bufoff;
       const unsigned dotr = dotr0 + bufoff;
 
@@ -398,20 +392,16 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       }
       __builtin_amdgcn_s_setprio(0);
 
-      // D<=64: next q-tile's glds issued after the global-load block
-      // (trap-b note in the fwd kernel); drained under the dV/dK MFMAs
-      if (KVREG_ && c0 + 32 < S) stage_qdo(cur ^ 1, c0 + 32);
-
       // per-tile q-row constants loaded ONCE per wave (lane j holds row
-      // c0+j's values) and broadcast per-reg via shfl — replaces 32
-      // scalar L1 loads per lane per tile with 2 loads + shuffles.
-      // D=128 keeps the L1 loads: the two extra live scalars tipped the
-      // register budget into deeper spill there.
-      float lse_lane = 0.f, del_lane = 0.f;
-      if constexpr (KVREG) {
-        lse_lane = lse[rhead + qrow];
-        del_lane = delta[rhead + qrow];
-      }
+      // c0+j's values) and broadcast per-reg via shfl — 2 loads instead
+      // of 32 scalar L1 loads per lane per tile, and they sit BEFORE
+      // the glds issue so no ordinary load drains the prefetch (trap b)
+      const float lse_lane = lse[rhead + qrow];
+      const float del_lane = delta[rhead + qrow];
+
+      // next q-tile's glds: issued after the tile's LAST ordinary
+      // global load; drained under the dV/dK MFMAs + end barrier
+      if (c0 + 32 < S) stage_qdo(cur ^ 1, c0 + 32);
 
       // p overwrites sacc in place (register reuse)
       #pragma unroll
@@ -420,16 +410,14 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
         const int qg = c0 + qrel;
         float sv = sacc[r] * scale;
         const bool masked = qg >= S || (causal && krow_g > qg);
-        const float lse_r = KVREG ? __shfl(lse_lane, qrel, 64)
-                                  : lse[rhead + min(qg, S - 1)];
+        const float lse_r = __shfl(lse_lane, qrel, 64);
         sacc[r] = masked ? 0.f : __expf(sv - lse_r);
       }
       // ds overwrites dpacc in place
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
-        const float del_r = KVREG ? __shfl(del_lane, qrel, 64)
-                                  : delta[rhead + min(c0 + qrel, S - 1)];
+        const float del_r = __shfl(del_lane, qrel, 64);
         dpacc[r] = sacc[r] * (dpacc[r] - del_r) * scale;
       }
 
@@ -459,8 +447,8 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
             *(const bf16x8_t*)dsf[1], cq.f[3], dkacc[t], 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
-      __syncthreads();  // tiles consumed; (KVREG_) next tile's glds drained
-      if (KVREG_) cur ^= 1;
+      __syncthreads();  // tiles consumed; next tile's glds drained
+      cur ^= 1;
     }
   }
 
