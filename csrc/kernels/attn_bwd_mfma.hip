@@ -206,10 +206,15 @@ attn_bwd_dq_mfma_kernel(const short* __restrict__ dout,
     const unsigned ktr = ktr0 + (unsigned)cur * (32 * D * 2);
 
     const int krow = min(c0 + lq, S - 1);
+    // K A-fragments from the staged tr-image (contiguous row runs of
+    // the tr layout — same trick as dkv); V has no image, stays global
+    const short* kimg = ktile + (long)cur * 32 * D;
+    const unsigned rowel = ((unsigned)(lq >> 2) * (D / 16)) * 64u +
+                           (unsigned)(lq & 3) * 16u + 8u * half;
     f32x16_t sacc = {}, dpacc = {};
     #pragma unroll
     for (int s = 0; s < NSL; ++s) {
-      bf16x8_t kf = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
+      bf16x8_t kf = *(const bf16x8_t*)(kimg + rowel + s * 64);
       bf16x8_t vf = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
       sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[s], sacc, 0, 0, 0);
       dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[s], dpacc, 0, 0, 0);
@@ -360,32 +365,38 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
       const unsigned qtr = qtr0 + bufoff;
       const unsigned dotr = dotr0 + bufoff;
 
-      // S = Q_tile . K^T and dP = dO . V^T: at D<=64 (fragments pinned
-      // in registers) the two independent accumulator chains interleave
-      // in ONE loop to hide dependent-MFMA latency (measured -6% on the
-      // GPT-2 bwd); at D=128 the doubled live operands spilled and
-      // measured slower, so the chains stay in two loops there.
+      // S = Q_tile . K^T and dP = dO . V^T.  The Q/dO A-fragments come
+      // from the STAGED tr-images: a row fragment (k, 16s+8h..+8) is a
+      // contiguous 16-B run of the tr layout at element
+      // (k>>2)*(D/16)*64 + s*64 + (k&3)*16 + 8h — so the per-tile
+      // global Q/dO operand loads (an L2 round trip + vmcnt wait per
+      // k-slice, a chunk of the 69%-parked time) become ds_reads of
+      // data the prefetch already placed.
       const int qrow = min(c0 + lk, S - 1);
+      const short* qimg = bwd_lds + (bufoff >> 1);
+      const short* doimg = qimg + 32 * D;
+      const unsigned rowel = ((unsigned)(lk >> 2) * (D / 16)) * 64u +
+                             (unsigned)(lk & 3) * 16u + 8u * half;
       f32x16_t sacc = {}, dpacc = {};
       __builtin_amdgcn_s_setprio(1);
       if constexpr (KVREG) {
         #pragma unroll
         for (int s = 0; s < NSL; ++s) {
-          bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
-          bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
+          bf16x8_t qf = *(const bf16x8_t*)(qimg + rowel + s * 64);
+          bf16x8_t dof = *(const bf16x8_t*)(doimg + rowel + s * 64);
           sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
           dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vtf[s], dpacc, 0, 0, 0);
         }
       } else {
         #pragma unroll
         for (int s = 0; s < NSL; ++s) {
-          bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
+          bf16x8_t qf = *(const bf16x8_t*)(qimg + rowel + s * 64);
           const bf16x8_t kv = *(const bf16x8_t*)(krow_ptr + 16 * s);
           sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
         }
         #pragma unroll
         for (int s = 0; s < NSL; ++s) {
-          bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
+          bf16x8_t dof = *(const bf16x8_t*)(doimg + rowel + s * 64);
           const bf16x8_t vv = *(const bf16x8_t*)(vrow_ptr + 16 * s);
           dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
         }
