@@ -162,6 +162,11 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
                            + (unsigned)sub * (32 * D * 2);
 
       const int krow = min(cs + lq, S - 1);
+      // K A-fragments from the staged tr-image (row fragments are
+      // contiguous 16-B runs of the tr layout); V has no image
+      const short* kimg8 = ktile + (long)cur * 64 * D + (long)sub * 32 * D;
+      const unsigned rowel = ((unsigned)(lq >> 2) * (D / 16)) * 64u +
+                             (unsigned)(lq & 3) * 16u + 8u * half;
       f32x16_t sacc = {}, dpacc = {};
       __builtin_amdgcn_s_setprio(1);
       // FULL unroll (a partial '#pragma unroll N' leaves a rolled loop
@@ -171,7 +176,7 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
       // compiler cannot buffer all 8 k-slices' loads at once
       #pragma unroll
       for (int s = 0; s < NSL; ++s) {
-        bf16x8_t kf = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
+        bf16x8_t kf = *(const bf16x8_t*)(kimg8 + rowel + s * 64);
         bf16x8_t vf = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
         const bf16x8_t dov =
             DOREG ? dof[s] : *(const bf16x8_t*)(dorow_ptr + 16 * s);
@@ -326,6 +331,13 @@ attn_bwd_dkv_8w_kernel(const short* __restrict__ dout,
         const unsigned dotr = qtr + (unsigned)(64 * D * 2);
 
         const int qrow = min(cs + lk, S - 1);
+        // Q/dO A-fragments from the staged tr-images (contiguous row
+        // runs of the tr layout) — no per-sub global operand loads
+        const short* qsub = bwd_lds + ((unsigned)cur * (2 * 64 * D))
+                            + (unsigned)sub * 32 * D;
+        const short* dosub = qsub + 64 * D;
+        const unsigned rowel = ((unsigned)(lk >> 2) * (D / 16)) * 64u +
+                               (unsigned)(lk & 3) * 16u + 8u * half;
         f32x16_t sacc = {}, dpacc = {};
         __builtin_amdgcn_s_setprio(1);
         if constexpr (KVREG) {
@@ -333,26 +345,27 @@ attn_bwd_dkv_8w_kernel(const short* __restrict__ dout,
           // dependent-MFMA latency; fits at D=64)
           #pragma unroll
           for (int s = 0; s < NSL; ++s) {
-            bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
-            bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
+            bf16x8_t qf = *(const bf16x8_t*)(qsub + rowel + s * 64);
+            bf16x8_t dof = *(const bf16x8_t*)(dosub + rowel + s * 64);
             sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
             dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vtf[s], dpacc, 0, 0, 0);
           }
         } else {
-          // D=128: split chains, unroll capped at 4 — a full unroll
-          // keeps all 8 k-slices' loads in flight (64 VGPRs of operand
-          // buffering) and tips the 256-reg budget into deep spill
-          #pragma unroll 4
+          // D=128: split chains with a mid-loop scheduling split (the
+          // scheduler otherwise buffers all k-slices' loads -> spill)
+          #pragma unroll
           for (int s = 0; s < NSL; ++s) {
-            bf16x8_t qf = *(const bf16x8_t*)(q + qb + (long)qrow * qs.ss + 16 * s + 8 * half);
+            bf16x8_t qf = *(const bf16x8_t*)(qsub + rowel + s * 64);
             const bf16x8_t kv = *(const bf16x8_t*)(krow_ptr + 16 * s);
             sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
+            if (s == NSL / 2 - 1) asm volatile("" ::: "memory");
           }
-          #pragma unroll 4
+          #pragma unroll
           for (int s = 0; s < NSL; ++s) {
-            bf16x8_t dof = *(const bf16x8_t*)(dout + dob + (long)qrow * dos.ss + 16 * s + 8 * half);
+            bf16x8_t dof = *(const bf16x8_t*)(dosub + rowel + s * 64);
             const bf16x8_t vv = *(const bf16x8_t*)(vrow_ptr + 16 * s);
             dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
+            if (s == NSL / 2 - 1) asm volatile("" ::: "memory");
           }
         }
         __builtin_amdgcn_s_setprio(0);
