@@ -51,6 +51,10 @@ DEVINL ColMap colmap(int C) {
     m.G = 256 / CV;
     m.vc = (int)threadIdx.x % CV;
     m.g = (int)threadIdx.x / CV;
+    // tail threads when CV does not divide 256 would re-walk other
+    // groups' rows — idle them (all ResNet C are powers of two, but
+    // the kernel must not silently double-count for other C)
+    if (m.g >= m.G) m.vc = -1;
   }
   return m;
 }

@@ -27,6 +27,10 @@ colsum_slice_kernel(const short* __restrict__ x, float* __restrict__ mid,
     G = 256 / CV;
     vc = (int)threadIdx.x % CV;
     g = (int)threadIdx.x / CV;
+    // when CV does not divide 256 the tail threads would re-walk other
+    // groups' rows (double count — caught by test_colsum_vs_fp32 at
+    // C=768): idle them instead
+    if (g >= G) vc = -1;
   }
   float s[8] = {};
   if (vc >= 0) {
@@ -44,17 +48,17 @@ colsum_slice_kernel(const short* __restrict__ x, float* __restrict__ mid,
       for (int j = 0; j < 8; ++j) out[vc * 8 + j] = s[j];
     return;
   }
-  if (g == 0) {
+  if (g == 0 && vc >= 0) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) acc[vc * 8 + j] = s[j];
   }
   __syncthreads();
-  if (g != 0) {
+  if (g != 0 && vc >= 0) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) atomicAdd(&acc[vc * 8 + j], s[j]);
   }
   __syncthreads();
-  if (g == 0) {
+  if (g == 0 && vc >= 0) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) out[vc * 8 + j] = acc[vc * 8 + j];
   }
