@@ -88,28 +88,28 @@ bn_stats_kernel(const short* __restrict__ x, float* __restrict__ mid,
       }
     return;
   }
-  // G>1 (C < 2048): fold the row-groups with LDS fp32 atomics
-  if (m.g == 0) {
+  // G>1 (C < 2048): DETERMINISTIC fold — per-group partials in
+  // [g][C] slots (G*C = 2048 floats per tensor), group 0 sums in
+  // fixed g order (atomicAdd would vary fp32 order run to run and
+  // break the bitwise-determinism race screen)
+  if (m.vc >= 0) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      acc[0][m.vc * 8 + j] = s[j];
-      acc[1][m.vc * 8 + j] = q[j];
+      acc[0][m.g * C + m.vc * 8 + j] = s[j];
+      acc[1][m.g * C + m.vc * 8 + j] = q[j];
     }
   }
   __syncthreads();
-  if (m.g != 0) {
+  if (m.g == 0 && m.vc >= 0) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&acc[0][m.vc * 8 + j], s[j]);
-      atomicAdd(&acc[1][m.vc * 8 + j], q[j]);
-    }
-  }
-  __syncthreads();
-  if (m.g == 0) {
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      out[m.vc * 8 + j] = acc[0][m.vc * 8 + j];
-      out[C + m.vc * 8 + j] = acc[1][m.vc * 8 + j];
+      float ts = 0.f, tq = 0.f;
+      for (int gg = 0; gg < m.G; ++gg) {
+        ts += acc[0][gg * C + m.vc * 8 + j];
+        tq += acc[1][gg * C + m.vc * 8 + j];
+      }
+      out[m.vc * 8 + j] = ts;
+      out[C + m.vc * 8 + j] = tq;
     }
   }
 }
@@ -243,27 +243,24 @@ bn_bwd_reduce_kernel(const short* __restrict__ dy, const short* __restrict__ x,
       }
     return;
   }
-  if (m.g == 0) {
+  if (m.vc >= 0) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      acc[0][m.vc * 8 + j] = sd[j];
-      acc[1][m.vc * 8 + j] = sx[j];
+      acc[0][m.g * C + m.vc * 8 + j] = sd[j];
+      acc[1][m.g * C + m.vc * 8 + j] = sx[j];
     }
   }
   __syncthreads();
-  if (m.g != 0) {
+  if (m.g == 0 && m.vc >= 0) {
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      atomicAdd(&acc[0][m.vc * 8 + j], sd[j]);
-      atomicAdd(&acc[1][m.vc * 8 + j], sx[j]);
-    }
-  }
-  __syncthreads();
-  if (m.g == 0) {
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      out[m.vc * 8 + j] = acc[0][m.vc * 8 + j];
-      out[C + m.vc * 8 + j] = acc[1][m.vc * 8 + j];
+      float ts = 0.f, tx = 0.f;
+      for (int gg = 0; gg < m.G; ++gg) {
+        ts += acc[0][gg * C + m.vc * 8 + j];
+        tx += acc[1][gg * C + m.vc * 8 + j];
+      }
+      out[m.vc * 8 + j] = ts;
+      out[C + m.vc * 8 + j] = tx;
     }
   }
 }

@@ -48,19 +48,22 @@ colsum_slice_kernel(const short* __restrict__ x, float* __restrict__ mid,
       for (int j = 0; j < 8; ++j) out[vc * 8 + j] = s[j];
     return;
   }
-  if (g == 0 && vc >= 0) {
+  // deterministic fold: per-group partials land in [g][C] slots (G*C
+  // is always 2048 floats) and group 0 sums them in FIXED g order —
+  // an LDS atomicAdd fold would vary fp32 addition order run to run
+  // and break the bitwise-determinism race screen
+  if (vc >= 0) {
     #pragma unroll
-    for (int j = 0; j < 8; ++j) acc[vc * 8 + j] = s[j];
+    for (int j = 0; j < 8; ++j) acc[g * C + vc * 8 + j] = s[j];
   }
   __syncthreads();
-  if (g != 0 && vc >= 0) {
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(&acc[vc * 8 + j], s[j]);
-  }
-  __syncthreads();
   if (g == 0 && vc >= 0) {
     #pragma unroll
-    for (int j = 0; j < 8; ++j) out[vc * 8 + j] = acc[vc * 8 + j];
+    for (int j = 0; j < 8; ++j) {
+      float t = 0.f;
+      for (int gg = 0; gg < G; ++gg) t += acc[gg * C + vc * 8 + j];
+      out[vc * 8 + j] = t;
+    }
   }
 }
 

@@ -414,6 +414,20 @@ def test_kernel_determinism(ext):
     y2, *_ = ext.layernorm_fwd(x, w, b, 1e-5)
     assert torch.equal(y1, y2)
 
+    # column reduces: the slice folds must be fixed-order (an LDS
+    # atomic fold would vary fp32 addition order run to run)
+    xc = _bf(torch.randn(16384, 768)).cuda()
+    assert torch.equal(ext.colsum(xc), ext.colsum(xc))
+    xb = _bf(torch.randn(8, 64, 28, 28)).cuda().to(memory_format=torch.channels_last)
+    wb = torch.rand(64, device="cuda") + 0.5
+    bb = torch.randn(64, device="cuda")
+    outs_bn = []
+    for _ in range(2):
+        y, mean, rstd = ext.bn_fwd(xb, None, wb, bb, None, None, 0.1, 1e-5, True)
+        outs_bn.append((y, mean, rstd))
+    for a, b2 in zip(outs_bn[0], outs_bn[1]):
+        assert torch.equal(a, b2), "nondeterministic BN stats"
+
 
 def test_lm_head_ce_matches_eager():
     from scaffold import ops
