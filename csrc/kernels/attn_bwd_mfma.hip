@@ -379,27 +379,18 @@ attn_bwd_dkv_mfma_kernel(const short* __restrict__ dout,
                              (unsigned)(lk & 3) * 16u + 8u * half;
       f32x16_t sacc = {}, dpacc = {};
       __builtin_amdgcn_s_setprio(1);
-      if constexpr (KVREG) {
-        #pragma unroll
-        for (int s = 0; s < NSL; ++s) {
-          bf16x8_t qf = *(const bf16x8_t*)(qimg + rowel + s * 64);
-          bf16x8_t dof = *(const bf16x8_t*)(doimg + rowel + s * 64);
-          sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
-          dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vtf[s], dpacc, 0, 0, 0);
-        }
-      } else {
-        #pragma unroll
-        for (int s = 0; s < NSL; ++s) {
-          bf16x8_t qf = *(const bf16x8_t*)(qimg + rowel + s * 64);
-          const bf16x8_t kv = *(const bf16x8_t*)(krow_ptr + 16 * s);
-          sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
-        }
-        #pragma unroll
-        for (int s = 0; s < NSL; ++s) {
-          bf16x8_t dof = *(const bf16x8_t*)(doimg + rowel + s * 64);
-          const bf16x8_t vv = *(const bf16x8_t*)(vrow_ptr + 16 * s);
-          dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
-        }
+      // both head dims: the two accumulator chains interleave (hides
+      // dependent-MFMA latency); with the operands now coming from LDS
+      // the D=128 split-chain workaround for global-load buffering is
+      // no longer needed
+      #pragma unroll
+      for (int s = 0; s < NSL; ++s) {
+        bf16x8_t qf = *(const bf16x8_t*)(qimg + rowel + s * 64);
+        bf16x8_t dof = *(const bf16x8_t*)(doimg + rowel + s * 64);
+        const bf16x8_t kv = KVREG ? ktf[s] : *(const bf16x8_t*)(krow_ptr + 16 * s);
+        const bf16x8_t vv = KVREG ? vtf[s] : *(const bf16x8_t*)(vrow_ptr + 16 * s);
+        sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kv, sacc, 0, 0, 0);
+        dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vv, dpacc, 0, 0, 0);
       }
       __builtin_amdgcn_s_setprio(0);
 
