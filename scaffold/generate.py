@@ -41,19 +41,34 @@ def _sdpa(q, k, v, causal):
 
 
 class KVCache:
-    def __init__(self, n_layers: int):
-        self.k: List[Optional[torch.Tensor]] = [None] * n_layers
-        self.v: List[Optional[torch.Tensor]] = [None] * n_layers
+    """Preallocated per-layer KV buffers.
+
+    append() slice-assigns into the reserved region and returns views of
+    the filled prefix — the r1 torch.cat form reallocated and copied the
+    WHOLE cache every decode step per layer (O(S^2) traffic and
+    allocator churn over a generation)."""
+
+    def __init__(self, n_layers: int, B: int, n_kv: int, max_len: int,
+                 head_dim: int, device, dtype):
+        self.n_layers = n_layers
+        self.k = [torch.empty(B, n_kv, max_len, head_dim, device=device,
+                              dtype=dtype) for _ in range(n_layers)]
+        self.v = [torch.empty_like(self.k[0]) for _ in range(n_layers)]
+        self.len = 0
 
     def append(self, i: int, k: torch.Tensor, v: torch.Tensor
                ) -> Tuple[torch.Tensor, torch.Tensor]:
-        self.k[i] = k if self.k[i] is None else torch.cat([self.k[i], k], dim=2)
-        self.v[i] = v if self.v[i] is None else torch.cat([self.v[i], v], dim=2)
-        return self.k[i], self.v[i]
+        S = k.shape[2]
+        start = self.len
+        self.k[i][:, :, start:start + S] = k
+        self.v[i][:, :, start:start + S] = v
+        if i == self.n_layers - 1:   # all layers appended for this step
+            self.len = start + S
+        return (self.k[i][:, :, :start + S], self.v[i][:, :, :start + S])
 
     @property
     def seq_len(self) -> int:
-        return 0 if self.k[0] is None else self.k[0].shape[2]
+        return self.len
 
 
 class Generator:
@@ -125,15 +140,22 @@ class Generator:
                  generator: Optional[torch.Generator] = None) -> torch.Tensor:
         """prompt (B, S0) int64 -> (B, S0 + max_new_tokens)."""
         n_layers = len(self.m.blocks)
-        cache = KVCache(n_layers)
-        out = prompt
+        B, S0 = prompt.shape
+        attn0 = self.m.blocks[0].attn
+        n_kv = attn0.n_kv if self.is_llama else attn0.n_heads
+        dtype = next(self.m.parameters()).dtype
+        cache = KVCache(n_layers, B, n_kv, S0 + max_new_tokens,
+                        attn0.head_dim, prompt.device, dtype)
+        out = torch.empty(B, S0 + max_new_tokens, dtype=prompt.dtype,
+                          device=prompt.device)
+        out[:, :S0] = prompt
         logits = self._forward_tokens(prompt, cache, 0)   # prefill
-        for _ in range(max_new_tokens):
+        for t in range(max_new_tokens):
             if temperature > 0:
                 probs = torch.softmax(logits.float() / temperature, dim=-1)
                 nxt = torch.multinomial(probs, 1, generator=generator)
             else:
                 nxt = logits.argmax(dim=-1, keepdim=True)
-            out = torch.cat([out, nxt], dim=1)
+            out[:, S0 + t] = nxt[:, 0]
             logits = self._forward_tokens(nxt, cache, cache.seq_len)
         return out

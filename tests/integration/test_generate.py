@@ -24,9 +24,14 @@ def _ref_greedy(gen: Generator, prompt, n):
     """no-cache reference: full forward from scratch every step."""
     from scaffold.generate import KVCache
 
+    m = gen.m
+    attn0 = m.blocks[0].attn
+    n_kv = attn0.n_kv if gen.is_llama else attn0.n_heads
+    dtype = next(m.parameters()).dtype
     out = prompt
     for _ in range(n):
-        cache = KVCache(len(gen.m.blocks))
+        cache = KVCache(len(m.blocks), prompt.shape[0], n_kv,
+                        out.shape[1], attn0.head_dim, prompt.device, dtype)
         logits = gen._forward_tokens(out, cache, 0)
         out = torch.cat([out, logits.argmax(-1, keepdim=True)], dim=1)
     return out
@@ -60,7 +65,10 @@ def test_decode_logits_match_training_forward():
     tokens = torch.randint(0, 128, (1, 9))
     from scaffold.generate import KVCache
 
-    logits_dec = g._forward_tokens(tokens, KVCache(len(m.blocks)), 0)
+    cache = KVCache(len(m.blocks), 1, m.blocks[0].attn.n_heads, 9,
+                    m.blocks[0].attn.head_dim, tokens.device,
+                    next(m.parameters()).dtype)
+    logits_dec = g._forward_tokens(tokens, cache, 0)
     # training-path logits for the LAST position: run model internals
     pos = torch.arange(9)
     x = m.wte(tokens) + m.wpe(pos)
