@@ -22,6 +22,15 @@ def bench(name, overrides, B, prompt, new):
     tps = B * new / dt
     print(f"{name}: B={B} prompt={prompt} new={new}: {dt*1e3:.0f} ms, "
           f"{tps:.0f} tokens/s ({dt/new*1e3:.2f} ms/step)", flush=True)
+    gen.generate_graphed(x, max_new_tokens=8)  # capture warmup
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    out = gen.generate_graphed(x, max_new_tokens=new)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    tps = B * new / dt
+    print(f"{name} GRAPHED: {dt*1e3:.0f} ms, {tps:.0f} tokens/s "
+          f"({dt/new*1e3:.2f} ms/step)", flush=True)
 
 bench("gpt2-small", [
     "model.name=gpt2", "model.vocab_size=50257", "model.n_layers=12",

@@ -159,3 +159,132 @@ class Generator:
             out[:, S0 + t] = nxt[:, 0]
             logits = self._forward_tokens(nxt, cache, cache.seq_len)
         return out
+
+    # ---- static-shape decode step (hipGraph-replayable) ---------------
+    def _sdpa_static(self, q, k, v, valid_to):
+        """Attention of a single query over a PADDED cache: rows at
+        index >= valid_to (a device scalar tensor) are masked, so all
+        shapes are replay-static."""
+        Hq, Hkv = q.shape[1], k.shape[1]
+        if Hkv != Hq:
+            rep = Hq // Hkv
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+        scale = 1.0 / math.sqrt(q.shape[-1])
+        s = (q.float() @ k.float().transpose(-1, -2)) * scale  # (B,H,1,L)
+        idx = torch.arange(k.shape[2], device=k.device)
+        s = s.masked_fill(idx >= valid_to, float("-inf"))
+        p = torch.softmax(s, dim=-1)
+        return (p @ v.float()).to(q.dtype)
+
+    def _block_static(self, blk, x, cache: KVCache, i: int, pos_dev):
+        """One decode block over the padded cache; pos_dev (device int64
+        scalar tensor) is the write position == tokens so far."""
+        if self.is_llama:
+            m = blk.attn
+            B = x.shape[0]
+            h = ops.rms_norm(x, blk.attn_norm.weight, blk.attn_norm.eps)
+            qkv = F.linear(h, m.wqkv.weight).view(
+                B, 1, m.n_heads + 2 * m.n_kv, m.head_dim)
+            q = qkv[:, :, :m.n_heads].transpose(1, 2)
+            k = qkv[:, :, m.n_heads:m.n_heads + m.n_kv].transpose(1, 2)
+            v = qkv[:, :, m.n_heads + m.n_kv:].transpose(1, 2)
+            cos = self.m.rope_cos.index_select(0, pos_dev).float()
+            sin = self.m.rope_sin.index_select(0, pos_dev).float()
+            q = (ops.rope_apply(q, cos, sin) if q.is_cuda
+                 else ops._rope_eager(q, cos, sin))
+            k = (ops.rope_apply(k, cos, sin) if k.is_cuda
+                 else ops._rope_eager(k, cos, sin))
+            cache.k[i].index_copy_(2, pos_dev, k)
+            cache.v[i].index_copy_(2, pos_dev, v)
+            o = self._sdpa_static(q, cache.k[i], cache.v[i], pos_dev + 1)
+            x = x + F.linear(o.transpose(1, 2).reshape(B, 1, -1), m.wo.weight)
+            h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
+            gu = F.linear(h, blk.w_gateup.weight)
+            Fh = gu.shape[-1] // 2
+            g = F.silu(gu[..., :Fh]) * gu[..., Fh:]
+            return x + F.linear(g, blk.w_down.weight)
+        m = blk.attn
+        B = x.shape[0]
+        h = ops.layer_norm(x, blk.ln1.weight, blk.ln1.bias, blk.ln1.eps)
+        qkv = F.linear(h, m.qkv.weight, m.qkv.bias).view(
+            B, 1, 3, m.n_heads, m.head_dim)
+        q, k, v = (qkv[:, :, j].transpose(1, 2) for j in range(3))
+        cache.k[i].index_copy_(2, pos_dev, k)
+        cache.v[i].index_copy_(2, pos_dev, v)
+        o = self._sdpa_static(q, cache.k[i], cache.v[i], pos_dev + 1)
+        x = x + F.linear(o.transpose(1, 2).reshape(B, 1, -1),
+                         m.proj.weight, m.proj.bias)
+        h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+        h = F.gelu(F.linear(h, blk.fc.weight, blk.fc.bias), approximate="tanh")
+        return x + F.linear(h, blk.proj.weight, blk.proj.bias)
+
+    def _decode_step_static(self, cur_tok, cache: KVCache, pos_dev, out_dev):
+        """ONE greedy decode step with fully static shapes: embeds the
+        static cur_tok buffer, runs the blocks over the padded cache,
+        argmaxes, writes the token at pos_dev+1 of out_dev, advances
+        cur_tok and pos_dev IN PLACE (graph-replayable)."""
+        m = self.m
+        if self.is_llama:
+            x = m.embed(cur_tok)
+        else:
+            x = m.wte(cur_tok) + m.wpe(pos_dev).unsqueeze(0)
+        for i, blk in enumerate(m.blocks):
+            x = self._block_static(blk, x, cache, i, pos_dev)
+        x = (ops.rms_norm(x, m.norm_f.weight, m.norm_f.eps) if self.is_llama
+             else ops.layer_norm(x, m.ln_f.weight, m.ln_f.bias, m.ln_f.eps))
+        logits = F.linear(x[:, -1], m.lm_head.weight)
+        nxt = logits.argmax(dim=-1, keepdim=True)
+        out_dev.scatter_(1, (pos_dev + 1).expand(nxt.shape), nxt)
+        cur_tok.copy_(nxt)
+        pos_dev.add_(1)
+
+    @torch.no_grad()
+    def generate_graphed(self, prompt: torch.Tensor, max_new_tokens: int,
+                         use_graph: bool = True) -> torch.Tensor:
+        """Greedy generation with a hipGraph-replayed decode step: after
+        the (dynamic-shape) prefill, every per-token forward is ONE
+        graph replay over static buffers — cache writes, attention
+        masking, argmax and position advance all run device-side
+        against a device position scalar.  use_graph=False runs the
+        same static-shape step eagerly (the CPU-testable path)."""
+        n_layers = len(self.m.blocks)
+        B, S0 = prompt.shape
+        attn0 = self.m.blocks[0].attn
+        n_kv = attn0.n_kv if self.is_llama else attn0.n_heads
+        dtype = next(self.m.parameters()).dtype
+        L = S0 + max_new_tokens
+        cache = KVCache(n_layers, B, n_kv, L, attn0.head_dim,
+                        prompt.device, dtype)
+        out = torch.zeros(B, L, dtype=prompt.dtype, device=prompt.device)
+        out[:, :S0] = prompt
+        logits = self._forward_tokens(prompt, cache, 0)   # prefill
+        nxt = logits.argmax(dim=-1, keepdim=True)
+        out[:, S0] = nxt[:, 0]
+        if max_new_tokens == 1:
+            return out
+        cur_tok = nxt.clone()
+        # pos_dev = position the NEXT step writes its kv/token at
+        pos_dev = torch.tensor([S0], dtype=torch.int64, device=prompt.device)
+        steps = max_new_tokens - 1
+        if not (use_graph and prompt.is_cuda):
+            for _ in range(steps):
+                self._decode_step_static(cur_tok, cache, pos_dev, out)
+            return out
+        # warm up twice on a side stream, then capture one step
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        did = 0
+        with torch.cuda.stream(s):
+            for _ in range(min(2, steps)):
+                self._decode_step_static(cur_tok, cache, pos_dev, out)
+                did += 1
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            # capture RECORDS without executing: state does not advance
+            self._decode_step_static(cur_tok, cache, pos_dev, out)
+        for _ in range(steps - did):
+            g.replay()
+        return out
+

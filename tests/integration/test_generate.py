@@ -79,3 +79,17 @@ def test_decode_logits_match_training_forward():
     x = ops.layer_norm(x, m.ln_f.weight, m.ln_f.bias, m.ln_f.eps)
     logits_train = F.linear(x[:, -1], m.lm_head.weight)
     assert torch.allclose(logits_dec, logits_train, atol=1e-5)
+
+
+def test_static_decode_matches_dynamic():
+    """generate_graphed's static-shape step (padded cache, device-side
+    position) == the dynamic-shape path, both families (CPU eager)."""
+    for fam, extra in [("gpt2", []),
+                       ("llama", ["model.n_kv_heads=2", "model.d_ff=128"])]:
+        m = _tiny(fam, extra) if extra else _tiny(fam)
+        g = Generator(m)
+        torch.manual_seed(4)
+        prompt = torch.randint(0, 128, (2, 7))
+        a = g.generate(prompt, max_new_tokens=9)
+        b = g.generate_graphed(prompt, max_new_tokens=9, use_graph=False)
+        assert torch.equal(a, b), fam

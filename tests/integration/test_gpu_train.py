@@ -169,3 +169,23 @@ def test_hip_graph_step_replay():
     l0 = t.model(*next(iter(t.data))).float().item()
     assert torch.isfinite(torch.tensor(l0))
     assert l0 < 5.0, l0  # ln(16)=2.77 start; must have learned something
+
+
+def test_generate_graphed_gpu_matches():
+    """hipGraph-replayed decode == plain greedy decode on GPU."""
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=128", "model.d_ff=256",
+        "model.vocab_size=512", "model.max_seq_len=128",
+    ])
+    torch.manual_seed(9)
+    m = build_model(cfg).to("cuda", torch.bfloat16).eval()
+    g = Generator(m)
+    prompt = torch.randint(0, 512, (2, 16), device="cuda")
+    a = g.generate(prompt, max_new_tokens=24)
+    b = g.generate_graphed(prompt, max_new_tokens=24)
+    torch.cuda.synchronize()
+    assert torch.equal(a, b)
