@@ -70,6 +70,8 @@ std::vector<torch::Tensor> bn_fwd(torch::Tensor x,
                                   c10::optional<torch::Tensor> running_mean,
                                   c10::optional<torch::Tensor> running_var,
                                   double momentum, double eps, bool relu);
+torch::Tensor decode_gemv(torch::Tensor x, torch::Tensor w,
+                          c10::optional<torch::Tensor> bias);
 std::vector<torch::Tensor> bn_bwd(torch::Tensor dy, torch::Tensor x,
                                   c10::optional<torch::Tensor> res,
                                   torch::Tensor weight, torch::Tensor bias,
@@ -120,6 +122,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("res"), py::arg("weight"), py::arg("bias"),
         py::arg("running_mean"), py::arg("running_var"), py::arg("momentum"),
         py::arg("eps"), py::arg("relu"));
+  m.def("decode_gemv", &decode_gemv,
+        "decode-regime GEMV y = x W^T (+bias), B<=8",
+        py::arg("x"), py::arg("w"), py::arg("bias") = py::none());
   m.def("bn_bwd", &bn_bwd, "fused bf16 NHWC BatchNorm bwd",
         py::arg("dy"), py::arg("x"), py::arg("res"), py::arg("weight"),
         py::arg("bias"), py::arg("mean"), py::arg("rstd"), py::arg("relu"));

@@ -39,6 +39,7 @@ _SOURCES = [
     "kernels/swiglu.hip",
     "kernels/colsum.hip",
     "kernels/batchnorm.hip",
+    "kernels/decode_gemv.hip",
     "kernels/probe.hip",
 ]
 

@@ -548,3 +548,22 @@ def test_colsum_vs_fp32(ext, R, C):
     assert torch.allclose(got, want, rtol=1e-2, atol=R * 2e-3), \
         (got - want).abs().max().item()
 
+
+
+# -------------------------------------------------------- decode GEMV
+@pytest.mark.parametrize("B,N,K,bias", [
+    (1, 512, 256, False), (4, 1000, 768, True), (8, 4096, 4096, False),
+])
+def test_decode_gemv_vs_fp32(ext, B, N, K, bias):
+    torch.manual_seed(33)
+    x = torch.randn(B, K)
+    w = torch.randn(N, K) * 0.05
+    b = torch.randn(N) if bias else None
+    got = ext.decode_gemv(_bf(x).cuda(), _bf(w).cuda(),
+                          _bf(b).cuda() if bias else None).cpu().float()
+    want = _bf(x).float() @ _bf(w).float().t()
+    if bias:
+        want = want + _bf(b).float()
+    torch.cuda.synchronize()
+    assert torch.allclose(got, want, rtol=2e-2, atol=2e-2), \
+        (got - want).abs().max().item()
