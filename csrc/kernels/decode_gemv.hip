@@ -31,33 +31,51 @@ decode_gemv_kernel(const float* __restrict__ x,  // [B][K] fp32 (pre-upcast)
   const int wave = (blockIdx.x * 256 + (int)threadIdx.x) >> 6;
   const int nwaves = (gridDim.x * 256) >> 6;
 
-  for (int n = wave; n < N; n += nwaves) {
-    const short* wrow = w + (long)n * K;
-    float acc[B] = {};
+  // TWO rows per wave iteration with a 4-deep unrolled k-loop: 8
+  // independent W loads in flight per lane (the v1 single-row
+  // sequential loop was latency-bound — each 512-chunk load waited the
+  // previous one's HBM round trip)
+  for (int n = wave * 2; n < N; n += nwaves * 2) {
+    const int n1 = min(n + 1, N - 1);
+    const short* wr0 = w + (long)n * K;
+    const short* wr1 = w + (long)n1 * K;
+    float acc0[B] = {}, acc1[B] = {};
+    #pragma unroll 4
     for (int k0 = l * 8; k0 < K; k0 += 64 * 8) {
-      bf16x8_t wv = *(const bf16x8_t*)(wrow + k0);
-      float wf[8];
+      bf16x8_t wv0 = *(const bf16x8_t*)(wr0 + k0);
+      bf16x8_t wv1 = *(const bf16x8_t*)(wr1 + k0);
+      float wf0[8], wf1[8];
       #pragma unroll
-      for (int j = 0; j < 8; ++j) wf[j] = bf2f(wv[j]);
+      for (int j = 0; j < 8; ++j) {
+        wf0[j] = bf2f(wv0[j]);
+        wf1[j] = bf2f(wv1[j]);
+      }
       #pragma unroll
       for (int b = 0; b < B; ++b) {
         const f32x4_t xlo = *(const f32x4_t*)(x + (long)b * K + k0);
         const f32x4_t xhi = *(const f32x4_t*)(x + (long)b * K + k0 + 4);
-        acc[b] = fmaf(wf[0], xlo[0], acc[b]);
-        acc[b] = fmaf(wf[1], xlo[1], acc[b]);
-        acc[b] = fmaf(wf[2], xlo[2], acc[b]);
-        acc[b] = fmaf(wf[3], xlo[3], acc[b]);
-        acc[b] = fmaf(wf[4], xhi[0], acc[b]);
-        acc[b] = fmaf(wf[5], xhi[1], acc[b]);
-        acc[b] = fmaf(wf[6], xhi[2], acc[b]);
-        acc[b] = fmaf(wf[7], xhi[3], acc[b]);
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          acc0[b] = fmaf(wf0[j], xlo[j], acc0[b]);
+          acc1[b] = fmaf(wf1[j], xlo[j], acc1[b]);
+        }
+        #pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          acc0[b] = fmaf(wf0[4 + j], xhi[j], acc0[b]);
+          acc1[b] = fmaf(wf1[4 + j], xhi[j], acc1[b]);
+        }
       }
     }
-    const float bv = bias ? bf2f(bias[n]) : 0.f;
+    const float bv0 = bias ? bf2f(bias[n]) : 0.f;
+    const float bv1 = bias ? bf2f(bias[n1]) : 0.f;
     #pragma unroll
     for (int b = 0; b < B; ++b) {
-      const float s = wave_reduce_sum(acc[b]);
-      if (l == 0) y[(long)b * N + n] = f2bf(s + bv);
+      const float s0 = wave_reduce_sum(acc0[b]);
+      const float s1 = wave_reduce_sum(acc1[b]);
+      if (l == 0) {
+        y[(long)b * N + n] = f2bf(s0 + bv0);
+        if (n1 != n) y[(long)b * N + n1] = f2bf(s1 + bv1);
+      }
     }
   }
 }
@@ -83,8 +101,8 @@ torch::Tensor decode_gemv(torch::Tensor x, torch::Tensor w,
     bp = (const short*)bias->data_ptr();
   }
   auto stream = scaffold_stream();
-  // 4 waves/block; enough waves that each handles a handful of rows
-  const int blocks = std::min(2048, (N + 3) / 4);
+  // 4 waves/block, 2 rows per wave iteration
+  const int blocks = std::min(2048, (N + 7) / 8);
   #define DG(BB)                                                               \
     hipLaunchKernelGGL((decode_gemv_kernel<BB>), dim3(blocks), dim3(256), 0,   \
                        stream, xf.data_ptr<float>(),                           \

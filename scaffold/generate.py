@@ -162,12 +162,17 @@ class Generator:
 
 
     def _lin1(self, x, weight, bias=None):
-        """Single-token linear: the hand decode GEMV streams each W row
-        once at the HBM rate (hipBLASLt on these skinny M<=8 shapes
-        measured ~1.7 TB/s inside the llama decode step); library path
-        otherwise."""
+        """Single-token linear.  MEASURED: the hand decode_gemv kernel
+        (2 rows/wave, 4-deep unroll) reached ~2 TB/s but hipBLASLt's
+        skinny-M GEMVs are still faster inside the graphed step (llama
+        graphed 437 vs 405 tok/s, gpt2 6117 vs 5374) — so the library
+        path stays; the kernel is parked behind SCAFFOLD_DECODE_GEMV=1
+        with its numerics test (NOTES.md r2 serving)."""
+        import os
+
         B = x.shape[0]
-        if x.is_cuda and B <= 8 and ops.has_ext():
+        if (x.is_cuda and B <= 8 and ops.has_ext()
+                and os.environ.get("SCAFFOLD_DECODE_GEMV") == "1"):
             y = ops.require_ext().decode_gemv(
                 x.reshape(B, -1).contiguous(), weight, bias)
             return y.view(B, 1, -1)
