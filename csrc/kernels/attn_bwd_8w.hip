@@ -495,3 +495,210 @@ void attn_bwd_dkv_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
   #undef LAUNCH_DKV8
   HIP_CHECK(hipGetLastError());
 }
+
+namespace {
+
+// ------------------------------------------------- dKV, ROLE-SPLIT
+// 8 waves over a 128-row kv block: waves 0-3 compute dV for rows
+// k0 + 4*lk + (w&3), waves 4-7 compute dK for the SAME rows.  The
+// per-wave register budget drops enough to PIN K and V fragments at
+// D=128 (the all-roles kernel spills 124 B/lane there), each role
+// reads only ONE tr-image in its accumulation (dO for dV, Q for dK),
+// and the whole llama-shaped grid fits in ONE residency round.  Cost:
+// S = QK^T is computed by both roles (+25% MFMA issue) — bought
+// against a kernel at <10% matrix-pipe occupancy.
+template <int D>
+__global__ void __launch_bounds__(512, 2)
+attn_bwd_dkv_rs_kernel(const short* __restrict__ dout,
+                       const short* __restrict__ q,
+                       const short* __restrict__ k,
+                       const short* __restrict__ v,
+                       const float* __restrict__ lse,
+                       const float* __restrict__ delta,
+                       short* __restrict__ dk, short* __restrict__ dv,
+                       int B, int H, int Hkv, int S, float scale,
+                       int causal, int nbx, Strides4 dos, Strides4 qs,
+                       Strides4 ks, Strides4 vs, Strides4 dks,
+                       Strides4 dvs) {
+  constexpr int NSL = D / 16;
+  constexpr int NDT = D / 32;
+  __shared__ __attribute__((aligned(16))) short bwd_lds[2 * 2 * 64 * D];
+
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int role = wid >> 2;          // 0 = dV, 1 = dK
+  const int wq = wid & 3;             // row-interleave slot
+  const int l = tid & 63;
+  const int lk = l & 31;
+  const int half = l >> 5;
+
+  int bx, bhk;
+  xcd_tile_map(nbx, &bx, &bhk);
+  const int b = bhk / Hkv, hk = bhk % Hkv;
+  const int group = H / Hkv;
+  const int k0 = bx * 128;
+  if (k0 >= S) return;
+
+  const long kbb = (long)b * ks.sb + (long)hk * ks.sh;
+  const long vbb = (long)b * vs.sb + (long)hk * vs.sh;
+  const int krow_g = k0 + 4 * lk + wq;      // interleaved kv row
+  const int krow = min(krow_g, S - 1);
+  const float scale2 = scale * LOG2E;
+
+  const unsigned qtr0 = tr_col_base<D>(bwd_lds, l);
+
+  // role-local pinned operands: dV waves pin K (for S); dK waves pin
+  // K AND V (for S and dP)
+  bf16x8_t ktf[NSL], vtf[NSL];
+  #pragma unroll
+  for (int s = 0; s < NSL; ++s) {
+    ktf[s] = *(const bf16x8_t*)(k + kbb + (long)krow * ks.ss + 16 * s + 8 * half);
+    if (role == 1)
+      vtf[s] = *(const bf16x8_t*)(v + vbb + (long)krow * vs.ss + 16 * s + 8 * half);
+  }
+
+  f32x16_t oacc[NDT];   // dV (role 0) or dK (role 1)
+  #pragma unroll
+  for (int t = 0; t < NDT; ++t) oacc[t] = {};
+
+  for (int g = 0; g < group; ++g) {
+    const int h = hk * group + g;
+    const long qb = (long)b * qs.sb + (long)h * qs.sh;
+    const long dob = (long)b * dos.sb + (long)h * dos.sh;
+    const long rhead = ((long)b * H + h) * S;
+    const int qstart = causal ? k0 : 0;
+    auto stage_qdo = [&](int buf, int c0s) {
+      short* base = bwd_lds + buf * (2 * 64 * D);
+      stage_tr64<D>(q + qb, qs.ss, c0s, S, base, tid);
+      stage_tr64<D>(dout + dob, dos.ss, c0s, S, base + 64 * D, tid);
+    };
+    stage_qdo(0, qstart);
+    __syncthreads();
+    int cur = 0;
+    for (int c0 = qstart; c0 < S; c0 += 64) {
+      #pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        const int cs = c0 + sub * 32;
+        if (cs >= S) break;
+        const unsigned bufoff = (unsigned)cur * (2 * 64 * D * 2)
+                                + (unsigned)sub * (32 * D * 2);
+        const unsigned qtr = qtr0 + bufoff;
+        const unsigned dotr = qtr + (unsigned)(64 * D * 2);
+
+        const int qrow = min(cs + lk, S - 1);
+        const short* qsub = bwd_lds + ((unsigned)cur * (2 * 64 * D))
+                            + (unsigned)sub * 32 * D;
+        const short* dosub = qsub + 64 * D;
+        const unsigned rowel = ((unsigned)(lk >> 2) * (D / 16)) * 64u +
+                               (unsigned)(lk & 3) * 16u + 8u * half;
+
+        f32x16_t sacc = {}, dpacc = {};
+        __builtin_amdgcn_s_setprio(1);
+        if (role == 0) {
+          #pragma unroll
+          for (int s = 0; s < NSL; ++s) {
+            bf16x8_t qf = *(const bf16x8_t*)(qsub + rowel + s * 64);
+            sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
+          }
+        } else {
+          #pragma unroll
+          for (int s = 0; s < NSL; ++s) {
+            bf16x8_t qf = *(const bf16x8_t*)(qsub + rowel + s * 64);
+            bf16x8_t dof = *(const bf16x8_t*)(dosub + rowel + s * 64);
+            sacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, ktf[s], sacc, 0, 0, 0);
+            dpacc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vtf[s], dpacc, 0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+
+        const float lse2_lane = lse[rhead + qrow] * LOG2E;
+        const float del_lane = delta[rhead + qrow];
+
+        if (sub == 1 && c0 + 64 < S) stage_qdo(cur ^ 1, c0 + 64);
+
+        // p (and for dK, ds) in place
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
+          const int qg = cs + qrel;
+          float sv = sacc[r] * scale2;
+          const bool masked = qg >= S || (causal && krow_g > qg);
+          const float lse_r = __shfl(lse2_lane, qrel, 64);
+          sacc[r] = masked ? 0.f : __builtin_amdgcn_exp2f(sv - lse_r);
+        }
+        if (role == 1) {
+          #pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int qrel = (r & 3) + 8 * (r >> 2) + 4 * half;
+            const float del_r = __shfl(del_lane, qrel, 64);
+            sacc[r] = sacc[r] * (dpacc[r] - del_r) * scale;
+          }
+        }
+
+        unsigned frag[2][4];
+        pack_frags((const float*)&sacc, frag);
+
+        // role 0: dV += P^T . dO_cols ; role 1: dK += dS^T . Q_cols
+        const unsigned tr = role == 0 ? dotr : qtr;
+        __builtin_amdgcn_s_setprio(1);
+        #pragma unroll
+        for (int t = 0; t < NDT; ++t) {
+          const unsigned qstep = 128u * (D / 16);
+          const unsigned o0 = tr + 128u * (unsigned)(2 * t);
+          const unsigned o1 = o0 + 128u * (unsigned)(4 * (D / 16));
+          bf16x8_pair cq = tr_read_pair(o0, o0 + qstep, o1, o1 + qstep);
+          oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *(const bf16x8_t*)frag[0], cq.f[0], oacc[t], 0, 0, 0);
+          oacc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *(const bf16x8_t*)frag[1], cq.f[1], oacc[t], 0, 0, 0);
+        }
+        __builtin_amdgcn_s_setprio(0);
+      }
+      __syncthreads();
+      cur ^= 1;
+    }
+  }
+
+  // epilogue: role 0 writes dv, role 1 writes dk
+  #pragma unroll
+  for (int t = 0; t < NDT; ++t)
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krel = (r & 3) + 8 * (r >> 2) + 4 * half;
+      const int kg = k0 + 4 * krel + wq;
+      if (kg < S) {
+        if (role == 0)
+          dv[(long)b * dvs.sb + (long)hk * dvs.sh + (long)kg * dvs.ss + 32 * t + lk]
+              = f2bf(oacc[t][r]);
+        else
+          dk[(long)b * dks.sb + (long)hk * dks.sh + (long)kg * dks.ss + 32 * t + lk]
+              = f2bf(oacc[t][r]);
+      }
+    }
+}
+
+}  // namespace
+
+void attn_bwd_dkv_rs(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                     torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                     torch::Tensor dk, torch::Tensor dv, int B, int H,
+                     int Hkv, int S, int D, double scale, bool causal) {
+  auto stream = scaffold_stream();
+  const int nbx = (S + 127) / 128;
+  dim3 grid(nbx * B * Hkv), block(512);
+  const Strides4 dos = strides_of(dout), qs = strides_of(q), ks = strides_of(k),
+                 vs = strides_of(v), dks = strides_of(dk), dvs = strides_of(dv);
+  #define LAUNCH_RS(DD)                                                        \
+    hipLaunchKernelGGL((attn_bwd_dkv_rs_kernel<DD>), grid, block, 0, stream,   \
+                       (const short*)dout.data_ptr(),                          \
+                       (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
+                       (const short*)v.data_ptr(), lse.data_ptr<float>(),      \
+                       delta.data_ptr<float>(), (short*)dk.data_ptr(),         \
+                       (short*)dv.data_ptr(), B, H, Hkv, S, (float)scale,      \
+                       causal ? 1 : 0, nbx, dos, qs, ks, vs, dks, dvs)
+  if (D == 64) LAUNCH_RS(64);
+  else if (D == 128) LAUNCH_RS(128);
+  else TORCH_CHECK(false, "attn_bwd_dkv_rs: unsupported D ", D);
+  #undef LAUNCH_RS
+  HIP_CHECK(hipGetLastError());
+}

@@ -481,14 +481,20 @@ void attn_bwd_dkv_8w(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                      torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
                      torch::Tensor dk, torch::Tensor dv, int B, int H,
                      int Hkv, int S, int D, double scale, bool causal);
+void attn_bwd_dkv_rs(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
+                     torch::Tensor v, torch::Tensor lse, torch::Tensor delta,
+                     torch::Tensor dk, torch::Tensor dv, int B, int H,
+                     int Hkv, int S, int D, double scale, bool causal);
 
 namespace {
-// SCAFFOLD_ATTN_BWD = 4w | 8w | auto (default: 8w for S >= 256).
-// Read per call so tests can exercise both kernels in one process.
+// SCAFFOLD_ATTN_BWD = 4w | 8w | rs | auto.  rs = role-split dkv
+// (dV and dK waves over the same rows) with the auto dq choice.
+// Read per call so tests can exercise the kernels in one process.
 int attn_bwd_mode() {
   const char* e = getenv("SCAFFOLD_ATTN_BWD");
   if (e && strcmp(e, "4w") == 0) return 4;
   if (e && strcmp(e, "8w") == 0) return 8;
+  if (e && strcmp(e, "rs") == 0) return 5;
   return 0;
 }
 }  // namespace
@@ -517,6 +523,31 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                        (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
                        dlse_ptr, delta.data_ptr<float>(), H, S, rows, dos, osr);
   const int mode = attn_bwd_mode();
+  if (mode == 5) {
+    // role-split dkv + the auto dq choice
+    if (S >= 256 && D <= 64) {
+      attn_bwd_dq_8w(dout, q, k, v, lse, delta, dq, B, H, Hkv, S, D, scale,
+                     causal);
+    } else {
+      #define LAUNCH_DQ4(DD)                                                   \
+        hipLaunchKernelGGL((attn_bwd_dq_mfma_kernel<DD>),                      \
+                           dim3(nbx4 * B * H), dim3(256), 0, stream,           \
+                           (const short*)dout.data_ptr(),                      \
+                           (const short*)q.data_ptr(),                         \
+                           (const short*)k.data_ptr(),                         \
+                           (const short*)v.data_ptr(), lse.data_ptr<float>(),  \
+                           delta.data_ptr<float>(), (short*)dq.data_ptr(),     \
+                           B, H, Hkv, S, (float)scale, causal ? 1 : 0, nbx4,   \
+                           dos, qs, ks, vs, dqs)
+      if (D == 64) LAUNCH_DQ4(64);
+      else LAUNCH_DQ4(128);
+      #undef LAUNCH_DQ4
+    }
+    attn_bwd_dkv_rs(dout, q, k, v, lse, delta, dk, dv, B, H, Hkv, S, D,
+                    scale, causal);
+    HIP_CHECK(hipGetLastError());
+    return;
+  }
   // auto: 8-wave at D=64 (measured ~even to +2%); 4-wave at D=128 where
   // the 8-wave dkv's 36 B/lane spill measured net-slower (NOTES.md r2)
   if ((mode == 8) || (mode == 0 && S >= 256 && D <= 64)) {
