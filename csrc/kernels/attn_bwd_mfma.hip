@@ -523,7 +523,9 @@ void attn_bwd_mfma(torch::Tensor dout, torch::Tensor q, torch::Tensor k,
                        (const short*)dout.data_ptr(), (const short*)o.data_ptr(),
                        dlse_ptr, delta.data_ptr<float>(), H, S, rows, dos, osr);
   const int mode = attn_bwd_mode();
-  if (mode == 5) {
+  // auto (re-measured r2): role-split dkv at D=128 (llama train 6.13 ->
+  // 5.30 ms); 8-wave at D=64 (rs measured 0.71 vs 8w 0.63 ms there)
+  if (mode == 5 || (mode == 0 && D == 128 && S >= 128)) {
     // role-split dkv + the auto dq choice
     if (S >= 256 && D <= 64) {
       attn_bwd_dq_8w(dout, q, k, v, lse, delta, dq, B, H, Hkv, S, D, scale,
