@@ -238,12 +238,17 @@ attn_bwd_dq_8w_kernel(const short* __restrict__ dout,
 
   if (qrow_g < S) {
     const long obase = (long)b * dqs.sb + (long)h * dqs.sh + (long)qrow_g * dqs.ss;
+    // packed 8-B stores per r-quad (consecutive d in this lane)
+    typedef short bf16x4_s __attribute__((ext_vector_type(4)));
     #pragma unroll
     for (int t = 0; t < NDT; ++t)
       #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int d = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * half;
-        dq[obase + d] = f2bf(dqacc[t][r]);
+      for (int rq = 0; rq < 4; ++rq) {
+        bf16x4_s pk;
+        #pragma unroll
+        for (int j = 0; j < 4; ++j)
+          pk[j] = f2bf(dqacc[t][rq * 4 + j]);
+        *(bf16x4_s*)(dq + obase + 32 * t + 8 * rq + 4 * half) = pk;
       }
   }
 }

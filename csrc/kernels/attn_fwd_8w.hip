@@ -47,7 +47,7 @@ DEVINL unsigned cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
-template <int D>
+template <int D, int PACK>
 __global__ void __launch_bounds__(512, 2)
 attn_fwd_8w_kernel(const short* __restrict__ q, const short* __restrict__ k,
                    const short* __restrict__ v, short* __restrict__ o,
@@ -314,13 +314,31 @@ attn_fwd_8w_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
   if (qrow_g < S) {
     const long obase = (long)b * os.sb + (long)h * os.sh + (long)qrow_g * os.ss;
-    #pragma unroll
-    for (int t = 0; t < NDT; ++t)
+    // packed store: each r-quad's four d-values are consecutive in
+    // THIS lane (d = 32t + (r&3) + 8(r>>2) + 4h), so one 8-B store per
+    // quad replaces 4 scalar stores (the store tail is issue-bound —
+    // guide T21 diagnostic)
+    typedef short bf16x4_s __attribute__((ext_vector_type(4)));
+    if (PACK) {
       #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int d = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * half;
-        o[obase + d] = f2bf(oacc[t][r] * inv_l);
-      }
+      for (int t = 0; t < NDT; ++t)
+        #pragma unroll
+        for (int rq = 0; rq < 4; ++rq) {
+          bf16x4_s pk;
+          #pragma unroll
+          for (int j = 0; j < 4; ++j)
+            pk[j] = f2bf(oacc[t][rq * 4 + j] * inv_l);
+          *(bf16x4_s*)(o + obase + 32 * t + 8 * rq + 4 * half) = pk;
+        }
+    } else {
+      #pragma unroll
+      for (int t = 0; t < NDT; ++t)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int d = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * half;
+          o[obase + d] = f2bf(oacc[t][r] * inv_l);
+        }
+    }
     if (half == 0)
       // lse contract is NATURAL log (backward consumes it):
       // max and sum ran in base 2, so lse = ln2 * (m2 + log2(l))
@@ -341,8 +359,17 @@ void attn_fwd_8w(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                  os = strides_of(o);
   // exp2-fold: scale * log2(e) — softmax runs in the base-2 domain
   const float scale2 = (float)scale * 1.4426950408889634f;
+  const char* pk_env = getenv("SCAFFOLD_ATTN_PACK");
+  const int pack = pk_env ? atoi(pk_env) : 1;
   #define LAUNCH_F8(DD)                                                        \
-    hipLaunchKernelGGL((attn_fwd_8w_kernel<DD>), grid, block, 0, stream,       \
+    if (pack)                                                                  \
+      hipLaunchKernelGGL((attn_fwd_8w_kernel<DD, 1>), grid, block, 0, stream,  \
+                       (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
+                       (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
+                       lse.data_ptr<float>(), B, H, Hkv, S, scale2,            \
+                       causal ? 1 : 0, nbx, qs, ks, vs, os);                   \
+    else                                                                       \
+      hipLaunchKernelGGL((attn_fwd_8w_kernel<DD, 0>), grid, block, 0, stream,  \
                        (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
                        (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
                        lse.data_ptr<float>(), B, H, Hkv, S, scale2,            \

@@ -44,7 +44,7 @@ DEVINL unsigned pack_bf16x2(float lo, float hi) {
 // allocator to fit 2 waves/SIMD (<=256 regs incl. AGPRs) — at D=128 the
 // default allocation was 362 regs -> 1 wave/SIMD and every barrier
 // stalled the whole SIMD (measured 11% slower than the 2-wave build).
-template <int D>
+template <int D, int PACK>
 __global__ void __launch_bounds__(256, 2)
 attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
                      const short* __restrict__ v, short* __restrict__ o,
@@ -263,13 +263,28 @@ attn_fwd_mfma_kernel(const short* __restrict__ q, const short* __restrict__ k,
   const float inv_l = (l_run > 0.f) ? 1.f / l_run : 0.f;
   if (qrow_g < S) {
     const long obase = (long)b * os.sb + (long)h * os.sh + (long)qrow_g * os.ss;
-    #pragma unroll
-    for (int t = 0; t < NDT; ++t)
+    // packed 8-B stores per r-quad (see the 8-wave kernel's note)
+    typedef short bf16x4_s __attribute__((ext_vector_type(4)));
+    if (PACK) {
       #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int d = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * half;
-        o[obase + d] = f2bf(oacc[t][r] * inv_l);
-      }
+      for (int t = 0; t < NDT; ++t)
+        #pragma unroll
+        for (int rq = 0; rq < 4; ++rq) {
+          bf16x4_s pk;
+          #pragma unroll
+          for (int j = 0; j < 4; ++j)
+            pk[j] = f2bf(oacc[t][rq * 4 + j] * inv_l);
+          *(bf16x4_s*)(o + obase + 32 * t + 8 * rq + 4 * half) = pk;
+        }
+    } else {
+      #pragma unroll
+      for (int t = 0; t < NDT; ++t)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int d = 32 * t + (r & 3) + 8 * (r >> 2) + 4 * half;
+          o[obase + d] = f2bf(oacc[t][r] * inv_l);
+        }
+    }
     if (half == 0)
       lse[((long)b * H + h) * S + qrow_g] = m_run + __logf(fmaxf(l_run, 1e-30f));
   }
@@ -285,8 +300,19 @@ void attn_fwd_mfma(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   dim3 grid(nbx * B * H), block(256);
   const Strides4 qs = strides_of(q), ks = strides_of(k), vs = strides_of(v),
                  os = strides_of(o);
+  const char* pk_env = getenv("SCAFFOLD_ATTN_PACK");
+  const int pack = pk_env ? atoi(pk_env) : 1;
   #define LAUNCH_F(DD)                                                         \
-    hipLaunchKernelGGL((attn_fwd_mfma_kernel<DD>), grid, block, 0, stream,     \
+    if (pack)                                                                  \
+      hipLaunchKernelGGL((attn_fwd_mfma_kernel<DD, 1>), grid, block, 0,        \
+                       stream,                                                 \
+                       (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
+                       (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
+                       lse.data_ptr<float>(), B, H, Hkv, S, (float)scale,      \
+                       causal ? 1 : 0, nbx, qs, ks, vs, os);                   \
+    else                                                                       \
+      hipLaunchKernelGGL((attn_fwd_mfma_kernel<DD, 0>), grid, block, 0,        \
+                       stream,                                                 \
                        (const short*)q.data_ptr(), (const short*)k.data_ptr(), \
                        (const short*)v.data_ptr(), (short*)o.data_ptr(),       \
                        lse.data_ptr<float>(), B, H, Hkv, S, (float)scale,      \
