@@ -43,6 +43,10 @@ class ModelConfig:
     rope_theta: float = 500000.0
     norm_eps: float = 1e-5
     tie_embeddings: bool = True
+    # recompute each transformer block in backward instead of storing
+    # its activations (trades ~1 extra block fwd for O(1) activation
+    # memory per block); LM families only, no-op in eval
+    act_ckpt: bool = False
 
 
 @dataclass

@@ -71,6 +71,7 @@ class GPT2(nn.Module):
         super().__init__()
         m = cfg.model
         d_ff = m.d_ff or 4 * m.d_model
+        self.act_ckpt = getattr(m, "act_ckpt", False)
         self.wte = nn.Embedding(m.vocab_size, m.d_model)
         self.wpe = nn.Embedding(m.max_seq_len, m.d_model)
         self.blocks = nn.ModuleList(
@@ -100,6 +101,12 @@ class GPT2(nn.Module):
         pos = torch.arange(S, device=tokens.device)
         x = self.wte(tokens) + self.wpe(pos)
         for blk in self.blocks:
-            x = blk(x)
+            if self.act_ckpt and self.training:
+                # no dropout anywhere in the family -> recompute is
+                # bitwise-identical (tests/unit/test_act_ckpt.py)
+                x = torch.utils.checkpoint.checkpoint(
+                    blk, x, use_reentrant=False)
+            else:
+                x = blk(x)
         # fused chunked lm_head+CE: the [T, V] logits never round-trip HBM
         return ops.lm_head_ce(self.ln_f(x), self.lm_head.weight, targets)

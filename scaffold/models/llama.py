@@ -74,6 +74,7 @@ class Llama(nn.Module):
     def __init__(self, cfg):
         super().__init__()
         m = cfg.model
+        self.act_ckpt = getattr(m, "act_ckpt", False)
         d_ff = m.d_ff or int(8 * m.d_model / 3 + 255) // 256 * 256
         n_kv = m.n_kv_heads or m.n_heads
         self.cfg = cfg
@@ -98,6 +99,10 @@ class Llama(nn.Module):
         sin = self.rope_sin[:S]
         x = self.embed(tokens)
         for blk in self.blocks:
-            x = blk(x, cos, sin)
+            if self.act_ckpt and self.training:
+                x = torch.utils.checkpoint.checkpoint(
+                    blk, x, cos, sin, use_reentrant=False)
+            else:
+                x = blk(x, cos, sin)
         # fused chunked lm_head+CE (ops.lm_head_ce)
         return ops.lm_head_ce(self.norm_f(x), self.lm_head.weight, targets)
