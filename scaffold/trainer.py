@@ -120,7 +120,17 @@ class Trainer:
                       flush=True)
                 self.optimizer.step_dev = None
                 use_graph = False
+        # test-only fault injection for the elastic-launcher tier
+        # (SURVEY §5.3): SCAFFOLD_FAULT_STEP/_RANK kill this rank
+        # mid-run, uncleanly, exactly once (the relaunched world clears
+        # the guard via SCAFFOLD_FAULT_DONE in its env).
+        fault_step = int(os.environ.get("SCAFFOLD_FAULT_STEP", "-1"))
+        fault_rank = int(os.environ.get("SCAFFOLD_FAULT_RANK", "0"))
         while self.step < total:
+            if (fault_step >= 0 and self.step == fault_step
+                    and self.ctx.rank == fault_rank
+                    and "SCAFFOLD_FAULT_DONE" not in os.environ):
+                os._exit(17)
             t0 = time.perf_counter()
             loss_val = (self._train_step_graph() if use_graph
                         else self._train_step(batch_iter, accum))
