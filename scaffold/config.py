@@ -59,6 +59,7 @@ class DataConfig:
     num_classes: int = 1000
     vector_dim: int = 256
     active_vocab: int = 0  # LM data: restrict token ids to [0,N) for learnable signal
+    path: str = ""  # token_file: flat binary of token ids (uint16/uint32 by vocab)
     seed: int = 1234
 
 

@@ -126,4 +126,68 @@ def build_data(cfg: TrainConfig, rank: int, device: torch.device, dtype: torch.d
         return SyntheticTokenData(cfg, rank, device, dtype)
     if kind == "synthetic_vector":
         return SyntheticVectorData(cfg, rank, device, dtype)
+    if kind == "token_file":
+        return TokenFileData(cfg, rank, device, dtype)
     raise ValueError(f"unknown data kind {kind!r}")
+
+
+class TokenFileData:
+    """Real-data LM path: a flat binary corpus of token ids, memory-
+    mapped and sampled as random (seq_len+1)-windows per rank.
+
+    File format is tokenizer-agnostic (nanoGPT-style): contiguous
+    uint16 ids when model.vocab_size <= 65536, else uint32 — anything
+    the wheelhouse `tokenizers`/`sentencepiece` emit can be dumped with
+    ndarray.tofile.  Window starts come from a CPU torch.Generator
+    seeded with the same (seed, rank, step) fold as the synthetic
+    streams, and its state rides the checkpoint's existing
+    `rng.data.{rank}` slot (scaffold/ckpt/sharded.py), so resume is
+    bit-exact and each rank sees a distinct stream.
+    """
+
+    def __init__(self, cfg: TrainConfig, rank: int, device: torch.device,
+                 dtype: torch.dtype):
+        import numpy as np
+
+        if not cfg.data.path:
+            raise ValueError("data.kind=token_file requires data.path")
+        self.cfg = cfg
+        self.rank = rank
+        self.device = device
+        nd = np.uint16 if cfg.model.vocab_size <= 65536 else np.uint32
+        self.arr = np.memmap(cfg.data.path, dtype=nd, mode="r")
+        need = cfg.data.seq_len + 1
+        if len(self.arr) < need + 1:
+            raise ValueError(
+                f"token file {cfg.data.path} has {len(self.arr)} tokens; "
+                f"need > {need}")
+        self.gen = torch.Generator()
+        self.reset()
+
+    def reset(self, step: int = 0) -> None:
+        self.gen.manual_seed(
+            self.cfg.data.seed * 1000003 + self.rank * 997 + step)
+
+    def state_dict(self) -> dict:
+        return {"gen_state": self.gen.get_state()}
+
+    def load_state_dict(self, sd: dict) -> None:
+        self.gen.set_state(sd["gen_state"].cpu())
+
+    def __iter__(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+        while True:
+            yield self.next_batch()
+
+    def next_batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
+        import numpy as np
+
+        d = self.cfg.data
+        S = d.seq_len
+        n = len(self.arr) - S - 1
+        starts = torch.randint(0, n, (d.batch_size,), generator=self.gen)
+        win = np.stack([np.asarray(self.arr[s:s + S + 1], dtype=np.int64)
+                        for s in starts.tolist()])
+        tokens = torch.from_numpy(win)
+        if self.device.type == "cuda":
+            tokens = tokens.pin_memory().to(self.device, non_blocking=True)
+        return tokens[:, :-1].contiguous(), tokens[:, 1:].contiguous()
