@@ -143,3 +143,70 @@ def load_hf_dir(model: torch.nn.Module, path: str) -> None:
         binp = os.path.join(path, "pytorch_model.bin")
         sd = torch.load(binp, map_location="cpu", weights_only=True)
     load_hf(model, sd)
+
+
+def _permute_rope(w: torch.Tensor, n_heads: int) -> torch.Tensor:
+    """Interleaved-pair row layout -> HF rotate_half (inverse of
+    _unpermute_rope)."""
+    out_dim, in_dim = w.shape
+    hd = out_dim // n_heads
+    return (w.view(n_heads, hd // 2, 2, in_dim)
+             .transpose(1, 2).reshape(out_dim, in_dim))
+
+
+@torch.no_grad()
+def to_hf_gpt2(model: GPT2) -> Dict[str, torch.Tensor]:
+    """scaffold GPT2 -> transformers GPT2LMHeadModel state dict
+    (inverse of from_hf_gpt2; roundtrip-tested)."""
+    sd = {k: v.detach().cpu() for k, v in model.state_dict().items()}
+    out = {
+        "transformer.wte.weight": sd["wte.weight"],
+        "transformer.wpe.weight": sd["wpe.weight"],
+        "transformer.ln_f.weight": sd["ln_f.weight"],
+        "transformer.ln_f.bias": sd["ln_f.bias"],
+        "lm_head.weight": sd["lm_head.weight"],
+    }
+    for i in range(len(model.blocks)):
+        h, b = f"transformer.h.{i}.", f"blocks.{i}."
+        out[h + "ln_1.weight"] = sd[b + "ln1.weight"]
+        out[h + "ln_1.bias"] = sd[b + "ln1.bias"]
+        out[h + "attn.c_attn.weight"] = sd[b + "attn.qkv.weight"].t().contiguous()
+        out[h + "attn.c_attn.bias"] = sd[b + "attn.qkv.bias"]
+        out[h + "attn.c_proj.weight"] = sd[b + "attn.proj.weight"].t().contiguous()
+        out[h + "attn.c_proj.bias"] = sd[b + "attn.proj.bias"]
+        out[h + "ln_2.weight"] = sd[b + "ln2.weight"]
+        out[h + "ln_2.bias"] = sd[b + "ln2.bias"]
+        out[h + "mlp.c_fc.weight"] = sd[b + "fc.weight"].t().contiguous()
+        out[h + "mlp.c_fc.bias"] = sd[b + "fc.bias"]
+        out[h + "mlp.c_proj.weight"] = sd[b + "proj.weight"].t().contiguous()
+        out[h + "mlp.c_proj.bias"] = sd[b + "proj.bias"]
+    return out
+
+
+@torch.no_grad()
+def to_hf_llama(model: Llama) -> Dict[str, torch.Tensor]:
+    """scaffold Llama -> transformers LlamaForCausalLM state dict."""
+    sd = {k: v.detach().cpu() for k, v in model.state_dict().items()}
+    a0 = model.blocks[0].attn
+    nh, nkv, hd = a0.n_heads, a0.n_kv, a0.head_dim
+    out = {
+        "model.embed_tokens.weight": sd["embed.weight"],
+        "model.norm.weight": sd["norm_f.weight"],
+        "lm_head.weight": sd["lm_head.weight"],
+    }
+    for i in range(len(model.blocks)):
+        h, b = f"model.layers.{i}.", f"blocks.{i}."
+        out[h + "input_layernorm.weight"] = sd[b + "attn_norm.weight"]
+        out[h + "post_attention_layernorm.weight"] = sd[b + "mlp_norm.weight"]
+        wqkv = sd[b + "attn.wqkv.weight"]
+        q, k, v = torch.split(wqkv, [nh * hd, nkv * hd, nkv * hd], dim=0)
+        out[h + "self_attn.q_proj.weight"] = _permute_rope(q, nh)
+        out[h + "self_attn.k_proj.weight"] = _permute_rope(k, nkv)
+        out[h + "self_attn.v_proj.weight"] = v.contiguous()
+        out[h + "self_attn.o_proj.weight"] = sd[b + "attn.wo.weight"]
+        gu = sd[b + "w_gateup.weight"]
+        f2 = gu.shape[0] // 2
+        out[h + "mlp.gate_proj.weight"] = gu[:f2].contiguous()
+        out[h + "mlp.up_proj.weight"] = gu[f2:].contiguous()
+        out[h + "mlp.down_proj.weight"] = sd[b + "w_down.weight"]
+    return out

@@ -124,3 +124,73 @@ def test_load_hf_rejects_shape_mismatch():
     ours = build_model(cfg)
     with pytest.raises(ValueError, match="shape"):
         load_hf(ours, hf.state_dict())
+
+
+def test_roundtrip_export_llama():
+    """ours -> HF layout -> transformers forward == ours; and
+    re-importing the export reproduces the original weights exactly."""
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    from scaffold.convert import to_hf_llama
+
+    torch.manual_seed(31)
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=128",
+        "model.vocab_size=128", "model.max_seq_len=32",
+        "model.rope_theta=500000.0",
+    ])
+    ours = build_model(cfg).eval()
+    exported = to_hf_llama(ours)
+
+    hf = LlamaForCausalLM(LlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=32, rope_theta=500000.0,
+        rms_norm_eps=1e-5, attention_bias=False, mlp_bias=False,
+        tie_word_embeddings=False, attn_implementation="eager",
+    )).eval()
+    missing, unexpected = hf.load_state_dict(exported, strict=False)
+    assert not unexpected, unexpected
+    assert all("rotary_emb" in m or "bias" not in m for m in missing), missing
+
+    tok = torch.randint(0, 128, (2, 32))
+    tgt = torch.randint(0, 128, (2, 32))
+    with torch.no_grad():
+        ref = _ref_loss(hf(tok).logits, tgt)
+    assert torch.allclose(_our_loss(ours, tok, tgt), ref,
+                          rtol=1e-4, atol=1e-5)
+
+    # import(export(x)) == x bitwise
+    clone = build_model(cfg)
+    load_hf(clone, exported)
+    for (n, a), (_, b) in zip(ours.named_parameters(),
+                              clone.named_parameters()):
+        assert torch.equal(a, b), n
+
+
+def test_roundtrip_export_gpt2():
+    from transformers import GPT2Config, GPT2LMHeadModel
+
+    from scaffold.convert import to_hf_gpt2
+
+    torch.manual_seed(37)
+    cfg = load_config(None, [
+        "model.name=gpt2", "model.n_layers=2", "model.n_heads=4",
+        "model.d_model=64", "model.vocab_size=128", "model.max_seq_len=32",
+    ])
+    ours = build_model(cfg).eval()
+    exported = to_hf_gpt2(ours)
+    hf = GPT2LMHeadModel(GPT2Config(
+        vocab_size=128, n_positions=32, n_embd=64, n_layer=2, n_head=4,
+        activation_function="gelu_new",
+        resid_pdrop=0.0, embd_pdrop=0.0, attn_pdrop=0.0,
+    )).eval()
+    missing, unexpected = hf.load_state_dict(exported, strict=False)
+    assert not unexpected, unexpected
+    tok = torch.randint(0, 128, (2, 32))
+    tgt = torch.randint(0, 128, (2, 32))
+    with torch.no_grad():
+        ref = _ref_loss(hf(tok).logits, tgt)
+    assert torch.allclose(_our_loss(ours, tok, tgt), ref,
+                          rtol=1e-4, atol=1e-5)
