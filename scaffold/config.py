@@ -43,6 +43,10 @@ class ModelConfig:
     rope_theta: float = 500000.0
     norm_eps: float = 1e-5
     tie_embeddings: bool = True
+    # gpt2: replace every block's dense FFN with a top-1-routed MoE of
+    # N local experts (0 = dense).  Expert-parallel sharding of the same
+    # layer across ranks is scaffold.dist.ep.ExpertParallelMoE
+    n_experts: int = 0
     # recompute each transformer block in backward instead of storing
     # its activations (trades ~1 extra block fwd for O(1) activation
     # memory per block); LM families only, no-op in eval

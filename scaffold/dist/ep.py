@@ -46,6 +46,14 @@ class ExpertParallelMoE(nn.Module):
             nn.Parameter(torch.randn(d_model, d_ff) * 0.02)
             for _ in range(self.e_local))
 
+    def _zero_tie(self, i_local: int) -> torch.Tensor:
+        # an expert that received no tokens this step must still appear
+        # in the graph: a bucketed-DDP reducer waits on EVERY param's
+        # grad hook, so an absent expert would stall the all-reduce
+        # (grad = exact zero either way)
+        return 0.0 * (self.w_up[i_local].flatten()[0]
+                      + self.w_down[i_local].flatten()[0])
+
     def _expert(self, i_local: int, x: torch.Tensor) -> torch.Tensor:
         h = torch.nn.functional.gelu(x @ self.w_up[i_local].t(),
                                      approximate="tanh")
@@ -103,6 +111,7 @@ class ExpertParallelMoE(nn.Module):
         for i in range(self.e_local):
             sel = (e_recv == self.e0 + i).nonzero(as_tuple=True)[0]
             if sel.numel() == 0:
+                y_recv = y_recv + self._zero_tie(i)
                 continue
             y_recv = y_recv.index_add(0, sel, self._expert(i, x_recv[sel]))
 
@@ -132,6 +141,7 @@ class ExpertParallelMoE(nn.Module):
         for i in range(self.e_local):
             sel = (top_e == self.e0 + i).nonzero(as_tuple=True)[0]
             if sel.numel() == 0:
+                y_partial = y_partial + self._zero_tie(i)
                 continue
             out = self._expert(i, x_all[sel]) * top_p[sel].unsqueeze(-1)
             y_partial = y_partial.index_add(0, sel, out)

@@ -49,19 +49,32 @@ class CausalSelfAttention(nn.Module):
 
 
 class Block(nn.Module):
-    def __init__(self, d_model: int, n_heads: int, d_ff: int, eps: float):
+    def __init__(self, d_model: int, n_heads: int, d_ff: int, eps: float,
+                 n_experts: int = 0):
         super().__init__()
         self.ln1 = LayerNorm(d_model, eps)
         self.attn = CausalSelfAttention(d_model, n_heads)
         self.ln2 = LayerNorm(d_model, eps)
-        self.fc = nn.Linear(d_model, d_ff)
-        self.proj = nn.Linear(d_ff, d_model)
+        if n_experts > 0:
+            # top-1-routed MoE FFN (model.n_experts); experts local to
+            # this rank — EP sharding composes via scaffold.dist.ep
+            from ..dist.ep import ExpertParallelMoE
+            self.moe = ExpertParallelMoE(d_model, d_ff, n_experts,
+                                         world=1, rank=0, dispatch="dense")
+        else:
+            self.moe = None
+            self.fc = nn.Linear(d_model, d_ff)
+            self.proj = nn.Linear(d_ff, d_model)
 
     def forward(self, x):
         # hot ops routed through scaffold.ops: K5 hand MFMA linears with
         # fused bias+GELU (fc) and fused residual-add (projections)
         x = self.attn(self.ln1(x), residual=x)
-        h = ops.linear_gelu(self.ln2(x), self.fc.weight, self.fc.bias)
+        h = self.ln2(x)
+        if self.moe is not None:
+            B, S, C = h.shape
+            return x + self.moe(h.reshape(B * S, C)).view(B, S, C)
+        h = ops.linear_gelu(h, self.fc.weight, self.fc.bias)
         x = ops.linear_residual(h, self.proj.weight, self.proj.bias, x)
         return x
 
@@ -75,7 +88,8 @@ class GPT2(nn.Module):
         self.wte = nn.Embedding(m.vocab_size, m.d_model)
         self.wpe = nn.Embedding(m.max_seq_len, m.d_model)
         self.blocks = nn.ModuleList(
-            Block(m.d_model, m.n_heads, d_ff, m.norm_eps) for _ in range(m.n_layers)
+            Block(m.d_model, m.n_heads, d_ff, m.norm_eps,
+                  getattr(m, "n_experts", 0)) for _ in range(m.n_layers)
         )
         self.ln_f = LayerNorm(m.d_model, m.norm_eps)
         self.lm_head = nn.Linear(m.d_model, m.vocab_size, bias=False)
