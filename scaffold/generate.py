@@ -134,9 +134,34 @@ class Generator:
             x = ops.layer_norm(x, m.ln_f.weight, m.ln_f.bias, m.ln_f.eps)
         return F.linear(x[:, -1], m.lm_head.weight)
 
+    @staticmethod
+    def _sample(logits: torch.Tensor, temperature: float, top_k: int,
+                top_p: float,
+                generator: Optional[torch.Generator]) -> torch.Tensor:
+        """(B, V) logits -> (B, 1) next ids.  temperature 0 = greedy;
+        top_k keeps the k best logits, top_p keeps the smallest nucleus
+        with cumulative probability >= top_p (both standard serving
+        truncations; composable)."""
+        if temperature <= 0:
+            return logits.argmax(dim=-1, keepdim=True)
+        lg = logits.float() / temperature
+        if top_k > 0 and top_k < lg.shape[-1]:
+            kth = lg.topk(top_k, dim=-1).values[:, -1:]
+            lg = lg.masked_fill(lg < kth, float("-inf"))
+        if 0.0 < top_p < 1.0:
+            srt, idx = lg.sort(dim=-1, descending=True)
+            cum = torch.softmax(srt, dim=-1).cumsum(dim=-1)
+            # keep tokens whose PRECEDING mass is < top_p (always >= 1)
+            drop_sorted = (cum - torch.softmax(srt, dim=-1)) >= top_p
+            drop = drop_sorted.scatter(1, idx, drop_sorted)
+            lg = lg.masked_fill(drop, float("-inf"))
+        probs = torch.softmax(lg, dim=-1)
+        return torch.multinomial(probs, 1, generator=generator)
+
     @torch.no_grad()
     def generate(self, prompt: torch.Tensor, max_new_tokens: int,
-                 temperature: float = 0.0,
+                 temperature: float = 0.0, top_k: int = 0,
+                 top_p: float = 0.0,
                  generator: Optional[torch.Generator] = None) -> torch.Tensor:
         """prompt (B, S0) int64 -> (B, S0 + max_new_tokens)."""
         n_layers = len(self.m.blocks)
@@ -151,11 +176,7 @@ class Generator:
         out[:, :S0] = prompt
         logits = self._forward_tokens(prompt, cache, 0)   # prefill
         for t in range(max_new_tokens):
-            if temperature > 0:
-                probs = torch.softmax(logits.float() / temperature, dim=-1)
-                nxt = torch.multinomial(probs, 1, generator=generator)
-            else:
-                nxt = logits.argmax(dim=-1, keepdim=True)
+            nxt = self._sample(logits, temperature, top_k, top_p, generator)
             out[:, S0 + t] = nxt[:, 0]
             logits = self._forward_tokens(nxt, cache, cache.seq_len)
         return out

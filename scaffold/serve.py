@@ -50,6 +50,8 @@ try:
         tokens: list       # [int] or [[int]]
         max_new_tokens: int = 32
         temperature: float = 0.0
+        top_k: int = 0
+        top_p: float = 0.0
 except ImportError:        # pydantic optional: InferenceWorker still works
     GenRequest = None
 
@@ -59,6 +61,8 @@ class _Request:
     tokens: torch.Tensor            # (S,) int64 CPU
     max_new_tokens: int
     temperature: float
+    top_k: int = 0
+    top_p: float = 0.0
     future: "asyncio.Future | None" = None
     event: threading.Event = field(default_factory=threading.Event)
     result: Optional[List[int]] = None
@@ -66,7 +70,8 @@ class _Request:
 
     @property
     def key(self):
-        return (int(self.tokens.numel()), round(float(self.temperature), 6))
+        return (int(self.tokens.numel()), round(float(self.temperature), 6),
+                self.top_k, round(float(self.top_p), 6))
 
 
 class InferenceWorker:
@@ -96,14 +101,15 @@ class InferenceWorker:
         self._thread.start()
 
     def submit(self, tokens: List[int], max_new_tokens: int,
-               temperature: float = 0.0) -> _Request:
+               temperature: float = 0.0, top_k: int = 0,
+               top_p: float = 0.0) -> _Request:
         if not tokens:
             raise ValueError("empty prompt")
         n = min(int(max_new_tokens), self.max_new_tokens_cap)
         if n < 1:
             raise ValueError("max_new_tokens must be >= 1")
         req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
-                       float(temperature))
+                       float(temperature), int(top_k), float(top_p))
         with self._cv:
             self._q.append(req)
             self._cv.notify()
@@ -147,7 +153,9 @@ class InferenceWorker:
         if self.use_graph and temp == 0.0:
             out = self.gen.generate_graphed(prompt, steps)
         else:
-            out = self.gen.generate(prompt, steps, temperature=temp)
+            out = self.gen.generate(prompt, steps, temperature=temp,
+                                    top_k=batch[0].top_k,
+                                    top_p=batch[0].top_p)
         s0 = prompt.shape[1]
         new = out[:, s0:].cpu()
         for i, r in enumerate(batch):
@@ -209,7 +217,8 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
             raise HTTPException(400, "tokens must be a non-empty id list")
         t0 = time.time()
         try:
-            reqs = [worker.submit(p, body.max_new_tokens, body.temperature)
+            reqs = [worker.submit(p, body.max_new_tokens, body.temperature,
+                                  body.top_k, body.top_p)
                     for p in prompts]
         except ValueError as e:
             raise HTTPException(400, str(e))

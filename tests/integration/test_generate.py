@@ -93,3 +93,51 @@ def test_static_decode_matches_dynamic():
         a = g.generate(prompt, max_new_tokens=9)
         b = g.generate_graphed(prompt, max_new_tokens=9, use_graph=False)
         assert torch.equal(a, b), fam
+
+
+def test_topk_topp_sampling():
+    """top_k=1 == greedy for any temperature; top_k/top_p masks keep
+    samples inside the allowed sets."""
+    import torch
+
+    from scaffold.generate import Generator
+
+    logits = torch.tensor([[4.0, 3.0, 2.0, 1.0, 0.0],
+                           [0.0, 1.0, 2.0, 3.0, 4.0]])
+    g = torch.Generator().manual_seed(0)
+    # top_k=1 at temperature>0 is greedy
+    nxt = Generator._sample(logits, 1.0, 1, 0.0, g)
+    assert nxt[:, 0].tolist() == [0, 4]
+    # top_k=2: only the two best ids can ever be drawn
+    seen = set()
+    for _ in range(64):
+        nxt = Generator._sample(logits, 2.0, 2, 0.0, g)
+        seen.update((int(nxt[0, 0]), int(nxt[1, 0]) - 3 + 10))
+    assert seen <= {0, 1, 10, 11}, seen
+    # top_p tiny: nucleus collapses to the argmax
+    nxt = Generator._sample(logits, 5.0, 0, 1e-6, g)
+    assert nxt[:, 0].tolist() == [0, 4]
+    # greedy path (temperature 0) ignores top_k/top_p
+    nxt = Generator._sample(logits, 0.0, 3, 0.5, None)
+    assert nxt[:, 0].tolist() == [0, 4]
+
+
+def test_generate_with_sampling_runs():
+    import torch
+
+    from scaffold.config import load_config
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    cfg = load_config(None, [
+        "model.name=gpt2", "model.n_layers=1", "model.n_heads=2",
+        "model.d_model=32", "model.vocab_size=64", "model.max_seq_len=32",
+    ])
+    torch.manual_seed(1)
+    m = build_model(cfg).eval()
+    g = torch.Generator().manual_seed(7)
+    out = Generator(m).generate(torch.randint(0, 64, (2, 4)), 8,
+                                temperature=0.8, top_k=10, top_p=0.9,
+                                generator=g)
+    assert out.shape == (2, 12)
+    assert out.max() < 64 and out.min() >= 0
