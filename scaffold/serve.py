@@ -52,8 +52,16 @@ try:
         temperature: float = 0.0
         top_k: int = 0
         top_p: float = 0.0
+
+    class TextRequest(_BaseModel):
+        prompt: object             # str or [str]
+        max_new_tokens: int = 32
+        temperature: float = 0.0
+        top_k: int = 0
+        top_p: float = 0.0
 except ImportError:        # pydantic optional: InferenceWorker still works
     GenRequest = None
+    TextRequest = None
 
 
 @dataclass
@@ -179,9 +187,11 @@ class InferenceWorker:
 
 
 def create_app(model: torch.nn.Module, max_batch: int = 8,
-               window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False):
+               window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False,
+               tokenizer=None):
     """FastAPI app around an InferenceWorker.  `model` is an eval-mode
-    LM from scaffold.models (tests inject a tiny CPU one)."""
+    LM from scaffold.models (tests inject a tiny CPU one).  Passing a
+    `tokenizers.Tokenizer` adds the text endpoint /v1/completions."""
     from contextlib import asynccontextmanager
 
     from fastapi import FastAPI, HTTPException
@@ -231,6 +241,35 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
         return {"tokens": [r.result for r in reqs],
                 "latency_ms": (time.time() - t0) * 1e3}
 
+    async def _run(reqs, t0):
+        loop = asyncio.get_running_loop()
+        for r in reqs:
+            await loop.run_in_executor(None, r.event.wait)
+        for r in reqs:
+            if r.error is not None:
+                raise HTTPException(500, f"generation failed: {r.error!r}")
+        return (time.time() - t0) * 1e3
+
+    if tokenizer is not None:
+        @app.post("/v1/completions")
+        async def completions(body: TextRequest):
+            prompts = [body.prompt] if isinstance(body.prompt, str) \
+                else list(body.prompt)
+            if not prompts or not all(isinstance(p, str) and p
+                                      for p in prompts):
+                raise HTTPException(400, "prompt must be non-empty text")
+            t0 = time.time()
+            try:
+                reqs = [worker.submit(tokenizer.encode(p).ids,
+                                      body.max_new_tokens, body.temperature,
+                                      body.top_k, body.top_p)
+                        for p in prompts]
+            except ValueError as e:
+                raise HTTPException(400, str(e))
+            ms = await _run(reqs, t0)
+            return {"text": [tokenizer.decode(r.result) for r in reqs],
+                    "tokens": [r.result for r in reqs], "latency_ms": ms}
+
     return app
 
 
@@ -247,6 +286,8 @@ def main(argv=None):
     ap.add_argument("--max-batch", type=int, default=8)
     ap.add_argument("--graph", action="store_true",
                     help="hipGraph-replayed decode (fixed shapes)")
+    ap.add_argument("--tokenizer", default="",
+                    help="tokenizer.json path: adds text /v1/completions")
     args = ap.parse_args(argv)
 
     import uvicorn
@@ -254,7 +295,12 @@ def main(argv=None):
     dev = "cuda" if torch.cuda.is_available() else "cpu"
     dt = torch.bfloat16 if dev == "cuda" else torch.float32
     model = build_model(cfg).to(dev, dt).eval()
-    app = create_app(model, max_batch=args.max_batch, use_graph=args.graph)
+    tok = None
+    if args.tokenizer:
+        from tokenizers import Tokenizer
+        tok = Tokenizer.from_file(args.tokenizer)
+    app = create_app(model, max_batch=args.max_batch, use_graph=args.graph,
+                     tokenizer=tok)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 
 

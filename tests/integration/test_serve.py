@@ -141,3 +141,31 @@ def test_worker_gpu_batched_matches_reference():
             assert r.result == ref[i, 4:].tolist()
     finally:
         w.close()
+
+
+def test_text_completions_endpoint():
+    from starlette.testclient import TestClient
+    tokenizers = pytest.importorskip("tokenizers")
+    from tokenizers import Tokenizer, models, pre_tokenizers, trainers
+
+    tok = Tokenizer(models.BPE(unk_token="<unk>"))
+    tok.pre_tokenizer = pre_tokenizers.Whitespace()
+    corpus = ["the quick brown fox jumps over the lazy dog"] * 8
+    tok.train_from_iterator(corpus, trainers.BpeTrainer(
+        vocab_size=96, special_tokens=["<unk>"]))
+    assert tok.get_vocab_size() <= 128  # fits the tiny model vocab
+
+    app = create_app(_tiny_model(), window_s=0.0, tokenizer=tok)
+    with TestClient(app) as client:
+        r = client.post("/v1/completions", json={
+            "prompt": "the quick brown fox", "max_new_tokens": 6})
+        assert r.status_code == 200, r.text
+        body = r.json()
+        assert len(body["text"]) == 1 and isinstance(body["text"][0], str)
+        assert len(body["tokens"][0]) == 6
+        # batch of texts
+        r2 = client.post("/v1/completions", json={
+            "prompt": ["lazy dog", "quick fox"], "max_new_tokens": 4})
+        assert r2.status_code == 200 and len(r2.json()["text"]) == 2
+        assert client.post("/v1/completions", json={
+            "prompt": "", "max_new_tokens": 4}).status_code == 400
