@@ -51,9 +51,14 @@ class KVCache:
     def __init__(self, n_layers: int, B: int, n_kv: int, max_len: int,
                  head_dim: int, device, dtype):
         self.n_layers = n_layers
-        self.k = [torch.empty(B, n_kv, max_len, head_dim, device=device,
+        # zeros, not empty: the padded-cache attention paths
+        # (_sdpa_static / _block_varpos) compute p @ v BEFORE masking
+        # limits which rows matter, and softmax-zero times an
+        # uninitialized inf/NaN is NaN — freed CPU pages really do hold
+        # such bit patterns (caught by the continuous-batching tests)
+        self.k = [torch.zeros(B, n_kv, max_len, head_dim, device=device,
                               dtype=dtype) for _ in range(n_layers)]
-        self.v = [torch.empty_like(self.k[0]) for _ in range(n_layers)]
+        self.v = [torch.zeros_like(self.k[0]) for _ in range(n_layers)]
         self.len = 0
 
     def append(self, i: int, k: torch.Tensor, v: torch.Tensor
@@ -278,6 +283,86 @@ class Generator:
         out_dev.scatter_(1, (pos_dev + 1).expand(nxt.shape), nxt)
         cur_tok.copy_(nxt)
         pos_dev.add_(1)
+
+    # ---- variable-position decode (continuous batching) ---------------
+    def _rope_rows(self, x: torch.Tensor, pos: torch.Tensor) -> torch.Tensor:
+        """Per-ROW RoPE: x (B, H, 1, D) where row b sits at position
+        pos[b] (unlike the shared-table paths, every slot of a
+        continuous batch is at a different depth)."""
+        cos = self.m.rope_cos.index_select(0, pos).float()   # (B, D/2)
+        sin = self.m.rope_sin.index_select(0, pos).float()
+        c = cos.view(-1, 1, 1, cos.shape[-1])
+        sn = sin.view(-1, 1, 1, sin.shape[-1])
+        xf = x.float()
+        x1, x2 = xf[..., 0::2], xf[..., 1::2]
+        out = torch.empty_like(xf)
+        out[..., 0::2] = x1 * c - x2 * sn
+        out[..., 1::2] = x1 * sn + x2 * c
+        return out.to(x.dtype)
+
+    def _block_varpos(self, blk, x, cache: KVCache, i: int,
+                      pos: torch.Tensor):
+        """One decode block where row b's token sits at position pos[b]
+        ((B,) int64): per-row cache scatter, per-row causal mask, per-row
+        RoPE.  The serving continuous batcher drives this; semantics per
+        slot are identical to _block_static at that slot's depth
+        (equivalence tested against Generator.generate)."""
+        B = x.shape[0]
+        if self.is_llama:
+            m = blk.attn
+            h = ops.rms_norm(x, blk.attn_norm.weight, blk.attn_norm.eps)
+            qkv = self._lin1(h, m.wqkv.weight).view(
+                B, 1, m.n_heads + 2 * m.n_kv, m.head_dim)
+            q = qkv[:, :, :m.n_heads].transpose(1, 2)
+            k = qkv[:, :, m.n_heads:m.n_heads + m.n_kv].transpose(1, 2)
+            v = qkv[:, :, m.n_heads + m.n_kv:].transpose(1, 2)
+            q = self._rope_rows(q, pos)
+            k = self._rope_rows(k, pos)
+            idx = pos.view(B, 1, 1, 1).expand(B, m.n_kv, 1, m.head_dim)
+            cache.k[i].scatter_(2, idx, k)
+            cache.v[i].scatter_(2, idx, v)
+            o = self._sdpa_static(q, cache.k[i], cache.v[i],
+                                  (pos + 1).view(B, 1, 1, 1))
+            x = x + self._lin1(o.transpose(1, 2).reshape(B, 1, -1),
+                               m.wo.weight)
+            h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
+            gu = self._lin1(h, blk.w_gateup.weight)
+            Fh = gu.shape[-1] // 2
+            g = F.silu(gu[..., :Fh]) * gu[..., Fh:]
+            return x + self._lin1(g, blk.w_down.weight)
+        m = blk.attn
+        h = ops.layer_norm(x, blk.ln1.weight, blk.ln1.bias, blk.ln1.eps)
+        qkv = self._lin1(h, m.qkv.weight, m.qkv.bias).view(
+            B, 1, 3, m.n_heads, m.head_dim)
+        q, k, v = (qkv[:, :, j].transpose(1, 2) for j in range(3))
+        idx = pos.view(B, 1, 1, 1).expand(B, m.n_heads, 1, m.head_dim)
+        cache.k[i].scatter_(2, idx, k)
+        cache.v[i].scatter_(2, idx, v)
+        o = self._sdpa_static(q, cache.k[i], cache.v[i],
+                              (pos + 1).view(B, 1, 1, 1))
+        x = x + self._lin1(o.transpose(1, 2).reshape(B, 1, -1),
+                           m.proj.weight, m.proj.bias)
+        h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+        h = F.gelu(self._lin1(h, blk.fc.weight, blk.fc.bias),
+                   approximate="tanh")
+        return x + self._lin1(h, blk.proj.weight, blk.proj.bias)
+
+    @torch.no_grad()
+    def decode_step_varpos(self, cur_tok: torch.Tensor, cache: KVCache,
+                           pos: torch.Tensor) -> torch.Tensor:
+        """cur_tok (B, 1) int64, pos (B,) int64 (per-row position of
+        cur_tok) -> next-token logits (B, V).  Writes cur_tok's kv at
+        pos[b]; caller owns sampling and position advance."""
+        m = self.m
+        if self.is_llama:
+            x = m.embed(cur_tok)
+        else:
+            x = m.wte(cur_tok) + m.wpe(pos).unsqueeze(1)
+        for i, blk in enumerate(m.blocks):
+            x = self._block_varpos(blk, x, cache, i, pos)
+        x = (ops.rms_norm(x, m.norm_f.weight, m.norm_f.eps) if self.is_llama
+             else ops.layer_norm(x, m.ln_f.weight, m.ln_f.bias, m.ln_f.eps))
+        return self._lin1(x, m.lm_head.weight)[:, 0]
 
     @torch.no_grad()
     def generate_graphed(self, prompt: torch.Tensor, max_new_tokens: int,

@@ -186,18 +186,192 @@ class InferenceWorker:
                 r.event.set()
 
 
+class ContinuousBatcher:
+    """vLLM-style continuous batching over the Generator's variable-
+    position decode step: a fixed pool of KV-cache SLOTS, each at its
+    own depth; finished requests retire and new ones are admitted
+    mid-decode without draining the batch.
+
+    Per-step work is one decode_step_varpos over all slots — on MI355X
+    the step is weight-read-bound (profiles/decode_batch_sweep_r2.txt:
+    step time flat in batch size), so keeping slots full converts
+    directly into tokens/s.  Admission runs a (dynamic-shape) prefill
+    for the new request and copies its kv rows into the slot.
+
+    API-compatible with InferenceWorker.submit; per-slot greedy or
+    sampled decoding (temperature/top_k/top_p per request)."""
+
+    def __init__(self, model: torch.nn.Module, max_slots: int = 8,
+                 max_len: int = 0, max_new_tokens_cap: int = 512,
+                 seed: int = 0):
+        from .generate import KVCache  # noqa: F401 (type reference)
+
+        self.gen = Generator(model.eval())
+        self.device = next(model.parameters()).device
+        m = model
+        self.is_llama = self.gen.is_llama
+        attn0 = m.blocks[0].attn
+        self.n_kv = attn0.n_kv if self.is_llama else attn0.n_heads
+        self.head_dim = attn0.head_dim
+        self.n_layers = len(m.blocks)
+        if not max_len:
+            max_len = (m.rope_cos.shape[0] if self.is_llama
+                       else m.wpe.num_embeddings)
+        self.max_len = max_len
+        self.max_slots = max_slots
+        self.max_new_tokens_cap = max_new_tokens_cap
+        self.dtype = next(m.parameters()).dtype
+        self._cache = None              # allocated on first request
+        self._slots = [None] * max_slots  # per-slot state dicts
+        self._q: deque[_Request] = deque()
+        self._cv = threading.Condition()
+        self._stop = False
+        self._sample_gen = torch.Generator()
+        self._sample_gen.manual_seed(seed)
+        self.stats = {"requests": 0, "steps": 0, "tokens_generated": 0,
+                      "started_at": time.time()}
+        self._thread = threading.Thread(target=self._loop, daemon=True,
+                                        name="scaffold-cont-batcher")
+        self._thread.start()
+
+    # InferenceWorker-compatible surface ------------------------------
+    def submit(self, tokens, max_new_tokens: int, temperature: float = 0.0,
+               top_k: int = 0, top_p: float = 0.0) -> _Request:
+        if not tokens:
+            raise ValueError("empty prompt")
+        n = min(int(max_new_tokens), self.max_new_tokens_cap)
+        if n < 1:
+            raise ValueError("max_new_tokens must be >= 1")
+        if len(tokens) + n > self.max_len:
+            raise ValueError(
+                f"prompt+max_new_tokens {len(tokens) + n} exceeds "
+                f"max_len {self.max_len}")
+        req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
+                       float(temperature), int(top_k), float(top_p))
+        with self._cv:
+            self._q.append(req)
+            self._cv.notify()
+        return req
+
+    def close(self):
+        with self._cv:
+            self._stop = True
+            self._cv.notify()
+        self._thread.join(timeout=10)
+
+    # worker side ------------------------------------------------------
+    def _ensure_cache(self):
+        if self._cache is None:
+            from .generate import KVCache
+            self._cache = KVCache(self.n_layers, self.max_slots, self.n_kv,
+                                  self.max_len, self.head_dim, self.device,
+                                  self.dtype)
+
+    @torch.no_grad()
+    def _admit(self, slot: int, req: _Request):
+        """Prefill the request and install it in `slot`."""
+        from .generate import KVCache
+
+        prompt = req.tokens.view(1, -1).to(self.device)
+        S0 = prompt.shape[1]
+        tmp = KVCache(self.n_layers, 1, self.n_kv, S0, self.head_dim,
+                      self.device, self.dtype)
+        logits = self.gen._forward_tokens(prompt, tmp, 0)
+        for i in range(self.n_layers):
+            self._cache.k[i][slot, :, :S0] = tmp.k[i][0]
+            self._cache.v[i][slot, :, :S0] = tmp.v[i][0]
+        first = Generator._sample(logits, req.temperature, req.top_k,
+                                  req.top_p, self._sample_gen)
+        self._slots[slot] = {
+            "req": req, "toks": [int(first[0, 0])],
+            "cur": int(first[0, 0]), "pos": S0,
+        }
+
+    def _finish(self, slot: int):
+        st = self._slots[slot]
+        st["req"].result = st["toks"]
+        self.stats["tokens_generated"] += len(st["toks"])
+        self.stats["requests"] += 1
+        st["req"].event.set()
+        self._slots[slot] = None
+
+    @torch.no_grad()
+    def _step(self):
+        """One decode step over every occupied slot."""
+        B = self.max_slots
+        cur = torch.zeros(B, 1, dtype=torch.int64, device=self.device)
+        # idle slots park on a scratch position (their rows are ignored)
+        pos = torch.full((B,), self.max_len - 1, dtype=torch.int64,
+                         device=self.device)
+        for b, st in enumerate(self._slots):
+            if st is not None:
+                cur[b, 0] = st["cur"]
+                pos[b] = st["pos"]
+        logits = self.gen.decode_step_varpos(cur, self._cache, pos)
+        self.stats["steps"] += 1
+        for b, st in enumerate(self._slots):
+            if st is None:
+                continue
+            st["pos"] += 1
+            r = st["req"]
+            nxt = Generator._sample(logits[b:b + 1], r.temperature,
+                                    r.top_k, r.top_p, self._sample_gen)
+            st["cur"] = int(nxt[0, 0])
+            st["toks"].append(st["cur"])
+            if len(st["toks"]) >= r.max_new_tokens:
+                self._finish(b)
+
+    def _loop(self):
+        while True:
+            with self._cv:
+                while (not self._q
+                       and not any(s is not None for s in self._slots)
+                       and not self._stop):
+                    self._cv.wait()
+                if self._stop:
+                    return
+                pending = []
+                for b in range(self.max_slots):
+                    if self._slots[b] is None and self._q:
+                        pending.append((b, self._q.popleft()))
+            try:
+                self._ensure_cache()
+                for b, req in pending:
+                    try:
+                        self._admit(b, req)
+                        if len(self._slots[b]["toks"]) >= \
+                                self._slots[b]["req"].max_new_tokens:
+                            self._finish(b)
+                    except Exception as e:   # bad request: surface it
+                        req.error = e
+                        req.event.set()
+                if any(s is not None for s in self._slots):
+                    self._step()
+            except Exception as e:          # engine fault: fail occupants
+                for b, st in enumerate(self._slots):
+                    if st is not None:
+                        st["req"].error = e
+                        st["req"].event.set()
+                        self._slots[b] = None
+
+
 def create_app(model: torch.nn.Module, max_batch: int = 8,
                window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False,
-               tokenizer=None):
-    """FastAPI app around an InferenceWorker.  `model` is an eval-mode
+               tokenizer=None, continuous: bool = False):
+    """FastAPI app around an inference worker.  `model` is an eval-mode
     LM from scaffold.models (tests inject a tiny CPU one).  Passing a
-    `tokenizers.Tokenizer` adds the text endpoint /v1/completions."""
+    `tokenizers.Tokenizer` adds the text endpoint /v1/completions.
+    continuous=True serves from the ContinuousBatcher (slot-level
+    admission/retirement) instead of the same-shape batch worker."""
     from contextlib import asynccontextmanager
 
     from fastapi import FastAPI, HTTPException
 
-    worker = InferenceWorker(model.eval(), max_batch=max_batch,
-                             window_s=window_s, use_graph=use_graph)
+    if continuous:
+        worker = ContinuousBatcher(model.eval(), max_slots=max_batch)
+    else:
+        worker = InferenceWorker(model.eval(), max_batch=max_batch,
+                                 window_s=window_s, use_graph=use_graph)
 
     @asynccontextmanager
     async def _lifespan(app):
@@ -215,7 +389,11 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
     def stats():
         s = dict(worker.stats)
         s["uptime_s"] = time.time() - s.pop("started_at")
-        s["mean_batch"] = (s["requests"] / s["batches"]) if s["batches"] else 0.0
+        if s.get("batches"):
+            s["mean_batch"] = s["requests"] / s["batches"]
+        if s.get("steps"):
+            # continuous batching: average decode-slot occupancy
+            s["mean_occupancy"] = s["tokens_generated"] / s["steps"]
         return s
 
     @app.post("/v1/generate")
@@ -288,6 +466,9 @@ def main(argv=None):
                     help="hipGraph-replayed decode (fixed shapes)")
     ap.add_argument("--tokenizer", default="",
                     help="tokenizer.json path: adds text /v1/completions")
+    ap.add_argument("--continuous", action="store_true",
+                    help="continuous batching (per-slot admission) "
+                         "instead of same-shape batch grouping")
     args = ap.parse_args(argv)
 
     import uvicorn
@@ -300,7 +481,7 @@ def main(argv=None):
         from tokenizers import Tokenizer
         tok = Tokenizer.from_file(args.tokenizer)
     app = create_app(model, max_batch=args.max_batch, use_graph=args.graph,
-                     tokenizer=tok)
+                     tokenizer=tok, continuous=args.continuous)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 
 
