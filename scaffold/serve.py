@@ -223,10 +223,19 @@ class ContinuousBatcher:
         self.dtype = next(m.parameters()).dtype
         self._cache = None              # allocated on first request
         self._slots = [None] * max_slots  # per-slot state dicts
+        # device-resident decode state: current token and generated-token
+        # ring per slot.  Keeping these on device means a decode step
+        # costs ZERO host syncs (the 1.76x-serial first cut synced 8x
+        # per step pulling sampled ids back; see NOTES r2 serving) —
+        # token values only come host-side once, at request finish.
+        self._cur = torch.zeros(max_slots, 1, dtype=torch.int64,
+                                device=self.device)
+        self._toks = torch.zeros(max_slots, max_new_tokens_cap,
+                                 dtype=torch.int64, device=self.device)
         self._q: deque[_Request] = deque()
         self._cv = threading.Condition()
         self._stop = False
-        self._sample_gen = torch.Generator()
+        self._sample_gen = torch.Generator(device=self.device)
         self._sample_gen.manual_seed(seed)
         self.stats = {"requests": 0, "steps": 0, "tokens_generated": 0,
                       "started_at": time.time()}
@@ -282,43 +291,48 @@ class ContinuousBatcher:
             self._cache.v[i][slot, :, :S0] = tmp.v[i][0]
         first = Generator._sample(logits, req.temperature, req.top_k,
                                   req.top_p, self._sample_gen)
-        self._slots[slot] = {
-            "req": req, "toks": [int(first[0, 0])],
-            "cur": int(first[0, 0]), "pos": S0,
-        }
+        self._cur[slot].copy_(first[0])
+        self._toks[slot, 0].copy_(first[0, 0])
+        self._slots[slot] = {"req": req, "count": 1, "pos": S0}
 
     def _finish(self, slot: int):
         st = self._slots[slot]
-        st["req"].result = st["toks"]
-        self.stats["tokens_generated"] += len(st["toks"])
+        # the one host sync of a request's lifetime
+        st["req"].result = self._toks[slot, :st["count"]].cpu().tolist()
+        self.stats["tokens_generated"] += st["count"]
         self.stats["requests"] += 1
         st["req"].event.set()
         self._slots[slot] = None
 
     @torch.no_grad()
     def _step(self):
-        """One decode step over every occupied slot."""
-        B = self.max_slots
-        cur = torch.zeros(B, 1, dtype=torch.int64, device=self.device)
+        """One decode step over every occupied slot — no host syncs:
+        positions/counts are host ints (host->device copies only),
+        sampled ids stay in the device-side _cur/_toks buffers."""
         # idle slots park on a scratch position (their rows are ignored)
-        pos = torch.full((B,), self.max_len - 1, dtype=torch.int64,
-                         device=self.device)
-        for b, st in enumerate(self._slots):
-            if st is not None:
-                cur[b, 0] = st["cur"]
-                pos[b] = st["pos"]
-        logits = self.gen.decode_step_varpos(cur, self._cache, pos)
+        host_pos = [st["pos"] if st is not None else self.max_len - 1
+                    for st in self._slots]
+        pos = torch.tensor(host_pos, dtype=torch.int64, device=self.device)
+        logits = self.gen.decode_step_varpos(self._cur, self._cache, pos)
         self.stats["steps"] += 1
-        for b, st in enumerate(self._slots):
-            if st is None:
-                continue
+        active = [b for b, st in enumerate(self._slots) if st is not None]
+        greedy = all(self._slots[b]["req"].temperature <= 0 for b in active)
+        if greedy:
+            nxt = logits.argmax(dim=-1, keepdim=True)
+        else:
+            nxt = torch.zeros_like(self._cur)
+            for b in active:
+                r = self._slots[b]["req"]
+                nxt[b] = Generator._sample(logits[b:b + 1], r.temperature,
+                                           r.top_k, r.top_p,
+                                           self._sample_gen)[0]
+        self._cur.copy_(nxt)
+        for b in active:
+            st = self._slots[b]
             st["pos"] += 1
-            r = st["req"]
-            nxt = Generator._sample(logits[b:b + 1], r.temperature,
-                                    r.top_k, r.top_p, self._sample_gen)
-            st["cur"] = int(nxt[0, 0])
-            st["toks"].append(st["cur"])
-            if len(st["toks"]) >= r.max_new_tokens:
+            self._toks[b, st["count"]].copy_(nxt[b, 0])
+            st["count"] += 1
+            if st["count"] >= st["req"].max_new_tokens:
                 self._finish(b)
 
     def _loop(self):
@@ -339,7 +353,7 @@ class ContinuousBatcher:
                 for b, req in pending:
                     try:
                         self._admit(b, req)
-                        if len(self._slots[b]["toks"]) >= \
+                        if self._slots[b]["count"] >= \
                                 self._slots[b]["req"].max_new_tokens:
                             self._finish(b)
                     except Exception as e:   # bad request: surface it
