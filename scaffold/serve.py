@@ -203,7 +203,7 @@ class ContinuousBatcher:
 
     def __init__(self, model: torch.nn.Module, max_slots: int = 8,
                  max_len: int = 0, max_new_tokens_cap: int = 512,
-                 seed: int = 0):
+                 seed: int = 0, use_graph: bool = False):
         from .generate import KVCache  # noqa: F401 (type reference)
 
         self.gen = Generator(model.eval())
@@ -230,8 +230,18 @@ class ContinuousBatcher:
         # token values only come host-side once, at request finish.
         self._cur = torch.zeros(max_slots, 1, dtype=torch.int64,
                                 device=self.device)
-        self._toks = torch.zeros(max_slots, max_new_tokens_cap,
+        # +1 scratch column: the static step scatters INACTIVE rows'
+        # (ignored) tokens there so the kernel stays branch-free
+        self._toks = torch.zeros(max_slots, max_new_tokens_cap + 1,
                                  dtype=torch.int64, device=self.device)
+        self._pos = torch.zeros(max_slots, dtype=torch.int64,
+                                device=self.device)
+        self._count = torch.zeros(max_slots, dtype=torch.int64,
+                                  device=self.device)
+        self._active = torch.zeros(max_slots, dtype=torch.int64,
+                                   device=self.device)
+        self.use_graph = use_graph
+        self._graph = None
         self._q: deque[_Request] = deque()
         self._cv = threading.Condition()
         self._stop = False
@@ -305,35 +315,96 @@ class ContinuousBatcher:
         self._slots[slot] = None
 
     @torch.no_grad()
+    def _step_static(self):
+        """The STATIC greedy step: reads/writes only the persistent
+        device buffers (_cur/_pos/_count/_active/_toks/cache), no host
+        reads — hipGraph-capturable (use_graph=True replays it)."""
+        logits = self.gen.decode_step_varpos(self._cur, self._cache,
+                                             self._pos)
+        nxt = logits.argmax(dim=-1, keepdim=True)
+        act = self._active.unsqueeze(1)
+        new = torch.where(act.bool(), nxt, self._cur)
+        self._cur.copy_(new)
+        # active rows land at their count; inactive rows at the scratch
+        # column (cap)
+        cap = self._toks.shape[1] - 1
+        idx = torch.where(self._active.bool(), self._count,
+                          torch.full_like(self._count, cap))
+        self._toks.scatter_(1, idx.unsqueeze(1), new)
+        self._count.add_(self._active)
+        self._pos.add_(self._active)
+
+    def _upload_state(self):
+        """Refresh the device buffers from the host slot mirrors (tiny
+        H2D copies, no device sync); the host stays the source of truth
+        so the eager sampled path and the static path can interleave."""
+        host_pos, host_cnt, host_act = [], [], []
+        for st in self._slots:
+            host_pos.append(st["pos"] if st is not None else self.max_len - 1)
+            host_cnt.append(st["count"] if st is not None else 0)
+            host_act.append(1 if st is not None else 0)
+        self._pos.copy_(torch.tensor(host_pos, dtype=torch.int64))
+        self._count.copy_(torch.tensor(host_cnt, dtype=torch.int64))
+        self._active.copy_(torch.tensor(host_act, dtype=torch.int64))
+
+    def _advance_host(self):
+        for b, st in enumerate(self._slots):
+            if st is None:
+                continue
+            st["pos"] += 1
+            st["count"] += 1
+            if st["count"] >= st["req"].max_new_tokens:
+                self._finish(b)
+
+    @torch.no_grad()
     def _step(self):
         """One decode step over every occupied slot — no host syncs:
         positions/counts are host ints (host->device copies only),
         sampled ids stay in the device-side _cur/_toks buffers."""
-        # idle slots park on a scratch position (their rows are ignored)
-        host_pos = [st["pos"] if st is not None else self.max_len - 1
-                    for st in self._slots]
-        pos = torch.tensor(host_pos, dtype=torch.int64, device=self.device)
-        logits = self.gen.decode_step_varpos(self._cur, self._cache, pos)
-        self.stats["steps"] += 1
         active = [b for b, st in enumerate(self._slots) if st is not None]
         greedy = all(self._slots[b]["req"].temperature <= 0 for b in active)
+        self._upload_state()
         if greedy:
-            nxt = logits.argmax(dim=-1, keepdim=True)
-        else:
-            nxt = torch.zeros_like(self._cur)
-            for b in active:
-                r = self._slots[b]["req"]
-                nxt[b] = Generator._sample(logits[b:b + 1], r.temperature,
-                                           r.top_k, r.top_p,
-                                           self._sample_gen)[0]
+            if self.use_graph and self.device.type == "cuda":
+                self._graphed_step()
+            else:
+                self._step_static()
+            self.stats["steps"] += 1
+            self._advance_host()
+            return
+        # sampled path (per-request temperature/top_k/top_p): eager
+        logits = self.gen.decode_step_varpos(self._cur, self._cache,
+                                             self._pos)
+        self.stats["steps"] += 1
+        nxt = self._cur.clone()
+        for b in active:
+            r = self._slots[b]["req"]
+            nxt[b] = Generator._sample(logits[b:b + 1], r.temperature,
+                                       r.top_k, r.top_p,
+                                       self._sample_gen)[0]
         self._cur.copy_(nxt)
         for b in active:
-            st = self._slots[b]
-            st["pos"] += 1
-            self._toks[b, st["count"]].copy_(nxt[b, 0])
-            st["count"] += 1
-            if st["count"] >= st["req"].max_new_tokens:
-                self._finish(b)
+            self._toks[b, self._slots[b]["count"]].copy_(nxt[b, 0])
+        self._advance_host()
+
+    def _graphed_step(self):
+        """Capture once (after two warmup executions — generate_graphed's
+        proven recipe), then replay; the graph reads the persistent
+        buffers that _upload_state refreshes."""
+        if self._graph is None:
+            if getattr(self, "_warmups", 0) < 2:
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    self._step_static()
+                torch.cuda.current_stream().wait_stream(side)
+                self._warmups = getattr(self, "_warmups", 0) + 1
+                return
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self._step_static()   # records WITHOUT executing
+            self._graph = g
+        self._graph.replay()
 
     def _loop(self):
         while True:
@@ -382,7 +453,8 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
     from fastapi import FastAPI, HTTPException
 
     if continuous:
-        worker = ContinuousBatcher(model.eval(), max_slots=max_batch)
+        worker = ContinuousBatcher(model.eval(), max_slots=max_batch,
+                                   use_graph=use_graph)
     else:
         worker = InferenceWorker(model.eval(), max_batch=max_batch,
                                  window_s=window_s, use_graph=use_graph)

@@ -125,3 +125,22 @@ def test_http_surface_continuous():
         assert all(len(t) == 5 for t in body["tokens"])
         s = client.get("/v1/stats").json()
         assert s["requests"] == 2 and "mean_occupancy" in s
+
+
+def test_sampled_requests_complete():
+    """Non-greedy requests take the eager sampled path; mixing sampled
+    and greedy requests in one pool must still complete correctly."""
+    m = _model()
+    cb = ContinuousBatcher(m, max_slots=3, seed=5)
+    try:
+        a = cb.submit([1, 2, 3], 8, temperature=0.9, top_k=20)
+        b = cb.submit([4, 5], 6, temperature=0.7, top_p=0.9)
+        c = cb.submit([6, 7, 8], 5)               # greedy alongside
+        for r in (a, b, c):
+            assert r.event.wait(timeout=60) and r.error is None
+        assert len(a.result) == 8 and len(b.result) == 6
+        assert all(0 <= t < 128 for t in a.result + b.result)
+        # the greedy request is unaffected by its sampled neighbours
+        assert c.result == _ref(m, [6, 7, 8], 5)
+    finally:
+        cb.close()
