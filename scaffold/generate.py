@@ -13,7 +13,7 @@ SwiGLU).  Greedy (temperature=0) or temperature sampling.
 from __future__ import annotations
 
 import math
-from typing import List, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 import torch.nn.functional as F
@@ -166,9 +166,11 @@ class Generator:
     @torch.no_grad()
     def generate(self, prompt: torch.Tensor, max_new_tokens: int,
                  temperature: float = 0.0, top_k: int = 0,
-                 top_p: float = 0.0,
+                 top_p: float = 0.0, eos_token_id: Optional[int] = None,
                  generator: Optional[torch.Generator] = None) -> torch.Tensor:
-        """prompt (B, S0) int64 -> (B, S0 + max_new_tokens)."""
+        """prompt (B, S0) int64 -> (B, S0 + max_new_tokens).  With
+        eos_token_id, rows freeze at EOS (subsequent positions repeat
+        it) and decoding stops early once every row has finished."""
         n_layers = len(self.m.blocks)
         B, S0 = prompt.shape
         attn0 = self.m.blocks[0].attn
@@ -180,9 +182,17 @@ class Generator:
                           device=prompt.device)
         out[:, :S0] = prompt
         logits = self._forward_tokens(prompt, cache, 0)   # prefill
+        done = torch.zeros(B, dtype=torch.bool, device=prompt.device)
         for t in range(max_new_tokens):
             nxt = self._sample(logits, temperature, top_k, top_p, generator)
+            if eos_token_id is not None:
+                nxt = torch.where(done.unsqueeze(1),
+                                  torch.full_like(nxt, eos_token_id), nxt)
+                done = done | (nxt[:, 0] == eos_token_id)
             out[:, S0 + t] = nxt[:, 0]
+            if eos_token_id is not None and bool(done.all()):
+                out[:, S0 + t + 1:] = eos_token_id
+                break
             logits = self._forward_tokens(nxt, cache, cache.seq_len)
         return out
 

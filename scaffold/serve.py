@@ -52,6 +52,7 @@ try:
         temperature: float = 0.0
         top_k: int = 0
         top_p: float = 0.0
+        eos_token_id: "int | None" = None
 
     class TextRequest(_BaseModel):
         prompt: object             # str or [str]
@@ -59,6 +60,7 @@ try:
         temperature: float = 0.0
         top_k: int = 0
         top_p: float = 0.0
+        eos_token_id: "int | None" = None
 except ImportError:        # pydantic optional: InferenceWorker still works
     GenRequest = None
     TextRequest = None
@@ -71,6 +73,7 @@ class _Request:
     temperature: float
     top_k: int = 0
     top_p: float = 0.0
+    eos_token_id: "int | None" = None
     future: "asyncio.Future | None" = None
     event: threading.Event = field(default_factory=threading.Event)
     result: Optional[List[int]] = None
@@ -79,7 +82,13 @@ class _Request:
     @property
     def key(self):
         return (int(self.tokens.numel()), round(float(self.temperature), 6),
-                self.top_k, round(float(self.top_p), 6))
+                self.top_k, round(float(self.top_p), 6), self.eos_token_id)
+
+    def trim_eos(self, toks):
+        """Cut at the first EOS (exclusive)."""
+        if self.eos_token_id is None or self.eos_token_id not in toks:
+            return toks
+        return toks[:toks.index(self.eos_token_id)]
 
 
 class InferenceWorker:
@@ -110,14 +119,15 @@ class InferenceWorker:
 
     def submit(self, tokens: List[int], max_new_tokens: int,
                temperature: float = 0.0, top_k: int = 0,
-               top_p: float = 0.0) -> _Request:
+               top_p: float = 0.0, eos_token_id=None) -> _Request:
         if not tokens:
             raise ValueError("empty prompt")
         n = min(int(max_new_tokens), self.max_new_tokens_cap)
         if n < 1:
             raise ValueError("max_new_tokens must be >= 1")
         req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
-                       float(temperature), int(top_k), float(top_p))
+                       float(temperature), int(top_k), float(top_p),
+                       eos_token_id)
         with self._cv:
             self._q.append(req)
             self._cv.notify()
@@ -158,17 +168,19 @@ class InferenceWorker:
         prompt = torch.stack([r.tokens for r in batch]).to(self.device)
         steps = max(r.max_new_tokens for r in batch)
         temp = batch[0].temperature
-        if self.use_graph and temp == 0.0:
+        eos = batch[0].eos_token_id
+        if self.use_graph and temp == 0.0 and eos is None:
             out = self.gen.generate_graphed(prompt, steps)
         else:
             out = self.gen.generate(prompt, steps, temperature=temp,
                                     top_k=batch[0].top_k,
-                                    top_p=batch[0].top_p)
+                                    top_p=batch[0].top_p,
+                                    eos_token_id=eos)
         s0 = prompt.shape[1]
         new = out[:, s0:].cpu()
         for i, r in enumerate(batch):
-            r.result = new[i, :r.max_new_tokens].tolist()
-            self.stats["tokens_generated"] += r.max_new_tokens
+            r.result = r.trim_eos(new[i, :r.max_new_tokens].tolist())
+            self.stats["tokens_generated"] += len(r.result)
         self.stats["batches"] += 1
         self.stats["requests"] += len(batch)
 
@@ -255,7 +267,8 @@ class ContinuousBatcher:
 
     # InferenceWorker-compatible surface ------------------------------
     def submit(self, tokens, max_new_tokens: int, temperature: float = 0.0,
-               top_k: int = 0, top_p: float = 0.0) -> _Request:
+               top_k: int = 0, top_p: float = 0.0,
+               eos_token_id=None) -> _Request:
         if not tokens:
             raise ValueError("empty prompt")
         n = min(int(max_new_tokens), self.max_new_tokens_cap)
@@ -266,7 +279,8 @@ class ContinuousBatcher:
                 f"prompt+max_new_tokens {len(tokens) + n} exceeds "
                 f"max_len {self.max_len}")
         req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
-                       float(temperature), int(top_k), float(top_p))
+                       float(temperature), int(top_k), float(top_p),
+                       eos_token_id)
         with self._cv:
             self._q.append(req)
             self._cv.notify()
@@ -307,9 +321,13 @@ class ContinuousBatcher:
 
     def _finish(self, slot: int):
         st = self._slots[slot]
-        # the one host sync of a request's lifetime
-        st["req"].result = self._toks[slot, :st["count"]].cpu().tolist()
-        self.stats["tokens_generated"] += st["count"]
+        # the one host sync of a request's lifetime; EOS is trimmed at
+        # harvest (the slot runs to max_new_tokens — early slot freeing
+        # would need a per-step device read, defeating the sync-free
+        # step)
+        st["req"].result = st["req"].trim_eos(
+            self._toks[slot, :st["count"]].cpu().tolist())
+        self.stats["tokens_generated"] += len(st["req"].result)
         self.stats["requests"] += 1
         st["req"].event.set()
         self._slots[slot] = None
@@ -492,7 +510,8 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
         t0 = time.time()
         try:
             reqs = [worker.submit(p, body.max_new_tokens, body.temperature,
-                                  body.top_k, body.top_p)
+                                  body.top_k, body.top_p,
+                                  eos_token_id=body.eos_token_id)
                     for p in prompts]
         except ValueError as e:
             raise HTTPException(400, str(e))
@@ -526,7 +545,8 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
             try:
                 reqs = [worker.submit(tokenizer.encode(p).ids,
                                       body.max_new_tokens, body.temperature,
-                                      body.top_k, body.top_p)
+                                      body.top_k, body.top_p,
+                                      eos_token_id=body.eos_token_id)
                         for p in prompts]
             except ValueError as e:
                 raise HTTPException(400, str(e))

@@ -144,3 +144,16 @@ def test_sampled_requests_complete():
         assert c.result == _ref(m, [6, 7, 8], 5)
     finally:
         cb.close()
+
+
+def test_eos_trim_at_harvest():
+    m = _model()
+    ref = _ref(m, [1, 2, 3], 10)
+    eos = ref[2]                      # third generated token
+    cb = ContinuousBatcher(m, max_slots=2)
+    try:
+        r = cb.submit([1, 2, 3], 10, eos_token_id=eos)
+        assert r.event.wait(60) and r.error is None
+        assert r.result == ref[:2]    # trimmed before the eos
+    finally:
+        cb.close()

@@ -141,3 +141,38 @@ def test_generate_with_sampling_runs():
                                 generator=g)
     assert out.shape == (2, 12)
     assert out.max() < 64 and out.min() >= 0
+
+
+def test_eos_early_stop_and_freeze():
+    import torch
+
+    from scaffold.config import load_config
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=128",
+        "model.vocab_size=128", "model.max_seq_len=64",
+    ])
+    torch.manual_seed(47)
+    m = build_model(cfg).eval()
+    g = Generator(m)
+    prompt = torch.tensor([[1, 2, 3], [4, 5, 6]])
+    ref = g.generate(prompt, 10)
+    # use each row's FIRST greedy continuation as EOS: both rows must
+    # freeze immediately and the tail repeat the eos id
+    eos = int(ref[0, 3])
+    out = g.generate(prompt, 10, eos_token_id=eos)
+    # wherever a row hits eos, everything after is eos
+    for b in range(2):
+        toks = out[b, 3:].tolist()
+        if eos in toks:
+            i = toks.index(eos)
+            assert all(t == eos for t in toks[i:]), toks
+    # rows identical to the unconstrained rollout up to their first eos
+    for b in range(2):
+        toks = out[b, 3:].tolist()
+        reft = ref[b, 3:].tolist()
+        n = toks.index(eos) if eos in toks else len(toks)
+        assert toks[:n] == reft[:n]

@@ -169,3 +169,17 @@ def test_text_completions_endpoint():
         assert r2.status_code == 200 and len(r2.json()["text"]) == 2
         assert client.post("/v1/completions", json={
             "prompt": "", "max_new_tokens": 4}).status_code == 400
+
+
+def test_worker_eos_trim():
+    m = _tiny_model()
+    ref = Generator(m).generate(
+        torch.tensor([[1, 2, 3, 4]], dtype=torch.int64), 8)[0, 4:].tolist()
+    eos = ref[3]
+    w = InferenceWorker(m, max_batch=4, window_s=0.0)
+    try:
+        r = w.submit([1, 2, 3, 4], 8, eos_token_id=eos)
+        assert r.event.wait(30) and r.error is None
+        assert r.result == ref[:3]
+    finally:
+        w.close()
