@@ -40,6 +40,16 @@ from .generate import Generator
 # batch assembly: requests arriving within this window join the batch
 DEFAULT_WINDOW_S = 0.002
 
+# bounded per-worker latency history for the stats endpoint
+LATENCY_HISTORY = 512
+
+
+def _percentile(sorted_vals, q):
+    if not sorted_vals:
+        return 0.0
+    i = min(len(sorted_vals) - 1, int(q * (len(sorted_vals) - 1) + 0.5))
+    return sorted_vals[i]
+
 try:
     from pydantic import BaseModel as _BaseModel
 
@@ -74,6 +84,7 @@ class _Request:
     top_k: int = 0
     top_p: float = 0.0
     eos_token_id: "int | None" = None
+    t_submit: float = 0.0
     future: "asyncio.Future | None" = None
     event: threading.Event = field(default_factory=threading.Event)
     result: Optional[List[int]] = None
@@ -113,6 +124,7 @@ class InferenceWorker:
         self._stop = False
         self.stats = {"requests": 0, "batches": 0, "tokens_generated": 0,
                       "started_at": time.time()}
+        self.latencies_ms: deque = deque(maxlen=LATENCY_HISTORY)
         self._thread = threading.Thread(target=self._loop, daemon=True,
                                         name="scaffold-serve-worker")
         self._thread.start()
@@ -127,7 +139,7 @@ class InferenceWorker:
             raise ValueError("max_new_tokens must be >= 1")
         req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
                        float(temperature), int(top_k), float(top_p),
-                       eos_token_id)
+                       eos_token_id, t_submit=time.time())
         with self._cv:
             self._q.append(req)
             self._cv.notify()
@@ -178,9 +190,11 @@ class InferenceWorker:
                                     eos_token_id=eos)
         s0 = prompt.shape[1]
         new = out[:, s0:].cpu()
+        now = time.time()
         for i, r in enumerate(batch):
             r.result = r.trim_eos(new[i, :r.max_new_tokens].tolist())
             self.stats["tokens_generated"] += len(r.result)
+            self.latencies_ms.append((now - r.t_submit) * 1e3)
         self.stats["batches"] += 1
         self.stats["requests"] += len(batch)
 
@@ -261,6 +275,7 @@ class ContinuousBatcher:
         self._sample_gen.manual_seed(seed)
         self.stats = {"requests": 0, "steps": 0, "tokens_generated": 0,
                       "started_at": time.time()}
+        self.latencies_ms: deque = deque(maxlen=LATENCY_HISTORY)
         self._thread = threading.Thread(target=self._loop, daemon=True,
                                         name="scaffold-cont-batcher")
         self._thread.start()
@@ -280,7 +295,7 @@ class ContinuousBatcher:
                 f"max_len {self.max_len}")
         req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
                        float(temperature), int(top_k), float(top_p),
-                       eos_token_id)
+                       eos_token_id, t_submit=time.time())
         with self._cv:
             self._q.append(req)
             self._cv.notify()
@@ -327,6 +342,7 @@ class ContinuousBatcher:
         # step)
         st["req"].result = st["req"].trim_eos(
             self._toks[slot, :st["count"]].cpu().tolist())
+        self.latencies_ms.append((time.time() - st["req"].t_submit) * 1e3)
         self.stats["tokens_generated"] += len(st["req"].result)
         self.stats["requests"] += 1
         st["req"].event.set()
@@ -498,6 +514,11 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
         if s.get("steps"):
             # continuous batching: average decode-slot occupancy
             s["mean_occupancy"] = s["tokens_generated"] / s["steps"]
+        lat = sorted(worker.latencies_ms)
+        if lat:
+            s["latency_ms"] = {"p50": _percentile(lat, 0.5),
+                               "p95": _percentile(lat, 0.95),
+                               "max": lat[-1]}
         return s
 
     @app.post("/v1/generate")

@@ -115,6 +115,8 @@ def test_http_surface():
         s = client.get("/v1/stats").json()
         assert s["requests"] == 3 and s["tokens_generated"] == 6 + 5 + 5
         assert s["mean_batch"] >= 1.0
+        assert s["latency_ms"]["p50"] > 0
+        assert s["latency_ms"]["p95"] >= s["latency_ms"]["p50"]
 
 
 @pytest.mark.gpu
