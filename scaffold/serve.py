@@ -141,6 +141,8 @@ class InferenceWorker:
                        float(temperature), int(top_k), float(top_p),
                        eos_token_id, t_submit=time.time())
         with self._cv:
+            if self._stop:
+                raise RuntimeError("worker is shut down")
             self._q.append(req)
             self._cv.notify()
         return req
@@ -150,6 +152,12 @@ class InferenceWorker:
             self._stop = True
             self._cv.notify()
         self._thread.join(timeout=5)
+        # no waiter may hang on a request the dead worker will never run
+        with self._cv:
+            while self._q:
+                r = self._q.popleft()
+                r.error = RuntimeError("worker shut down")
+                r.event.set()
 
     # -- worker side --------------------------------------------------
 
@@ -297,6 +305,8 @@ class ContinuousBatcher:
                        float(temperature), int(top_k), float(top_p),
                        eos_token_id, t_submit=time.time())
         with self._cv:
+            if self._stop:
+                raise RuntimeError("batcher is shut down")
             self._q.append(req)
             self._cv.notify()
         return req
@@ -448,6 +458,18 @@ class ContinuousBatcher:
                        and not self._stop):
                     self._cv.wait()
                 if self._stop:
+                    # fail everything still queued or mid-decode so no
+                    # waiter hangs on a worker that will never run again
+                    err = RuntimeError("batcher shut down")
+                    for r in self._q:
+                        r.error = err
+                        r.event.set()
+                    self._q.clear()
+                    for b, st in enumerate(self._slots):
+                        if st is not None:
+                            st["req"].error = err
+                            st["req"].event.set()
+                            self._slots[b] = None
                     return
                 pending = []
                 for b in range(self.max_slots):

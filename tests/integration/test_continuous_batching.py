@@ -157,3 +157,18 @@ def test_eos_trim_at_harvest():
         assert r.result == ref[:2]    # trimmed before the eos
     finally:
         cb.close()
+
+
+def test_close_fails_outstanding_requests():
+    """Shutdown mid-decode must fail waiters, not hang them."""
+    m = _model()
+    cb = ContinuousBatcher(m, max_slots=1)
+    r1 = cb.submit([1, 2, 3], 60)        # long-running occupant
+    r2 = cb.submit([4, 5, 6], 4)         # queued behind it
+    time.sleep(0.05)
+    cb.close()
+    assert r1.event.wait(10) and r2.event.wait(10)
+    assert (r1.result is not None) or (r1.error is not None)
+    assert r2.error is not None or r2.result is not None
+    with pytest.raises(RuntimeError, match="shut down"):
+        cb.submit([7], 2)
