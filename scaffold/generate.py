@@ -147,7 +147,10 @@ class Generator:
         top_k keeps the k best logits, top_p keeps the smallest nucleus
         with cumulative probability >= top_p (both standard serving
         truncations; composable)."""
-        if temperature <= 0:
+        # tiny temperatures overflow logits/T to inf (softmax -> NaN ->
+        # multinomial crash; found by the hypothesis edge-case sweep):
+        # below 1e-5 the distribution is one-hot anyway, so go greedy
+        if temperature <= 1e-5:
             return logits.argmax(dim=-1, keepdim=True)
         lg = logits.float() / temperature
         if top_k > 0 and top_k < lg.shape[-1]:
@@ -156,8 +159,12 @@ class Generator:
         if 0.0 < top_p < 1.0:
             srt, idx = lg.sort(dim=-1, descending=True)
             cum = torch.softmax(srt, dim=-1).cumsum(dim=-1)
-            # keep tokens whose PRECEDING mass is < top_p (always >= 1)
+            # keep tokens whose PRECEDING mass is < top_p; the top-1
+            # token is kept UNCONDITIONALLY (a sub-float32 top_p
+            # underflows the comparison to >= 0 and would drop every
+            # token -> NaN probs; found by the hypothesis sweep)
             drop_sorted = (cum - torch.softmax(srt, dim=-1)) >= top_p
+            drop_sorted[:, 0] = False
             drop = drop_sorted.scatter(1, idx, drop_sorted)
             lg = lg.masked_fill(drop, float("-inf"))
         probs = torch.softmax(lg, dim=-1)

@@ -89,3 +89,62 @@ def test_config_roundtrip(n_layers, bsz, steps, clip):
     assert cfg2.data.batch_size == bsz
     assert cfg2.steps == steps
     assert abs(cfg2.optim.grad_clip - clip) < 1e-12
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    b=st.integers(1, 4),
+    v=st.integers(4, 200),
+    top_k=st.integers(0, 220),
+    top_p=st.floats(0.0, 1.0),
+    temperature=st.floats(0.0, 3.0),
+    seed=st.integers(0, 2**16),
+)
+def test_sample_respects_truncations(b, v, top_k, top_p, temperature, seed):
+    """Generator._sample: ids always valid; temperature 0 is argmax;
+    with top_k the draw is inside the k best logits."""
+    from scaffold.generate import Generator
+
+    g = torch.Generator().manual_seed(seed)
+    logits = torch.randn(b, v, generator=g)
+    out = Generator._sample(logits, temperature, top_k, top_p,
+                            torch.Generator().manual_seed(seed + 1))
+    assert out.shape == (b, 1)
+    assert int(out.min()) >= 0 and int(out.max()) < v
+    if temperature <= 1e-5:
+        assert torch.equal(out[:, 0], logits.argmax(dim=-1))
+    elif 0 < top_k < v:
+        kth = logits.topk(top_k, dim=-1).values[:, -1]
+        chosen = logits.gather(1, out)[:, 0]
+        assert bool((chosen >= kth - 1e-6).all())
+
+
+@settings(max_examples=100, deadline=None)
+@given(
+    names_sizes=st.lists(
+        st.tuples(st.integers(0, 10**9), st.integers(1, 10**8)),
+        min_size=1, max_size=64, unique_by=lambda t: t[0]),
+    world=st.integers(1, 8),
+)
+def test_shard_partition_covers_once(names_sizes, world):
+    """ckpt layout: every tensor lands on exactly one shard."""
+    from scaffold.ckpt.layout import partition_by_size
+
+    named = [(f"t{n}", sz) for n, sz in names_sizes]
+    assign = partition_by_size(named, world)
+    assert sorted(assign.keys()) == sorted(n for n, _ in named)
+    assert all(0 <= r < world for r in assign.values())
+
+
+@settings(max_examples=100, deadline=None)
+@given(vals=st.lists(st.floats(0, 1e6), min_size=0, max_size=50),
+       q=st.floats(0.0, 1.0))
+def test_percentile_bounds(vals, q):
+    from scaffold.serve import _percentile
+
+    sv = sorted(vals)
+    p = _percentile(sv, q)
+    if not sv:
+        assert p == 0.0
+    else:
+        assert sv[0] <= p <= sv[-1]
