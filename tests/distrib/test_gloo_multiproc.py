@@ -55,16 +55,17 @@ def _worker(rank, world, file_store_path, result_q):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
-def test_ddp_matches_large_batch(tmp_path):
+@pytest.mark.timeout(240)
+@pytest.mark.parametrize("world", [2, 4])
+def test_ddp_matches_large_batch(tmp_path, world):
     from scaffold.trainer import Trainer
 
     ctxmp = mp.get_context("spawn")
     q = ctxmp.Queue()
     store_path = str(tmp_path / "filestore")
     procs = [
-        ctxmp.Process(target=_worker, args=(r, WORLD, store_path, q))
-        for r in range(WORLD)
+        ctxmp.Process(target=_worker, args=(r, world, store_path, q))
+        for r in range(world)
     ]
     for p in procs:
         p.start()
@@ -76,14 +77,14 @@ def test_ddp_matches_large_batch(tmp_path):
         assert p.exitcode == 0
 
     # single-process reference: same model seed, the whole global batch
-    cfg = _cfg(batch_size=WORLD * 8)
+    cfg = _cfg(batch_size=world * 8)
     t = Trainer(cfg)
     ref_params = torch.cat([p.detach().flatten() for p in t.raw_model.parameters()])
     assert torch.allclose(ref_params, dist_params), "init params differ (broadcast/seed)"
 
     g = torch.Generator().manual_seed(42)
-    X = torch.randn(WORLD * 8, 64, generator=g)
-    Y = torch.randn(WORLD * 8, 8, generator=g)
+    X = torch.randn(world * 8, 64, generator=g)
+    Y = torch.randn(world * 8, 8, generator=g)
     t.ddp.zero_grad()
     t.model(X, Y).backward()
     t.ddp.finish()
