@@ -29,7 +29,10 @@ def _named_state_tensors(model, optimizer) -> List[Tuple[str, torch.Tensor]]:
     for name, b in model.named_buffers():
         out.append((f"buffer.{name}", b.data))
     if optimizer is not None:
-        pnames = [n for n, _ in model.named_parameters()]
+        # ZeRO-1 subset optimizers carry their own name identities
+        pnames = (optimizer.param_names
+                  if getattr(optimizer, "param_names", None) is not None
+                  else [n for n, _ in model.named_parameters()])
         for i, t in enumerate(optimizer.master):
             out.append((f"opt.master.{pnames[i]}", t))
         for i, t in enumerate(optimizer.exp_avg):
@@ -90,6 +93,15 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
     rank = ctx.rank
 
     named = _named_state_tensors(model, optimizer)
+    # ZeRO-1: optimizer state is RANK-PINNED (each rank owns a shard of
+    # masters/moments by construction), so it bypasses the size-balanced
+    # partition; every rank can still build the full manifest because
+    # the ownership map is deterministic and masters mirror param shapes
+    zero1 = getattr(optimizer, "shard_owners", None) is not None
+    pinned = []
+    if zero1:
+        pinned = [(n, t) for n, t in named if n.startswith("opt.")]
+        named = [(n, t) for n, t in named if not n.startswith("opt.")]
     # dedupe aliased storages (fp32 params alias their optimizer master;
     # tied embeddings alias lm_head) — safetensors forbids shared tensors
     seen_ptr: Dict[int, str] = {}
@@ -117,6 +129,22 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
             shard[name] = t.detach()
     for name, target in aliases.items():
         tensor_meta[name] = {"alias": target}
+    if zero1:
+        shapes = {n: list(p.shape) for n, p in model.named_parameters()}
+        for pname, r in optimizer.shard_owners.items():
+            for kind in ("master", "exp_avg", "exp_avg_sq"):
+                tensor_meta[f"opt.{kind}.{pname}"] = {
+                    "shard": r, "shape": shapes[pname], "dtype": "float32"}
+        taken = {t.untyped_storage().data_ptr() for t in shard.values()}
+        for name, t in pinned:
+            t = t.detach()
+            # fp32 masters alias the working param (already in this
+            # shard when the save-partition assigned it here): clone so
+            # safetensors never sees shared storage
+            if t.untyped_storage().data_ptr() in taken:
+                t = t.clone()
+            shard[name] = t
+            taken.add(t.untyped_storage().data_ptr())
 
     # rank-local RNG + data state (every shard)
     shard[f"rng.torch.{rank}"] = torch.random.get_rng_state()
@@ -178,7 +206,10 @@ def _dest_tensors(model, optimizer, manifest) -> List[Tuple[str, torch.Tensor]]:
         if key in manifest["tensors"]:
             out.append((key, b.data))
     if optimizer is not None:
-        for i, name in enumerate(pnames):
+        onames = (optimizer.param_names
+                  if getattr(optimizer, "param_names", None) is not None
+                  else pnames)
+        for i, name in enumerate(onames):
             out.append((f"opt.master.{name}", optimizer.master[i]))
             out.append((f"opt.exp_avg.{name}", optimizer.exp_avg[i]))
             out.append((f"opt.exp_avg_sq.{name}", optimizer.exp_avg_sq[i]))
@@ -249,6 +280,7 @@ def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = 
             # Dedup by storage (fp32 params alias their master) — the
             # alias structure is identical on every rank, so the
             # broadcast sequences stay aligned.
+            zero1 = getattr(optimizer, "shard_owners", None) is not None
             seen_storage = set()
             for name, dst in _dest_tensors(model, optimizer, m):
                 rname = resolve(name)
@@ -257,6 +289,12 @@ def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = 
                 if ptr in seen_storage:
                     continue
                 seen_storage.add(ptr)
+                if zero1 and name.startswith("opt."):
+                    # rank-pinned state: same world => it sits in OUR
+                    # shard; other ranks don't even have a destination,
+                    # so no collective may be issued for it
+                    stager.copy_(dst, own.get_tensor(rname))
+                    continue
                 if owner == ctx.rank:
                     stager.copy_(dst, own.get_tensor(rname))
                 dist.broadcast(dst, src=owner)
@@ -288,7 +326,10 @@ def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = 
             if key in m["tensors"]:
                 b.data.copy_(fetch(key).to(b.device, b.dtype))
         if optimizer is not None:
-            for i, name in enumerate(pnames):
+            onames = (optimizer.param_names
+                      if getattr(optimizer, "param_names", None) is not None
+                      else pnames)
+            for i, name in enumerate(onames):
                 optimizer.master[i].copy_(fetch(f"opt.master.{name}").to(optimizer.master[i].device))
                 optimizer.exp_avg[i].copy_(fetch(f"opt.exp_avg.{name}").to(optimizer.exp_avg[i].device))
                 optimizer.exp_avg_sq[i].copy_(fetch(f"opt.exp_avg_sq.{name}").to(optimizer.exp_avg_sq[i].device))

@@ -24,7 +24,16 @@ class MasterAdamW:
         eps: float = 1e-8,
         weight_decay: float = 0.01,
         fused: bool = True,
+        param_names=None,
+        shard_owners=None,
     ):
+        # param_names: checkpoint identities aligned with `params` (set
+        # when the optimizer covers a SUBSET of the model — ZeRO-1);
+        # shard_owners: the full trainable-name -> owner-rank map (same
+        # deterministic partition on every rank), carried so the
+        # checkpoint layer can build a consistent manifest.
+        self.param_names = list(param_names) if param_names is not None else None
+        self.shard_owners = dict(shard_owners) if shard_owners is not None else None
         self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
         if not self.params:
             raise ValueError("no trainable parameters")
@@ -142,13 +151,35 @@ class MasterAdamW:
                 p.data.copy_(m.to(p.dtype))
 
 
-def build_optimizer(cfg, model: torch.nn.Module) -> MasterAdamW:
+def build_optimizer(cfg, model: torch.nn.Module, ctx=None) -> MasterAdamW:
     o = cfg.optim
+    zero1 = (getattr(cfg.dist, "zero1", False) and ctx is not None
+             and ctx.initialized and ctx.world_size > 1)
+    if not zero1:
+        return MasterAdamW(
+            model.parameters(),
+            lr=o.lr,
+            betas=tuple(o.betas),
+            eps=o.eps,
+            weight_decay=o.weight_decay,
+            fused=o.fused,
+        )
+    # ZeRO-1: deterministic size-balanced ownership over the trainable
+    # params (same map on every rank); this rank's optimizer holds
+    # masters/moments ONLY for its shard
+    from .ckpt import layout
+
+    trainable = [(n, p) for n, p in model.named_parameters() if p.requires_grad]
+    owners = layout.partition_by_size(
+        [(n, p.numel()) for n, p in trainable], ctx.world_size)
+    own = [(n, p) for n, p in trainable if owners[n] == ctx.rank]
     return MasterAdamW(
-        model.parameters(),
+        [p for _, p in own],
         lr=o.lr,
         betas=tuple(o.betas),
         eps=o.eps,
         weight_decay=o.weight_decay,
         fused=o.fused,
+        param_names=[n for n, _ in own],
+        shard_owners=owners,
     )

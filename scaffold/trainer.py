@@ -74,7 +74,16 @@ class Trainer:
             overlap=cfg.dist.overlap,
         )
         self.model = self.ddp
-        self.optimizer: MasterAdamW = build_optimizer(cfg, raw_model)
+        self.optimizer: MasterAdamW = build_optimizer(cfg, raw_model, self.ctx)
+        # ZeRO-1 (dist.zero1): after the sharded optimizer step, owners
+        # push their updated working params to the group (prebuilt
+        # (param, owner) list; owners map is deterministic per rank)
+        self._zero1_bcast = None
+        if self.optimizer.shard_owners is not None:
+            self._zero1_bcast = [
+                (p, self.optimizer.shard_owners[n])
+                for n, p in raw_model.named_parameters() if p.requires_grad
+            ]
         self.data = data if data is not None else build_data(
             cfg, self.ctx.rank, self.device, self.dtype
         )
@@ -178,7 +187,8 @@ class Trainer:
                 and self.cfg.optim.schedule == "const"
                 and self.cfg.optim.warmup_steps == 0
                 and self.cfg.optim.grad_clip == 0
-                and self.cfg.grad_accum <= 1)
+                and self.cfg.grad_accum <= 1
+                and self._zero1_bcast is None)  # collectives in step
 
     def _capture_step(self, batch_iter):
         """Capture ONE whole training step (synthetic-batch generation,
@@ -257,6 +267,13 @@ class Trainer:
             if gnorm > self.cfg.optim.grad_clip:
                 scale *= self.cfg.optim.grad_clip / (gnorm + 1e-12)
         self.optimizer.step(grad_scale=scale)
+        if self._zero1_bcast is not None:
+            import torch.distributed as dist
+
+            works = [dist.broadcast(p.data, src=owner, async_op=True)
+                     for p, owner in self._zero1_bcast]
+            for w in works:
+                w.wait()
         bd.mark("opt")
         bd.end_step()
         return loss_val
