@@ -82,3 +82,23 @@ def test_launch_order_is_index_order_despite_ragged_readiness():
     sizes = [r.numel for r in reduces]
     expected = [b.flat.numel() for b in ddp.buckets]
     assert sizes == expected, (sizes, expected)
+
+
+def test_latest_checkpoint_skips_incomplete_dirs(tmp_path):
+    """Atomic-commit contract: a step dir with shards but NO manifest
+    (a save that died mid-write) is never selected for resume."""
+    import os
+
+    from scaffold.ckpt import layout
+
+    base = tmp_path / "ck"
+    for step, complete in ((2, True), (4, True), (6, False)):
+        d = base / f"step_{step:08d}"
+        os.makedirs(d)
+        (d / layout.shard_file(0)).write_bytes(b"x")
+        if complete:
+            layout.write_manifest(str(d), step=step, world_size=1,
+                                  cfg_hash="h", tensors={})
+    assert layout.latest_checkpoint(str(base)).endswith("step_00000004")
+    # and an empty/alien dir yields no checkpoint at all
+    assert layout.latest_checkpoint(str(tmp_path / "nope")) == ""
