@@ -63,6 +63,7 @@ try:
         top_k: int = 0
         top_p: float = 0.0
         eos_token_id: "int | None" = None
+        stream: bool = False
 
     class TextRequest(_BaseModel):
         prompt: object             # str or [str]
@@ -85,6 +86,11 @@ class _Request:
     top_p: float = 0.0
     eos_token_id: "int | None" = None
     t_submit: float = 0.0
+    # streaming (continuous batcher only): sampled ids are pushed to
+    # this queue in chunks of `stream_every` tokens, then a None
+    # sentinel; the final `result` is still set at finish
+    stream: "object | None" = None
+    stream_every: int = 4
     future: "asyncio.Future | None" = None
     event: threading.Event = field(default_factory=threading.Event)
     result: Optional[List[int]] = None
@@ -291,7 +297,8 @@ class ContinuousBatcher:
     # InferenceWorker-compatible surface ------------------------------
     def submit(self, tokens, max_new_tokens: int, temperature: float = 0.0,
                top_k: int = 0, top_p: float = 0.0,
-               eos_token_id=None) -> _Request:
+               eos_token_id=None, stream=None,
+               stream_every: int = 4) -> _Request:
         if not tokens:
             raise ValueError("empty prompt")
         n = min(int(max_new_tokens), self.max_new_tokens_cap)
@@ -303,7 +310,8 @@ class ContinuousBatcher:
                 f"max_len {self.max_len}")
         req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
                        float(temperature), int(top_k), float(top_p),
-                       eos_token_id, t_submit=time.time())
+                       eos_token_id, t_submit=time.time(),
+                       stream=stream, stream_every=max(1, int(stream_every)))
         with self._cv:
             if self._stop:
                 raise RuntimeError("batcher is shut down")
@@ -391,6 +399,29 @@ class ContinuousBatcher:
         self._count.copy_(torch.tensor(host_cnt, dtype=torch.int64))
         self._active.copy_(torch.tensor(host_act, dtype=torch.int64))
 
+    def _flush_stream(self, b, st, final=False):
+        """Push newly generated ids to the request's stream queue.
+        Costs ONE device read per flush — streaming requests trade a
+        sync every `stream_every` tokens for liveness; non-streaming
+        slots keep the sync-free step."""
+        r = st["req"]
+        if r.stream is None or st.get("stream_done"):
+            return
+        seen = st.get("seen", 0)
+        due = final or (st["count"] - seen) >= r.stream_every
+        if not due or st["count"] <= seen:
+            return
+        chunk = self._toks[b, seen:st["count"]].cpu().tolist()
+        st["seen"] = st["count"]
+        if r.eos_token_id is not None and r.eos_token_id in chunk:
+            chunk = chunk[:chunk.index(r.eos_token_id)]
+            st["stream_done"] = True
+        for t in chunk:
+            r.stream.put(t)
+        if final or st.get("stream_done"):
+            r.stream.put(None)
+            st["stream_done"] = True
+
     def _advance_host(self):
         for b, st in enumerate(self._slots):
             if st is None:
@@ -398,7 +429,10 @@ class ContinuousBatcher:
             st["pos"] += 1
             st["count"] += 1
             if st["count"] >= st["req"].max_new_tokens:
+                self._flush_stream(b, st, final=True)
                 self._finish(b)
+            else:
+                self._flush_stream(b, st)
 
     @torch.no_grad()
     def _step(self):
@@ -468,6 +502,8 @@ class ContinuousBatcher:
                     for b, st in enumerate(self._slots):
                         if st is not None:
                             st["req"].error = err
+                            if st["req"].stream is not None:
+                                st["req"].stream.put(None)
                             st["req"].event.set()
                             self._slots[b] = None
                     return
@@ -482,6 +518,8 @@ class ContinuousBatcher:
                         self._admit(b, req)
                         if self._slots[b]["count"] >= \
                                 self._slots[b]["req"].max_new_tokens:
+                            self._flush_stream(b, self._slots[b],
+                                               final=True)
                             self._finish(b)
                     except Exception as e:   # bad request: surface it
                         req.error = e
@@ -492,6 +530,8 @@ class ContinuousBatcher:
                 for b, st in enumerate(self._slots):
                     if st is not None:
                         st["req"].error = e
+                        if st["req"].stream is not None:
+                            st["req"].stream.put(None)
                         st["req"].event.set()
                         self._slots[b] = None
 
@@ -550,6 +590,40 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
             prompts = [prompts]
         if not prompts:
             raise HTTPException(400, "tokens must be a non-empty id list")
+        if body.stream:
+            import json
+            import queue as _queue
+
+            from starlette.responses import StreamingResponse
+
+            if len(prompts) != 1:
+                raise HTTPException(400, "stream=true takes ONE prompt")
+            if not isinstance(worker, ContinuousBatcher):
+                raise HTTPException(400,
+                                    "streaming needs --continuous mode")
+            q = _queue.Queue()
+            try:
+                req = worker.submit(prompts[0], body.max_new_tokens,
+                                    body.temperature, body.top_k,
+                                    body.top_p,
+                                    eos_token_id=body.eos_token_id,
+                                    stream=q)
+            except ValueError as e:
+                raise HTTPException(400, str(e))
+
+            async def _events():
+                loop = asyncio.get_running_loop()
+                while True:
+                    tok = await loop.run_in_executor(None, q.get)
+                    if tok is None:
+                        break
+                    yield f"data: {json.dumps({'token': tok})}\n\n"
+                if req.error is not None:
+                    yield f"data: {json.dumps({'error': repr(req.error)})}\n\n"
+                yield "data: [DONE]\n\n"
+
+            return StreamingResponse(_events(),
+                                     media_type="text/event-stream")
         t0 = time.time()
         try:
             reqs = [worker.submit(p, body.max_new_tokens, body.temperature,

@@ -172,3 +172,74 @@ def test_close_fails_outstanding_requests():
     assert r2.error is not None or r2.result is not None
     with pytest.raises(RuntimeError, match="shut down"):
         cb.submit([7], 2)
+
+
+def test_streaming_tokens_match_result():
+    """Streamed chunks concatenate to exactly the final result, with a
+    None sentinel; EOS cuts the stream."""
+    import queue
+
+    m = _model()
+    ref = _ref(m, [1, 2, 3], 12)
+    cb = ContinuousBatcher(m, max_slots=2)
+    try:
+        q = queue.Queue()
+        r = cb.submit([1, 2, 3], 12, stream=q, stream_every=3)
+        got = []
+        while True:
+            t = q.get(timeout=60)
+            if t is None:
+                break
+            got.append(t)
+        assert r.event.wait(30) and r.error is None
+        assert got == r.result == ref
+
+        # eos mid-stream: stream stops before the eos id
+        eos = ref[5]
+        q2 = queue.Queue()
+        r2 = cb.submit([1, 2, 3], 12, eos_token_id=eos, stream=q2,
+                       stream_every=2)
+        got2 = []
+        while True:
+            t = q2.get(timeout=60)
+            if t is None:
+                break
+            got2.append(t)
+        assert r2.event.wait(30)
+        assert got2 == ref[:5]
+        assert r2.result == ref[:5]
+    finally:
+        cb.close()
+
+
+def test_http_sse_stream():
+    import json
+
+    from starlette.testclient import TestClient
+
+    from scaffold.serve import create_app
+
+    m = _model()
+    ref = _ref(m, [1, 2, 3], 8)
+    app = create_app(m, continuous=True, max_batch=2)
+    with TestClient(app) as client:
+        with client.stream("POST", "/v1/generate", json={
+                "tokens": [1, 2, 3], "max_new_tokens": 8,
+                "stream": True}) as r:
+            assert r.status_code == 200
+            assert "text/event-stream" in r.headers["content-type"]
+            toks = []
+            for line in r.iter_lines():
+                if not line.startswith("data: "):
+                    continue
+                payload = line[len("data: "):]
+                if payload == "[DONE]":
+                    break
+                toks.append(json.loads(payload)["token"])
+        assert toks == ref
+        # non-continuous worker rejects streaming
+    app2 = create_app(m, continuous=False)
+    with TestClient(app2) as client:
+        assert client.post("/v1/generate", json={
+            "tokens": [1, 2, 3], "max_new_tokens": 4,
+            "stream": True}).status_code == 400
