@@ -123,8 +123,10 @@ class Generator:
         return x + F.linear(g, blk.w_down.weight)
 
     def _forward_tokens(self, tokens: torch.Tensor, cache: KVCache,
-                        pos0: int) -> torch.Tensor:
-        """tokens (B, S) at positions [pos0, pos0+S) -> logits (B, V)."""
+                        pos0: int, all_logits: bool = False) -> torch.Tensor:
+        """tokens (B, S) at positions [pos0, pos0+S) -> logits (B, V),
+        or (B, S, V) for every position when all_logits (the speculative
+        verify pass needs the target's prediction AFTER each draft)."""
         m = self.m
         if self.is_llama:
             x = m.embed(tokens)
@@ -137,6 +139,8 @@ class Generator:
             for i, blk in enumerate(m.blocks):
                 x = self._gpt2_block(blk, x, cache, i, pos0)
             x = ops.layer_norm(x, m.ln_f.weight, m.ln_f.bias, m.ln_f.eps)
+        if all_logits:
+            return F.linear(x, m.lm_head.weight)
         return F.linear(x[:, -1], m.lm_head.weight)
 
     @staticmethod
@@ -300,6 +304,91 @@ class Generator:
         out_dev.scatter_(1, (pos_dev + 1).expand(nxt.shape), nxt)
         cur_tok.copy_(nxt)
         pos_dev.add_(1)
+
+    @torch.no_grad()
+    def generate_speculative(self, prompt: torch.Tensor,
+                             max_new_tokens: int, draft: "Generator",
+                             k: int = 4,
+                             stats: Optional[dict] = None) -> torch.Tensor:
+        """Greedy speculative decoding: `draft` proposes k tokens, ONE
+        target forward over the k-token window verifies them, and the
+        longest matching prefix is accepted plus the target's own next
+        token — output is EXACTLY Generator.generate's greedy rollout
+        regardless of draft quality (only the number of target forwards
+        changes).  B=1 (per-row acceptance lengths diverge in a batch).
+
+        On MI355X the target decode step is weight-read-bound (NOTES r2
+        serving), so accepting m tokens per verify cuts target weight
+        reads ~m-fold; the draft adds its own (much smaller) reads.
+        `stats["accepted"]/["rounds"]` expose the acceptance rate."""
+        assert prompt.shape[0] == 1, "speculative decode is B=1"
+        n_layers = len(self.m.blocks)
+        attn0 = self.m.blocks[0].attn
+        n_kv = attn0.n_kv if self.is_llama else attn0.n_heads
+        dtype = next(self.m.parameters()).dtype
+        S0 = prompt.shape[1]
+        L = S0 + max_new_tokens + k + 1   # headroom for a rolled-back tail
+        cache = KVCache(n_layers, 1, n_kv, L, attn0.head_dim,
+                        prompt.device, dtype)
+        d_attn = draft.m.blocks[0].attn
+        d_kv = d_attn.n_kv if draft.is_llama else d_attn.n_heads
+        d_dtype = next(draft.m.parameters()).dtype
+        dcache = KVCache(len(draft.m.blocks), 1, d_kv, L, d_attn.head_dim,
+                         prompt.device, d_dtype)
+
+        out = [int(t) for t in prompt[0]]
+        # prefill both; the target's prefill logits give token S0
+        logits = self._forward_tokens(prompt, cache, 0)
+        draft._forward_tokens(prompt, dcache, 0)
+        out.append(int(logits.argmax()))
+        produced = 1
+        rounds = accepted_total = 0
+        while produced < max_new_tokens:
+            kk = min(k, max_new_tokens - produced)
+            base = len(out) - 1          # index of the newest token
+            # --- draft proposes kk tokens from its own cache
+            dtoks = []
+            cur = torch.tensor([[out[base]]], device=prompt.device)
+            # draft's cache currently holds out[:dcache.len]; catch it up
+            # to the target (everything except the newest token)
+            if dcache.len < base:
+                catchup = torch.tensor([out[dcache.len:base]],
+                                       device=prompt.device)
+                draft._forward_tokens(catchup, dcache, dcache.len)
+            for _ in range(kk):
+                dlog = draft._forward_tokens(cur, dcache, dcache.len)
+                nxt = int(dlog.argmax())
+                dtoks.append(nxt)
+                cur = torch.tensor([[nxt]], device=prompt.device)
+            # --- one target forward over [newest, d1..d_{kk-1}]
+            window = torch.tensor([[out[base]] + dtoks[:-1]],
+                                  device=prompt.device)
+            tlog = self._forward_tokens(window, cache, cache.len,
+                                        all_logits=True)[0]
+            targets = tlog.argmax(dim=-1).tolist()   # target's choice AFTER each
+            m_acc = 0
+            while m_acc < kk and targets[m_acc] == dtoks[m_acc]:
+                m_acc += 1
+            # accept d1..d_m; if a mismatch stopped us, the target's own
+            # token at the mismatch position is the correction
+            new = dtoks[:m_acc]
+            if m_acc < kk:
+                new = new + [targets[m_acc]]
+            new = new[:max_new_tokens - produced]
+            out.extend(new)
+            produced += len(new)
+            rounds += 1
+            accepted_total += m_acc
+            # roll back the caches past what is actually accepted: the
+            # target consumed the whole window; keep entries only for
+            # tokens that are now part of `out` minus the newest
+            cache.len = len(out) - 1
+            dcache.len = min(dcache.len, len(out) - 1)
+        if stats is not None:
+            stats["rounds"] = rounds
+            stats["accepted"] = accepted_total
+        return torch.tensor([out[:S0 + max_new_tokens]],
+                            device=prompt.device)
 
     # ---- variable-position decode (continuous batching) ---------------
     def _rope_rows(self, x: torch.Tensor, pos: torch.Tensor) -> torch.Tensor:

@@ -176,3 +176,67 @@ def test_eos_early_stop_and_freeze():
         reft = ref[b, 3:].tolist()
         n = toks.index(eos) if eos in toks else len(toks)
         assert toks[:n] == reft[:n]
+
+
+def test_speculative_matches_greedy_exactly():
+    """Speculative decode == plain greedy regardless of draft quality:
+    same-family draft, cross-quality draft, and k sweep."""
+    import torch
+
+    from scaffold.config import load_config
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    def make(seed, layers=2, d=64):
+        cfg = load_config(None, [
+            "model.name=llama", f"model.n_layers={layers}",
+            "model.n_heads=4", "model.n_kv_heads=2", f"model.d_model={d}",
+            f"model.d_ff={2 * d}", "model.vocab_size=128",
+            "model.max_seq_len=96",
+        ])
+        torch.manual_seed(seed)
+        return build_model(cfg).eval()
+
+    target = Generator(make(51, layers=2, d=64))
+    prompt = torch.randint(0, 128, (1, 6))
+    ref = target.generate(prompt, 20)
+
+    # a GOOD draft (the target itself) and a BAD draft (random tiny)
+    for draft_m, tag in ((make(51, 2, 64), "self"), (make(99, 1, 32), "rand")):
+        for k in (1, 3, 5):
+            stats = {}
+            out = target.generate_speculative(prompt, 20,
+                                              Generator(draft_m), k=k,
+                                              stats=stats)
+            assert torch.equal(out, ref), (tag, k)
+            assert stats["rounds"] >= 1
+    # the self-draft must accept nearly everything
+    stats = {}
+    target.generate_speculative(prompt, 20, Generator(make(51, 2, 64)),
+                                k=4, stats=stats)
+    assert stats["accepted"] >= stats["rounds"] * 3, stats
+
+
+def test_speculative_gpt2_family():
+    import torch
+
+    from scaffold.config import load_config
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    cfg = load_config(None, [
+        "model.name=gpt2", "model.n_layers=2", "model.n_heads=4",
+        "model.d_model=64", "model.vocab_size=128", "model.max_seq_len=96",
+    ])
+    torch.manual_seed(53)
+    target = Generator(build_model(cfg).eval())
+    torch.manual_seed(77)
+    cfg2 = load_config(None, [
+        "model.name=gpt2", "model.n_layers=1", "model.n_heads=2",
+        "model.d_model=32", "model.vocab_size=128", "model.max_seq_len=96",
+    ])
+    draft = Generator(build_model(cfg2).eval())
+    prompt = torch.randint(0, 128, (1, 5))
+    ref = target.generate(prompt, 16)
+    out = target.generate_speculative(prompt, 16, draft, k=4)
+    assert torch.equal(out, ref)
