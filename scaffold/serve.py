@@ -243,7 +243,8 @@ class ContinuousBatcher:
 
     def __init__(self, model: torch.nn.Module, max_slots: int = 8,
                  max_len: int = 0, max_new_tokens_cap: int = 512,
-                 seed: int = 0, use_graph: bool = False):
+                 seed: int = 0, use_graph: bool = False,
+                 prefix_cache_entries: int = 8):
         from .generate import KVCache  # noqa: F401 (type reference)
 
         self.gen = Generator(model.eval())
@@ -282,6 +283,15 @@ class ContinuousBatcher:
                                    device=self.device)
         self.use_graph = use_graph
         self._graph = None
+        # exact-prefix KV cache: completed prefills keyed by their token
+        # tuple; a new request reuses the longest cached prefix of its
+        # prompt and prefills only the tail (system-prompt reuse).  LRU
+        # by entry count; entries hold per-layer (k, v) host... kept on
+        # DEVICE (serving has 288 GB of HBM; an 8-entry x 4k-token llama
+        # cache is ~2 GB)
+        self._prefix_entries = max(0, int(prefix_cache_entries))
+        self._prefix_cache = {}          # tuple(tokens) -> [(k, v) x layer]
+        self._prefix_lru = deque()
         self._q: deque[_Request] = deque()
         self._cv = threading.Condition()
         self._stop = False
@@ -333,16 +343,66 @@ class ContinuousBatcher:
                                   self.max_len, self.head_dim, self.device,
                                   self.dtype)
 
+    def _prefix_lookup(self, toks: tuple):
+        """Longest cached strict prefix of `toks` (or `toks` itself)."""
+        best = None
+        for key in self._prefix_cache:
+            if len(key) <= len(toks) and toks[:len(key)] == key:
+                if best is None or len(key) > len(best):
+                    best = key
+        return best
+
+    def _prefix_store(self, toks: tuple, kv):
+        if self._prefix_entries == 0 or toks in self._prefix_cache:
+            return
+        while len(self._prefix_lru) >= self._prefix_entries:
+            old = self._prefix_lru.popleft()
+            self._prefix_cache.pop(old, None)
+        self._prefix_cache[toks] = kv
+        self._prefix_lru.append(toks)
+
     @torch.no_grad()
     def _admit(self, slot: int, req: _Request):
-        """Prefill the request and install it in `slot`."""
+        """Prefill the request and install it in `slot`; a cached
+        prefix of the prompt skips its share of the prefill."""
         from .generate import KVCache
 
         prompt = req.tokens.view(1, -1).to(self.device)
         S0 = prompt.shape[1]
+        toks = tuple(int(t) for t in req.tokens)
+        hit = self._prefix_lookup(toks)
         tmp = KVCache(self.n_layers, 1, self.n_kv, S0, self.head_dim,
                       self.device, self.dtype)
-        logits = self.gen._forward_tokens(prompt, tmp, 0)
+        if hit is not None and len(hit) < S0:
+            # seed the cache with the reused prefix, prefill the tail
+            for i, (hk, hv) in enumerate(self._prefix_cache[hit]):
+                tmp.k[i][0, :, :len(hit)] = hk
+                tmp.v[i][0, :, :len(hit)] = hv
+            tmp.len = len(hit)
+            logits = self.gen._forward_tokens(prompt[:, len(hit):], tmp,
+                                              len(hit))
+            self.stats["prefix_hits"] = self.stats.get("prefix_hits", 0) + 1
+            self.stats["prefix_tokens_reused"] = (
+                self.stats.get("prefix_tokens_reused", 0) + len(hit))
+        elif hit is not None:
+            # full-prompt hit still needs last-token logits: reuse all
+            # but the final token, run one-step prefill for it
+            n = len(hit) - 1
+            for i, (hk, hv) in enumerate(self._prefix_cache[hit]):
+                tmp.k[i][0, :, :n] = hk[:, :n]
+                tmp.v[i][0, :, :n] = hv[:, :n]
+            tmp.len = n
+            logits = self.gen._forward_tokens(prompt[:, n:], tmp, n)
+            self.stats["prefix_hits"] = self.stats.get("prefix_hits", 0) + 1
+            self.stats["prefix_tokens_reused"] = (
+                self.stats.get("prefix_tokens_reused", 0) + n)
+        else:
+            logits = self.gen._forward_tokens(prompt, tmp, 0)
+        if self._prefix_entries:
+            self._prefix_store(
+                toks, [(tmp.k[i][0, :, :S0].clone(),
+                        tmp.v[i][0, :, :S0].clone())
+                       for i in range(self.n_layers)])
         for i in range(self.n_layers):
             self._cache.k[i][slot, :, :S0] = tmp.k[i][0]
             self._cache.v[i][slot, :, :S0] = tmp.v[i][0]

@@ -243,3 +243,50 @@ def test_http_sse_stream():
         assert client.post("/v1/generate", json={
             "tokens": [1, 2, 3], "max_new_tokens": 4,
             "stream": True}).status_code == 400
+
+
+def test_prefix_cache_exactness_and_hits():
+    """Prefix-cache admission: repeated prompts and extended prompts
+    reuse cached KV yet produce byte-identical results; LRU bounded."""
+    m = _model()
+    cb = ContinuousBatcher(m, max_slots=2, prefix_cache_entries=4)
+    try:
+        sys_prompt = [7, 8, 9, 10, 11, 12]
+        # cold: full prefill
+        a = cb.submit(sys_prompt, 6)
+        assert a.event.wait(60) and a.error is None
+        assert cb.stats.get("prefix_hits", 0) == 0
+        # same prompt again: full-prompt hit
+        b = cb.submit(sys_prompt, 6)
+        assert b.event.wait(60) and b.error is None
+        assert b.result == a.result == _ref(m, sys_prompt, 6)
+        assert cb.stats["prefix_hits"] == 1
+        # extended prompt: prefix hit, tail prefilled
+        ext = sys_prompt + [3, 4]
+        c = cb.submit(ext, 6)
+        assert c.event.wait(60) and c.error is None
+        assert c.result == _ref(m, ext, 6)
+        assert cb.stats["prefix_hits"] == 2
+        assert cb.stats["prefix_tokens_reused"] >= len(sys_prompt) * 2 - 1
+        # LRU stays bounded
+        for i in range(8):
+            r = cb.submit([20 + i, 21 + i, 22 + i], 2)
+            assert r.event.wait(60)
+        assert len(cb._prefix_cache) <= 4
+    finally:
+        cb.close()
+
+
+def test_prefix_cache_disabled():
+    m = _model()
+    cb = ContinuousBatcher(m, max_slots=2, prefix_cache_entries=0)
+    try:
+        p = [5, 6, 7]
+        a = cb.submit(p, 4)
+        b = cb.submit(p, 4)
+        assert a.event.wait(60) and b.event.wait(60)
+        assert a.result == b.result == _ref(m, p, 4)
+        assert cb.stats.get("prefix_hits", 0) == 0
+        assert len(cb._prefix_cache) == 0
+    finally:
+        cb.close()
