@@ -308,7 +308,7 @@ class Generator:
     @torch.no_grad()
     def generate_speculative(self, prompt: torch.Tensor,
                              max_new_tokens: int, draft: "Generator",
-                             k: int = 4,
+                             k: int = 4, eos_token_id: Optional[int] = None,
                              stats: Optional[dict] = None) -> torch.Tensor:
         """Greedy speculative decoding: `draft` proposes k tokens, ONE
         target forward over the k-token window verifies them, and the
@@ -343,7 +343,8 @@ class Generator:
         out.append(int(logits.argmax()))
         produced = 1
         rounds = accepted_total = 0
-        while produced < max_new_tokens:
+        done = eos_token_id is not None and out[-1] == eos_token_id
+        while produced < max_new_tokens and not done:
             kk = min(k, max_new_tokens - produced)
             base = len(out) - 1          # index of the newest token
             # --- draft proposes kk tokens from its own cache
@@ -375,6 +376,9 @@ class Generator:
             if m_acc < kk:
                 new = new + [targets[m_acc]]
             new = new[:max_new_tokens - produced]
+            if eos_token_id is not None and eos_token_id in new:
+                new = new[:new.index(eos_token_id) + 1]
+                done = True
             out.extend(new)
             produced += len(new)
             rounds += 1
@@ -387,6 +391,9 @@ class Generator:
         if stats is not None:
             stats["rounds"] = rounds
             stats["accepted"] = accepted_total
+        if eos_token_id is not None and produced < max_new_tokens:
+            # match generate(): frozen rows repeat eos to full length
+            out.extend([eos_token_id] * (max_new_tokens - produced))
         return torch.tensor([out[:S0 + max_new_tokens]],
                             device=prompt.device)
 

@@ -240,3 +240,33 @@ def test_speculative_gpt2_family():
     ref = target.generate(prompt, 16)
     out = target.generate_speculative(prompt, 16, draft, k=4)
     assert torch.equal(out, ref)
+
+
+def test_speculative_eos_matches_generate():
+    import torch
+
+    from scaffold.config import load_config
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=128",
+        "model.vocab_size=128", "model.max_seq_len=96",
+    ])
+    torch.manual_seed(51)
+    target = Generator(build_model(cfg).eval())
+    prompt = torch.randint(0, 128, (1, 6))
+    ref_free = target.generate(prompt, 20)
+    eos = int(ref_free[0, 6 + 4])         # 5th generated token as eos
+    ref = target.generate(prompt, 20, eos_token_id=eos)
+    torch.manual_seed(99)
+    cfg2 = load_config(None, [
+        "model.name=llama", "model.n_layers=1", "model.n_heads=2",
+        "model.n_kv_heads=1", "model.d_model=32", "model.d_ff=64",
+        "model.vocab_size=128", "model.max_seq_len=96",
+    ])
+    draft = Generator(build_model(cfg2).eval())
+    out = target.generate_speculative(prompt, 20, draft, k=3,
+                                      eos_token_id=eos)
+    assert torch.equal(out, ref)

@@ -98,3 +98,23 @@ def test_rejects_missing_or_short_file(tmp_path):
     np.zeros(8, dtype=np.uint16).tofile(p)
     with pytest.raises(ValueError, match="need >"):
         build_data(_cfg(str(p)), 0, torch.device("cpu"), torch.float32)
+
+
+def test_uint32_corpus_for_large_vocab(tmp_path):
+    """vocab > 65536 switches the memmap dtype to uint32."""
+    rng = np.random.default_rng(1)
+    arr = rng.integers(0, 100_000, size=2048, dtype=np.uint32)
+    p = tmp_path / "big.bin"
+    arr.tofile(p)
+    cfg = load_config(None, [
+        "model.name=gpt2", "model.n_layers=1", "model.n_heads=2",
+        "model.d_model=32", "model.vocab_size=128256",
+        "model.max_seq_len=16",
+        "data.kind=token_file", f"data.path={p}",
+        "data.batch_size=2", "data.seq_len=16", "device=cpu", "dtype=fp32",
+    ])
+    d = build_data(cfg, 0, torch.device("cpu"), torch.float32)
+    x, y = d.next_batch()
+    assert x.dtype == torch.int64
+    assert int(x.max()) < 100_000
+    assert torch.equal(x[:, 1:], y[:, :-1])
