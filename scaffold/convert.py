@@ -210,3 +210,44 @@ def to_hf_llama(model: Llama) -> Dict[str, torch.Tensor]:
         out[h + "mlp.up_proj.weight"] = gu[f2:].contiguous()
         out[h + "mlp.down_proj.weight"] = sd[b + "w_down.weight"]
     return out
+
+
+def save_model(model: torch.nn.Module, path: str) -> None:
+    """Single-file safetensors weight export (serving deployments —
+    trainer state stays with the sharded checkpoint engine).  Tied
+    tensors (GPT-2 wte/lm_head) are stored once and re-tied on load."""
+    import safetensors.torch as st
+
+    sd = {}
+    seen = {}
+    tied = {}
+    for k, v in model.state_dict().items():
+        ptr = v.untyped_storage().data_ptr()
+        if ptr in seen:
+            tied[k] = seen[ptr]
+        else:
+            seen[ptr] = k
+            sd[k] = v.detach().cpu().contiguous()
+    st.save_file(sd, path, metadata={"tied": repr(tied)} if tied else {})
+
+
+def load_model(model: torch.nn.Module, path: str) -> None:
+    """Inverse of save_model; dtype/device follow the destination."""
+    import ast
+
+    import safetensors.torch as st
+    from safetensors import safe_open
+
+    with safe_open(path, framework="pt", device="cpu") as f:
+        meta = f.metadata() or {}
+    tied = ast.literal_eval(meta.get("tied", "{}"))
+    sd = st.load_file(path)
+    for k, src in tied.items():
+        sd[k] = sd[src]
+    own = model.state_dict()
+    missing = [k for k in own if k not in sd]
+    if missing:
+        raise KeyError(f"load_model: missing {missing}")
+    with torch.no_grad():
+        for k, t in own.items():
+            t.copy_(sd[k].to(t.device, t.dtype))

@@ -194,3 +194,24 @@ def test_roundtrip_export_gpt2():
         ref = _ref_loss(hf(tok).logits, tgt)
     assert torch.allclose(_our_loss(ours, tok, tgt), ref,
                           rtol=1e-4, atol=1e-5)
+
+
+def test_save_load_model_single_file(tmp_path):
+    """Single-file weight export round-trips, including tied tensors."""
+    from scaffold.convert import load_model, save_model
+
+    cfg = load_config(None, [
+        "model.name=gpt2", "model.n_layers=2", "model.n_heads=4",
+        "model.d_model=64", "model.vocab_size=128", "model.max_seq_len=32",
+    ])
+    torch.manual_seed(61)
+    a = build_model(cfg).eval()
+    p = str(tmp_path / "m.safetensors")
+    save_model(a, p)
+    torch.manual_seed(62)
+    b = build_model(cfg).eval()
+    load_model(b, p)
+    for (n, x), (_, y) in zip(a.state_dict().items(), b.state_dict().items()):
+        assert torch.equal(x, y), n
+    # tying preserved: loading into b kept lm_head sharing wte storage
+    assert b.lm_head.weight.data_ptr() == b.wte.weight.data_ptr()
