@@ -47,6 +47,10 @@ class ModelConfig:
     # N local experts (0 = dense).  Expert-parallel sharding of the same
     # layer across ranks is scaffold.dist.ep.ExpertParallelMoE
     n_experts: int = 0
+    # LoRA fine-tuning: rank > 0 freezes the model and attaches
+    # low-rank adapters to every Linear except lm_head (scaffold/lora.py)
+    lora_r: int = 0
+    lora_alpha: float = 16.0
     # recompute each transformer block in backward instead of storing
     # its activations (trades ~1 extra block fwd for O(1) activation
     # memory per block); LM families only, no-op in eval

@@ -156,13 +156,19 @@ def build_optimizer(cfg, model: torch.nn.Module, ctx=None) -> MasterAdamW:
     zero1 = (getattr(cfg.dist, "zero1", False) and ctx is not None
              and ctx.initialized and ctx.world_size > 1)
     if not zero1:
+        # explicit trainable identities: with frozen params (LoRA) the
+        # optimizer's positional lists no longer align with
+        # model.named_parameters(), so the checkpoint layer needs names
+        trainable = [(n, p) for n, p in model.named_parameters()
+                     if p.requires_grad]
         return MasterAdamW(
-            model.parameters(),
+            [p for _, p in trainable],
             lr=o.lr,
             betas=tuple(o.betas),
             eps=o.eps,
             weight_decay=o.weight_decay,
             fused=o.fused,
+            param_names=[n for n, _ in trainable],
         )
     # ZeRO-1: deterministic size-balanced ownership over the trainable
     # params (same map on every rank); this rank's optimizer holds

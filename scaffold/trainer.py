@@ -62,6 +62,11 @@ class Trainer:
         if self.channels_last:
             # NHWC is MIOpen's fast conv layout on CDNA
             raw_model = raw_model.to(memory_format=torch.channels_last)
+        if getattr(cfg.model, "lora_r", 0) > 0:
+            from .lora import apply_lora
+
+            apply_lora(raw_model, r=cfg.model.lora_r,
+                       alpha=cfg.model.lora_alpha)
         self.raw_model = raw_model
 
         from .dist.comm import make_comm
