@@ -107,6 +107,10 @@ class CkptConfig:
     dir: str = ""
     interval: int = 0  # steps between checkpoints; 0 = only final
     resume: str = ""  # path to a step_NNNNNN dir to resume from
+    # initial weights for fine-tuning (NOT a resume: optimizer/RNG/data
+    # start fresh): a .safetensors file (scaffold.convert.save_model
+    # layout), an int8 quantized export, or an HF-layout directory
+    init_weights: str = ""
     save_final: bool = False
     async_d2h: bool = True  # pinned hipMemcpyAsync staging on a side stream (C6)
 

@@ -62,6 +62,10 @@ class Trainer:
         if self.channels_last:
             # NHWC is MIOpen's fast conv layout on CDNA
             raw_model = raw_model.to(memory_format=torch.channels_last)
+        if cfg.ckpt.init_weights:
+            # fine-tune initialization (BEFORE LoRA freeze and before
+            # the DDP wrap-time broadcast, which then distributes it)
+            self._load_init_weights(raw_model, cfg.ckpt.init_weights)
         if getattr(cfg.model, "lora_r", 0) > 0:
             from .lora import apply_lora
 
@@ -98,6 +102,28 @@ class Trainer:
         self.cfg_hash = config_hash(cfg)
         if cfg.ckpt.resume:
             self._resume(cfg.ckpt.resume)
+
+    @staticmethod
+    def _load_init_weights(model: torch.nn.Module, path: str) -> None:
+        """Dispatch on the artifact kind: HF directory, int8 quantized
+        export, or plain safetensors weights."""
+        import os as _os
+
+        from . import convert
+
+        if _os.path.isdir(path):
+            convert.load_hf_dir(model, path)
+            return
+        from safetensors import safe_open
+
+        with safe_open(path, framework="pt", device="cpu") as f:
+            quantized = any(k.endswith("::int8") for k in f.keys())
+        if quantized:
+            from .quant import load_quantized
+
+            load_quantized(model, path)
+        else:
+            convert.load_model(model, path)
 
     def _to_mixed(self, model: torch.nn.Module) -> torch.nn.Module:
         """bf16 working weights; BatchNorm stays fp32 (numerics)."""
