@@ -752,13 +752,37 @@ def main(argv=None):
     ap.add_argument("--continuous", action="store_true",
                     help="continuous batching (per-slot admission) "
                          "instead of same-shape batch grouping")
+    ap.add_argument("--weights", default="",
+                    help="initial weights: safetensors file, int8 "
+                         "quantized export, or HF-layout directory "
+                         "(default: random init)")
+    ap.add_argument("--lora", default="",
+                    help="LoRA adapter state to merge into the weights "
+                         "(requires --lora-r to rebuild the adapters)")
+    ap.add_argument("--lora-r", type=int, default=0)
+    ap.add_argument("--lora-alpha", type=float, default=16.0)
     args = ap.parse_args(argv)
 
     import uvicorn
     cfg = load_config(None, args.overrides)
     dev = "cuda" if torch.cuda.is_available() else "cpu"
     dt = torch.bfloat16 if dev == "cuda" else torch.float32
-    model = build_model(cfg).to(dev, dt).eval()
+    model = build_model(cfg)
+    if args.weights:
+        from .trainer import Trainer
+
+        Trainer._load_init_weights(model, args.weights)
+    if args.lora:
+        import safetensors.torch as _st
+
+        from .lora import apply_lora, load_lora_state_dict, merge_lora
+
+        if args.lora_r <= 0:
+            ap.error("--lora needs --lora-r (the adapters' rank)")
+        apply_lora(model, r=args.lora_r, alpha=args.lora_alpha)
+        load_lora_state_dict(model, _st.load_file(args.lora))
+        merge_lora(model)          # serve the baked weights
+    model = model.to(dev, dt).eval()
     tok = None
     if args.tokenizer:
         from tokenizers import Tokenizer

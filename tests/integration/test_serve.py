@@ -185,3 +185,58 @@ def test_worker_eos_trim():
         assert r.result == ref[:3]
     finally:
         w.close()
+
+
+def test_cli_weights_and_lora_flags(tmp_path):
+    """serve CLI wiring: --weights + --lora produce a served model whose
+    outputs equal the merged fine-tune (probed via create_app worker)."""
+    import safetensors.torch as st
+
+    from scaffold.convert import save_model
+    from scaffold.lora import apply_lora, lora_state_dict, merge_lora
+    from scaffold.models import build_model
+    from scaffold.config import load_config
+
+    over = [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=128",
+        "model.vocab_size=128", "model.max_seq_len=64",
+    ]
+    torch.manual_seed(81)
+    base = build_model(load_config(None, over)).eval()
+    wp = str(tmp_path / "w.safetensors")
+    save_model(base, wp)
+    apply_lora(base, r=4, alpha=8)
+    with torch.no_grad():
+        for p in base.parameters():
+            if p.requires_grad:
+                p.add_(torch.randn_like(p) * 0.03)
+    ap = str(tmp_path / "a.safetensors")
+    st.save_file(lora_state_dict(base), ap)
+    merge_lora(base)
+
+    # rebuild through the CLI path (main() minus uvicorn.run)
+    import scaffold.serve as serve_mod
+
+    captured = {}
+
+    def fake_run(app, **kw):
+        captured["app"] = app
+
+    import uvicorn as _uv
+    real_run = _uv.run
+    _uv.run = fake_run
+    try:
+        serve_mod.main(over + ["--weights", wp, "--lora", ap,
+                               "--lora-r", "4", "--lora-alpha", "8"])
+    finally:
+        _uv.run = real_run
+    worker = captured["app"].state.worker
+    try:
+        ref = Generator(base).generate(
+            torch.tensor([[1, 2, 3, 4]], dtype=torch.int64), 8)[0, 4:]
+        r = worker.submit([1, 2, 3, 4], 8)
+        assert r.event.wait(60) and r.error is None
+        assert r.result == ref.tolist()
+    finally:
+        worker.close()
