@@ -120,6 +120,8 @@ class InferenceWorker:
                  window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False,
                  max_new_tokens_cap: int = 512):
         self.gen = Generator(model)
+        self.max_len = (model.rope_cos.shape[0] if self.gen.is_llama
+                        else model.wpe.num_embeddings)
         self.device = next(model.parameters()).device
         self.max_batch = max_batch
         self.window_s = window_s
@@ -143,6 +145,10 @@ class InferenceWorker:
         n = min(int(max_new_tokens), self.max_new_tokens_cap)
         if n < 1:
             raise ValueError("max_new_tokens must be >= 1")
+        if len(tokens) + n > self.max_len:
+            raise ValueError(
+                f"prompt+max_new_tokens {len(tokens) + n} exceeds "
+                f"max_len {self.max_len}")
         req = _Request(torch.tensor(tokens, dtype=torch.int64), n,
                        float(temperature), int(top_k), float(top_p),
                        eos_token_id, t_submit=time.time())
@@ -642,6 +648,37 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
                                "p95": _percentile(lat, 0.95),
                                "max": lat[-1]}
         return s
+
+    @app.get("/metrics")
+    def metrics():
+        from starlette.responses import PlainTextResponse
+
+        st_ = dict(worker.stats)
+        up = time.time() - st_.pop("started_at")
+        lines = [
+            "# TYPE scaffold_requests_total counter",
+            f"scaffold_requests_total {st_.get('requests', 0)}",
+            "# TYPE scaffold_tokens_generated_total counter",
+            f"scaffold_tokens_generated_total {st_.get('tokens_generated', 0)}",
+            "# TYPE scaffold_uptime_seconds gauge",
+            f"scaffold_uptime_seconds {up:.3f}",
+        ]
+        for k in ("batches", "steps", "prefix_hits",
+                  "prefix_tokens_reused"):
+            if k in st_:
+                lines += [f"# TYPE scaffold_{k}_total counter",
+                          f"scaffold_{k}_total {st_[k]}"]
+        lat = sorted(worker.latencies_ms)
+        if lat:
+            lines += [
+                "# TYPE scaffold_request_latency_ms summary",
+                f'scaffold_request_latency_ms{{quantile="0.5"}} '
+                f"{_percentile(lat, 0.5):.3f}",
+                f'scaffold_request_latency_ms{{quantile="0.95"}} '
+                f"{_percentile(lat, 0.95):.3f}",
+            ]
+        return PlainTextResponse("\n".join(lines) + "\n",
+                                 media_type="text/plain; version=0.0.4")
 
     @app.post("/v1/generate")
     async def generate(body: GenRequest):

@@ -78,7 +78,12 @@ def test_worker_error_surfaces_not_fatal():
     m = _tiny_model()
     w = InferenceWorker(m, max_batch=4, window_s=0.0)
     try:
-        bad = w.submit([1, 2], 200)   # 2+200 > max_seq_len=64 -> op error
+        # context overflow is rejected synchronously at submit
+        with pytest.raises(ValueError, match="exceeds"):
+            w.submit([1, 2], 200)
+        # a RUNTIME failure (out-of-vocab id -> embedding index error)
+        # surfaces on the request without killing the worker loop
+        bad = w.submit([500, 501], 4)
         assert bad.event.wait(30)
         assert bad.error is not None
         ok = w.submit([1, 2, 3], 4)   # loop survived
@@ -240,3 +245,28 @@ def test_cli_weights_and_lora_flags(tmp_path):
         assert r.result == ref.tolist()
     finally:
         worker.close()
+
+
+def test_worker_context_length_validation():
+    m = _tiny_model()           # max_seq_len=64
+    w = InferenceWorker(m, window_s=0.0)
+    try:
+        with pytest.raises(ValueError, match="exceeds"):
+            w.submit([1] * 60, 10)
+    finally:
+        w.close()
+
+
+def test_prometheus_metrics_endpoint():
+    from starlette.testclient import TestClient
+
+    app = create_app(_tiny_model(), window_s=0.0)
+    with TestClient(app) as client:
+        client.post("/v1/generate", json={"tokens": [1, 2, 3],
+                                          "max_new_tokens": 4})
+        r = client.get("/metrics")
+        assert r.status_code == 200
+        body = r.text
+        assert "scaffold_requests_total 1" in body
+        assert "scaffold_tokens_generated_total 4" in body
+        assert 'scaffold_request_latency_ms{quantile="0.5"}' in body
