@@ -44,6 +44,10 @@ DEFAULT_WINDOW_S = 0.002
 LATENCY_HISTORY = 512
 
 
+class OverloadedError(RuntimeError):
+    """Queue full — the HTTP layer maps this to 503."""
+
+
 def _percentile(sorted_vals, q):
     if not sorted_vals:
         return 0.0
@@ -118,7 +122,8 @@ class InferenceWorker:
 
     def __init__(self, model: torch.nn.Module, max_batch: int = 8,
                  window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False,
-                 max_new_tokens_cap: int = 512):
+                 max_new_tokens_cap: int = 512, max_queue: int = 256):
+        self.max_queue = max_queue
         self.gen = Generator(model)
         self.max_len = (model.rope_cos.shape[0] if self.gen.is_llama
                         else model.wpe.num_embeddings)
@@ -155,6 +160,9 @@ class InferenceWorker:
         with self._cv:
             if self._stop:
                 raise RuntimeError("worker is shut down")
+            if len(self._q) >= self.max_queue:
+                raise OverloadedError(
+                    f"queue full ({self.max_queue} pending)")
             self._q.append(req)
             self._cv.notify()
         return req
@@ -250,7 +258,8 @@ class ContinuousBatcher:
     def __init__(self, model: torch.nn.Module, max_slots: int = 8,
                  max_len: int = 0, max_new_tokens_cap: int = 512,
                  seed: int = 0, use_graph: bool = False,
-                 prefix_cache_entries: int = 8):
+                 prefix_cache_entries: int = 8, max_queue: int = 256):
+        self.max_queue = max_queue
         from .generate import KVCache  # noqa: F401 (type reference)
 
         self.gen = Generator(model.eval())
@@ -331,6 +340,9 @@ class ContinuousBatcher:
         with self._cv:
             if self._stop:
                 raise RuntimeError("batcher is shut down")
+            if len(self._q) >= self.max_queue:
+                raise OverloadedError(
+                    f"queue full ({self.max_queue} pending)")
             self._q.append(req)
             self._cv.notify()
         return req
@@ -727,6 +739,8 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
                                   body.top_k, body.top_p,
                                   eos_token_id=body.eos_token_id)
                     for p in prompts]
+        except OverloadedError as e:
+            raise HTTPException(503, str(e))
         except ValueError as e:
             raise HTTPException(400, str(e))
         loop = asyncio.get_running_loop()
@@ -762,6 +776,8 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
                                       body.top_k, body.top_p,
                                       eos_token_id=body.eos_token_id)
                         for p in prompts]
+            except OverloadedError as e:
+                raise HTTPException(503, str(e))
             except ValueError as e:
                 raise HTTPException(400, str(e))
             ms = await _run(reqs, t0)

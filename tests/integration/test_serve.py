@@ -270,3 +270,17 @@ def test_prometheus_metrics_endpoint():
         assert "scaffold_requests_total 1" in body
         assert "scaffold_tokens_generated_total 4" in body
         assert 'scaffold_request_latency_ms{quantile="0.5"}' in body
+
+
+def test_queue_bound_rejects_with_overload():
+    from scaffold.serve import OverloadedError
+
+    m = _tiny_model()
+    w = InferenceWorker(m, window_s=5.0, max_queue=2)  # long window: queue holds
+    try:
+        w.submit([1, 2], 2)
+        w.submit([3, 4], 2)
+        with pytest.raises(OverloadedError):
+            w.submit([5, 6], 2)
+    finally:
+        w.close()
