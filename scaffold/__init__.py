@@ -23,5 +23,36 @@ from .config import TrainConfig, load_config, config_hash
 from .runtime import init_distributed, DistContext
 from .trainer import Trainer
 
-__version__ = "0.1.0"
-__all__ = ["TrainConfig", "load_config", "config_hash", "init_distributed", "DistContext", "Trainer"]
+__version__ = "0.2.0"
+
+# lazy attributes (PEP 562): the serving / interop / fine-tune surface
+# pulls optional deps (fastapi, tokenizers) or large modules only on use
+_LAZY = {
+    "Generator": ("scaffold.generate", "Generator"),
+    "KVCache": ("scaffold.generate", "KVCache"),
+    "InferenceWorker": ("scaffold.serve", "InferenceWorker"),
+    "ContinuousBatcher": ("scaffold.serve", "ContinuousBatcher"),
+    "create_app": ("scaffold.serve", "create_app"),
+    "load_hf": ("scaffold.convert", "load_hf"),
+    "load_hf_dir": ("scaffold.convert", "load_hf_dir"),
+    "save_model": ("scaffold.convert", "save_model"),
+    "load_model": ("scaffold.convert", "load_model"),
+    "apply_lora": ("scaffold.lora", "apply_lora"),
+    "merge_lora": ("scaffold.lora", "merge_lora"),
+    "save_quantized": ("scaffold.quant", "save_quantized"),
+    "load_quantized": ("scaffold.quant", "load_quantized"),
+    "run_elastic": ("scaffold.launch", "run_elastic"),
+}
+
+
+def __getattr__(name):
+    if name in _LAZY:
+        import importlib
+
+        mod, attr = _LAZY[name]
+        return getattr(importlib.import_module(mod), attr)
+    raise AttributeError(f"module 'scaffold' has no attribute {name!r}")
+
+
+__all__ = ["TrainConfig", "load_config", "config_hash", "init_distributed",
+           "DistContext", "Trainer", *sorted(_LAZY)]
