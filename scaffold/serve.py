@@ -1,27 +1,28 @@
-"""HTTP serving front end for the scaffold's LM families.
+"""HTTP serving for the scaffold's LM families (guide: docs/SERVING.md).
 
-A single `InferenceWorker` thread owns the model and drains a request
-queue in dynamic batches: pending requests with the same prompt length
-and temperature are stacked into one `Generator.generate` call (the
-KV-cache path, decode GEMVs batched across requests), then each
-response is sliced back out.  Grouping by prompt length keeps the
-batch un-padded — the Generator's cache layout has no padding mask, so
-mixing lengths would silently change positions (SURVEY §5.7 seam note
-applies to serving too: the op boundary stays (B, S) dense).
+Two engines behind one submit() surface:
 
-MI355X notes: serving decode is weight-read-bound (NOTES.md r2 serving
-ladder), so batching B requests into one decode sweep multiplies
-tokens/s by ~B until the batch GEMMs leave the skinny-M regime.  For
-fixed-shape deployments `use_graph=True` replays the hipGraph-captured
-decode step (2.5x on gpt2-small at B=1).
+- `InferenceWorker` — same-shape dynamic batching: requests with equal
+  (prompt length, sampling params, eos) arriving within a short window
+  are stacked into ONE `Generator.generate` call.  Grouping by prompt
+  length keeps the batch un-padded (the cache layout has no padding
+  mask, so mixed lengths would silently change positions).
+- `ContinuousBatcher` — vLLM-style slot pool over the variable-position
+  decode step: per-slot admission/retirement mid-decode, device-resident
+  slot state (zero host syncs per step), optional hipGraph-replayed
+  greedy step, exact-prefix KV reuse on admission, SSE streaming.
 
-Endpoints:
-  POST /v1/generate   {"tokens": [[int]] | [int], "max_new_tokens": N,
-                       "temperature": T}  ->  {"tokens": [[int]], ...}
-  GET  /healthz       liveness
-  GET  /v1/stats      batching + throughput counters
+MI355X notes: decode is weight-read-bound (NOTES.md r2 serving ladder,
+profiles/decode_batch_sweep_r2.txt: step time flat in batch size), so
+slot occupancy converts directly into tokens/s — measured 2.03x serial
+at 8 slots (profiles/serve_continuous_r2.txt).
 
-Run: `python -m scaffold.serve model.name=llama ... [--port 8000]`
+Endpoints (see docs/SERVING.md for bodies): POST /v1/generate (ids,
+optional SSE stream), POST /v1/completions (text, with --tokenizer),
+GET /healthz, GET /v1/stats, GET /metrics (Prometheus).
+
+Run: `python -m scaffold.serve model.name=llama3_8b [--continuous]
+[--graph] [--weights w.safetensors] [--lora a.safetensors --lora-r 8]`
 """
 
 from __future__ import annotations
@@ -95,7 +96,6 @@ class _Request:
     # sentinel; the final `result` is still set at finish
     stream: "object | None" = None
     stream_every: int = 4
-    future: "asyncio.Future | None" = None
     event: threading.Event = field(default_factory=threading.Event)
     result: Optional[List[int]] = None
     error: Optional[Exception] = None
@@ -260,8 +260,6 @@ class ContinuousBatcher:
                  seed: int = 0, use_graph: bool = False,
                  prefix_cache_entries: int = 8, max_queue: int = 256):
         self.max_queue = max_queue
-        from .generate import KVCache  # noqa: F401 (type reference)
-
         self.gen = Generator(model.eval())
         self.device = next(model.parameters()).device
         m = model
