@@ -484,16 +484,18 @@ class ContinuousBatcher:
         if r.stream is None or st.get("stream_done"):
             return
         seen = st.get("seen", 0)
-        due = final or (st["count"] - seen) >= r.stream_every
-        if not due or st["count"] <= seen:
+        if not final and (st["count"] - seen) < r.stream_every:
             return
-        chunk = self._toks[b, seen:st["count"]].cpu().tolist()
-        st["seen"] = st["count"]
-        if r.eos_token_id is not None and r.eos_token_id in chunk:
-            chunk = chunk[:chunk.index(r.eos_token_id)]
-            st["stream_done"] = True
-        for t in chunk:
-            r.stream.put(t)
+        if st["count"] > seen:
+            chunk = self._toks[b, seen:st["count"]].cpu().tolist()
+            st["seen"] = st["count"]
+            if r.eos_token_id is not None and r.eos_token_id in chunk:
+                chunk = chunk[:chunk.index(r.eos_token_id)]
+                st["stream_done"] = True
+            for t in chunk:
+                r.stream.put(t)
+        # the sentinel MUST go out on final even if nothing new was
+        # generated since the last flush — an SSE reader blocks on it
         if final or st.get("stream_done"):
             r.stream.put(None)
             st["stream_done"] = True
@@ -573,6 +575,8 @@ class ContinuousBatcher:
                     err = RuntimeError("batcher shut down")
                     for r in self._q:
                         r.error = err
+                        if r.stream is not None:
+                            r.stream.put(None)   # release SSE readers
                         r.event.set()
                     self._q.clear()
                     for b, st in enumerate(self._slots):
