@@ -443,24 +443,30 @@ class ContinuousBatcher:
         self._slots[slot] = None
 
     @torch.no_grad()
-    def _step_static(self):
+    def _step_static(self, hi: Optional[int] = None):
         """The STATIC greedy step: reads/writes only the persistent
         device buffers (_cur/_pos/_count/_active/_toks/cache), no host
-        reads — hipGraph-capturable (use_graph=True replays it)."""
-        logits = self.gen.decode_step_varpos(self._cur, self._cache,
-                                             self._pos)
+        reads — hipGraph-capturable (use_graph=True replays the
+        full-width form).  `hi` restricts the step to the first `hi`
+        slots (zero-copy views; eager path only)."""
+        if hi is None:
+            hi = self.max_slots
+            cache = self._cache
+        else:
+            cache = self._SlicedCache(self._cache, hi)
+        cur, pos = self._cur[:hi], self._pos[:hi]
+        count, act1 = self._count[:hi], self._active[:hi]
+        logits = self.gen.decode_step_varpos(cur, cache, pos)
         nxt = logits.argmax(dim=-1, keepdim=True)
-        act = self._active.unsqueeze(1)
-        new = torch.where(act.bool(), nxt, self._cur)
-        self._cur.copy_(new)
+        new = torch.where(act1.unsqueeze(1).bool(), nxt, cur)
+        cur.copy_(new)
         # active rows land at their count; inactive rows at the scratch
         # column (cap)
         cap = self._toks.shape[1] - 1
-        idx = torch.where(self._active.bool(), self._count,
-                          torch.full_like(self._count, cap))
-        self._toks.scatter_(1, idx.unsqueeze(1), new)
-        self._count.add_(self._active)
-        self._pos.add_(self._active)
+        idx = torch.where(act1.bool(), count, torch.full_like(count, cap))
+        self._toks[:hi].scatter_(1, idx.unsqueeze(1), new)
+        count.add_(act1)
+        pos.add_(act1)
 
     def _upload_state(self):
         """Refresh the device buffers from the host slot mirrors (tiny
@@ -512,33 +518,50 @@ class ContinuousBatcher:
             else:
                 self._flush_stream(b, st)
 
+    class _SlicedCache:
+        """Zero-copy view of the first `hi` slots of a KVCache (the
+        varpos step only needs .k/.v row views)."""
+
+        def __init__(self, cache, hi):
+            self.k = [t[:hi] for t in cache.k]
+            self.v = [t[:hi] for t in cache.v]
+
     @torch.no_grad()
     def _step(self):
         """One decode step over every occupied slot — no host syncs:
         positions/counts are host ints (host->device copies only),
-        sampled ids stay in the device-side _cur/_toks buffers."""
+        sampled ids stay in the device-side _cur/_toks buffers.
+
+        Eager steps compute only the OCCUPIED PREFIX of the slot pool
+        (admission always picks the lowest free index, so a half-empty
+        pool costs half a step, not a full one); the graphed step keeps
+        the full static width so ONE capture serves every occupancy."""
         active = [b for b, st in enumerate(self._slots) if st is not None]
         greedy = all(self._slots[b]["req"].temperature <= 0 for b in active)
         self._upload_state()
+        if greedy and self.use_graph and self.device.type == "cuda":
+            self._graphed_step()
+            self.stats["steps"] += 1
+            self._advance_host()
+            return
+        hi = max(active) + 1
         if greedy:
-            if self.use_graph and self.device.type == "cuda":
-                self._graphed_step()
-            else:
-                self._step_static()
+            self._step_static(hi)
             self.stats["steps"] += 1
             self._advance_host()
             return
         # sampled path (per-request temperature/top_k/top_p): eager
-        logits = self.gen.decode_step_varpos(self._cur, self._cache,
-                                             self._pos)
+        logits = self.gen.decode_step_varpos(
+            self._cur[:hi], self._SlicedCache(self._cache, hi),
+            self._pos[:hi])
         self.stats["steps"] += 1
-        nxt = self._cur.clone()
+        nxt = self._cur[:hi].clone()
         for b in active:
             r = self._slots[b]["req"]
             nxt[b] = Generator._sample(logits[b:b + 1], r.temperature,
                                        r.top_k, r.top_p,
                                        self._sample_gen)[0]
-        self._cur.copy_(nxt)
+        self._cur[:hi].copy_(nxt)
         for b in active:
             self._toks[b, self._slots[b]["count"]].copy_(nxt[b, 0])
         self._advance_host()
