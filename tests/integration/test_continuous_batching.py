@@ -290,3 +290,43 @@ def test_prefix_cache_disabled():
         assert len(cb._prefix_cache) == 0
     finally:
         cb.close()
+
+
+def test_everything_on_serving_flow():
+    """Compose the serving features in one pool: streaming + eos +
+    prefix-cache + sampling + mixed lengths + slot reuse."""
+    import queue
+
+    m = _model()
+    cb = ContinuousBatcher(m, max_slots=2, prefix_cache_entries=4, seed=3)
+    try:
+        sysp = [2, 3, 4, 5]
+        ref = _ref(m, sysp, 14)
+        eos = ref[6]
+        cut = ref.index(eos)          # trim happens at FIRST occurrence
+        # 1: plain greedy (cold, populates prefix cache)
+        r1 = cb.submit(sysp, 14)
+        # 2: streamed greedy with eos on the SAME prompt (prefix hit)
+        q2 = queue.Queue()
+        r2 = cb.submit(sysp, 14, eos_token_id=eos, stream=q2,
+                       stream_every=2)
+        # 3: sampled request on an extended prompt
+        r3 = cb.submit(sysp + [9, 9], 8, temperature=0.8, top_k=16)
+        # 4: another greedy, different length
+        r4 = cb.submit([30, 31, 32], 5)
+        got2 = []
+        while True:
+            t = q2.get(timeout=60)
+            if t is None:
+                break
+            got2.append(t)
+        for r in (r1, r2, r3, r4):
+            assert r.event.wait(timeout=120) and r.error is None
+        assert r1.result == ref
+        assert r2.result == ref[:cut] == got2
+        assert len(r3.result) == 8 and all(0 <= t < 128 for t in r3.result)
+        assert r4.result == _ref(m, [30, 31, 32], 5)
+        assert cb.stats["prefix_hits"] >= 1
+        assert cb.stats["requests"] == 4
+    finally:
+        cb.close()

@@ -270,3 +270,41 @@ def test_speculative_eos_matches_generate():
     out = target.generate_speculative(prompt, 20, draft, k=3,
                                       eos_token_id=eos)
     assert torch.equal(out, ref)
+
+
+def test_speculative_randomized_sweep():
+    """Deterministic randomized stress: many (prompt_len, k, max_new)
+    combinations against plain greedy — guards the cache-rollback
+    arithmetic across round boundaries and truncation edges."""
+    import random
+
+    import torch
+
+    from scaffold.config import load_config
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=128",
+        "model.vocab_size=128", "model.max_seq_len=128",
+    ])
+    torch.manual_seed(57)
+    target = Generator(build_model(cfg).eval())
+    torch.manual_seed(58)
+    cfg_d = load_config(None, [
+        "model.name=llama", "model.n_layers=1", "model.n_heads=2",
+        "model.n_kv_heads=1", "model.d_model=32", "model.d_ff=64",
+        "model.vocab_size=128", "model.max_seq_len=128",
+    ])
+    draft = Generator(build_model(cfg_d).eval())
+
+    rng = random.Random(0)
+    for _ in range(12):
+        s0 = rng.randint(1, 12)
+        n = rng.randint(1, 25)
+        k = rng.randint(1, 6)
+        prompt = torch.randint(0, 128, (1, s0))
+        ref = target.generate(prompt, n)
+        out = target.generate_speculative(prompt, n, draft, k=k)
+        assert torch.equal(out, ref), (s0, n, k)
