@@ -114,3 +114,49 @@ def test_zero1_reshard_to_single_process(tmp_path):
     assert t.optimizer.step_count == 4
     # moments carry real signal (loaded, not fresh zeros)
     assert any(m.abs().sum() > 0 for m in t.optimizer.exp_avg)
+
+
+def _accum_worker(rank, world, store_path, zero1, accum, q):
+    from scaffold.runtime import DistContext
+    from scaffold.trainer import Trainer
+
+    store = dist.FileStore(store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
+                      backend="gloo", device=torch.device("cpu"),
+                      initialized=True)
+    cfg = _cfg(zero1, steps=4)
+    cfg.grad_accum = accum
+    t = Trainer(cfg, ctx=ctx)
+    it = iter(t.data)
+    for _ in range(4):
+        t._train_step(it, accum)
+        t.step += 1
+    params = torch.cat([p.detach().flatten()
+                        for p in t.raw_model.parameters()])
+    if rank == 0:
+        q.put(params.numpy())
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_zero1_with_grad_accumulation(tmp_path):
+    """zero1 x grad_accum: broadcasts fire once per OPTIMIZER step, so
+    the accumulated run must stay bitwise-equal to plain DDP's."""
+
+    def run(zero1):
+        ctxm = mp.get_context("spawn")
+        q = ctxm.Queue()
+        procs = [ctxm.Process(
+            target=_accum_worker,
+            args=(r, WORLD, str(tmp_path / f"s{zero1}"), zero1, 2, q))
+            for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        out = q.get(timeout=200)
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        return out
+
+    assert np.array_equal(run(False), run(True))
