@@ -106,3 +106,81 @@ class PipelineStage:
         if self.is_last:
             return float(sum(l.detach() for l in losses) / n_micro)
         return 0.0
+
+    def run_batch_1f1b(self, micro_inputs: List[torch.Tensor] = None,
+                       micro_labels: List[torch.Tensor] = None,
+                       n_micro: int = None) -> float:
+        """One-forward-one-backward schedule: after a rank-dependent
+        warmup (world-1-rank forwards), every forward is immediately
+        paired with the backward of the OLDEST outstanding micro-batch,
+        so at most world-rank activations are ever live per stage —
+        GPipe's run_batch keeps all n_micro.  Same math (micro-grads
+        accumulate; fp32 summation ORDER differs from GPipe's reversed
+        wave, so equality with the unpipelined model is allclose, not
+        bitwise).  `peak_live` records the measured activation high-water
+        mark for the memory claim."""
+        if n_micro is None:
+            n_micro = len(micro_inputs) if micro_inputs else len(micro_labels)
+        pending_in: List[torch.Tensor] = []   # FIFO of boundary leaves
+        pending_out: List[torch.Tensor] = []  # FIFO of boundary outputs
+        loss_sum = 0.0
+        self.peak_live = 0
+        fwd_done = 0
+        bwd_done = 0
+        # 1F1B interleaves sends in BOTH directions: a blocking
+        # (rendezvous) send deadlocks head-to-head with the neighbour's
+        # grad send (measured: rank r in send-F(m+1) vs rank r+1 in
+        # send-B(m)).  isend + deferred wait keeps the schedule's
+        # pairwise channel order without the rendezvous coupling; the
+        # sent tensors (detached y, x.grad) are never mutated afterwards.
+        sends = []
+
+        def _isend(t, dst):
+            sends.append(dist.isend(t.contiguous(), dst=dst,
+                                    group=self.group))
+
+        def one_fwd():
+            nonlocal fwd_done
+            m = fwd_done
+            if self.is_first:
+                x = micro_inputs[m]
+            else:
+                x = _recv(self.act_shape, self.act_dtype, self.rank - 1,
+                          self.device, self.group)
+                x = x.requires_grad_(True)
+            pending_in.append(x)
+            if self.is_last:
+                pending_out.append(self.module(x, micro_labels[m]))
+            else:
+                y = self.module(x)
+                pending_out.append(y)
+                _isend(y.detach(), self.rank + 1)
+            fwd_done += 1
+            self.peak_live = max(self.peak_live, len(pending_in))
+
+        def one_bwd():
+            nonlocal bwd_done, loss_sum
+            x = pending_in.pop(0)
+            out = pending_out.pop(0)
+            if self.is_last:
+                loss_sum += float(out.detach())
+                (out / n_micro).backward()
+            else:
+                dy = _recv(self.act_shape, self.act_dtype, self.rank + 1,
+                           self.device, self.group)
+                out.backward(dy)
+            if not self.is_first:
+                _isend(x.grad, self.rank - 1)
+            bwd_done += 1
+
+        warmup = min(self.world - 1 - self.rank, n_micro)
+        for _ in range(warmup):
+            one_fwd()
+        while fwd_done < n_micro:
+            one_fwd()
+            one_bwd()
+        while bwd_done < n_micro:
+            one_bwd()
+        for w in sends:
+            w.wait()
+        return loss_sum / n_micro if self.is_last else 0.0
