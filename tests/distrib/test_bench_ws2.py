@@ -40,3 +40,14 @@ def test_bench_ws2_gloo(tmp_path):
     assert d["n_gpus"] == 2
     assert d["config"]["parallelism"] == "dp2"
     assert d["value"] > 0
+    # every field the driver's contract names must be present and typed
+    for key, typ in (("metric", str), ("value", (int, float)),
+                     ("unit", str), ("n_gpus", int), ("steps", int),
+                     ("warmup", int), ("ms_per_step", (int, float)),
+                     ("higher_is_better", bool), ("scaling", str),
+                     ("dtype", str), ("data", str), ("config", dict)):
+        assert key in d and isinstance(d[key], typ), key
+    assert "vs_baseline" in d          # may be null
+    assert d["scaling"] in ("weak", "strong")
+    assert d["data"] == "synthetic"
+    assert {"model", "global_batch"} <= set(d["config"])
