@@ -110,3 +110,22 @@ def test_metrics_jsonl(tmp_path):
     assert len(recs) == 2
     assert recs[0]["loss"] == 2.5 and recs[0]["comm_ms"] == 1.25
     assert recs[1]["samples_per_sec_node"] > 0
+
+
+@pytest.mark.parametrize("argv", [
+    ["train.py", "--help"],
+    ["bench.py", "--help"],
+    ["-m", "scaffold.serve", "--help"],
+    ["-m", "scaffold.launch", "--help"],
+])
+def test_cli_help_exits_zero(argv):
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    r = subprocess.run([sys.executable, *argv], cwd=repo,
+                       capture_output=True, text=True, timeout=120)
+    assert r.returncode == 0, r.stderr[-500:]
+    assert "usage" in (r.stdout + r.stderr).lower()
