@@ -112,6 +112,9 @@ def test_metrics_jsonl(tmp_path):
     assert recs[1]["samples_per_sec_node"] > 0
 
 
+import pytest
+
+
 @pytest.mark.parametrize("argv", [
     ["train.py", "--help"],
     ["bench.py", "--help"],
