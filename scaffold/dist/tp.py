@@ -151,3 +151,55 @@ class TpMLP(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         h = torch.nn.functional.gelu(self.fc(x), approximate="tanh")
         return self.proj(h)
+
+
+@torch.no_grad()
+def shard_llama_tp(model, rank: int, world: int, group=None):
+    """Shard a built Llama IN PLACE for tensor-parallel INFERENCE:
+    qkv heads and gate|up columns split across ranks (column-parallel),
+    wo / w_down input-sliced (row-parallel); embeddings, norms and
+    lm_head stay replicated.  The Generator completes each block with
+    an all_reduce after the two row-parallel projections (it reduces
+    when `attn._tp_group`-style markers are present), so per-rank decode
+    touches 1/world of the attention+MLP weights — the llama-70B-class
+    single-node serving layout (7 xGMI links serve the 2 all_reduces
+    per block per token).
+
+    Exactness: tests/distrib/test_tp.py proves ws=2 sharded decode ==
+    the unsharded rollout token-for-token.  Training-side TP remains
+    the layer-level Column/RowParallelLinear modules above.
+    """
+    for blk in model.blocks:
+        m = blk.attn
+        assert m.n_heads % world == 0 and m.n_kv % world == 0, \
+            f"heads {m.n_heads}/{m.n_kv} not divisible by tp={world}"
+        hl = m.n_heads // world
+        kl = m.n_kv // world
+        d = m.head_dim
+        w = m.wqkv.weight
+        q = w[rank * hl * d:(rank + 1) * hl * d]
+        k0 = m.n_heads * d
+        k = w[k0 + rank * kl * d:k0 + (rank + 1) * kl * d]
+        v0 = k0 + m.n_kv * d
+        v = w[v0 + rank * kl * d:v0 + (rank + 1) * kl * d]
+        m.wqkv.weight = torch.nn.Parameter(
+            torch.cat([q, k, v], dim=0).contiguous())
+        m.wo.weight = torch.nn.Parameter(
+            m.wo.weight[:, rank * hl * d:(rank + 1) * hl * d].contiguous())
+        m.n_heads = hl
+        m.n_kv = kl
+        m._tp_group = group
+        m._tp_world = world
+
+        gu = blk.w_gateup.weight
+        F_ = gu.shape[0] // 2
+        fl = F_ // world
+        gate = gu[rank * fl:(rank + 1) * fl]
+        up = gu[F_ + rank * fl:F_ + (rank + 1) * fl]
+        blk.w_gateup.weight = torch.nn.Parameter(
+            torch.cat([gate, up], dim=0).contiguous())
+        blk.w_down.weight = torch.nn.Parameter(
+            blk.w_down.weight[:, rank * fl:(rank + 1) * fl].contiguous())
+        blk._tp_group = group
+        blk._tp_world = world
+    return model

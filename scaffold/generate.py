@@ -76,6 +76,18 @@ class KVCache:
         return self.len
 
 
+def _tp_reduce(t: torch.Tensor, holder) -> torch.Tensor:
+    """Sum the row-parallel partial across the TP group when `holder`
+    carries the shard markers (dist.tp.shard_llama_tp); no-op
+    otherwise."""
+    if getattr(holder, "_tp_world", 1) > 1:
+        import torch.distributed as dist
+
+        if dist.is_initialized():
+            dist.all_reduce(t, group=getattr(holder, "_tp_group", None))
+    return t
+
+
 class Generator:
     """Greedy / temperature decoding for the scaffold's LM families."""
 
@@ -115,12 +127,13 @@ class Generator:
         k = ops._rope_eager(k, cos, sin) if not k.is_cuda else ops.rope_apply(k, cos, sin)
         k, v = cache.append(i, k, v)
         o = _sdpa(q, k, v, causal=True)
-        x = x + F.linear(o.transpose(1, 2).reshape(B, S, -1), m.wo.weight)
+        x = x + _tp_reduce(
+            F.linear(o.transpose(1, 2).reshape(B, S, -1), m.wo.weight), m)
         h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
         gu = F.linear(h, blk.w_gateup.weight)
         Fh = gu.shape[-1] // 2
         g = F.silu(gu[..., :Fh]) * gu[..., Fh:]
-        return x + F.linear(g, blk.w_down.weight)
+        return x + _tp_reduce(F.linear(g, blk.w_down.weight), blk)
 
     def _forward_tokens(self, tokens: torch.Tensor, cache: KVCache,
                         pos0: int, all_logits: bool = False) -> torch.Tensor:
@@ -436,13 +449,14 @@ class Generator:
             cache.v[i].scatter_(2, idx, v)
             o = self._sdpa_static(q, cache.k[i], cache.v[i],
                                   (pos + 1).view(B, 1, 1, 1))
-            x = x + self._lin1(o.transpose(1, 2).reshape(B, 1, -1),
-                               m.wo.weight)
+            x = x + _tp_reduce(
+                self._lin1(o.transpose(1, 2).reshape(B, 1, -1),
+                           m.wo.weight), m)
             h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
             gu = self._lin1(h, blk.w_gateup.weight)
             Fh = gu.shape[-1] // 2
             g = F.silu(gu[..., :Fh]) * gu[..., Fh:]
-            return x + self._lin1(g, blk.w_down.weight)
+            return x + _tp_reduce(self._lin1(g, blk.w_down.weight), blk)
         m = blk.attn
         h = ops.layer_norm(x, blk.ln1.weight, blk.ln1.bias, blk.ln1.eps)
         qkv = self._lin1(h, m.qkv.weight, m.qkv.bias).view(
