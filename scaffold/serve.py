@@ -122,9 +122,17 @@ class InferenceWorker:
 
     def __init__(self, model: torch.nn.Module, max_batch: int = 8,
                  window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False,
-                 max_new_tokens_cap: int = 512, max_queue: int = 256):
+                 max_new_tokens_cap: int = 512, max_queue: int = 256,
+                 draft_model: "torch.nn.Module | None" = None,
+                 draft_k: int = 4):
         self.max_queue = max_queue
         self.gen = Generator(model)
+        # speculative decoding (greedy B=1 requests only — per-row
+        # acceptance lengths diverge in a batch): a small draft model
+        # proposes draft_k tokens per target verify
+        self.draft = Generator(draft_model.eval()) if draft_model is not None \
+            else None
+        self.draft_k = draft_k
         self.max_len = (model.rope_cos.shape[0] if self.gen.is_llama
                         else model.wpe.num_embeddings)
         self.device = next(model.parameters()).device
@@ -209,7 +217,11 @@ class InferenceWorker:
         steps = max(r.max_new_tokens for r in batch)
         temp = batch[0].temperature
         eos = batch[0].eos_token_id
-        if self.use_graph and temp == 0.0 and eos is None:
+        if (self.draft is not None and len(batch) == 1 and temp == 0.0):
+            out = self.gen.generate_speculative(
+                prompt, steps, self.draft, k=self.draft_k,
+                eos_token_id=eos)
+        elif self.use_graph and temp == 0.0 and eos is None:
             out = self.gen.generate_graphed(prompt, steps)
         else:
             out = self.gen.generate(prompt, steps, temperature=temp,
@@ -641,7 +653,8 @@ class ContinuousBatcher:
 
 def create_app(model: torch.nn.Module, max_batch: int = 8,
                window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False,
-               tokenizer=None, continuous: bool = False):
+               tokenizer=None, continuous: bool = False,
+               draft_model=None, draft_k: int = 4):
     """FastAPI app around an inference worker.  `model` is an eval-mode
     LM from scaffold.models (tests inject a tiny CPU one).  Passing a
     `tokenizers.Tokenizer` adds the text endpoint /v1/completions.
@@ -656,7 +669,8 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
                                    use_graph=use_graph)
     else:
         worker = InferenceWorker(model.eval(), max_batch=max_batch,
-                                 window_s=window_s, use_graph=use_graph)
+                                 window_s=window_s, use_graph=use_graph,
+                                 draft_model=draft_model, draft_k=draft_k)
 
     @asynccontextmanager
     async def _lifespan(app):
@@ -839,6 +853,13 @@ def main(argv=None):
                          "(requires --lora-r to rebuild the adapters)")
     ap.add_argument("--lora-r", type=int, default=0)
     ap.add_argument("--lora-alpha", type=float, default=16.0)
+    ap.add_argument("--draft", default="",
+                    help="draft model weights for speculative decoding "
+                         "(safetensors; needs --draft-config overrides "
+                         "as a comma-joined list, e.g. "
+                         "'model.name=llama,model.n_layers=4')")
+    ap.add_argument("--draft-config", default="")
+    ap.add_argument("--draft-k", type=int, default=4)
     args = ap.parse_args(argv)
 
     import uvicorn
@@ -865,8 +886,18 @@ def main(argv=None):
     if args.tokenizer:
         from tokenizers import Tokenizer
         tok = Tokenizer.from_file(args.tokenizer)
+    draft = None
+    if args.draft:
+        from .trainer import Trainer
+
+        dcfg = load_config(None, args.draft_config.split(",")
+                           if args.draft_config else [])
+        draft = build_model(dcfg)
+        Trainer._load_init_weights(draft, args.draft)
+        draft = draft.to(dev, dt).eval()
     app = create_app(model, max_batch=args.max_batch, use_graph=args.graph,
-                     tokenizer=tok, continuous=args.continuous)
+                     tokenizer=tok, continuous=args.continuous,
+                     draft_model=draft, draft_k=args.draft_k)
     uvicorn.run(app, host=args.host, port=args.port, log_level="warning")
 
 

@@ -284,3 +284,27 @@ def test_queue_bound_rejects_with_overload():
             w.submit([5, 6], 2)
     finally:
         w.close()
+
+
+def test_worker_speculative_path():
+    """A worker with a draft model serves greedy B=1 requests through
+    generate_speculative — results identical to plain decode."""
+    torch.manual_seed(83)
+    m = _tiny_model()
+    cfg_d = load_config(None, [
+        "model.name=llama", "model.n_layers=1", "model.n_heads=2",
+        "model.n_kv_heads=1", "model.d_model=32", "model.d_ff=64",
+        "model.vocab_size=128", "model.max_seq_len=64",
+    ])
+    torch.manual_seed(84)
+    draft = build_model(cfg_d).eval()
+    ref = Generator(m).generate(
+        torch.tensor([[1, 2, 3, 4]], dtype=torch.int64), 10)[0, 4:].tolist()
+
+    w = InferenceWorker(m, window_s=0.0, draft_model=draft, draft_k=3)
+    try:
+        r = w.submit([1, 2, 3, 4], 10)
+        assert r.event.wait(60) and r.error is None
+        assert r.result == ref
+    finally:
+        w.close()
