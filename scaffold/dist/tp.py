@@ -169,6 +169,8 @@ def shard_llama_tp(model, rank: int, world: int, group=None):
     the unsharded rollout token-for-token.  Training-side TP remains
     the layer-level Column/RowParallelLinear modules above.
     """
+    if not hasattr(model.blocks[0].attn, "wqkv"):
+        return shard_gpt2_tp(model, rank, world, group)
     for blk in model.blocks:
         m = blk.attn
         assert m.n_heads % world == 0 and m.n_kv % world == 0, \
@@ -200,6 +202,50 @@ def shard_llama_tp(model, rank: int, world: int, group=None):
             torch.cat([gate, up], dim=0).contiguous())
         blk.w_down.weight = torch.nn.Parameter(
             blk.w_down.weight[:, rank * fl:(rank + 1) * fl].contiguous())
+        blk._tp_group = group
+        blk._tp_world = world
+    return model
+
+
+@torch.no_grad()
+def shard_gpt2_tp(model, rank: int, world: int, group=None):
+    """GPT-2 variant of shard_llama_tp.  Differences: the qkv weight
+    packs [q|k|v] each d_model rows with a BIAS (column-parallel bias
+    shards with its rows), and the row-parallel projections (attn.proj,
+    mlp proj) carry biases that must be added AFTER the all_reduce —
+    the Generator adds them post-reduction when the markers are set
+    (a partial+bias on every rank would count the bias world times)."""
+    for blk in model.blocks:
+        m = blk.attn
+        assert m.n_heads % world == 0, \
+            f"heads {m.n_heads} not divisible by tp={world}"
+        hl = m.n_heads // world
+        d = m.head_dim
+        dm = m.n_heads * d
+        w, b = m.qkv.weight, m.qkv.bias
+        rows = []
+        brows = []
+        for part in range(3):                 # q, k, v blocks
+            lo = part * dm + rank * hl * d
+            rows.append(w[lo:lo + hl * d])
+            brows.append(b[lo:lo + hl * d])
+        m.qkv.weight = torch.nn.Parameter(torch.cat(rows).contiguous())
+        m.qkv.bias = torch.nn.Parameter(torch.cat(brows).contiguous())
+        m.proj.weight = torch.nn.Parameter(
+            m.proj.weight[:, rank * hl * d:(rank + 1) * hl * d].contiguous())
+        # proj.bias stays replicated; added post-reduce by the Generator
+        m.n_heads = hl
+        m._tp_group = group
+        m._tp_world = world
+
+        F_ = blk.fc.weight.shape[0]
+        fl = F_ // world
+        blk.fc.weight = torch.nn.Parameter(
+            blk.fc.weight[rank * fl:(rank + 1) * fl].contiguous())
+        blk.fc.bias = torch.nn.Parameter(
+            blk.fc.bias[rank * fl:(rank + 1) * fl].contiguous())
+        blk.proj.weight = torch.nn.Parameter(
+            blk.proj.weight[:, rank * fl:(rank + 1) * fl].contiguous())
         blk._tp_group = group
         blk._tp_world = world
     return model

@@ -106,8 +106,18 @@ class Generator:
         q, k, v = (qkv[:, :, j].transpose(1, 2) for j in range(3))
         k, v = cache.append(i, k, v)
         o = _sdpa(q, k, v, causal=True)
-        x = x + F.linear(o.transpose(1, 2).reshape(B, S, C),
-                         m.proj.weight, m.proj.bias)
+        o2 = o.transpose(1, 2).reshape(B, S, -1)
+        if getattr(m, "_tp_world", 1) > 1:
+            # row-parallel: reduce the partial, THEN add the replicated
+            # bias (inside-linear bias would be summed world times)
+            x = x + (_tp_reduce(F.linear(o2, m.proj.weight), m)
+                     + m.proj.bias)
+            h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+            h = F.gelu(F.linear(h, blk.fc.weight, blk.fc.bias),
+                       approximate="tanh")
+            return x + (_tp_reduce(F.linear(h, blk.proj.weight), blk)
+                        + blk.proj.bias)
+        x = x + F.linear(o2, m.proj.weight, m.proj.bias)
         h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
         h = F.gelu(F.linear(h, blk.fc.weight, blk.fc.bias), approximate="tanh")
         return x + F.linear(h, blk.proj.weight, blk.proj.bias)
@@ -467,8 +477,16 @@ class Generator:
         cache.v[i].scatter_(2, idx, v)
         o = self._sdpa_static(q, cache.k[i], cache.v[i],
                               (pos + 1).view(B, 1, 1, 1))
-        x = x + self._lin1(o.transpose(1, 2).reshape(B, 1, -1),
-                           m.proj.weight, m.proj.bias)
+        o2 = o.transpose(1, 2).reshape(B, 1, -1)
+        if getattr(m, "_tp_world", 1) > 1:
+            x = x + (_tp_reduce(self._lin1(o2, m.proj.weight), m)
+                     + m.proj.bias)
+            h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+            h = F.gelu(self._lin1(h, blk.fc.weight, blk.fc.bias),
+                       approximate="tanh")
+            return x + (_tp_reduce(self._lin1(h, blk.proj.weight), blk)
+                        + blk.proj.bias)
+        x = x + self._lin1(o2, m.proj.weight, m.proj.bias)
         h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
         h = F.gelu(self._lin1(h, blk.fc.weight, blk.fc.bias),
                    approximate="tanh")
