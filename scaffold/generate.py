@@ -521,6 +521,12 @@ class Generator:
         n_layers = len(self.m.blocks)
         B, S0 = prompt.shape
         attn0 = self.m.blocks[0].attn
+        if getattr(attn0, "_tp_world", 1) > 1:
+            # the static step has no TP reductions — failing loudly
+            # beats silently wrong tokens; TP decode runs through
+            # generate() / ContinuousBatcher (both reduce)
+            raise NotImplementedError(
+                "generate_graphed does not support TP-sharded models")
         n_kv = attn0.n_kv if self.is_llama else attn0.n_heads
         dtype = next(self.m.parameters()).dtype
         L = S0 + max_new_tokens

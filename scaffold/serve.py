@@ -581,7 +581,12 @@ class ContinuousBatcher:
     def _graphed_step(self):
         """Capture once (after two warmup executions — generate_graphed's
         proven recipe), then replay; the graph reads the persistent
-        buffers that _upload_state refreshes."""
+        buffers that _upload_state refreshes.  TP-sharded models fall
+        back to the eager step: capturing the in-step all_reduces is
+        unvalidated on RCCL."""
+        if getattr(self.gen.m.blocks[0].attn, "_tp_world", 1) > 1:
+            self._step_static()
+            return
         if self._graph is None:
             if getattr(self, "_warmups", 0) < 2:
                 side = torch.cuda.Stream()

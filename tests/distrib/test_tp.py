@@ -266,3 +266,29 @@ def test_tp_gpt2_decode_matches_full(tmp_path):
     full = build_model(cfg).eval()
     ref = Generator(full).generate(torch.arange(1, 6).view(1, 5), 12)
     assert np.array_equal(out_tp, ref.numpy())
+
+
+def test_graphed_paths_guard_tp(tmp_path):
+    """A TP-sharded model must FAIL LOUDLY on generate_graphed (the
+    static step has no reductions) instead of emitting wrong tokens."""
+    import torch.distributed as dist
+
+    from scaffold.config import load_config
+    from scaffold.dist.tp import shard_llama_tp
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    store = dist.FileStore(str(tmp_path / "fs"), 1)
+    dist.init_process_group("gloo", store=store, rank=0, world_size=1)
+    try:
+        cfg = load_config(None, [
+            "model.name=llama", "model.n_layers=1", "model.n_heads=4",
+            "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=128",
+            "model.vocab_size=64", "model.max_seq_len=32",
+        ])
+        m = build_model(cfg).eval()
+        shard_llama_tp(m, 0, 2)       # markers set (world=2 semantics)
+        with pytest.raises(NotImplementedError, match="TP"):
+            Generator(m).generate_graphed(torch.tensor([[1, 2]]), 4)
+    finally:
+        dist.destroy_process_group()
