@@ -330,3 +330,43 @@ def test_everything_on_serving_flow():
         assert cb.stats["requests"] == 4
     finally:
         cb.close()
+
+
+@pytest.mark.timeout(300)
+def test_concurrent_submitters_stress():
+    """3 submitter threads x 10 randomized greedy requests against 3
+    slots: every result must equal its isolated rollout (no cross-slot
+    or cross-thread contamination under real contention)."""
+    import random
+    import threading
+
+    m = _model()
+    cb = ContinuousBatcher(m, max_slots=3, prefix_cache_entries=4)
+    results = {}
+    lock = threading.Lock()
+
+    def submitter(tid):
+        rng = random.Random(tid)
+        for i in range(10):
+            s0 = rng.randint(1, 10)
+            n = rng.randint(1, 12)
+            prompt = [rng.randint(0, 127) for _ in range(s0)]
+            r = cb.submit(prompt, n)
+            assert r.event.wait(timeout=120) and r.error is None
+            with lock:
+                results[(tid, i)] = (prompt, n, r.result)
+
+    try:
+        threads = [threading.Thread(target=submitter, args=(t,))
+                   for t in range(3)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=200)
+            assert not t.is_alive()
+    finally:
+        cb.close()
+
+    assert len(results) == 30
+    for prompt, n, got in results.values():
+        assert got == _ref(m, prompt, n), (prompt, n)
