@@ -308,3 +308,46 @@ def test_worker_speculative_path():
         assert r.result == ref
     finally:
         w.close()
+
+
+@pytest.mark.timeout(120)
+def test_real_uvicorn_server_over_tcp(tmp_path):
+    """The stack outside TestClient: a real uvicorn server on a TCP
+    port, queried with httpx."""
+    import socket
+    import threading
+    import time as _time
+
+    import httpx
+    import uvicorn
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+
+    app = create_app(_tiny_model(), window_s=0.0)
+    config = uvicorn.Config(app, host="127.0.0.1", port=port,
+                            log_level="error")
+    server = uvicorn.Server(config)
+    th = threading.Thread(target=server.run, daemon=True)
+    th.start()
+    try:
+        deadline = _time.time() + 30
+        url = f"http://127.0.0.1:{port}"
+        while _time.time() < deadline:
+            try:
+                if httpx.get(url + "/healthz", timeout=2).status_code == 200:
+                    break
+            except Exception:
+                _time.sleep(0.1)
+        else:
+            raise AssertionError("server never became healthy")
+        r = httpx.post(url + "/v1/generate",
+                       json={"tokens": [1, 2, 3], "max_new_tokens": 5},
+                       timeout=30)
+        assert r.status_code == 200, r.text
+        assert len(r.json()["tokens"][0]) == 5
+        assert httpx.get(url + "/metrics", timeout=10).status_code == 200
+    finally:
+        server.should_exit = True
+        th.join(timeout=10)
