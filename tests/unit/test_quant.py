@@ -57,3 +57,18 @@ def test_roundtrip_close_and_smaller(tmp_path):
     assert torch.allclose(la, lb, rtol=0.05, atol=0.05), (la, lb)
     # tie preserved structurally
     assert m2.lm_head.weight.data_ptr() == m2.wte.weight.data_ptr()
+
+
+@pytest.mark.gpu
+def test_quantized_load_gpu(tmp_path):
+    """int8 export loads into a bf16 CUDA model; logits stay close."""
+    m = _model()
+    p = str(tmp_path / "q.safetensors")
+    save_quantized(m, p)
+    mg = _model().to("cuda", torch.bfloat16)
+    load_quantized(mg, p)
+    tok = torch.randint(0, 128, (2, 16), device="cuda")
+    la = mg(tok, tok).float().cpu()
+    lb = m(tok, tok).float()
+    torch.cuda.synchronize()
+    assert torch.allclose(la, lb, rtol=0.1, atol=0.1), (la, lb)
