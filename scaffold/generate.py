@@ -2,12 +2,16 @@
 of the scaffold).
 
 Prefill runs the prompt through the normal block math (the flash
-attention op on GPU) while recording K/V; each decode step then attends
-the single new query against the cache with plain library ops (the
-per-step work is a handful of GEMV-shaped products — the library path
-per SURVEY §2.2; a fused decode kernel is a round-2 item).  Supports
-both model families (GPT-2 LayerNorm/GELU, Llama RMSNorm/RoPE/GQA/
-SwiGLU).  Greedy (temperature=0) or temperature sampling.
+attention op on GPU) while recording K/V; each decode step attends the
+single new query against the cache.  The decode GEMVs take the library
+path — the hand decode_gemv kernel measured below hipBLASLt in-graph
+and is parked behind SCAFFOLD_DECODE_GEMV=1 (NOTES r2 serving ladder).
+Both model families (GPT-2 LayerNorm/GELU, Llama RMSNorm/RoPE/GQA/
+SwiGLU); greedy / temperature / top-k / top-p sampling with optional
+EOS; hipGraph-replayed static decode (generate_graphed); per-row
+variable-position decode for continuous batching (decode_step_varpos);
+speculative decoding (generate_speculative); TP-sharded models reduce
+through _tp_reduce markers (dist.tp.shard_llama_tp).
 """
 
 from __future__ import annotations
