@@ -8,6 +8,14 @@ host thread serializes a chunk to the shard file only after its event
 fires.  On CPU the same code degenerates to plain copies.
 
 Format: safetensors per shard + manifest.json (layout.py).
+
+Load is streaming same-world (each rank reads only its own shard from
+disk, the full state completes over collectives through a pinned H2D
+double-buffer) or by-name resharding across world sizes.  ZeRO-1
+optimizers (optimizer.shard_owners set) save their state RANK-PINNED:
+each rank's shard carries its owned masters/moments, the manifest stays
+globally consistent because the ownership partition is deterministic,
+and the same-world loader reads them locally without collectives.
 """
 
 from __future__ import annotations
