@@ -41,6 +41,14 @@ class ModelConfig:
     d_ff: Optional[int] = None  # None -> 4*d_model (gpt2) / preset (llama)
     max_seq_len: int = 1024
     rope_theta: float = 500000.0
+    # RoPE frequency scaling for long-context checkpoints:
+    # "none" | "linear" (divide frequencies by rope_factor) | "llama3"
+    # (the Llama-3.1 wavelength-dependent rule; matches transformers)
+    rope_scaling: str = "none"
+    rope_factor: float = 8.0
+    rope_low_freq_factor: float = 1.0
+    rope_high_freq_factor: float = 4.0
+    rope_orig_max_pos: int = 8192
     norm_eps: float = 1e-5
     tie_embeddings: bool = True
     # gpt2: replace every block's dense FFN with a top-1-routed MoE of

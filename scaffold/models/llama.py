@@ -86,7 +86,13 @@ class Llama(nn.Module):
         self.norm_f = RMSNorm(m.d_model, m.norm_eps)
         self.lm_head = nn.Linear(m.d_model, m.vocab_size, bias=False)
         head_dim = m.d_model // m.n_heads
-        cos, sin = ops.rope_tables(m.max_seq_len, head_dim, m.rope_theta, "cpu")
+        cos, sin = ops.rope_tables(
+            m.max_seq_len, head_dim, m.rope_theta, "cpu",
+            scaling=getattr(m, "rope_scaling", "none"),
+            factor=getattr(m, "rope_factor", 8.0),
+            low_freq_factor=getattr(m, "rope_low_freq_factor", 1.0),
+            high_freq_factor=getattr(m, "rope_high_freq_factor", 4.0),
+            orig_max_pos=getattr(m, "rope_orig_max_pos", 8192))
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
         for p in self.parameters():

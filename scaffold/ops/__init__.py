@@ -141,11 +141,36 @@ def rope_apply(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.T
     return _rope_eager(x, cos, sin)
 
 
-def rope_tables(seq_len: int, dim: int, theta: float, device, dtype=torch.float32
+def rope_tables(seq_len: int, dim: int, theta: float, device, dtype=torch.float32,
+                scaling: str = "none", factor: float = 8.0,
+                low_freq_factor: float = 1.0, high_freq_factor: float = 4.0,
+                orig_max_pos: int = 8192
                 ) -> Tuple[torch.Tensor, torch.Tensor]:
     """Host-precomputed cos/sin tables (S, D/2) — Appendix B guidance:
-    never compute trig per element on device."""
+    never compute trig per element on device.
+
+    scaling="linear" divides every frequency by `factor`;
+    scaling="llama3" applies the Llama-3.1 wavelength-dependent rule
+    (long wavelengths scaled by 1/factor, short kept, smooth ramp in
+    between) — parity with transformers' implementation is tested in
+    tests/integration/test_hf_parity.py."""
     inv = 1.0 / (theta ** (torch.arange(0, dim, 2, dtype=torch.float32) / dim))
+    if scaling == "linear":
+        inv = inv / factor
+    elif scaling == "llama3":
+        import math as _math
+
+        low_wl = orig_max_pos / low_freq_factor
+        high_wl = orig_max_pos / high_freq_factor
+        wavelen = 2 * _math.pi / inv
+        inv_scaled = torch.where(wavelen > low_wl, inv / factor, inv)
+        smooth = ((orig_max_pos / wavelen - low_freq_factor)
+                  / (high_freq_factor - low_freq_factor))
+        mid = (1 - smooth) * inv / factor + smooth * inv
+        in_ramp = (wavelen <= low_wl) & (wavelen >= high_wl)
+        inv = torch.where(in_ramp, mid, inv_scaled)
+    elif scaling != "none":
+        raise ValueError(f"unknown rope scaling {scaling!r}")
     t = torch.arange(seq_len, dtype=torch.float32)
     ang = torch.outer(t, inv)
     return ang.cos().to(device=device, dtype=dtype), ang.sin().to(device=device, dtype=dtype)

@@ -215,3 +215,49 @@ def test_save_load_model_single_file(tmp_path):
         assert torch.equal(x, y), n
     # tying preserved: loading into b kept lm_head sharing wte storage
     assert b.lm_head.weight.data_ptr() == b.wte.weight.data_ptr()
+
+
+def test_llama31_rope_scaling_matches_transformers():
+    """Llama-3.1-style rope scaling: our wavelength-dependent rule ==
+    transformers' (loss parity on converted weights with scaling ON)."""
+    from transformers import LlamaConfig, LlamaForCausalLM
+
+    torch.manual_seed(101)
+    hf = LlamaForCausalLM(LlamaConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, rope_theta=500000.0,
+        rms_norm_eps=1e-5, attention_bias=False, mlp_bias=False,
+        tie_word_embeddings=False, attn_implementation="eager",
+        rope_scaling={"rope_type": "llama3", "factor": 8.0,
+                      "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+                      "original_max_position_embeddings": 16},
+    )).eval()
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=128",
+        "model.vocab_size=128", "model.max_seq_len=64",
+        "model.rope_theta=500000.0", "model.rope_scaling=llama3",
+        "model.rope_factor=8.0", "model.rope_low_freq_factor=1.0",
+        "model.rope_high_freq_factor=4.0", "model.rope_orig_max_pos=16",
+    ])
+    ours = build_model(cfg)
+    load_hf(ours, hf.state_dict())
+    tok = torch.randint(0, 128, (2, 48))     # beyond orig_max_pos=16
+    tgt = torch.randint(0, 128, (2, 48))
+    with torch.no_grad():
+        ref = _ref_loss(hf(tok).logits, tgt)
+    got = _our_loss(ours, tok, tgt)
+    assert torch.allclose(got, ref, rtol=1e-4, atol=1e-5), (got, ref)
+
+
+def test_linear_rope_scaling_changes_tables():
+    from scaffold import ops
+
+    c0, s0 = ops.rope_tables(16, 32, 1e4, "cpu")
+    c1, s1 = ops.rope_tables(16, 32, 1e4, "cpu", scaling="linear",
+                             factor=2.0)
+    assert not torch.allclose(c0, c1)
+    # position 2k with factor 2 == position k unscaled
+    assert torch.allclose(c1[10], c0[5], atol=1e-6)
+    assert torch.allclose(s1[10], s0[5], atol=1e-6)
