@@ -16,12 +16,12 @@ from scaffold.config import load_config
 WORLD = 2
 
 
-def _cfg(zero1, ckpt_dir="", resume="", steps=5):
+def _cfg(zero1, ckpt_dir="", resume="", steps=5, dtype="fp32"):
     over = [
         "model.name=gpt2", "model.n_layers=2", "model.n_heads=2",
         "model.d_model=32", "model.vocab_size=64", "model.max_seq_len=16",
         "data.kind=synthetic_tokens", "data.batch_size=4", "data.seq_len=16",
-        "data.active_vocab=8", "dtype=fp32", "device=cpu",
+        f"data.active_vocab=8", f"dtype={dtype}", "device=cpu",
         "log.interval=1000", "optim.lr=0.01", f"steps={steps}",
         f"dist.zero1={'true' if zero1 else 'false'}",
     ]
@@ -33,7 +33,8 @@ def _cfg(zero1, ckpt_dir="", resume="", steps=5):
     return load_config(None, over)
 
 
-def _worker(rank, world, store_path, zero1, ckpt_dir, resume, steps, q):
+def _worker(rank, world, store_path, zero1, ckpt_dir, resume, steps, q,
+            dtype="fp32"):
     from scaffold.runtime import DistContext
     from scaffold.trainer import Trainer
 
@@ -42,9 +43,9 @@ def _worker(rank, world, store_path, zero1, ckpt_dir, resume, steps, q):
     ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
                       backend="gloo", device=torch.device("cpu"),
                       initialized=True)
-    t = Trainer(_cfg(zero1, ckpt_dir, resume, steps), ctx=ctx)
+    t = Trainer(_cfg(zero1, ckpt_dir, resume, steps, dtype), ctx=ctx)
     t.fit()
-    params = torch.cat([p.detach().flatten()
+    params = torch.cat([p.detach().flatten().float()
                         for p in t.raw_model.parameters()])
     own_master = sum(m.numel() for m in t.optimizer.master)
     total = sum(p.numel() for p in t.raw_model.parameters())
@@ -53,13 +54,14 @@ def _worker(rank, world, store_path, zero1, ckpt_dir, resume, steps, q):
     dist.destroy_process_group()
 
 
-def _run(zero1, tmp_path, tag, ckpt_dir="", resume="", steps=5):
+def _run(zero1, tmp_path, tag, ckpt_dir="", resume="", steps=5,
+         dtype="fp32"):
     ctxm = mp.get_context("spawn")
     q = ctxm.Queue()
     procs = [ctxm.Process(
         target=_worker,
         args=(r, WORLD, str(tmp_path / f"store_{tag}"), zero1, ckpt_dir,
-              resume, steps, q))
+              resume, steps, q, dtype))
         for r in range(WORLD)]
     for p in procs:
         p.start()
@@ -78,6 +80,16 @@ def test_zero1_matches_plain_ddp(tmp_path):
     assert own0 == total0
     assert 0 < own1 < total1
     # and training is exactly equivalent (fp32 CPU: bitwise)
+    assert np.array_equal(p0, p1)
+
+
+@pytest.mark.timeout(300)
+def test_zero1_matches_plain_ddp_bf16(tmp_path):
+    """bf16 working params (fp32 masters): still bitwise — owner-only
+    arithmetic equals everyone's, broadcasts are lossless."""
+    # bf16 eager CPU math is slow — 3 steps suffice for the claim
+    p0, _, _, _ = _run(False, tmp_path, "ddpb", steps=3, dtype="bf16")
+    p1, _, _, _ = _run(True, tmp_path, "z1b", steps=3, dtype="bf16")
     assert np.array_equal(p0, p1)
 
 
