@@ -172,3 +172,51 @@ def test_zero1_with_grad_accumulation(tmp_path):
         return out
 
     assert np.array_equal(run(False), run(True))
+
+
+def _lora_worker(rank, world, store_path, zero1, q):
+    from scaffold.runtime import DistContext
+    from scaffold.trainer import Trainer
+
+    store = dist.FileStore(store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
+                      backend="gloo", device=torch.device("cpu"),
+                      initialized=True)
+    cfg = _cfg(zero1, steps=4)
+    cfg.model.lora_r = 4
+    t = Trainer(cfg, ctx=ctx)
+    t.fit()
+    params = torch.cat([p.detach().flatten().float()
+                        for p in t.raw_model.parameters()])
+    n_opt = len(t.optimizer.params)
+    n_trainable = sum(1 for p in t.raw_model.parameters() if p.requires_grad)
+    if rank == 0:
+        q.put((params.numpy(), n_opt, n_trainable))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_zero1_composes_with_lora(tmp_path):
+    """ZeRO-1 over a LoRA model shards only the ADAPTER params and
+    stays bitwise-equivalent to plain DDP LoRA."""
+
+    def run(zero1, tag):
+        ctxm = mp.get_context("spawn")
+        q = ctxm.Queue()
+        procs = [ctxm.Process(target=_lora_worker,
+                              args=(r, WORLD, str(tmp_path / tag), zero1, q))
+                 for r in range(WORLD)]
+        for p in procs:
+            p.start()
+        out = q.get(timeout=200)
+        for p in procs:
+            p.join(timeout=60)
+            assert p.exitcode == 0
+        return out
+
+    p0, n0, tr0 = run(False, "l_ddp")
+    p1, n1, tr1 = run(True, "l_z1")
+    assert n0 == tr0                    # plain: optimizer covers all adapters
+    assert 0 < n1 < tr1                 # zero1: a strict shard of them
+    assert np.array_equal(p0, p1)
