@@ -578,6 +578,9 @@ class ContinuousBatcher:
             self._toks[b, self._slots[b]["count"]].copy_(nxt[b, 0])
         self._advance_host()
 
+    def _on_stop(self):
+        """Engine-thread shutdown hook (see _loop); default no-op."""
+
     def _graphed_step(self):
         """Capture once (after two warmup executions — generate_graphed's
         proven recipe), then replay; the graph reads the persistent
@@ -610,6 +613,11 @@ class ContinuousBatcher:
                        and not self._stop):
                     self._cv.wait()
                 if self._stop:
+                    # subclass hook (TP serving broadcasts STOP to the
+                    # follower ranks from HERE — the engine thread — so
+                    # it is ordered after any in-flight step's
+                    # collectives)
+                    self._on_stop()
                     # fail everything still queued or mid-decode so no
                     # waiter hangs on a worker that will never run again
                     err = RuntimeError("batcher shut down")
