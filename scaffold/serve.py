@@ -667,7 +667,7 @@ class ContinuousBatcher:
 def create_app(model: torch.nn.Module, max_batch: int = 8,
                window_s: float = DEFAULT_WINDOW_S, use_graph: bool = False,
                tokenizer=None, continuous: bool = False,
-               draft_model=None, draft_k: int = 4):
+               draft_model=None, draft_k: int = 4, worker=None):
     """FastAPI app around an inference worker.  `model` is an eval-mode
     LM from scaffold.models (tests inject a tiny CPU one).  Passing a
     `tokenizers.Tokenizer` adds the text endpoint /v1/completions.
@@ -677,13 +677,15 @@ def create_app(model: torch.nn.Module, max_batch: int = 8,
 
     from fastapi import FastAPI, HTTPException
 
-    if continuous:
-        worker = ContinuousBatcher(model.eval(), max_slots=max_batch,
-                                   use_graph=use_graph)
-    else:
-        worker = InferenceWorker(model.eval(), max_batch=max_batch,
-                                 window_s=window_s, use_graph=use_graph,
-                                 draft_model=draft_model, draft_k=draft_k)
+    if worker is None:
+        if continuous:
+            worker = ContinuousBatcher(model.eval(), max_slots=max_batch,
+                                       use_graph=use_graph)
+        else:
+            worker = InferenceWorker(model.eval(), max_batch=max_batch,
+                                     window_s=window_s, use_graph=use_graph,
+                                     draft_model=draft_model,
+                                     draft_k=draft_k)
 
     @asynccontextmanager
     async def _lifespan(app):

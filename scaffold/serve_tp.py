@@ -139,3 +139,67 @@ class _Follower(ContinuousBatcher):
 def run_follower(model, group=None, **kw):
     """Blocking follower loop for ranks > 0 (returns on STOP)."""
     _Follower(model, group=group, **kw).serve_commands()
+
+
+def main(argv=None):
+    """Launch TP serving under a multi-process launcher:
+
+        torchrun --nproc-per-node 2 -m scaffold.serve_tp \\
+            model.name=llama3_8b --port 8000
+
+    Rank 0 serves HTTP over its shard; the other ranks follow.  Uses
+    the same config overrides / --weights flags as scaffold.serve."""
+    import argparse
+    import os
+
+    from .config import load_config
+    from .dist.tp import shard_llama_tp
+    from .models import build_model
+
+    ap = argparse.ArgumentParser(description="tensor-parallel serving")
+    ap.add_argument("overrides", nargs="*")
+    ap.add_argument("--host", default="127.0.0.1")
+    ap.add_argument("--port", type=int, default=8000)
+    ap.add_argument("--max-slots", type=int, default=8)
+    ap.add_argument("--weights", default="")
+    args = ap.parse_args(argv)
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29531")
+    dist.init_process_group(backend, rank=rank, world_size=world)
+    try:
+        dev = "cuda" if torch.cuda.is_available() else "cpu"
+        if dev == "cuda":
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        dt = torch.bfloat16 if dev == "cuda" else torch.float32
+        cfg = load_config(None, args.overrides)
+        torch.manual_seed(cfg.seed)     # identical init on every rank
+        model = build_model(cfg)
+        if args.weights:
+            from .trainer import Trainer
+
+            Trainer._load_init_weights(model, args.weights)
+        model = model.to(dev, dt).eval()
+        shard_llama_tp(model, rank, world)
+        if world == 1 or rank == 0:
+            import uvicorn
+
+            from .serve import create_app
+
+            worker = (TpContinuousBatcher(model, max_slots=args.max_slots)
+                      if world > 1 else None)
+            app = create_app(model, continuous=True,
+                             max_batch=args.max_slots, worker=worker)
+            uvicorn.run(app, host=args.host, port=args.port,
+                        log_level="warning")
+        else:
+            run_follower(model, max_slots=args.max_slots)
+    finally:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

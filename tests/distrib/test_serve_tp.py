@@ -73,3 +73,24 @@ def test_tp_serving_matches_unsharded(tmp_path):
     for prompt, n, got in zip(PROMPTS, NNEW, results):
         ref = g.generate(torch.tensor([prompt], dtype=torch.int64), n)
         assert got == ref[0, len(prompt):].tolist(), (prompt, n)
+
+
+def test_create_app_accepts_injected_worker():
+    """The endpoints must serve the INJECTED worker (regression for the
+    closed-over-worker bug in the first TP CLI cut)."""
+    from starlette.testclient import TestClient
+
+    from scaffold.models import build_model
+    from scaffold.serve import ContinuousBatcher, create_app
+
+    cfg = load_config(None, OVER)
+    torch.manual_seed(5)
+    m = build_model(cfg).eval()
+    w = ContinuousBatcher(m, max_slots=2)
+    app = create_app(m, worker=w)
+    assert app.state.worker is w
+    with TestClient(app) as client:
+        r = client.post("/v1/generate", json={"tokens": [1, 2, 3],
+                                              "max_new_tokens": 4})
+        assert r.status_code == 200
+        assert w.stats["requests"] == 1   # THE injected worker served it
