@@ -45,11 +45,13 @@ class TpContinuousBatcher(ContinuousBatcher):
         self._tp_pgroup = group
         super().__init__(model, **kw)
 
-    def submit(self, tokens, max_new_tokens, temperature=0.0, **kw):
+    def submit(self, tokens, max_new_tokens, temperature=0.0,
+               top_k=0, top_p=0.0, **kw):
         if temperature > 0:
             raise ValueError("TP serving is greedy-only (per-rank "
                              "sampling RNG would diverge)")
-        return super().submit(tokens, max_new_tokens, temperature, **kw)
+        return super().submit(tokens, max_new_tokens, temperature,
+                              top_k, top_p, **kw)
 
     def _announce(self, payload):
         t = torch.tensor(payload, dtype=torch.int64)

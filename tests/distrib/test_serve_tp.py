@@ -94,3 +94,71 @@ def test_create_app_accepts_injected_worker():
                                               "max_new_tokens": 4})
         assert r.status_code == 200
         assert w.stats["requests"] == 1   # THE injected worker served it
+
+
+@pytest.mark.timeout(300)
+def test_tp_serving_cli_end_to_end(tmp_path):
+    """The real CLI under a 2-rank env rendezvous: rank 0 uvicorn
+    answers over TCP, SIGTERM drains cleanly and the STOP broadcast
+    releases the follower (its clean exit proves the protocol)."""
+    import os
+    import signal
+    import socket
+    import subprocess
+    import sys
+    import time
+
+    import httpx
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        http_port = s.getsockname()[1]
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        pg_port = s.getsockname()[1]
+
+    repo = os.path.dirname(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({"RANK": str(rank), "WORLD_SIZE": "2",
+                    "MASTER_ADDR": "127.0.0.1",
+                    "MASTER_PORT": str(pg_port)})
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "scaffold.serve_tp", *OVER,
+             "--port", str(http_port), "--max-slots", "2"],
+            cwd=repo, env=env))
+    try:
+        url = f"http://127.0.0.1:{http_port}"
+        deadline = time.time() + 90
+        while time.time() < deadline:
+            try:
+                if httpx.get(url + "/healthz", timeout=2).status_code == 200:
+                    break
+            except Exception:
+                time.sleep(0.3)
+        else:
+            raise AssertionError("TP server never became healthy")
+        r = httpx.post(url + "/v1/generate",
+                       json={"tokens": [1, 2, 3], "max_new_tokens": 6},
+                       timeout=60)
+        assert r.status_code == 200, r.text
+        got = r.json()["tokens"][0]
+        assert len(got) == 6
+
+        from scaffold.generate import Generator
+        from scaffold.models import build_model
+
+        cfg = load_config(None, OVER)
+        torch.manual_seed(cfg.seed)
+        full = build_model(cfg).eval()
+        ref = Generator(full).generate(
+            torch.tensor([[1, 2, 3]], dtype=torch.int64), 6)
+        assert got == ref[0, 3:].tolist()
+    finally:
+        procs[0].send_signal(signal.SIGTERM)
+        rc0 = procs[0].wait(timeout=60)
+        rc1 = procs[1].wait(timeout=60)   # released by the STOP broadcast
+    assert rc1 == 0, "follower did not exit cleanly (STOP never arrived)"
+    assert rc0 in (0, -signal.SIGTERM)
