@@ -162,3 +162,66 @@ def test_tp_serving_cli_end_to_end(tmp_path):
         rc1 = procs[1].wait(timeout=60)   # released by the STOP broadcast
     assert rc1 == 0, "follower did not exit cleanly (STOP never arrived)"
     assert rc0 in (0, -signal.SIGTERM)
+
+
+def _stream_worker(rank, world, store_path, q):
+    import queue as pyqueue
+
+    from scaffold.dist.tp import shard_llama_tp
+    from scaffold.models import build_model
+    from scaffold.serve_tp import TpContinuousBatcher, run_follower
+
+    store = dist.FileStore(store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    cfg = load_config(None, OVER)
+    torch.manual_seed(111)
+    m = build_model(cfg).eval()
+    shard_llama_tp(m, rank, world)
+    if rank == 0:
+        cb = TpContinuousBatcher(m, max_slots=2)
+        try:
+            sq = pyqueue.Queue()
+            r1 = cb.submit([1, 2, 3], 10, stream=sq, stream_every=3)
+            streamed = []
+            while True:
+                t = sq.get(timeout=120)
+                if t is None:
+                    break
+                streamed.append(t)
+            assert r1.event.wait(60) and r1.error is None
+            # prefix-cache reuse on a repeated prompt, through TP
+            r2 = cb.submit([1, 2, 3], 10)
+            assert r2.event.wait(120) and r2.error is None
+            q.put((streamed, r1.result, r2.result,
+                   cb.stats.get("prefix_hits", 0)))
+        finally:
+            cb.close()
+    else:
+        run_follower(m, max_slots=2)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp_serving_streaming_and_prefix_cache(tmp_path):
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_stream_worker,
+                          args=(r, 2, str(tmp_path / "stps"), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    streamed, r1, r2, hits = q.get(timeout=250)
+    for p in procs:
+        p.join(timeout=90)
+        assert p.exitcode == 0
+
+    cfg = load_config(None, OVER)
+    torch.manual_seed(111)
+    ref = Generator(build_model(cfg).eval()).generate(
+        torch.tensor([[1, 2, 3]], dtype=torch.int64), 10)[0, 3:].tolist()
+    assert streamed == r1 == r2 == ref
+    assert hits >= 1
