@@ -91,7 +91,8 @@ class RcclBucketComm(Comm):
 
     def __init__(self, device: torch.device, group=None):
         super().__init__(group)
-        assert self.is_distributed, "RcclBucketComm needs torch.distributed"
+        if not self.is_distributed:
+            raise RuntimeError("RcclBucketComm needs torch.distributed")
         from ..ops.backend import require_ext
 
         self._ext = require_ext()

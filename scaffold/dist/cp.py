@@ -151,8 +151,8 @@ def ulysses_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     q/k/v: (B, H|Hkv, S_local, D)."""
     if world <= 1:
         return ops.attention(q, k, v, causal)
-    assert q.shape[1] % world == 0 and k.shape[1] % world == 0, \
-        "ulysses: world must divide both H and Hkv"
+    if q.shape[1] % world or k.shape[1] % world:
+        raise ValueError("ulysses: world must divide both H and Hkv")
 
     from torch.distributed import nn as dist_nn
 

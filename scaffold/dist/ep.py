@@ -29,8 +29,10 @@ class ExpertParallelMoE(nn.Module):
     def __init__(self, d_model: int, d_ff: int, n_experts: int, world: int,
                  rank: int, group=None, dispatch: str = "alltoall"):
         super().__init__()
-        assert n_experts % world == 0
-        assert dispatch in ("alltoall", "dense")
+        if n_experts % world:
+            raise ValueError(f"n_experts {n_experts} not divisible by world {world}")
+        if dispatch not in ("alltoall", "dense"):
+            raise ValueError(f"unknown dispatch {dispatch!r}")
         self.world = world
         self.rank = rank
         self.group = group

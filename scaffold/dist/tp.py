@@ -80,7 +80,8 @@ class ColumnParallelLinear(nn.Module):
     def __init__(self, in_features: int, out_features: int, world: int,
                  rank: int, bias: bool = True, group=None):
         super().__init__()
-        assert out_features % world == 0, "TP: world must divide out_features"
+        if out_features % world:
+            raise ValueError("TP: world must divide out_features")
         self.in_features = in_features
         self.out_local = out_features // world
         self.world = world
@@ -112,7 +113,8 @@ class RowParallelLinear(nn.Module):
     def __init__(self, in_features: int, out_features: int, world: int,
                  rank: int, bias: bool = True, group=None):
         super().__init__()
-        assert in_features % world == 0, "TP: world must divide in_features"
+        if in_features % world:
+            raise ValueError("TP: world must divide in_features")
         self.in_local = in_features // world
         self.world = world
         self.rank = rank
@@ -173,8 +175,9 @@ def shard_llama_tp(model, rank: int, world: int, group=None):
         return shard_gpt2_tp(model, rank, world, group)
     for blk in model.blocks:
         m = blk.attn
-        assert m.n_heads % world == 0 and m.n_kv % world == 0, \
-            f"heads {m.n_heads}/{m.n_kv} not divisible by tp={world}"
+        if m.n_heads % world or m.n_kv % world:
+            raise ValueError(
+                f"heads {m.n_heads}/{m.n_kv} not divisible by tp={world}")
         hl = m.n_heads // world
         kl = m.n_kv // world
         d = m.head_dim
@@ -217,8 +220,9 @@ def shard_gpt2_tp(model, rank: int, world: int, group=None):
     (a partial+bias on every rank would count the bias world times)."""
     for blk in model.blocks:
         m = blk.attn
-        assert m.n_heads % world == 0, \
-            f"heads {m.n_heads} not divisible by tp={world}"
+        if m.n_heads % world:
+            raise ValueError(
+                f"heads {m.n_heads} not divisible by tp={world}")
         hl = m.n_heads // world
         d = m.head_dim
         dm = m.n_heads * d

@@ -98,7 +98,8 @@ class Generator:
     def __init__(self, model: torch.nn.Module):
         self.m = model
         self.is_llama = isinstance(model, Llama)
-        assert self.is_llama or isinstance(model, GPT2), type(model)
+        if not (self.is_llama or isinstance(model, GPT2)):
+            raise TypeError(f"Generator: unsupported model {type(model).__name__}")
 
     # ---- per-family block math with external KV ------------------------
     def _gpt2_block(self, blk, x, cache: KVCache, i: int, pos0: int):
@@ -348,7 +349,9 @@ class Generator:
         serving), so accepting m tokens per verify cuts target weight
         reads ~m-fold; the draft adds its own (much smaller) reads.
         `stats["accepted"]/["rounds"]` expose the acceptance rate."""
-        assert prompt.shape[0] == 1, "speculative decode is B=1"
+        if prompt.shape[0] != 1:
+            raise ValueError("speculative decode is B=1 (per-row "
+                             "acceptance lengths diverge in a batch)")
         n_layers = len(self.m.blocks)
         attn0 = self.m.blocks[0].attn
         n_kv = attn0.n_kv if self.is_llama else attn0.n_heads
