@@ -30,7 +30,9 @@ class LayerNorm(nn.Module):
 class CausalSelfAttention(nn.Module):
     def __init__(self, d_model: int, n_heads: int):
         super().__init__()
-        assert d_model % n_heads == 0
+        if d_model % n_heads:
+            raise ValueError(f"d_model {d_model} not divisible by "
+                             f"n_heads {n_heads}")
         self.n_heads = n_heads
         self.head_dim = d_model // n_heads
         self.qkv = nn.Linear(d_model, 3 * d_model)

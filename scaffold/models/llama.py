@@ -30,6 +30,10 @@ class LlamaAttention(nn.Module):
 
     def __init__(self, d_model: int, n_heads: int, n_kv_heads: int):
         super().__init__()
+        if d_model % n_heads or n_heads % n_kv_heads:
+            raise ValueError(
+                f"d_model {d_model} / n_heads {n_heads} / n_kv "
+                f"{n_kv_heads}: need n_heads | d_model and n_kv | n_heads")
         self.n_heads = n_heads
         self.n_kv = n_kv_heads
         self.head_dim = d_model // n_heads
