@@ -33,7 +33,8 @@ def _named_state_tensors(model, optimizer) -> List[Tuple[str, torch.Tensor]]:
     """Flat name->tensor list: model params+buffers, optimizer masters/moments."""
     out: List[Tuple[str, torch.Tensor]] = []
     for name, p in model.named_parameters():
-        out.append((f"model.{name}", p.data))
+        # rank-pinned params (EP experts) carry global identities
+        out.append((f"model.{getattr(p, '_ckpt_name', name)}", p.data))
     for name, b in model.named_buffers():
         out.append((f"buffer.{name}", b.data))
     if optimizer is not None:
@@ -110,6 +111,21 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
     if zero1:
         pinned = [(n, t) for n, t in named if n.startswith("opt.")]
         named = [(n, t) for n, t in named if not n.startswith("opt.")]
+    # EP experts: rank-LOCAL params (and their optimizer state) keyed
+    # by global expert id — pin them to this rank; the manifest covers
+    # every rank's experts via the model's _pinned_manifest map
+    ep_map = getattr(model, "_pinned_manifest", None)
+    if ep_map:
+        ep_names = set(ep_map)
+
+        def _pin_base(n):
+            base = n.split(".", 1)[1] if n.startswith("model.") else \
+                n.split(".", 2)[2] if n.startswith("opt.") else None
+            return base if base in ep_names else None
+
+        more = [(n, t) for n, t in named if _pin_base(n)]
+        pinned += more
+        named = [(n, t) for n, t in named if not _pin_base(n)]
     # dedupe aliased storages (fp32 params alias their optimizer master;
     # tied embeddings alias lm_head) — safetensors forbids shared tensors
     seen_ptr: Dict[int, str] = {}
@@ -143,6 +159,17 @@ def save_sharded(ckpt_dir: str, *, step: int, model, optimizer, data, ctx,
             for kind in ("master", "exp_avg", "exp_avg_sq"):
                 tensor_meta[f"opt.{kind}.{pname}"] = {
                     "shard": r, "shape": shapes[pname], "dtype": "float32"}
+    if ep_map:
+        pdtype = str(next(model.parameters()).dtype).replace("torch.", "")
+        for base, (owner_r, shape) in ep_map.items():
+            tensor_meta[f"model.{base}"] = {
+                "shard": owner_r, "shape": shape, "dtype": pdtype}
+            if optimizer is not None:
+                for kind in ("master", "exp_avg", "exp_avg_sq"):
+                    tensor_meta[f"opt.{kind}.{base}"] = {
+                        "shard": owner_r, "shape": shape,
+                        "dtype": "float32"}
+    if pinned:
         taken = {t.untyped_storage().data_ptr() for t in shard.values()}
         for name, t in pinned:
             t = t.detach()
@@ -207,6 +234,7 @@ def _dest_tensors(model, optimizer, manifest) -> List[Tuple[str, torch.Tensor]]:
     out: List[Tuple[str, torch.Tensor]] = []
     pnames = []
     for name, p in model.named_parameters():
+        name = getattr(p, "_ckpt_name", name)   # EP experts: global ids
         pnames.append(name)
         out.append((f"model.{name}", p.data))
     for name, b in model.named_buffers():
@@ -265,6 +293,14 @@ def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = 
     saved_world = m["world_size"]
 
     def resolve(name: str) -> str:
+        if name not in m["tensors"] and ".moe." in name:
+            # reshard across EP worlds: a local-expert model asks for
+            # slot index i; an EP checkpoint stored global ids gN.
+            # With world=1 (or replicated experts) slot i IS global i.
+            head, idx = name.rsplit(".", 1)
+            alt = f"{head}.g{idx}"
+            if alt in m["tensors"]:
+                name = alt
         info = m["tensors"][name]
         return info["alias"] if "alias" in info else name
 
@@ -289,6 +325,16 @@ def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = 
             # alias structure is identical on every rank, so the
             # broadcast sequences stay aligned.
             zero1 = getattr(optimizer, "shard_owners", None) is not None
+            ep_bases = set(getattr(model, "_pinned_manifest", {}) or {})
+
+            def _rank_pinned(n):
+                if zero1 and n.startswith("opt.") and not ep_bases:
+                    return True
+                base = (n.split(".", 1)[1] if n.startswith("model.")
+                        else n.split(".", 2)[2] if n.startswith("opt.")
+                        else None)
+                return (zero1 and n.startswith("opt.")) or base in ep_bases
+
             seen_storage = set()
             for name, dst in _dest_tensors(model, optimizer, m):
                 rname = resolve(name)
@@ -297,10 +343,11 @@ def load_sharded(ckpt_dir: str, *, model, optimizer, data, ctx, cfg_hash: str = 
                 if ptr in seen_storage:
                     continue
                 seen_storage.add(ptr)
-                if zero1 and name.startswith("opt."):
-                    # rank-pinned state: same world => it sits in OUR
-                    # shard; other ranks don't even have a destination,
-                    # so no collective may be issued for it
+                if _rank_pinned(name):
+                    # rank-pinned state (zero1 opt shards, EP experts):
+                    # same world => it sits in OUR shard; other ranks
+                    # don't even have a destination, so no collective
+                    # may be issued for it
                     stager.copy_(dst, own.get_tensor(rname))
                     continue
                 if owner == ctx.rank:

@@ -108,6 +108,12 @@ class DistConfig:
     # updated working params.  Cuts optimizer memory by ~world_size
     # (Llama-8B: 96 GB -> 12 GB/rank at 8 GPUs).
     zero1: bool = False
+    # expert parallelism for the MoE family (model.n_experts > 0):
+    # shard the experts across the WORLD (n_experts % world == 0),
+    # routed tokens exchanged by all_to_all; backbone/gate stay
+    # data-parallel.  Mutually exclusive with zero1 (expert params are
+    # rank-local, so a name-keyed optimizer partition would collide).
+    ep_moe: bool = False
 
 
 @dataclass

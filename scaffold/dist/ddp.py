@@ -110,12 +110,22 @@ class BucketedDDP(nn.Module):
         self._sync_enabled = True
         self._require_finish = False
 
-        self.trainable = [p for p in module.parameters() if p.requires_grad]
+        # params marked _no_ddp are RANK-LOCAL by design (expert-parallel
+        # MoE shards: each rank owns different experts) — they join
+        # neither the buckets nor the wrap-time broadcast
+        self.trainable = [p for p in module.parameters()
+                          if p.requires_grad
+                          and not getattr(p, "_no_ddp", False)]
+        self.local_params = [p for p in module.parameters()
+                             if p.requires_grad
+                             and getattr(p, "_no_ddp", False)]
 
         # 1) rank-0 params/buffers win (SURVEY §3.1 broadcast at wrap)
         if self.comm.is_distributed:
             with torch.no_grad():
                 for t in list(module.parameters()) + list(module.buffers()):
+                    if getattr(t, "_no_ddp", False):
+                        continue
                     if t.is_floating_point() or t.dtype in (torch.int64, torch.int32):
                         self.comm.broadcast(t.data, src=0)
 
@@ -219,9 +229,15 @@ class BucketedDDP(nn.Module):
         self._next_launch = 0
 
     def zero_grad(self) -> None:
-        """Zero the flat buffers (cheap contiguous memset per bucket)."""
+        """Zero the flat buffers (cheap contiguous memset per bucket).
+        Rank-local params (_no_ddp, e.g. EP experts) own ordinary .grad
+        tensors — drop them too, or they silently ACCUMULATE across
+        steps (found via a resume-divergence hunt: the accumulated
+        history is exactly what a checkpoint cannot restore)."""
         for b in self.buckets:
             b.flat.zero_()
+        for p in self.local_params:
+            p.grad = None
 
     @contextlib.contextmanager
     def no_sync(self):

@@ -27,7 +27,8 @@ class ExpertParallelMoE(nn.Module):
     """Top-1 MoE FFN; `n_experts` must be divisible by `world`."""
 
     def __init__(self, d_model: int, d_ff: int, n_experts: int, world: int,
-                 rank: int, group=None, dispatch: str = "alltoall"):
+                 rank: int, group=None, dispatch: str = "alltoall",
+                 init_seed: int = 0):
         super().__init__()
         if n_experts % world:
             raise ValueError(f"n_experts {n_experts} not divisible by world {world}")
@@ -41,12 +42,21 @@ class ExpertParallelMoE(nn.Module):
         self.e_local = n_experts // world
         self.e0 = rank * self.e_local
         self.gate = nn.Linear(d_model, n_experts, bias=False)  # replicated
+        # expert weights draw from PER-GLOBAL-EXPERT generators, not the
+        # ambient (rank-seeded) stream: expert g is bit-identical no
+        # matter which rank owns it or what the world size is, so an
+        # EP-sharded model == the single-process model weight-for-weight
+        # (and the ambient stream position stays world-invariant too)
+        def _init(shape, gid):
+            g = torch.Generator().manual_seed(init_seed * 1000003 + gid)
+            return nn.Parameter(torch.randn(*shape, generator=g) * 0.02)
+
         self.w_up: nn.ParameterList = nn.ParameterList(
-            nn.Parameter(torch.randn(d_ff, d_model) * 0.02)
-            for _ in range(self.e_local))
+            _init((d_ff, d_model), self.e0 + i)
+            for i in range(self.e_local))
         self.w_down: nn.ParameterList = nn.ParameterList(
-            nn.Parameter(torch.randn(d_model, d_ff) * 0.02)
-            for _ in range(self.e_local))
+            _init((d_model, d_ff), 10**6 + self.e0 + i)
+            for i in range(self.e_local))
 
     def _zero_tie(self, i_local: int) -> torch.Tensor:
         # an expert that received no tokens this step must still appear
@@ -109,7 +119,14 @@ class ExpertParallelMoE(nn.Module):
                                output_split_sizes=recv_l,
                                input_split_sizes=send_l, group=self.group)
 
-        y_recv = torch.zeros_like(x_recv)
+        # x_recv * 0, NOT zeros_like: when ZERO tokens land on this
+        # rank's experts there is no index_add, and a graph-free zero
+        # would let autograd PRUNE the feature-a2a backward on THIS
+        # rank while the peers (whose experts did get tokens) still run
+        # it — divergent collective sequences that corrupt/abort the
+        # group (found via per-rank collective tracing when a collapsed
+        # gate routed everything to one rank)
+        y_recv = x_recv * 0.0
         for i in range(self.e_local):
             sel = (e_recv == self.e0 + i).nonzero(as_tuple=True)[0]
             if sel.numel() == 0:
@@ -139,7 +156,7 @@ class ExpertParallelMoE(nn.Module):
         probs = torch.softmax(self.gate(x_all), dim=-1)      # (T_all, E)
         top_p, top_e = probs.max(dim=-1)                     # top-1 routing
 
-        y_partial = torch.zeros_like(x_all)
+        y_partial = x_all * 0.0   # keep the graph tie (see alltoall path)
         for i in range(self.e_local):
             sel = (top_e == self.e0 + i).nonzero(as_tuple=True)[0]
             if sel.numel() == 0:

@@ -52,17 +52,28 @@ class CausalSelfAttention(nn.Module):
 
 class Block(nn.Module):
     def __init__(self, d_model: int, n_heads: int, d_ff: int, eps: float,
-                 n_experts: int = 0):
+                 n_experts: int = 0, ep=None, init_seed: int = 0):
         super().__init__()
         self.ln1 = LayerNorm(d_model, eps)
         self.attn = CausalSelfAttention(d_model, n_heads)
         self.ln2 = LayerNorm(d_model, eps)
         if n_experts > 0:
-            # top-1-routed MoE FFN (model.n_experts); experts local to
-            # this rank — EP sharding composes via scaffold.dist.ep
+            # top-1-routed MoE FFN (model.n_experts).  Without an ep
+            # context the experts are local; with one (dist.ep_moe)
+            # they shard across the group and the routed tokens travel
+            # by all_to_all — expert params are marked _no_ddp so the
+            # bucket engine leaves them rank-local.
             from ..dist.ep import ExpertParallelMoE
-            self.moe = ExpertParallelMoE(d_model, d_ff, n_experts,
-                                         world=1, rank=0, dispatch="dense")
+            world, rank, group = (ep if ep is not None else (1, 0, None))
+            self.moe = ExpertParallelMoE(
+                d_model, d_ff, n_experts, world=world, rank=rank,
+                group=group,
+                dispatch="alltoall" if world > 1 else "dense",
+                init_seed=init_seed)
+            if world > 1:
+                for plist in (self.moe.w_up, self.moe.w_down):
+                    for p_ in plist:
+                        p_._no_ddp = True
         else:
             self.moe = None
             self.fc = nn.Linear(d_model, d_ff)
@@ -89,10 +100,34 @@ class GPT2(nn.Module):
         self.act_ckpt = getattr(m, "act_ckpt", False)
         self.wte = nn.Embedding(m.vocab_size, m.d_model)
         self.wpe = nn.Embedding(m.max_seq_len, m.d_model)
+        ep = getattr(cfg, "_ep_ctx", None)
         self.blocks = nn.ModuleList(
             Block(m.d_model, m.n_heads, d_ff, m.norm_eps,
-                  getattr(m, "n_experts", 0)) for _ in range(m.n_layers)
+                  getattr(m, "n_experts", 0), ep,
+                  init_seed=getattr(cfg, "seed", 0) * 4099 + li)
+            for li in range(m.n_layers)
         )
+        if ep is not None and ep[0] > 1:  # (world, rank, group)
+            # checkpoint identity for rank-local experts: LOCAL slot i
+            # holds GLOBAL expert e0+i, so shards/manifest key them by
+            # global id (same name never maps to different content on
+            # different ranks); _pinned_manifest lets rank 0 write a
+            # complete manifest for experts it does not hold
+            world, _rank, _group = ep
+            self._pinned_manifest = {}
+            for li, blk in enumerate(self.blocks):
+                moe = blk.moe
+                for kind, plist in (("w_up", moe.w_up),
+                                    ("w_down", moe.w_down)):
+                    for i, p_ in enumerate(plist):
+                        gid = moe.e0 + i
+                        p_._ckpt_name = f"blocks.{li}.moe.{kind}.g{gid}"
+                        p_._ckpt_owner = _rank
+                    e_local = moe.e_local
+                    for gid in range(moe.n_experts):
+                        self._pinned_manifest[
+                            f"blocks.{li}.moe.{kind}.g{gid}"] = (
+                            gid // e_local, list(plist[0].shape))
         self.ln_f = LayerNorm(m.d_model, m.norm_eps)
         self.lm_head = nn.Linear(m.d_model, m.vocab_size, bias=False)
         if m.tie_embeddings:

@@ -159,8 +159,8 @@ def build_optimizer(cfg, model: torch.nn.Module, ctx=None) -> MasterAdamW:
         # explicit trainable identities: with frozen params (LoRA) the
         # optimizer's positional lists no longer align with
         # model.named_parameters(), so the checkpoint layer needs names
-        trainable = [(n, p) for n, p in model.named_parameters()
-                     if p.requires_grad]
+        trainable = [(getattr(p, "_ckpt_name", n), p)
+                     for n, p in model.named_parameters() if p.requires_grad]
         return MasterAdamW(
             [p for _, p in trainable],
             lr=o.lr,

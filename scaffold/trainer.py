@@ -48,6 +48,21 @@ class Trainer:
 
         set_seed(cfg.seed, self.ctx.rank)
 
+        if getattr(cfg.dist, "ep_moe", False) and self.ctx.world_size > 1:
+            if getattr(cfg.dist, "zero1", False):
+                raise ValueError("dist.ep_moe and dist.zero1 are mutually "
+                                 "exclusive (expert params are rank-local)")
+            # expert-parallel MoE: build_model shards experts by rank.
+            # The routed-token all_to_alls run during BACKWARD and would
+            # interleave with the bucket all-reduces in a readiness-
+            # dependent (rank-INconsistent) order — a collective
+            # mismatch on a single communicator (gloo detects it; RCCL
+            # would corrupt).  A dedicated EP group gives the a2a its
+            # own communicator, making the two streams order-independent.
+            import torch.distributed as dist
+
+            ep_group = dist.new_group(list(range(self.ctx.world_size)))
+            cfg._ep_ctx = (self.ctx.world_size, self.ctx.rank, ep_group)
         raw_model = model if model is not None else build_model(cfg)
         raw_model = raw_model.to(self.device)
         if self.dtype is not torch.float32:
