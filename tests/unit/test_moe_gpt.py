@@ -262,3 +262,44 @@ def test_ep_moe_ckpt_resume_and_reshard(tmp_path):
     saved_g3 = shards["model.blocks.0.moe.w_up.g3"]
     assert torch.equal(t.raw_model.blocks[0].moe.w_up[3].detach(),
                        saved_g3)
+
+
+def _ep_ckpt_act_worker(rank, world, store_path, q):
+    import torch.distributed as dist
+
+    from scaffold.runtime import DistContext
+    from scaffold.trainer import Trainer
+
+    store = dist.FileStore(store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
+                      backend="gloo", device=torch.device("cpu"),
+                      initialized=True)
+    cfg = _cfg(**{"dist.ep_moe": "true", "model.act_ckpt": "true",
+                  "steps": 4})
+    t = Trainer(cfg, ctx=ctx)
+    t.fit()
+    if rank == 0:
+        q.put(t.step)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ep_moe_composes_with_act_ckpt(tmp_path):
+    """Recompute-in-backward re-issues the EP a2a collectives at
+    deterministic graph points: the combination must run to completion
+    (ordering is the risk, not math — exactness is covered above)."""
+    import torch.multiprocessing as mp
+
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_ep_ckpt_act_worker,
+                          args=(r, 2, str(tmp_path / "epac"), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    assert q.get(timeout=200) == 4
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
