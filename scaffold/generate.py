@@ -124,6 +124,9 @@ class Generator:
                         + blk.proj.bias)
         x = x + F.linear(o2, m.proj.weight, m.proj.bias)
         h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+        if getattr(blk, "moe", None) is not None:
+            B2, S2, C2 = h.shape
+            return x + blk.moe(h.reshape(B2 * S2, C2)).view(B2, S2, C2)
         h = F.gelu(F.linear(h, blk.fc.weight, blk.fc.bias), approximate="tanh")
         return x + F.linear(h, blk.proj.weight, blk.proj.bias)
 
@@ -309,6 +312,9 @@ class Generator:
         x = x + self._lin1(o.transpose(1, 2).reshape(B, 1, -1),
                            m.proj.weight, m.proj.bias)
         h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+        if getattr(blk, "moe", None) is not None:
+            B2 = h.shape[0]
+            return x + blk.moe(h.reshape(B2, -1)).view(B2, 1, -1)
         h = F.gelu(self._lin1(h, blk.fc.weight, blk.fc.bias),
                    approximate="tanh")
         return x + self._lin1(h, blk.proj.weight, blk.proj.bias)
@@ -495,6 +501,9 @@ class Generator:
                         + blk.proj.bias)
         x = x + self._lin1(o2, m.proj.weight, m.proj.bias)
         h = ops.layer_norm(x, blk.ln2.weight, blk.ln2.bias, blk.ln2.eps)
+        if getattr(blk, "moe", None) is not None:
+            B2 = h.shape[0]
+            return x + blk.moe(h.reshape(B2, -1)).view(B2, 1, -1)
         h = F.gelu(self._lin1(h, blk.fc.weight, blk.fc.bias),
                    approximate="tanh")
         return x + self._lin1(h, blk.proj.weight, blk.proj.bias)
