@@ -219,6 +219,10 @@ def shard_gpt2_tp(model, rank: int, world: int, group=None):
     the Generator adds them post-reduction when the markers are set
     (a partial+bias on every rank would count the bias world times)."""
     for blk in model.blocks:
+        if getattr(blk, "moe", None) is not None:
+            raise NotImplementedError(
+                "TP sharding of MoE blocks is not supported — shard the "
+                "EXPERTS instead (dist.ep_moe / scaffold.dist.ep)")
         m = blk.attn
         if m.n_heads % world:
             raise ValueError(

@@ -348,3 +348,28 @@ def test_moe_model_decodes():
         assert r.result == out[0, 5:].tolist()
     finally:
         cb.close()
+
+
+def test_speculative_on_moe_model():
+    import torch
+
+    from scaffold.config import load_config
+    from scaffold.generate import Generator
+    from scaffold.models import build_model
+
+    over = [
+        "model.name=gpt2", "model.n_layers=2", "model.n_heads=2",
+        "model.d_model=32", "model.vocab_size=64", "model.max_seq_len=64",
+        "model.n_experts=4",
+    ]
+    torch.manual_seed(131)
+    target = Generator(build_model(load_config(None, over)).eval())
+    torch.manual_seed(132)
+    draft = Generator(build_model(load_config(None, [
+        "model.name=gpt2", "model.n_layers=1", "model.n_heads=2",
+        "model.d_model=32", "model.vocab_size=64", "model.max_seq_len=64",
+    ])).eval())
+    prompt = torch.randint(0, 64, (1, 5))
+    ref = target.generate(prompt, 14)
+    out = target.generate_speculative(prompt, 14, draft, k=3)
+    assert torch.equal(out, ref)
