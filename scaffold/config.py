@@ -55,6 +55,8 @@ class ModelConfig:
     # N local experts (0 = dense).  Expert-parallel sharding of the same
     # layer across ranks is scaffold.dist.ep.ExpertParallelMoE
     n_experts: int = 0
+    # experts per token (Mixtral-style weighted mixing when > 1)
+    moe_top_k: int = 1
     # LoRA fine-tuning: rank > 0 freezes the model and attaches
     # low-rank adapters to every Linear except lm_head (scaffold/lora.py)
     lora_r: int = 0

@@ -28,12 +28,15 @@ class ExpertParallelMoE(nn.Module):
 
     def __init__(self, d_model: int, d_ff: int, n_experts: int, world: int,
                  rank: int, group=None, dispatch: str = "alltoall",
-                 init_seed: int = 0):
+                 init_seed: int = 0, top_k: int = 1):
         super().__init__()
         if n_experts % world:
             raise ValueError(f"n_experts {n_experts} not divisible by world {world}")
         if dispatch not in ("alltoall", "dense"):
             raise ValueError(f"unknown dispatch {dispatch!r}")
+        if not 1 <= top_k <= n_experts:
+            raise ValueError(f"top_k {top_k} out of range 1..{n_experts}")
+        self.top_k = top_k
         self.world = world
         self.rank = rank
         self.group = group
@@ -94,13 +97,22 @@ class ExpertParallelMoE(nn.Module):
         from torch.distributed import nn as dist_nn
 
         probs = torch.softmax(self.gate(x_local), dim=-1)
-        top_p, top_e = probs.max(dim=-1)                     # (T,)
-        dest = top_e // self.e_local                         # owner rank
+        # top-k: each token travels to its k chosen experts (k-fold
+        # transport — the standard cost of Mixtral-style routing); the
+        # returned outputs are mixed with renormalized gate weights
+        top_p, top_e = probs.topk(self.top_k, dim=-1)        # (T, k)
+        top_p = top_p / top_p.sum(dim=-1, keepdim=True)
+        T = x_local.shape[0]
+        k = self.top_k
+        flat_e = top_e.reshape(-1)                           # (T*k,)
+        flat_p = top_p.reshape(-1)
+        src_tok = (torch.arange(T * k, device=x_local.device) // k)
+        dest = flat_e // self.e_local                        # owner rank
 
         order = torch.argsort(dest, stable=True)
         inv = torch.argsort(order)
-        x_s = x_local[order]
-        e_s = top_e[order]
+        x_s = x_local[src_tok[order]]
+        e_s = flat_e[order]
 
         send_counts = torch.bincount(dest, minlength=self.world)
         recv_counts = torch.empty_like(send_counts)
@@ -138,7 +150,8 @@ class ExpertParallelMoE(nn.Module):
         y_back = dist_nn.functional.all_to_all_single(
             y_back, y_recv.contiguous(), output_split_sizes=send_l,
             input_split_sizes=recv_l, group=self.group)
-        return top_p.unsqueeze(-1) * y_back[inv]
+        mixed = flat_p.unsqueeze(-1) * y_back[inv]           # (T*k, d)
+        return mixed.view(T, k, -1).sum(dim=1)
 
     def _forward_dense(self, x_local: torch.Tensor) -> torch.Tensor:
         import torch.distributed as dist
@@ -154,16 +167,20 @@ class ExpertParallelMoE(nn.Module):
             x_all = x_local
 
         probs = torch.softmax(self.gate(x_all), dim=-1)      # (T_all, E)
-        top_p, top_e = probs.max(dim=-1)                     # top-1 routing
+        # top-k routing: each token mixes its k best experts' outputs,
+        # weighted by their (renormalized) gate probabilities
+        top_p, top_e = probs.topk(self.top_k, dim=-1)        # (T_all, k)
+        top_p = top_p / top_p.sum(dim=-1, keepdim=True)
 
         y_partial = x_all * 0.0   # keep the graph tie (see alltoall path)
         for i in range(self.e_local):
-            sel = (top_e == self.e0 + i).nonzero(as_tuple=True)[0]
-            if sel.numel() == 0:
+            sel_t, sel_k = (top_e == self.e0 + i).nonzero(as_tuple=True)
+            if sel_t.numel() == 0:
                 y_partial = y_partial + self._zero_tie(i)
                 continue
-            out = self._expert(i, x_all[sel]) * top_p[sel].unsqueeze(-1)
-            y_partial = y_partial.index_add(0, sel, out)
+            out = self._expert(i, x_all[sel_t]) \
+                * top_p[sel_t, sel_k].unsqueeze(-1)
+            y_partial = y_partial.index_add(0, sel_t, out)
 
         if self.world > 1 and dist.is_initialized():
             from torch.distributed import nn as dist_nn

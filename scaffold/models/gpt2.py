@@ -52,7 +52,8 @@ class CausalSelfAttention(nn.Module):
 
 class Block(nn.Module):
     def __init__(self, d_model: int, n_heads: int, d_ff: int, eps: float,
-                 n_experts: int = 0, ep=None, init_seed: int = 0):
+                 n_experts: int = 0, ep=None, init_seed: int = 0,
+                 moe_top_k: int = 1):
         super().__init__()
         self.ln1 = LayerNorm(d_model, eps)
         self.attn = CausalSelfAttention(d_model, n_heads)
@@ -69,7 +70,7 @@ class Block(nn.Module):
                 d_model, d_ff, n_experts, world=world, rank=rank,
                 group=group,
                 dispatch="alltoall" if world > 1 else "dense",
-                init_seed=init_seed)
+                init_seed=init_seed, top_k=moe_top_k)
             if world > 1:
                 for plist in (self.moe.w_up, self.moe.w_down):
                     for p_ in plist:
@@ -104,7 +105,8 @@ class GPT2(nn.Module):
         self.blocks = nn.ModuleList(
             Block(m.d_model, m.n_heads, d_ff, m.norm_eps,
                   getattr(m, "n_experts", 0), ep,
-                  init_seed=getattr(cfg, "seed", 0) * 4099 + li)
+                  init_seed=getattr(cfg, "seed", 0) * 4099 + li,
+                  moe_top_k=getattr(m, "moe_top_k", 1))
             for li in range(m.n_layers)
         )
         if ep is not None and ep[0] > 1:  # (world, rank, group)

@@ -27,12 +27,13 @@ def _seed_weights(moe, base):
     return X, dy
 
 
-def _worker(rank, world, store_path, q_ret, dispatch="dense"):
+def _worker(rank, world, store_path, q_ret, dispatch="dense", top_k=1):
     from scaffold.dist.ep import ExpertParallelMoE
 
     store = dist.FileStore(store_path, world)
     dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
-    moe = ExpertParallelMoE(DM, DF, E, world, rank, dispatch=dispatch)
+    moe = ExpertParallelMoE(DM, DF, E, world, rank, dispatch=dispatch,
+                            top_k=top_k)
     X, dy = _seed_weights(moe, rank * (E // world))
     xl = X[rank * T:(rank + 1) * T].clone().requires_grad_(True)
     y = moe(xl)
@@ -47,13 +48,16 @@ def _worker(rank, world, store_path, q_ret, dispatch="dense"):
 
 @pytest.mark.timeout(180)
 @pytest.mark.parametrize("dispatch", ["dense", "alltoall"])
-def test_ep_moe_matches_full(tmp_path, dispatch):
+@pytest.mark.parametrize("top_k", [1, 2])
+def test_ep_moe_matches_full(tmp_path, dispatch, top_k):
     from scaffold.dist.ep import ExpertParallelMoE
 
     ctxmp = mp.get_context("spawn")
     ret = ctxmp.Queue()
-    procs = [ctxmp.Process(target=_worker,
-                           args=(r, WORLD, str(tmp_path / f"fs{dispatch}"), ret, dispatch))
+    procs = [ctxmp.Process(
+        target=_worker,
+        args=(r, WORLD, str(tmp_path / f"fs{dispatch}{top_k}"), ret,
+              dispatch, top_k))
              for r in range(WORLD)]
     for p in procs:
         p.start()
@@ -66,7 +70,7 @@ def test_ep_moe_matches_full(tmp_path, dispatch):
         assert p.exitcode == 0
 
     # single-process reference: world=1 holds ALL experts
-    ref = ExpertParallelMoE(DM, DF, E, 1, 0)
+    ref = ExpertParallelMoE(DM, DF, E, 1, 0, top_k=top_k)
     X, dy = _seed_weights(ref, 0)
     x = X.clone().requires_grad_(True)
     y_ref = ref(x)

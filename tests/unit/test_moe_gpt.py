@@ -303,3 +303,34 @@ def test_ep_moe_composes_with_act_ckpt(tmp_path):
     for p in procs:
         p.join(timeout=60)
         assert p.exitcode == 0
+
+
+def test_top2_moe_gpt_trains_and_decodes():
+    """Mixtral-style top-2 routing through the model: trains, and the
+    KV-cached decode equals recompute-from-scratch."""
+    import torch as _t
+
+    from scaffold.generate import Generator, KVCache
+    from scaffold.models import build_model
+    from scaffold.trainer import Trainer
+
+    t = Trainer(_cfg(steps=25, **{"model.moe_top_k": 2}))
+    it = iter(t.data)
+    losses = [t._train_step(it, 1) for _ in range(25)]
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert min(losses[-5:]) < losses[0] - 0.4, (losses[0], losses[-5:])
+    assert t.raw_model.blocks[0].moe.top_k == 2
+
+    m = t.raw_model.eval()
+    g = Generator(m)
+    prompt = torch.randint(0, 64, (1, 5))
+    out = g.generate(prompt, 8)
+    toks = prompt.clone()
+    with torch.no_grad():
+        for _ in range(8):
+            cache = KVCache(2, 1, 2, toks.shape[1], 16, toks.device,
+                            _t.float32)
+            logits = g._forward_tokens(toks, cache, 0)
+            toks = _t.cat([toks, logits.argmax(-1, keepdim=True)], dim=1)
+    assert torch.equal(out, toks)
+    m.train()
