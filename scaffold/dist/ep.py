@@ -28,7 +28,7 @@ class ExpertParallelMoE(nn.Module):
 
     def __init__(self, d_model: int, d_ff: int, n_experts: int, world: int,
                  rank: int, group=None, dispatch: str = "alltoall",
-                 init_seed: int = 0, top_k: int = 1):
+                 init_seed: int = 0, top_k: int = 1, act: str = "gelu"):
         super().__init__()
         if n_experts % world:
             raise ValueError(f"n_experts {n_experts} not divisible by world {world}")
@@ -36,7 +36,10 @@ class ExpertParallelMoE(nn.Module):
             raise ValueError(f"unknown dispatch {dispatch!r}")
         if not 1 <= top_k <= n_experts:
             raise ValueError(f"top_k {top_k} out of range 1..{n_experts}")
+        if act not in ("gelu", "swiglu"):
+            raise ValueError(f"unknown expert act {act!r}")
         self.top_k = top_k
+        self.act = act
         self.world = world
         self.rank = rank
         self.group = group
@@ -54,8 +57,11 @@ class ExpertParallelMoE(nn.Module):
             g = torch.Generator().manual_seed(init_seed * 1000003 + gid)
             return nn.Parameter(torch.randn(*shape, generator=g) * 0.02)
 
+        # swiglu experts pack gate|up rows in one matrix (the same
+        # layout as the dense Llama block's w_gateup)
+        up_rows = 2 * d_ff if act == "swiglu" else d_ff
         self.w_up: nn.ParameterList = nn.ParameterList(
-            _init((d_ff, d_model), self.e0 + i)
+            _init((up_rows, d_model), self.e0 + i)
             for i in range(self.e_local))
         self.w_down: nn.ParameterList = nn.ParameterList(
             _init((d_model, d_ff), 10**6 + self.e0 + i)
@@ -70,8 +76,13 @@ class ExpertParallelMoE(nn.Module):
                       + self.w_down[i_local].flatten()[0])
 
     def _expert(self, i_local: int, x: torch.Tensor) -> torch.Tensor:
-        h = torch.nn.functional.gelu(x @ self.w_up[i_local].t(),
-                                     approximate="tanh")
+        if self.act == "swiglu":
+            gu = x @ self.w_up[i_local].t()
+            F_ = gu.shape[-1] // 2
+            h = torch.nn.functional.silu(gu[..., :F_]) * gu[..., F_:]
+        else:
+            h = torch.nn.functional.gelu(x @ self.w_up[i_local].t(),
+                                         approximate="tanh")
         return h @ self.w_down[i_local].t()
 
     def forward(self, x_local: torch.Tensor) -> torch.Tensor:
@@ -188,3 +199,25 @@ class ExpertParallelMoE(nn.Module):
             y_all = dist_nn.functional.all_reduce(y_partial, group=self.group)
             return y_all[self.rank * T:(self.rank + 1) * T]
         return y_partial
+
+
+def mark_ep_checkpoint_identities(model, world: int, rank: int) -> None:
+    """Tag sharded expert params with global checkpoint identities and
+    build the model-level manifest map the checkpoint layer consumes
+    (see scaffold/ckpt/sharded.py): LOCAL slot i on this rank holds
+    GLOBAL expert e0+i, so shards key by global id and rank 0 can write
+    a complete manifest for experts it does not hold."""
+    model._pinned_manifest = {}
+    for li, blk in enumerate(model.blocks):
+        moe = getattr(blk, "moe", None)
+        if moe is None:
+            continue
+        for kind, plist in (("w_up", moe.w_up), ("w_down", moe.w_down)):
+            for i, p_ in enumerate(plist):
+                gid = moe.e0 + i
+                p_._ckpt_name = f"blocks.{li}.moe.{kind}.g{gid}"
+                p_._ckpt_owner = rank
+            for gid in range(moe.n_experts):
+                model._pinned_manifest[
+                    f"blocks.{li}.moe.{kind}.g{gid}"] = (
+                    gid // moe.e_local, list(plist[0].shape))

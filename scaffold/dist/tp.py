@@ -174,6 +174,10 @@ def shard_llama_tp(model, rank: int, world: int, group=None):
     if not hasattr(model.blocks[0].attn, "wqkv"):
         return shard_gpt2_tp(model, rank, world, group)
     for blk in model.blocks:
+        if getattr(blk, "moe", None) is not None:
+            raise NotImplementedError(
+                "TP sharding of MoE blocks is not supported — shard the "
+                "EXPERTS instead (dist.ep_moe / scaffold.dist.ep)")
         m = blk.attn
         if m.n_heads % world or m.n_kv % world:
             raise ValueError(

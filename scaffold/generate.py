@@ -148,6 +148,9 @@ class Generator:
         x = x + _tp_reduce(
             F.linear(o.transpose(1, 2).reshape(B, S, -1), m.wo.weight), m)
         h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
+        if getattr(blk, "moe", None) is not None:
+            B2, S2, C2 = h.shape
+            return x + blk.moe(h.reshape(B2 * S2, C2)).view(B2, S2, C2)
         gu = F.linear(h, blk.w_gateup.weight)
         Fh = gu.shape[-1] // 2
         g = F.silu(gu[..., :Fh]) * gu[..., Fh:]
@@ -296,6 +299,8 @@ class Generator:
             o = self._sdpa_static(q, cache.k[i], cache.v[i], pos_dev + 1)
             x = x + self._lin1(o.transpose(1, 2).reshape(B, 1, -1), m.wo.weight)
             h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
+            if getattr(blk, "moe", None) is not None:
+                return x + blk.moe(h.reshape(B, -1)).view(B, 1, -1)
             gu = self._lin1(h, blk.w_gateup.weight)
             Fh = gu.shape[-1] // 2
             g = F.silu(gu[..., :Fh]) * gu[..., Fh:]
@@ -476,6 +481,8 @@ class Generator:
                 self._lin1(o.transpose(1, 2).reshape(B, 1, -1),
                            m.wo.weight), m)
             h = ops.rms_norm(x, blk.mlp_norm.weight, blk.mlp_norm.eps)
+            if getattr(blk, "moe", None) is not None:
+                return x + blk.moe(h.reshape(B, -1)).view(B, 1, -1)
             gu = self._lin1(h, blk.w_gateup.weight)
             Fh = gu.shape[-1] // 2
             g = F.silu(gu[..., :Fh]) * gu[..., Fh:]

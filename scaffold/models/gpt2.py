@@ -110,26 +110,9 @@ class GPT2(nn.Module):
             for li in range(m.n_layers)
         )
         if ep is not None and ep[0] > 1:  # (world, rank, group)
-            # checkpoint identity for rank-local experts: LOCAL slot i
-            # holds GLOBAL expert e0+i, so shards/manifest key them by
-            # global id (same name never maps to different content on
-            # different ranks); _pinned_manifest lets rank 0 write a
-            # complete manifest for experts it does not hold
-            world, _rank, _group = ep
-            self._pinned_manifest = {}
-            for li, blk in enumerate(self.blocks):
-                moe = blk.moe
-                for kind, plist in (("w_up", moe.w_up),
-                                    ("w_down", moe.w_down)):
-                    for i, p_ in enumerate(plist):
-                        gid = moe.e0 + i
-                        p_._ckpt_name = f"blocks.{li}.moe.{kind}.g{gid}"
-                        p_._ckpt_owner = _rank
-                    e_local = moe.e_local
-                    for gid in range(moe.n_experts):
-                        self._pinned_manifest[
-                            f"blocks.{li}.moe.{kind}.g{gid}"] = (
-                            gid // e_local, list(plist[0].shape))
+            from ..dist.ep import mark_ep_checkpoint_identities
+
+            mark_ep_checkpoint_identities(self, ep[0], ep[1])
         self.ln_f = LayerNorm(m.d_model, m.norm_eps)
         self.lm_head = nn.Linear(m.d_model, m.vocab_size, bias=False)
         if m.tie_embeddings:

@@ -56,19 +56,41 @@ class LlamaAttention(nn.Module):
 
 
 class LlamaBlock(nn.Module):
-    def __init__(self, d_model: int, n_heads: int, n_kv_heads: int, d_ff: int, eps: float):
+    def __init__(self, d_model: int, n_heads: int, n_kv_heads: int,
+                 d_ff: int, eps: float, n_experts: int = 0, ep=None,
+                 init_seed: int = 0, moe_top_k: int = 1):
         super().__init__()
         self.attn_norm = RMSNorm(d_model, eps)
         self.attn = LlamaAttention(d_model, n_heads, n_kv_heads)
         self.mlp_norm = RMSNorm(d_model, eps)
-        # gate and up fused into one [2*d_ff, d] GEMM; the packed SwiGLU
-        # kernel consumes the gate|up halves in place
-        self.w_gateup = nn.Linear(d_model, 2 * d_ff, bias=False)
-        self.w_down = nn.Linear(d_ff, d_model, bias=False)
+        if n_experts > 0:
+            # Mixtral-class routed FFN: SwiGLU experts (packed gate|up,
+            # the same layout as the dense w_gateup), sharded across
+            # ranks when an ep context is present (dist.ep_moe)
+            from ..dist.ep import ExpertParallelMoE
+            world, rank, group = (ep if ep is not None else (1, 0, None))
+            self.moe = ExpertParallelMoE(
+                d_model, d_ff, n_experts, world=world, rank=rank,
+                group=group,
+                dispatch="alltoall" if world > 1 else "dense",
+                init_seed=init_seed, top_k=moe_top_k, act="swiglu")
+            if world > 1:
+                for plist in (self.moe.w_up, self.moe.w_down):
+                    for p_ in plist:
+                        p_._no_ddp = True
+        else:
+            self.moe = None
+            # gate and up fused into one [2*d_ff, d] GEMM; the packed
+            # SwiGLU kernel consumes the gate|up halves in place
+            self.w_gateup = nn.Linear(d_model, 2 * d_ff, bias=False)
+            self.w_down = nn.Linear(d_ff, d_model, bias=False)
 
     def forward(self, x, cos, sin):
         x = self.attn(self.attn_norm(x), cos, sin, residual=x)
         h = self.mlp_norm(x)
+        if self.moe is not None:
+            B, S, C = h.shape
+            return x + self.moe(h.reshape(B * S, C)).view(B, S, C)
         g = ops.swiglu_packed(ops.linear(h, self.w_gateup.weight))
         x = ops.linear_residual(g, self.w_down.weight, None, x)
         return x
@@ -83,9 +105,13 @@ class Llama(nn.Module):
         n_kv = m.n_kv_heads or m.n_heads
         self.cfg = cfg
         self.embed = nn.Embedding(m.vocab_size, m.d_model)
+        ep = getattr(cfg, "_ep_ctx", None)
         self.blocks = nn.ModuleList(
-            LlamaBlock(m.d_model, m.n_heads, n_kv, d_ff, m.norm_eps)
-            for _ in range(m.n_layers)
+            LlamaBlock(m.d_model, m.n_heads, n_kv, d_ff, m.norm_eps,
+                       getattr(m, "n_experts", 0), ep,
+                       init_seed=getattr(cfg, "seed", 0) * 4099 + li,
+                       moe_top_k=getattr(m, "moe_top_k", 1))
+            for li in range(m.n_layers)
         )
         self.norm_f = RMSNorm(m.d_model, m.norm_eps)
         self.lm_head = nn.Linear(m.d_model, m.vocab_size, bias=False)
@@ -99,9 +125,13 @@ class Llama(nn.Module):
             orig_max_pos=getattr(m, "rope_orig_max_pos", 8192))
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
-        for p in self.parameters():
-            if p.dim() > 1:
+        for n_, p in self.named_parameters():
+            if p.dim() > 1 and ".moe." not in n_:
                 nn.init.normal_(p, std=0.02)
+        if ep is not None and ep[0] > 1:
+            from ..dist.ep import mark_ep_checkpoint_identities
+
+            mark_ep_checkpoint_identities(self, ep[0], ep[1])
 
     def forward(self, tokens: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
         B, S = tokens.shape

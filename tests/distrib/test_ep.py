@@ -14,8 +14,9 @@ T, DM, DF, E = 8, 16, 32, 4  # T tokens per rank
 
 def _seed_weights(moe, base):
     g = torch.Generator().manual_seed(777)
+    up_rows = moe.w_up[0].shape[0]        # DF (gelu) or 2*DF (swiglu)
     gate = torch.randn(E, DM, generator=g) * 0.3
-    ups = [torch.randn(DF, DM, generator=g) * 0.1 for _ in range(E)]
+    ups = [torch.randn(up_rows, DM, generator=g) * 0.1 for _ in range(E)]
     downs = [torch.randn(DM, DF, generator=g) * 0.1 for _ in range(E)]
     X = torch.randn(WORLD * T, DM, generator=g)
     dy = torch.randn(WORLD * T, DM, generator=g)
@@ -27,13 +28,14 @@ def _seed_weights(moe, base):
     return X, dy
 
 
-def _worker(rank, world, store_path, q_ret, dispatch="dense", top_k=1):
+def _worker(rank, world, store_path, q_ret, dispatch="dense", top_k=1,
+            act="gelu"):
     from scaffold.dist.ep import ExpertParallelMoE
 
     store = dist.FileStore(store_path, world)
     dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
     moe = ExpertParallelMoE(DM, DF, E, world, rank, dispatch=dispatch,
-                            top_k=top_k)
+                            top_k=top_k, act=act)
     X, dy = _seed_weights(moe, rank * (E // world))
     xl = X[rank * T:(rank + 1) * T].clone().requires_grad_(True)
     y = moe(xl)
@@ -46,18 +48,19 @@ def _worker(rank, world, store_path, q_ret, dispatch="dense", top_k=1):
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(180)
+@pytest.mark.timeout(300)
 @pytest.mark.parametrize("dispatch", ["dense", "alltoall"])
-@pytest.mark.parametrize("top_k", [1, 2])
-def test_ep_moe_matches_full(tmp_path, dispatch, top_k):
+@pytest.mark.parametrize("top_k,act", [(1, "gelu"), (2, "gelu"),
+                                       (2, "swiglu")])
+def test_ep_moe_matches_full(tmp_path, dispatch, top_k, act):
     from scaffold.dist.ep import ExpertParallelMoE
 
     ctxmp = mp.get_context("spawn")
     ret = ctxmp.Queue()
     procs = [ctxmp.Process(
         target=_worker,
-        args=(r, WORLD, str(tmp_path / f"fs{dispatch}{top_k}"), ret,
-              dispatch, top_k))
+        args=(r, WORLD, str(tmp_path / f"fs{dispatch}{top_k}{act}"), ret,
+              dispatch, top_k, act))
              for r in range(WORLD)]
     for p in procs:
         p.start()
@@ -70,7 +73,7 @@ def test_ep_moe_matches_full(tmp_path, dispatch, top_k):
         assert p.exitcode == 0
 
     # single-process reference: world=1 holds ALL experts
-    ref = ExpertParallelMoE(DM, DF, E, 1, 0, top_k=top_k)
+    ref = ExpertParallelMoE(DM, DF, E, 1, 0, top_k=top_k, act=act)
     X, dy = _seed_weights(ref, 0)
     x = X.clone().requires_grad_(True)
     y_ref = ref(x)
