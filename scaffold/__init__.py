@@ -42,6 +42,10 @@ _LAZY = {
     "save_quantized": ("scaffold.quant", "save_quantized"),
     "load_quantized": ("scaffold.quant", "load_quantized"),
     "run_elastic": ("scaffold.launch", "run_elastic"),
+    "shard_llama_tp": ("scaffold.dist.tp", "shard_llama_tp"),
+    "TpContinuousBatcher": ("scaffold.serve_tp", "TpContinuousBatcher"),
+    "run_follower": ("scaffold.serve_tp", "run_follower"),
+    "ExpertParallelMoE": ("scaffold.dist.ep", "ExpertParallelMoE"),
 }
 
 
