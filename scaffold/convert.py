@@ -110,7 +110,12 @@ def load_hf(model: torch.nn.Module, sd: Dict[str, torch.Tensor]) -> None:
         mapped = from_hf_gpt2(sd, len(model.blocks))
     elif isinstance(model, Llama):
         a0 = model.blocks[0].attn
-        mapped = from_hf_llama(sd, len(model.blocks), a0.n_heads, a0.n_kv)
+        if any("mlp.experts.gate_up_proj" in k for k in sd):
+            mapped = from_hf_mixtral(sd, len(model.blocks), a0.n_heads,
+                                     a0.n_kv)
+        else:
+            mapped = from_hf_llama(sd, len(model.blocks), a0.n_heads,
+                                   a0.n_kv)
     else:
         raise TypeError(f"load_hf: unsupported model {type(model).__name__}")
     own = dict(model.named_parameters())
@@ -251,3 +256,39 @@ def load_model(model: torch.nn.Module, path: str) -> None:
     with torch.no_grad():
         for k, t in own.items():
             t.copy_(sd[k].to(t.device, t.dtype))
+
+
+def from_hf_mixtral(sd: Dict[str, torch.Tensor], n_layers: int, n_heads: int,
+                    n_kv_heads: int) -> Dict[str, torch.Tensor]:
+    """transformers MixtralForCausalLM -> scaffold Llama-MoE names.
+
+    Attention maps like Llama (rotate-half unpermute).  transformers'
+    current Mixtral stores the experts FUSED: gate_up_proj (E, 2F, d)
+    whose rows chunk to gate|up — the exact layout of our packed
+    w_up — and down_proj (E, d, F); the router (full-logit softmax ->
+    top-k -> renormalize) matches this repo's routing bit-for-bit, so
+    parity holds end to end (tested)."""
+    lm_head = sd.get("lm_head.weight")
+    sd = _strip_prefix(sd, "model.")
+    out = {
+        "embed.weight": sd["embed_tokens.weight"],
+        "norm_f.weight": sd["norm.weight"],
+        "lm_head.weight": lm_head if lm_head is not None
+        else sd["embed_tokens.weight"],
+    }
+    for i in range(n_layers):
+        h, b = f"layers.{i}.", f"blocks.{i}."
+        out[b + "attn_norm.weight"] = sd[h + "input_layernorm.weight"]
+        out[b + "mlp_norm.weight"] = sd[h + "post_attention_layernorm.weight"]
+        q = _unpermute_rope(sd[h + "self_attn.q_proj.weight"], n_heads)
+        k = _unpermute_rope(sd[h + "self_attn.k_proj.weight"], n_kv_heads)
+        v = sd[h + "self_attn.v_proj.weight"]
+        out[b + "attn.wqkv.weight"] = torch.cat([q, k, v], dim=0)
+        out[b + "attn.wo.weight"] = sd[h + "self_attn.o_proj.weight"]
+        out[b + "moe.gate.weight"] = sd[h + "mlp.gate.weight"]
+        gu = sd[h + "mlp.experts.gate_up_proj"]      # (E, 2F, d)
+        dn = sd[h + "mlp.experts.down_proj"]         # (E, d, F)
+        for e in range(gu.shape[0]):
+            out[b + f"moe.w_up.{e}"] = gu[e].contiguous()
+            out[b + f"moe.w_down.{e}"] = dn[e].contiguous()
+    return out

@@ -261,3 +261,34 @@ def test_linear_rope_scaling_changes_tables():
     # position 2k with factor 2 == position k unscaled
     assert torch.allclose(c1[10], c0[5], atol=1e-6)
     assert torch.allclose(s1[10], s0[5], atol=1e-6)
+
+
+def test_mixtral_matches_transformers():
+    """Mixtral interop: converted weights + our top-2 SwiGLU-expert
+    routing reproduce transformers' MixtralForCausalLM loss."""
+    from transformers import MixtralConfig, MixtralForCausalLM
+
+    torch.manual_seed(141)
+    hf = MixtralForCausalLM(MixtralConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2,
+        max_position_embeddings=32, rope_theta=1e6,
+        rms_norm_eps=1e-5, attention_bias=False,
+        tie_word_embeddings=False, attn_implementation="eager",
+        router_jitter_noise=0.0,
+    )).eval()
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=96",
+        "model.vocab_size=128", "model.max_seq_len=32",
+        "model.rope_theta=1e6", "model.n_experts=4", "model.moe_top_k=2",
+    ])
+    ours = build_model(cfg)
+    load_hf(ours, hf.state_dict())
+    tok = torch.randint(0, 128, (2, 32))
+    tgt = torch.randint(0, 128, (2, 32))
+    with torch.no_grad():
+        ref = _ref_loss(hf(tok).logits, tgt)
+    got = _our_loss(ours, tok, tgt)
+    assert torch.allclose(got, ref, rtol=1e-4, atol=1e-5), (got, ref)
