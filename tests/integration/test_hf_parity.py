@@ -292,3 +292,34 @@ def test_mixtral_matches_transformers():
         ref = _ref_loss(hf(tok).logits, tgt)
     got = _our_loss(ours, tok, tgt)
     assert torch.allclose(got, ref, rtol=1e-4, atol=1e-5), (got, ref)
+
+
+def test_mixtral_generate_matches_transformers_greedy():
+    from transformers import MixtralConfig, MixtralForCausalLM
+
+    from scaffold.generate import Generator
+
+    torch.manual_seed(151)
+    hf = MixtralForCausalLM(MixtralConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2,
+        max_position_embeddings=64, rope_theta=1e6,
+        rms_norm_eps=1e-5, attention_bias=False,
+        tie_word_embeddings=False, attn_implementation="eager",
+        router_jitter_noise=0.0,
+    )).eval()
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=96",
+        "model.vocab_size=128", "model.max_seq_len=64",
+        "model.rope_theta=1e6", "model.n_experts=4", "model.moe_top_k=2",
+    ])
+    ours = build_model(cfg).eval()
+    load_hf(ours, hf.state_dict())
+    prompt = torch.randint(0, 128, (2, 8))
+    with torch.no_grad():
+        ref = hf.generate(prompt, max_new_tokens=10, do_sample=False,
+                          pad_token_id=0)
+        got = Generator(ours).generate(prompt, max_new_tokens=10)
+    assert torch.equal(got, ref)
