@@ -292,3 +292,38 @@ def from_hf_mixtral(sd: Dict[str, torch.Tensor], n_layers: int, n_heads: int,
             out[b + f"moe.w_up.{e}"] = gu[e].contiguous()
             out[b + f"moe.w_down.{e}"] = dn[e].contiguous()
     return out
+
+
+@torch.no_grad()
+def to_hf_mixtral(model: Llama) -> Dict[str, torch.Tensor]:
+    """scaffold Llama-MoE -> transformers MixtralForCausalLM state dict
+    (fused expert tensors; inverse of from_hf_mixtral)."""
+    if getattr(model.blocks[0], "moe", None) is None:
+        raise TypeError("to_hf_mixtral needs a Llama-MoE model "
+                        "(model.n_experts > 0)")
+    sd = {k: v.detach().cpu() for k, v in model.state_dict().items()}
+    a0 = model.blocks[0].attn
+    nh, nkv, hd = a0.n_heads, a0.n_kv, a0.head_dim
+    out = {
+        "model.embed_tokens.weight": sd["embed.weight"],
+        "model.norm.weight": sd["norm_f.weight"],
+        "lm_head.weight": sd["lm_head.weight"],
+    }
+    for i, blk in enumerate(model.blocks):
+        h, b = f"model.layers.{i}.", f"blocks.{i}."
+        out[h + "input_layernorm.weight"] = sd[b + "attn_norm.weight"]
+        out[h + "post_attention_layernorm.weight"] = sd[b + "mlp_norm.weight"]
+        wqkv = sd[b + "attn.wqkv.weight"]
+        q, k, v = torch.split(wqkv, [nh * hd, nkv * hd, nkv * hd], dim=0)
+        out[h + "self_attn.q_proj.weight"] = _permute_rope(q, nh)
+        out[h + "self_attn.k_proj.weight"] = _permute_rope(k, nkv)
+        out[h + "self_attn.v_proj.weight"] = v.contiguous()
+        out[h + "self_attn.o_proj.weight"] = sd[b + "attn.wo.weight"]
+        out[h + "mlp.gate.weight"] = sd[b + "moe.gate.weight"]
+        ups = torch.stack([sd[b + f"moe.w_up.{e}"]
+                           for e in range(blk.moe.n_experts)])
+        downs = torch.stack([sd[b + f"moe.w_down.{e}"]
+                             for e in range(blk.moe.n_experts)])
+        out[h + "mlp.experts.gate_up_proj"] = ups
+        out[h + "mlp.experts.down_proj"] = downs
+    return out

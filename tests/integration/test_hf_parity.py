@@ -323,3 +323,42 @@ def test_mixtral_generate_matches_transformers_greedy():
                           pad_token_id=0)
         got = Generator(ours).generate(prompt, max_new_tokens=10)
     assert torch.equal(got, ref)
+
+
+def test_roundtrip_export_mixtral():
+    from transformers import MixtralConfig, MixtralForCausalLM
+
+    from scaffold.convert import to_hf_mixtral
+
+    torch.manual_seed(161)
+    cfg = load_config(None, [
+        "model.name=llama", "model.n_layers=2", "model.n_heads=4",
+        "model.n_kv_heads=2", "model.d_model=64", "model.d_ff=96",
+        "model.vocab_size=128", "model.max_seq_len=32",
+        "model.rope_theta=1e6", "model.n_experts=4", "model.moe_top_k=2",
+    ])
+    ours = build_model(cfg).eval()
+    exported = to_hf_mixtral(ours)
+    hf = MixtralForCausalLM(MixtralConfig(
+        vocab_size=128, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2,
+        max_position_embeddings=32, rope_theta=1e6,
+        rms_norm_eps=1e-5, attention_bias=False,
+        tie_word_embeddings=False, attn_implementation="eager",
+        router_jitter_noise=0.0,
+    )).eval()
+    missing, unexpected = hf.load_state_dict(exported, strict=False)
+    assert not unexpected, unexpected
+    tok = torch.randint(0, 128, (2, 32))
+    tgt = torch.randint(0, 128, (2, 32))
+    with torch.no_grad():
+        ref = _ref_loss(hf(tok).logits, tgt)
+    assert torch.allclose(_our_loss(ours, tok, tgt), ref,
+                          rtol=1e-4, atol=1e-5)
+    # import(export(x)) == x
+    clone = build_model(cfg)
+    load_hf(clone, exported)
+    for (n, a), (_, b) in zip(ours.named_parameters(),
+                              clone.named_parameters()):
+        assert torch.equal(a, b), n
