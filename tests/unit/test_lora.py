@@ -176,3 +176,24 @@ def test_lora_gpu_identity_and_step():
              if not p.requires_grad]
     for a, b in zip(frozen, after):
         assert torch.equal(a, b)
+
+
+def test_lora_on_moe_model_freezes_experts():
+    """LoRA over a MoE model: experts (raw Parameters, not nn.Linear)
+    freeze with the backbone; adapters train; loss still decreases."""
+    from scaffold.trainer import Trainer
+
+    cfg = _cfg(lora=8, steps=40, **{"model.n_experts": 4,
+                                    "optim.lr": 0.05,
+                                    "model.lora_alpha": 32})
+    t = Trainer(cfg)
+    moe = t.raw_model.blocks[0].moe
+    assert not any(p.requires_grad
+                   for p in list(moe.w_up) + list(moe.w_down))
+    frozen = [p.detach().clone()
+              for p in list(moe.w_up) + list(moe.w_down)]
+    it = iter(t.data)
+    losses = [t._train_step(it, 1) for _ in range(40)]
+    assert min(losses[-5:]) < losses[0] - 0.05, (losses[0], losses[-5:])
+    for a, b in zip(frozen, list(moe.w_up) + list(moe.w_down)):
+        assert torch.equal(a, b.detach())
