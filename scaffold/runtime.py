@@ -10,7 +10,6 @@ import datetime
 import os
 import random
 from dataclasses import dataclass
-from typing import Optional
 
 import numpy as np
 import torch
