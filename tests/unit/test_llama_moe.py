@@ -107,3 +107,29 @@ def test_llama_moe_ep_resume_bitexact(tmp_path):
     pb = _run(tmp_path, "b2", 6, ckpt_dir=str(tmp_path / "cb2"),
               resume=mid)
     assert np.array_equal(pa, pb)
+
+
+@pytest.mark.timeout(600)
+def test_llama_moe_reshard_to_single_process(tmp_path):
+    """EP ws=2 Llama-MoE checkpoint loads into a ws=1 local-expert
+    model (global-id name fallback), optimizer state included."""
+    from scaffold.ckpt import layout
+    from scaffold.trainer import Trainer
+
+    _run(tmp_path, "r", 4, ckpt_dir=str(tmp_path / "cr"))
+    ck = layout.latest_checkpoint(str(tmp_path / "cr"))
+    t = Trainer(_cfg(**{"ckpt.resume": ck, "steps": 4}))
+    assert t.step == 4
+    assert t.raw_model.blocks[0].moe.world == 1
+    assert any(m.abs().sum() > 0 for m in t.optimizer.exp_avg)
+    # every global expert materialized locally
+    import safetensors.torch as st
+    import pathlib
+    shards = {}
+    for f in sorted(pathlib.Path(ck).iterdir()):
+        if f.name.startswith("shard_"):
+            shards.update(st.load_file(str(f)))
+    for gid in range(4):
+        saved = shards[f"model.blocks.0.moe.w_up.g{gid}"]
+        assert torch.equal(
+            t.raw_model.blocks[0].moe.w_up[gid].detach(), saved)
