@@ -9,6 +9,11 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def pytest_configure(config):
     config.addinivalue_line("markers", "gpu: needs a real MI355X GPU (run via gpurun)")
+    # starlette's TestClient warns about its own httpx dependency —
+    # environment-level noise, not ours
+    config.addinivalue_line(
+        "filterwarnings",
+        "ignore::starlette.exceptions.StarletteDeprecationWarning")
 
 
 def pytest_collection_modifyitems(config, items):
@@ -36,3 +41,4 @@ def cpu_cfg():
             "log.interval=1000",
         ],
     )
+
