@@ -363,6 +363,12 @@ class Generator:
         if prompt.shape[0] != 1:
             raise ValueError("speculative decode is B=1 (per-row "
                              "acceptance lengths diverge in a batch)")
+        if getattr(self.m.blocks[0].attn, "_tp_world", 1) > 1:
+            # a lone rank issuing the TP all_reduces would deadlock the
+            # group; speculative x TP needs the command-replay protocol
+            raise NotImplementedError(
+                "speculative decode over a TP-sharded target is not "
+                "supported")
         n_layers = len(self.m.blocks)
         attn0 = self.m.blocks[0].attn
         n_kv = attn0.n_kv if self.is_llama else attn0.n_heads

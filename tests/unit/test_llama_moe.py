@@ -133,3 +133,40 @@ def test_llama_moe_reshard_to_single_process(tmp_path):
         saved = shards[f"model.blocks.0.moe.w_up.g{gid}"]
         assert torch.equal(
             t.raw_model.blocks[0].moe.w_up[gid].detach(), saved)
+
+
+def _ep_act_worker(rank, world, store_path, q):
+    import torch.distributed as dist
+
+    from scaffold.runtime import DistContext
+    from scaffold.trainer import Trainer
+
+    store = dist.FileStore(store_path, world)
+    dist.init_process_group("gloo", store=store, rank=rank, world_size=world)
+    ctx = DistContext(rank=rank, world_size=world, local_rank=rank,
+                      backend="gloo", device=torch.device("cpu"),
+                      initialized=True)
+    t = Trainer(_cfg(**{"dist.ep_moe": "true", "model.act_ckpt": "true",
+                        "steps": 4}), ctx=ctx)
+    t.fit()
+    if rank == 0:
+        q.put(t.step)
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_llama_moe_ep_with_act_ckpt(tmp_path):
+    import torch.multiprocessing as mp
+
+    ctxm = mp.get_context("spawn")
+    q = ctxm.Queue()
+    procs = [ctxm.Process(target=_ep_act_worker,
+                          args=(r, 2, str(tmp_path / "epa"), q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    assert q.get(timeout=200) == 4
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
